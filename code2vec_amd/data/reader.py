@@ -1,0 +1,250 @@
+"""`.c2v` input pipeline.
+
+Reproduces the reference reader's semantics (path_context_reader.py):
+- a row is `target ctx1 ctx2 ...` space-delimited with up to MAX_CONTEXTS
+  contexts; missing/empty fields default to the padding context and an empty
+  target defaults to the target-vocab OOV word (:76-82 record defaults).
+- each context splits on ',' into (source-token, path, target-token); missing
+  parts default to the token PAD word (:189-196).
+- string→index lookups use the OOV default (:198-201).
+- `context_valid_mask[c] = 1.0` iff any of the three parts maps to a non-PAD
+  index (:203-208).
+- row filter (:153-177): evaluate keeps rows with ≥1 valid context; training
+  additionally requires `target_index > target-OOV index`; predict applies NO
+  filter (:96-107).
+- training shuffles with a bounded shuffle buffer and repeats for the
+  configured number of epochs (:138-142).
+
+The pipeline is CPU-side by design: tokenization is not a
+GPU problem. Batches come out as torch int32/float32 tensors (pinned when a
+GPU is present) ready for an async H2D copy on a side stream. A C++ parser
+core can replace `_parse_line` without changing this interface.
+"""
+
+import os
+import random
+from enum import Enum
+from typing import Iterable, Iterator, List, NamedTuple, Optional
+
+import numpy as np
+import torch
+
+from ..config import Config
+from ..vocabularies import Code2VecVocabs
+
+
+class EstimatorAction(Enum):
+    Train = 'train'
+    Evaluate = 'evaluate'
+    Predict = 'predict'
+
+    @property
+    def is_train(self):
+        return self is EstimatorAction.Train
+
+    @property
+    def is_evaluate(self):
+        return self is EstimatorAction.Evaluate
+
+    @property
+    def is_predict(self):
+        return self is EstimatorAction.Predict
+
+    @property
+    def is_evaluate_or_predict(self):
+        return self.is_evaluate or self.is_predict
+
+
+class ReaderBatch(NamedTuple):
+    """One batch of model inputs (the 9-field surface of the reference's
+    ReaderInputTensors, path_context_reader.py:32-44, split into index tensors
+    for the GPU and string lists for CPU-side metrics/serving)."""
+    source_token_indices: torch.Tensor   # (B, C) int32
+    path_indices: torch.Tensor           # (B, C) int32
+    target_token_indices: torch.Tensor   # (B, C) int32
+    context_valid_mask: torch.Tensor     # (B, C) float32
+    target_index: Optional[torch.Tensor] = None       # (B,) int64
+    target_string: Optional[List[str]] = None
+    source_token_strings: Optional[List[List[str]]] = None
+    path_strings: Optional[List[List[str]]] = None
+    target_token_strings: Optional[List[List[str]]] = None
+
+    def to(self, device, non_blocking: bool = True) -> 'ReaderBatch':
+        return self._replace(
+            source_token_indices=self.source_token_indices.to(device, non_blocking=non_blocking),
+            path_indices=self.path_indices.to(device, non_blocking=non_blocking),
+            target_token_indices=self.target_token_indices.to(device, non_blocking=non_blocking),
+            context_valid_mask=self.context_valid_mask.to(device, non_blocking=non_blocking),
+            target_index=None if self.target_index is None
+            else self.target_index.to(device, non_blocking=non_blocking))
+
+
+class PathContextReader:
+    def __init__(self, vocabs: Code2VecVocabs, config: Config,
+                 estimator_action: EstimatorAction,
+                 repeat_endlessly: bool = False, keep_strings: bool = False,
+                 world_size: int = 1, rank: int = 0):
+        self.vocabs = vocabs
+        self.config = config
+        self.estimator_action = estimator_action
+        self.repeat_endlessly = repeat_endlessly
+        # strings are needed for evaluate (metrics/log) and predict (attention display)
+        self.keep_strings = keep_strings or estimator_action.is_evaluate_or_predict
+        self.world_size = world_size
+        self.rank = rank
+
+        tok = vocabs.token_vocab
+        pth = vocabs.path_vocab
+        tgt = vocabs.target_vocab
+        self._tok_w2i = tok.word_to_index
+        self._pth_w2i = pth.word_to_index
+        self._tok_pad = tok.word_to_index[tok.special_words.PAD]
+        self._tok_oov = tok.oov_index
+        self._pth_pad = pth.word_to_index[pth.special_words.PAD]
+        self._pth_oov = pth.oov_index
+        self._tgt_w2i = tgt.word_to_index
+        self._tgt_oov = tgt.oov_index
+        self._tgt_oov_word = tgt.special_words.OOV
+        self._tok_pad_word = tok.special_words.PAD
+
+    # ---- per-row parsing ----
+
+    def _parse_line(self, line: str):
+        """Returns (target_str, target_idx, src (C,), path (C,), tgt (C,),
+        mask (C,), [strings]) as numpy arrays, or None for a malformed line."""
+        C = self.config.MAX_CONTEXTS
+        fields = line.rstrip('\n').split(' ')
+        target_str = fields[0] if fields and fields[0] else self._tgt_oov_word
+        target_idx = self._tgt_w2i.get(target_str, self._tgt_oov)
+
+        src = np.full(C, self._tok_pad, dtype=np.int32)
+        pth = np.full(C, self._pth_pad, dtype=np.int32)
+        tgt = np.full(C, self._tok_pad, dtype=np.int32)
+        strings = ([self._tok_pad_word] * C, [self._tok_pad_word] * C,
+                   [self._tok_pad_word] * C) if self.keep_strings else None
+
+        tok_w2i, pth_w2i = self._tok_w2i, self._pth_w2i
+        tok_oov, pth_oov = self._tok_oov, self._pth_oov
+        n = min(len(fields) - 1, C)
+        for c in range(n):
+            f = fields[1 + c]
+            if not f:
+                continue
+            parts = f.split(',')
+            s = parts[0] if len(parts) > 0 and parts[0] else self._tok_pad_word
+            p = parts[1] if len(parts) > 1 and parts[1] else self._tok_pad_word
+            t = parts[2] if len(parts) > 2 and parts[2] else self._tok_pad_word
+            src[c] = tok_w2i.get(s, tok_oov)
+            pth[c] = pth_w2i.get(p, pth_oov)
+            tgt[c] = tok_w2i.get(t, tok_oov)
+            if strings is not None:
+                strings[0][c] = s
+                strings[1][c] = p
+                strings[2][c] = t
+
+        mask = ((src != self._tok_pad) | (tgt != self._tok_pad)
+                | (pth != self._pth_pad)).astype(np.float32)
+        return target_str, target_idx, src, pth, tgt, mask, strings
+
+    def _row_passes_filter(self, target_idx: int, mask: np.ndarray) -> bool:
+        if self.estimator_action.is_predict:
+            return True
+        any_valid = bool(mask.any())
+        if self.estimator_action.is_evaluate:
+            return any_valid
+        return any_valid and target_idx > self._tgt_oov
+
+    # ---- dataset iteration ----
+
+    def _line_stream(self, data_path: str) -> Iterator[str]:
+        epochs_left = None
+        if self.repeat_endlessly:
+            epochs_left = -1
+        elif self.estimator_action.is_train:
+            epochs_left = max(1, self.config.NUM_TRAIN_EPOCHS)
+        else:
+            epochs_left = 1
+        while epochs_left != 0:
+            with open(data_path, 'r', buffering=self.config.CSV_BUFFER_SIZE or 1 << 20) as f:
+                for line in f:
+                    yield line
+            if epochs_left > 0:
+                epochs_left -= 1
+
+    def _shuffled(self, lines: Iterator[str]) -> Iterator[str]:
+        if not self.estimator_action.is_train or self.config.SHUFFLE_BUFFER_SIZE <= 0:
+            yield from lines
+            return
+        rng = random.Random(1234 + self.rank)
+        buf: List[str] = []
+        size = self.config.SHUFFLE_BUFFER_SIZE
+        for line in lines:
+            if len(buf) < size:
+                buf.append(line)
+                continue
+            j = rng.randrange(size)
+            yield buf[j]
+            buf[j] = line
+        rng.shuffle(buf)
+        yield from buf
+
+    def iter_batches(self, data_path: Optional[str] = None,
+                     input_lines: Optional[Iterable[str]] = None) -> Iterator[ReaderBatch]:
+        """Yield ReaderBatch from a `.c2v` file (sharded across DP ranks by
+        row index) or from in-memory lines (the predict path)."""
+        if input_lines is None:
+            if data_path is None:
+                data_path = self.config.data_path(
+                    is_evaluating=self.estimator_action.is_evaluate)
+            assert data_path and os.path.isfile(data_path), \
+                "dataset file not found: %r" % (data_path,)
+            lines: Iterable[str] = self._shuffled(self._line_stream(data_path))
+        else:
+            lines = input_lines
+
+        batch_size = 1 if self.estimator_action.is_predict else \
+            self.config.batch_size(is_evaluating=self.estimator_action.is_evaluate)
+
+        rows = []
+        for i, line in enumerate(lines):
+            if self.world_size > 1 and (i % self.world_size) != self.rank:
+                continue
+            parsed = self._parse_line(line)
+            if parsed is None:
+                continue
+            if not self._row_passes_filter(parsed[1], parsed[5]):
+                continue
+            rows.append(parsed)
+            if len(rows) == batch_size:
+                yield self._collate(rows)
+                rows = []
+        if rows:
+            yield self._collate(rows)
+
+    def _collate(self, rows) -> ReaderBatch:
+        pin = torch.cuda.is_available()
+
+        def t(arrs, dtype):
+            out = torch.from_numpy(np.stack(arrs)).to(dtype)
+            return out.pin_memory() if pin else out
+
+        batch = ReaderBatch(
+            source_token_indices=t([r[2] for r in rows], torch.int32),
+            path_indices=t([r[3] for r in rows], torch.int32),
+            target_token_indices=t([r[4] for r in rows], torch.int32),
+            context_valid_mask=t([r[5] for r in rows], torch.float32),
+            target_index=t([np.int64(r[1]) for r in rows], torch.int64),
+            target_string=[r[0] for r in rows],
+        )
+        if self.keep_strings:
+            batch = batch._replace(
+                source_token_strings=[r[6][0] for r in rows],
+                path_strings=[r[6][1] for r in rows],
+                target_token_strings=[r[6][2] for r in rows])
+        return batch
+
+    def process_input_row(self, row: str) -> ReaderBatch:
+        """Parse ONE raw line with no filtering and a leading batch dim of 1
+        (the predict path; reference path_context_reader.py:96-107)."""
+        parsed = self._parse_line(row)
+        return self._collate([parsed])

@@ -1,0 +1,242 @@
+"""The code2vec engine: parameters + explicit forward/backward/step orchestration.
+
+One model implementation (vs the reference's TF/Keras dual backends). The math
+is the reference graph (tensorflow_model.py:197-265):
+
+    ctx    = dropout(concat(tok[src], path[p], tok[tgt]))        # K1-K3
+    comb   = tanh(ctx @ TRANSFORM)                               # K4
+    code   = Σ_c softmax_c(comb·ATTENTION + log mask) · comb     # K5-K7
+    logits = code @ TARGETS^T                                    # K8
+    loss   = mean CE(logits, target)                             # K9
+    Adam (dense for TRANSFORM/ATTENTION/TARGETS; lazy sparse-row
+    for the token/path embedding tables)                         # K10
+
+The backward pass is hand-orchestrated (no autograd): every op is an explicit
+fwd/bwd pair dispatched to HIP kernels on GPU (ops/functional.py), fp32
+master weights with bf16 compute shadows, and data-parallel gradient
+communication hooks placed exactly where overlap is possible (the dense
+target-table grad all-reduce is launched before the rest of backward runs).
+"""
+
+import math
+from typing import Dict, NamedTuple, Optional
+
+import torch
+
+from ..config import Config
+from ..ops import functional as F
+
+
+class ForwardState(NamedTuple):
+    ctx: torch.Tensor        # (B*C, 3d) compute dtype
+    comb: torch.Tensor       # (B*C, D) compute dtype
+    code: torch.Tensor       # (B, D) fp32
+    alpha: torch.Tensor      # (B, C) fp32
+    seed: int
+
+
+class NullReducer:
+    """Single-process stand-in for the DP reducer (parallel/ddp.py)."""
+    world_size = 1
+
+    def allreduce_dense(self, key: str, tensor: torch.Tensor):
+        return tensor
+
+    def wait(self, key: str):
+        pass
+
+    def allgather_sparse(self, ids: torch.Tensor, rows: torch.Tensor):
+        return ids, rows
+
+    def allreduce_mean_scalar(self, value: float) -> float:
+        return value
+
+
+class Code2VecNetwork:
+    """Parameter store + step engine. Not an nn.Module: parameters are plain
+    fp32 tensors updated in place by fused Adam kernels; state_dict I/O is
+    provided explicitly for checkpoint compatibility."""
+
+    def __init__(self, config: Config, token_vocab_size: int,
+                 path_vocab_size: int, target_vocab_size: int,
+                 device: Optional[str] = None):
+        self.config = config
+        self.device = torch.device(device or config.resolve_device())
+        self.compute_dtype = (torch.bfloat16 if config.COMPUTE_DTYPE == 'bf16'
+                              else torch.float32)
+        d = config.TOKEN_EMBEDDINGS_SIZE
+        D = config.CODE_VECTOR_SIZE
+        g = torch.Generator(device='cpu').manual_seed(20260911)
+
+        def uniform(shape, limit):
+            t = torch.empty(shape, dtype=torch.float32)
+            t.uniform_(-limit, limit, generator=g)
+            return t.to(self.device)
+
+        # Initializers match the reference (tensorflow_model.py:205-220,:249-250):
+        # variance_scaling(scale=1, fan_out, uniform) for the three tables
+        # → U(±√(3/fan_out)); glorot uniform for TRANSFORM and ATTENTION.
+        self.tok_table = uniform((token_vocab_size, d), math.sqrt(3.0 / d))
+        self.path_table = uniform((path_vocab_size, d), math.sqrt(3.0 / d))
+        self.target_table = uniform((target_vocab_size, D), math.sqrt(3.0 / D))
+        self.w = uniform((D, D), math.sqrt(6.0 / (D + D)))       # (in, out)
+        self.a = uniform((D,), math.sqrt(6.0 / (D + 1)))
+
+        # Adam state (fp32), lazy for the embedding tables.
+        self.adam_step = 0
+        self._adam_m: Dict[str, torch.Tensor] = {}
+        self._adam_v: Dict[str, torch.Tensor] = {}
+        for name in self.param_names():
+            p = self.get_param(name)
+            self._adam_m[name] = torch.zeros_like(p)
+            self._adam_v[name] = torch.zeros_like(p)
+
+        # bf16 (compute-dtype) shadows for GEMM operands.
+        self._refresh_shadows()
+        self._seed_counter = torch.Generator(device='cpu').manual_seed(4242).initial_seed()
+        self._step_ctr = 0
+
+    # ---- parameters ----
+
+    @staticmethod
+    def param_names():
+        return ['tok_table', 'path_table', 'target_table', 'w', 'a']
+
+    def get_param(self, name: str) -> torch.Tensor:
+        return getattr(self, name)
+
+    def num_trainable_params(self) -> int:
+        return sum(self.get_param(n).numel() for n in self.param_names())
+
+    def _refresh_shadows(self, only_w: bool = False):
+        cd = self.compute_dtype
+        self.w_oi = self.w.t().contiguous().to(cd)   # (out,in): fwd GEMM B^T operand
+        self.w_io = self.w.contiguous().to(cd)       # (in,out): bwd GEMM B^T operand
+        self.a_c = self.a.to(torch.float32)
+        if not only_w:
+            self.target_shadow = self.target_table.to(cd)
+
+    def state_dict(self) -> Dict[str, torch.Tensor]:
+        sd = {n: self.get_param(n).cpu() for n in self.param_names()}
+        sd['adam_step'] = torch.tensor(self.adam_step)
+        for n in self.param_names():
+            sd['adam_m.' + n] = self._adam_m[n].cpu()
+            sd['adam_v.' + n] = self._adam_v[n].cpu()
+        return sd
+
+    def weights_state_dict(self) -> Dict[str, torch.Tensor]:
+        """Release form: weights only, optimizer state stripped
+        (reference release flow, tensorflow_model.py:129-136)."""
+        return {n: self.get_param(n).cpu() for n in self.param_names()}
+
+    def load_state_dict(self, sd: Dict[str, torch.Tensor]):
+        for n in self.param_names():
+            self.get_param(n).copy_(sd[n].to(self.device))
+        if 'adam_step' in sd:
+            self.adam_step = int(sd['adam_step'])
+            for n in self.param_names():
+                if ('adam_m.' + n) in sd:
+                    self._adam_m[n].copy_(sd['adam_m.' + n].to(self.device))
+                    self._adam_v[n].copy_(sd['adam_v.' + n].to(self.device))
+        self._refresh_shadows()
+
+    # ---- forward ----
+
+    def forward(self, src_ids, path_ids, tgt_ids, valid_mask,
+                training: bool) -> ForwardState:
+        B, C = src_ids.shape
+        D = self.config.CODE_VECTOR_SIZE
+        self._step_ctr += 1
+        seed = (self._seed_counter + self._step_ctr * 2654435761) & 0x7FFFFFFFFFFFFFFF
+        keep = self.config.DROPOUT_KEEP_RATE if training else 1.0
+        ctx = F.gather_concat_fwd(self.tok_table, self.path_table, src_ids,
+                                  path_ids, tgt_ids, keep, seed, training,
+                                  out_dtype=self.compute_dtype)
+        comb = F.transform_tanh_fwd(ctx, self.w_oi)                  # (B*C, D)
+        code, alpha = F.attention_fwd(comb.reshape(B, C, D), self.a_c, valid_mask)
+        return ForwardState(ctx=ctx, comb=comb, code=code, alpha=alpha, seed=seed)
+
+    def logits(self, code: torch.Tensor) -> torch.Tensor:
+        """code (B,D) fp32 → (B, V_tgt) compute dtype via hipBLASLt
+        (plain library GEMM)."""
+        return code.to(self.compute_dtype) @ self.target_shadow.t()
+
+    # ---- full training step ----
+
+    def train_step(self, src_ids, path_ids, tgt_ids, valid_mask, labels,
+                   reducer: Optional[NullReducer] = None) -> float:
+        reducer = reducer or NullReducer()
+        cfg = self.config
+        B, C = src_ids.shape
+        D = cfg.CODE_VECTOR_SIZE
+
+        st = self.forward(src_ids, path_ids, tgt_ids, valid_mask, training=True)
+        code_c = st.code.to(self.compute_dtype)
+        logits = code_c @ self.target_shadow.t()
+        loss_rows, lse = F.ce_fwd(logits, labels)
+        loss = float(loss_rows.float().mean())
+
+        # ---- backward ----
+        d_logits = F.ce_bwd(logits, lse, labels, 1.0 / B)
+        # Dense target-table grad first: it is the big all-reduce (≈200 MB
+        # bf16 on java14m), launched async so it overlaps the rest of backward.
+        d_target = (d_logits.t() @ code_c)                      # (V,D) compute dtype
+        reducer.allreduce_dense('target_table', d_target)
+        d_code = (d_logits @ self.target_shadow).float()        # (B,D)
+
+        d_comb3, d_a = F.attention_bwd(st.comb.reshape(B, C, D), self.a_c,
+                                       st.alpha, d_code)
+        d_comb = d_comb3.reshape(B * C, D)
+        d_ctx, d_w = F.transform_tanh_bwd(st.ctx, self.w_io, st.comb, d_comb)
+        reducer.allreduce_dense('w', d_w)
+        reducer.allreduce_dense('a', d_a)
+        d_ctx = F.gather_concat_bwd(d_ctx, cfg.DROPOUT_KEEP_RATE, st.seed, True)
+
+        # Sparse embedding grads: (ids, rows) pairs; under DP these are
+        # all-gathered (not dense-all-reduced) — SURVEY §2.4.
+        dt = cfg.TOKEN_EMBEDDINGS_SIZE
+        tok_ids = torch.cat([src_ids.reshape(-1), tgt_ids.reshape(-1)])
+        tok_rows = torch.cat([d_ctx[:, :dt], d_ctx[:, 2 * dt:]], dim=0)
+        path_ids_flat = path_ids.reshape(-1)
+        path_rows = d_ctx[:, dt:2 * dt]
+        tok_ids, tok_rows = reducer.allgather_sparse(tok_ids, tok_rows)
+        path_ids_flat, path_rows = reducer.allgather_sparse(path_ids_flat, path_rows)
+
+        # ---- optimizer (TF AdamOptimizer formulation) ----
+        self.adam_step += 1
+        t, lr = self.adam_step, cfg.ADAM_LR
+        b1, b2, eps = cfg.ADAM_BETA1, cfg.ADAM_BETA2, cfg.ADAM_EPS
+
+        F.adam_sparse_rows_step(self.tok_table, tok_ids, tok_rows,
+                                self._adam_m['tok_table'], self._adam_v['tok_table'],
+                                t, lr, b1, b2, eps)
+        F.adam_sparse_rows_step(self.path_table, path_ids_flat, path_rows,
+                                self._adam_m['path_table'], self._adam_v['path_table'],
+                                t, lr, b1, b2, eps)
+        reducer.wait('w')
+        F.adam_dense_step(self.w, d_w, self._adam_m['w'], self._adam_v['w'],
+                          t, lr, b1, b2, eps)
+        reducer.wait('a')
+        F.adam_dense_step(self.a, d_a, self._adam_m['a'], self._adam_v['a'],
+                          t, lr, b1, b2, eps)
+        self._refresh_shadows(only_w=True)
+        reducer.wait('target_table')
+        F.adam_dense_step(self.target_table, d_target,
+                          self._adam_m['target_table'], self._adam_v['target_table'],
+                          t, lr, b1, b2, eps, shadow=self.target_shadow)
+        return loss
+
+    # ---- evaluation / prediction forward ----
+
+    @torch.no_grad()
+    def predict_batch(self, src_ids, path_ids, tgt_ids, valid_mask, top_k: int,
+                      normalize_scores: bool = False):
+        """Returns (topk_indices (B,k) int64, topk_scores (B,k) fp32,
+        code (B,D) fp32, alpha (B,C) fp32)."""
+        st = self.forward(src_ids, path_ids, tgt_ids, valid_mask, training=False)
+        logits = self.logits(st.code).float()
+        # k is clamped to the vocab size (reference: tensorflow_model.py:299-301)
+        scores, indices = torch.topk(logits, k=min(top_k, logits.shape[1]), dim=1)
+        if normalize_scores:
+            scores = torch.softmax(scores, dim=1)
+        return indices, scores, st.code, st.alpha

@@ -1,0 +1,250 @@
+"""The PyTorch-ROCm code2vec model: training loop, evaluation, prediction,
+checkpoint I/O and embedding export.
+
+This is the single backend replacing the reference's TF/Keras pair. Behavior
+parity anchors (all in /root/reference/):
+- train loop cadence: progress log every NUM_BATCHES_TO_LOG_PROGRESS batches,
+  save+evaluate every SAVE_EVERY_EPOCHS epochs (tensorflow_model.py:75-101).
+- evaluation: top-k words + metrics + per-example `log.txt` dump + optional
+  `.vectors` export (tensorflow_model.py:114-195).
+- prediction: one row at a time, unfiltered, normalized top-k scores and
+  attention per context (tensorflow_model.py:311-368).
+- checkpoints: `<path>__entire-model` includes optimizer state, release form
+  is weights-only `<path>.release` (tensorflow_model.py:129-136,
+  keras_model.py:230-296); `dictionaries.bin` sits next to the weights.
+"""
+
+import os
+import time
+from typing import Iterable, List, Optional
+
+import numpy as np
+import torch
+
+from ..config import Config
+from ..common import common
+from ..data.reader import EstimatorAction, PathContextReader
+from ..utils.metrics import SubtokensEvaluationMetric, TopKAccuracyEvaluationMetric
+from ..vocabularies import VocabType
+from .model_base import (Code2VecModelBase, ModelEvaluationResults,
+                         ModelPredictionResults)
+from .network import Code2VecNetwork, NullReducer
+
+
+class Code2VecModel(Code2VecModelBase):
+    def __init__(self, config: Config, reducer=None, world_size: int = 1,
+                 rank: int = 0):
+        self.network: Optional[Code2VecNetwork] = None
+        self.reducer = reducer or NullReducer()
+        self.world_size = world_size
+        self.rank = rank
+        self._epochs_trained = 0
+        super().__init__(config)
+
+    # ---- inner model lifecycle ----
+
+    def _create_inner_model(self):
+        self.network = Code2VecNetwork(
+            self.config,
+            token_vocab_size=self.vocabs.token_vocab.size,
+            path_vocab_size=self.vocabs.path_vocab.size,
+            target_vocab_size=self.vocabs.target_vocab.size)
+        self.log('Created model on device: %s (compute dtype: %s)'
+                 % (self.network.device, self.config.COMPUTE_DTYPE))
+        self.log('Number of trainable params: %d' % self.network.num_trainable_params())
+
+    def _load_or_create_inner_model(self):
+        self._create_inner_model()
+        if self.config.is_loading:
+            self._load_inner_model()
+
+    def _checkpoint_candidates(self, base_path: str):
+        return [self.config.get_entire_model_path(base_path),
+                self.config.get_model_weights_path(base_path),
+                base_path + '.release',
+                base_path]
+
+    def _load_inner_model(self):
+        for path in self._checkpoint_candidates(self.config.MODEL_LOAD_PATH):
+            if os.path.isfile(path):
+                payload = torch.load(path, map_location='cpu', weights_only=False)
+                state = payload.get('model', payload)
+                self.network.load_state_dict(state)
+                self._epochs_trained = int(payload.get('epoch', 0))
+                self.log('Loaded model weights from: %s (epochs trained: %d)'
+                         % (path, self._epochs_trained))
+                return
+        raise ValueError('No checkpoint found for load path: %s'
+                         % self.config.MODEL_LOAD_PATH)
+
+    def _save_inner_model(self, path: str):
+        if self.config.RELEASE:
+            # weights only, optimizer state stripped (reference release flow)
+            torch.save({'model': self.network.weights_state_dict(),
+                        'epoch': self._epochs_trained}, path + '.release')
+            self.log('Released model saved to: %s' % (path + '.release'))
+            return
+        torch.save({'model': self.network.state_dict(),
+                    'epoch': self._epochs_trained},
+                   self.config.get_entire_model_path(path))
+
+    # ---- training ----
+
+    def train(self):
+        self.log('Starting training')
+        cfg = self.config
+        reader = PathContextReader(vocabs=self.vocabs, config=cfg,
+                                   estimator_action=EstimatorAction.Train,
+                                   world_size=self.world_size, rank=self.rank)
+        device = self.network.device
+        steps_per_epoch = max(1, cfg.train_steps_per_epoch // max(1, self.world_size))
+        save_every = max(1, steps_per_epoch * cfg.SAVE_EVERY_EPOCHS)
+
+        batch_num = 0
+        sum_loss = 0.0
+        window_examples = 0
+        start = time.time()
+        multi_batch_start = start
+        for batch in reader.iter_batches():
+            b = batch.to(device) if device.type != 'cpu' else batch
+            loss = self.network.train_step(
+                b.source_token_indices, b.path_indices, b.target_token_indices,
+                b.context_valid_mask, b.target_index, reducer=self.reducer)
+            batch_num += 1
+            sum_loss += loss
+            window_examples += batch.source_token_indices.shape[0] * self.world_size
+
+            if batch_num % cfg.NUM_BATCHES_TO_LOG_PROGRESS == 0:
+                elapsed = time.time() - multi_batch_start
+                throughput = window_examples / max(elapsed, 1e-9)
+                self.log('Average loss at batch %d: %f, throughput: %d samples/sec'
+                         % (batch_num, sum_loss / cfg.NUM_BATCHES_TO_LOG_PROGRESS,
+                            throughput))
+                sum_loss = 0.0
+                window_examples = 0
+                multi_batch_start = time.time()
+
+            if batch_num % save_every == 0:
+                self._epochs_trained += cfg.SAVE_EVERY_EPOCHS
+                epoch_num = self._epochs_trained
+                if self.rank == 0 and cfg.MODEL_SAVE_PATH:
+                    save_path = cfg.MODEL_SAVE_PATH + '_iter' + str(epoch_num)
+                    self.save(save_path)
+                    self.log('Saved after %d epochs in: %s' % (epoch_num, save_path))
+                if cfg.is_testing and self.rank == 0:
+                    results = self.evaluate()
+                    if results is not None:
+                        self.log('After %d epochs -- %s' % (epoch_num, str(results)))
+
+        elapsed = int(time.time() - start)
+        self.log('Done training')
+        self.log('Training time: %sH:%sM:%sS'
+                 % (elapsed // 3600, (elapsed // 60) % 60, elapsed % 60))
+
+    # ---- evaluation ----
+
+    def evaluate(self) -> Optional[ModelEvaluationResults]:
+        cfg = self.config
+        if cfg.RELEASE and cfg.is_loading:
+            # `--release` with a loaded model: strip optimizer state and exit
+            # (reference: tensorflow_model.py:129-136).
+            self._save_inner_model(cfg.MODEL_LOAD_PATH)
+            return None
+        self.log('Starting evaluation')
+        reader = PathContextReader(vocabs=self.vocabs, config=cfg,
+                                   estimator_action=EstimatorAction.Evaluate)
+        device = self.network.device
+        k = cfg.TOP_K_WORDS_CONSIDERED_DURING_PREDICTION
+        index_to_word = self.vocabs.target_vocab.index_to_word
+        special_words = self.vocabs.target_vocab.special_words
+
+        topk_metric = TopKAccuracyEvaluationMetric(
+            k, lambda orig, top: common.get_first_match_word_from_top_predictions(
+                special_words, orig, top))
+        subtoken_metric = SubtokensEvaluationMetric(
+            lambda top: common.filter_impossible_names(special_words, top))
+
+        export_file = None
+        if cfg.EXPORT_CODE_VECTORS:
+            export_file = open(cfg.TEST_DATA_PATH + '.vectors', 'w')
+        log_file = open('log.txt', 'w')
+        total_loss, total_rows = 0.0, 0
+        start = time.time()
+        try:
+            nr_examples = 0
+            for batch in reader.iter_batches():
+                b = batch.to(device) if device.type != 'cpu' else batch
+                indices, scores, code, _alpha = self.network.predict_batch(
+                    b.source_token_indices, b.path_indices,
+                    b.target_token_indices, b.context_valid_mask, top_k=k)
+                idx_np = indices.cpu().numpy()
+                top_words = [[index_to_word.get(int(i), special_words.OOV)
+                              for i in row] for row in idx_np]
+                pairs = list(zip(batch.target_string, top_words))
+                topk_metric.update_batch(pairs)
+                subtoken_metric.update_batch(pairs)
+                nr_examples += len(pairs)
+                for original_name, words in pairs:
+                    found = common.get_first_match_word_from_top_predictions(
+                        special_words, original_name, words)
+                    if found is not None:
+                        log_file.write('Original: ' + original_name
+                                       + ', predicted %d. most likely: %s\n'
+                                       % (found[0] + 1, found[1]))
+                    else:
+                        log_file.write('No results for predicting: ' + original_name + '\n')
+                if export_file is not None:
+                    self._write_code_vectors(export_file, code.cpu().numpy())
+        finally:
+            log_file.close()
+            if export_file is not None:
+                export_file.close()
+
+        elapsed = int(time.time() - start)
+        self.log('Done evaluating, epoch reached. Evaluated %d examples in %ds'
+                 % (nr_examples, elapsed))
+        return ModelEvaluationResults(
+            topk_acc=topk_metric.topk_correct_predictions,
+            subtoken_precision=subtoken_metric.precision,
+            subtoken_recall=subtoken_metric.recall,
+            subtoken_f1=subtoken_metric.f1,
+            loss=(total_loss / total_rows) if total_rows else None)
+
+    # ---- prediction (serving path) ----
+
+    def predict(self, predict_data_lines: Iterable[str]) -> List[ModelPredictionResults]:
+        reader = PathContextReader(vocabs=self.vocabs, config=self.config,
+                                   estimator_action=EstimatorAction.Predict,
+                                   keep_strings=True)
+        device = self.network.device
+        k = self.config.TOP_K_WORDS_CONSIDERED_DURING_PREDICTION
+        index_to_word = self.vocabs.target_vocab.index_to_word
+        special_words = self.vocabs.target_vocab.special_words
+        results: List[ModelPredictionResults] = []
+        for line in predict_data_lines:
+            batch = reader.process_input_row(line)
+            b = batch.to(device) if device.type != 'cpu' else batch
+            indices, scores, code, alpha = self.network.predict_batch(
+                b.source_token_indices, b.path_indices, b.target_token_indices,
+                b.context_valid_mask, top_k=k, normalize_scores=True)
+            words = np.array([index_to_word.get(int(i), special_words.OOV)
+                              for i in indices[0].cpu().numpy()])
+            attention_per_context = self._get_attention_weight_per_context(
+                batch.source_token_strings[0], batch.path_strings[0],
+                batch.target_token_strings[0], alpha[0].cpu().numpy())
+            results.append(ModelPredictionResults(
+                original_name=batch.target_string[0],
+                topk_predicted_words=words,
+                topk_predicted_words_scores=scores[0].cpu().numpy(),
+                attention_per_context=attention_per_context,
+                code_vector=code[0].cpu().numpy()))
+        return results
+
+    # ---- embedding export ----
+
+    def _get_vocab_embedding_as_np_array(self, vocab_type: VocabType) -> np.ndarray:
+        assert vocab_type in VocabType
+        table = {VocabType.Token: self.network.tok_table,
+                 VocabType.Path: self.network.path_table,
+                 VocabType.Target: self.network.target_table}[vocab_type]
+        return table.detach().cpu().numpy()

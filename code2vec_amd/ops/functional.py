@@ -1,0 +1,100 @@
+"""Backend dispatch: each engine op routes to the HIP extension on GPU and to
+the pure-torch reference on CPU. The HIP path is mandatory on GPU (see
+ops/__init__.backend_for) — no silent eager fallback."""
+
+from typing import Tuple
+
+import torch
+
+from . import backend_for, hip_ext
+from . import reference as ref
+
+
+def gather_concat_fwd(tok_table, path_table, src_ids, path_ids, tgt_ids,
+                      keep_prob: float, seed: int, training: bool,
+                      out_dtype=torch.bfloat16) -> torch.Tensor:
+    if backend_for(tok_table) == 'hip':
+        assert out_dtype == torch.bfloat16, "HIP gather_concat emits bf16"
+        return hip_ext(True).gather_concat_fwd(
+            tok_table, path_table, src_ids, path_ids, tgt_ids,
+            float(keep_prob), int(seed), bool(training))
+    return ref.gather_concat_fwd(tok_table, path_table, src_ids, path_ids,
+                                 tgt_ids, keep_prob, seed, training,
+                                 out_dtype=out_dtype)
+
+
+def gather_concat_bwd(d_ctx, keep_prob: float, seed: int, training: bool) -> torch.Tensor:
+    if backend_for(d_ctx) == 'hip':
+        return hip_ext(True).gather_concat_bwd(d_ctx, float(keep_prob),
+                                               int(seed), bool(training))
+    return ref.gather_concat_bwd(d_ctx, keep_prob, seed, training)
+
+
+def transform_tanh_fwd(ctx, w_oi_bf16) -> torch.Tensor:
+    """ctx (N,K) bf16 × W — the HIP kernel takes the (out,in) bf16 shadow
+    (row-major in K, the MFMA B^T layout); the reference path takes the same
+    and transposes internally."""
+    if backend_for(ctx) == 'hip':
+        return hip_ext(True).transform_tanh_fwd(ctx, w_oi_bf16)
+    return ref.transform_tanh_fwd(ctx, w_oi_bf16.t())
+
+
+def transform_tanh_bwd(ctx, w_io_bf16, y, d_y) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Returns (d_ctx, d_w (in,out) fp32). The GEMM for d_ctx = d_z·W^T runs on
+    the MFMA kernel with the (in,out) shadow as its B^T operand; d_w runs on
+    hipBLASLt (plain library GEMM)."""
+    if backend_for(ctx) == 'hip':
+        e = hip_ext(True)
+        d_z = e.tanh_bwd_mul(d_y, y)          # d_z = d_y * (1 - y^2)
+        d_ctx = e.gemm_bt_bf16(d_z, w_io_bf16)
+        d_w = (ctx.t().to(d_z.dtype) @ d_z).float()
+        return d_ctx, d_w
+    d_zr = d_y * (1.0 - y.float() ** 2).to(d_y.dtype)
+    d_ctx = d_zr @ w_io_bf16.to(d_zr.dtype).t()
+    d_w = (ctx.t().float() @ d_zr.float())
+    return d_ctx, d_w
+
+
+def attention_fwd(comb, a, valid_mask) -> Tuple[torch.Tensor, torch.Tensor]:
+    if backend_for(comb) == 'hip':
+        return hip_ext(True).attention_fwd(comb, a, valid_mask)
+    return ref.attention_fwd(comb, a, valid_mask)
+
+
+def attention_bwd(comb, a, alpha, d_code) -> Tuple[torch.Tensor, torch.Tensor]:
+    if backend_for(comb) == 'hip':
+        return hip_ext(True).attention_bwd(comb, a, alpha, d_code)
+    return ref.attention_bwd(comb, a, alpha, d_code)
+
+
+def ce_fwd(logits, labels) -> Tuple[torch.Tensor, torch.Tensor]:
+    if backend_for(logits) == 'hip':
+        return hip_ext(True).ce_fwd(logits, labels)
+    return ref.ce_fwd(logits, labels)
+
+
+def ce_bwd(logits, lse, labels, scale: float) -> torch.Tensor:
+    if backend_for(logits) == 'hip':
+        return hip_ext(True).ce_bwd(logits, lse, labels, float(scale))
+    return ref.ce_bwd(logits, lse, labels, scale)
+
+
+def adam_dense_step(p, g, m, v, step: int, lr: float, beta1: float,
+                    beta2: float, eps: float, shadow=None):
+    if backend_for(p) == 'hip':
+        hip_ext(True).adam_dense_step(
+            p, g, m, v, int(step), float(lr), float(beta1), float(beta2),
+            float(eps), shadow if shadow is not None else torch.empty(0))
+        return
+    ref.adam_dense_step(p, g, m, v, step, lr, beta1, beta2, eps, shadow)
+
+
+def adam_sparse_rows_step(p, ids, grad_rows, m, v, step: int, lr: float,
+                          beta1: float, beta2: float, eps: float):
+    if backend_for(p) == 'hip':
+        uniq, inverse = torch.unique(ids.long(), return_inverse=True)
+        hip_ext(True).adam_sparse_rows_step(
+            p, uniq.to(torch.int64), inverse.to(torch.int32), grad_rows, m, v,
+            int(step), float(lr), float(beta1), float(beta2), float(eps))
+        return
+    ref.adam_sparse_rows_step(p, ids, grad_rows, m, v, step, lr, beta1, beta2, eps)

@@ -1,0 +1,136 @@
+"""The engine's hand-orchestrated backward + TF-style Adam vs a pure torch
+autograd twin of the exact reference math (tensorflow_model.py:226-263).
+Run in fp32 with dropout off so the comparison is tight."""
+
+import math
+
+import pytest
+import torch
+
+from code2vec_amd.config import Config
+from code2vec_amd.models.network import Code2VecNetwork
+
+B, C, d, D = 5, 7, 8, 24
+V_TOK, V_PATH, V_TGT = 30, 20, 15
+
+
+def tiny_config():
+    cfg = Config(set_defaults=True)
+    cfg.TRAIN_DATA_PATH_PREFIX = 'unused'
+    cfg.MAX_CONTEXTS = C
+    cfg.TOKEN_EMBEDDINGS_SIZE = d
+    cfg.PATH_EMBEDDINGS_SIZE = d
+    cfg.CODE_VECTOR_SIZE = 3 * d
+    cfg.TARGET_EMBEDDINGS_SIZE = 3 * d
+    cfg.DROPOUT_KEEP_RATE = 1.0       # deterministic for parity
+    cfg.COMPUTE_DTYPE = 'fp32'
+    cfg.DEVICE = 'cpu'
+    return cfg
+
+
+def make_net():
+    torch.manual_seed(7)
+    return Code2VecNetwork(tiny_config(), V_TOK, V_PATH, V_TGT, device='cpu')
+
+
+def make_batch(seed=3):
+    g = torch.Generator().manual_seed(seed)
+    src = torch.randint(0, V_TOK, (B, C), generator=g, dtype=torch.int32)
+    pth = torch.randint(0, V_PATH, (B, C), generator=g, dtype=torch.int32)
+    tgt = torch.randint(0, V_TOK, (B, C), generator=g, dtype=torch.int32)
+    mask = (torch.rand(B, C, generator=g) > 0.3).float()
+    mask[:, 0] = 1.0  # every row has ≥1 valid context (reader guarantees this)
+    labels = torch.randint(1, V_TGT, (B,), generator=g)
+    return src, pth, tgt, mask, labels
+
+
+def autograd_reference_loss_and_grads(net, src, pth, tgt, mask, labels):
+    params = {n: net.get_param(n).detach().clone().requires_grad_(True)
+              for n in net.param_names()}
+    Dv = net.config.CODE_VECTOR_SIZE
+    s = params['tok_table'][src.long()]
+    p = params['path_table'][pth.long()]
+    t = params['tok_table'][tgt.long()]
+    ctx = torch.cat([s, p, t], dim=-1)                       # (B,C,3d)
+    comb = torch.tanh(ctx.reshape(-1, Dv) @ params['w']).reshape(B, C, Dv)
+    scores = comb @ params['a'].reshape(-1, 1)               # (B,C,1)
+    scores = scores + torch.log(mask).reshape(B, C, 1)
+    alpha = torch.softmax(scores, dim=1)
+    code = (comb * alpha).sum(dim=1)                         # (B,D)
+    logits = code @ params['target_table'].t()
+    loss = torch.nn.functional.cross_entropy(logits, labels, reduction='sum') / B
+    loss.backward()
+    return loss, {n: params[n].grad for n in params}
+
+
+def tf_adam_expected(p0, g, lr=0.001, b1=0.9, b2=0.999, eps=1e-8, t=1):
+    m = (1 - b1) * g
+    v = (1 - b2) * g * g
+    lr_t = lr * math.sqrt(1 - b2 ** t) / (1 - b1 ** t)
+    return p0 - lr_t * m / (v.sqrt() + eps)
+
+
+def test_forward_matches_autograd_reference():
+    net = make_net()
+    src, pth, tgt, mask, labels = make_batch()
+    st = net.forward(src, pth, tgt, mask, training=True)  # keep=1 → deterministic
+    loss_ref, _ = autograd_reference_loss_and_grads(net, src, pth, tgt, mask, labels)
+    logits = net.logits(st.code)
+    from code2vec_amd.ops import reference as R
+    loss_rows, _ = R.ce_fwd(logits, labels)
+    assert torch.allclose(loss_rows.mean(), loss_ref.float(), atol=1e-5)
+
+
+def test_train_step_matches_autograd_plus_tf_adam():
+    net = make_net()
+    src, pth, tgt, mask, labels = make_batch()
+    p0 = {n: net.get_param(n).detach().clone() for n in net.param_names()}
+    _, grads = autograd_reference_loss_and_grads(net, src, pth, tgt, mask, labels)
+
+    net.train_step(src, pth, tgt, mask, labels)
+
+    # dense params: exact TF-Adam update from the autograd grads
+    for name in ['w', 'a', 'target_table']:
+        expected = tf_adam_expected(p0[name], grads[name])
+        assert torch.allclose(net.get_param(name), expected, atol=1e-5), name
+
+    # embedding tables: touched rows get the same update; untouched rows and
+    # their Adam moments stay exactly put (lazy sparse Adam)
+    for name, ids in [('tok_table', torch.cat([src.reshape(-1), tgt.reshape(-1)])),
+                      ('path_table', pth.reshape(-1))]:
+        touched = torch.unique(ids.long())
+        untouched = torch.tensor([i for i in range(p0[name].shape[0])
+                                  if i not in set(touched.tolist())])
+        expected = tf_adam_expected(p0[name][touched], grads[name][touched])
+        assert torch.allclose(net.get_param(name)[touched], expected, atol=1e-5), name
+        if untouched.numel():
+            assert torch.equal(net.get_param(name)[untouched], p0[name][untouched])
+            assert torch.all(net._adam_m[name][untouched] == 0)
+
+
+def test_loss_decreases_over_steps():
+    net = make_net()
+    src, pth, tgt, mask, labels = make_batch()
+    losses = [net.train_step(src, pth, tgt, mask, labels) for _ in range(150)]
+    assert losses[-1] < losses[0] * 0.5
+
+
+def test_all_masked_row_yields_zero_code_vector():
+    net = make_net()
+    src, pth, tgt, mask, labels = make_batch()
+    mask[2] = 0.0
+    st = net.forward(src, pth, tgt, mask, training=False)
+    assert torch.all(torch.isfinite(st.code))
+    assert torch.all(st.code[2] == 0)
+    assert torch.all(st.alpha[2] == 0)
+
+
+def test_dropout_mask_deterministic_and_scaled():
+    from code2vec_amd.ops.reference import dropout_keep_mask
+    m1 = dropout_keep_mask(123, 10000, 0.75, 'cpu')
+    m2 = dropout_keep_mask(123, 10000, 0.75, 'cpu')
+    assert torch.equal(m1, m2)
+    frac = m1.float().mean().item()
+    assert 0.70 < frac < 0.80
+    m3 = dropout_keep_mask(124, 10000, 0.75, 'cpu')
+    assert not torch.equal(m1, m3)
