@@ -164,7 +164,9 @@ class Code2VecNetwork:
     # ---- full training step ----
 
     def train_step(self, src_ids, path_ids, tgt_ids, valid_mask, labels,
-                   reducer: Optional[NullReducer] = None) -> float:
+                   reducer: Optional[NullReducer] = None) -> torch.Tensor:
+        """One full optimizer step. Returns the (device-resident, unsynced)
+        scalar mean loss so callers control when to pay the H2D sync."""
         reducer = reducer or NullReducer()
         cfg = self.config
         B, C = src_ids.shape
@@ -174,7 +176,7 @@ class Code2VecNetwork:
         code_c = st.code.to(self.compute_dtype)
         logits = code_c @ self.target_shadow.t()
         loss_rows, lse = F.ce_fwd(logits, labels)
-        loss = float(loss_rows.float().mean())
+        loss = loss_rows.float().mean()
 
         # ---- backward ----
         d_logits = F.ce_bwd(logits, lse, labels, 1.0 / B)

@@ -111,7 +111,7 @@ class Code2VecModel(Code2VecModelBase):
                 b.source_token_indices, b.path_indices, b.target_token_indices,
                 b.context_valid_mask, b.target_index, reducer=self.reducer)
             batch_num += 1
-            sum_loss += loss
+            sum_loss += float(loss)
             window_examples += batch.source_token_indices.shape[0] * self.world_size
 
             if batch_num % cfg.NUM_BATCHES_TO_LOG_PROGRESS == 0:

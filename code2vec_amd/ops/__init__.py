@@ -23,13 +23,25 @@ def _try_load_extension():
     global _EXT, _EXT_ERR
     if _EXT is not None or _EXT_ERR is not None:
         return _EXT
-    try:
-        import importlib
-        _EXT = importlib.import_module('code2vec_amd.ops._c2v_hip')
-    except Exception as e:  # noqa: BLE001
-        _EXT_ERR = e
-        _EXT = None
-    return _EXT
+    import importlib.util
+    here = os.path.dirname(os.path.abspath(__file__))
+    candidates = [os.path.join(here, '_c2v_hip.so'),
+                  os.path.join(here, '_build', '_c2v_hip.so')]
+    last_err = None
+    for path in candidates:
+        if not os.path.isfile(path):
+            continue
+        try:
+            spec = importlib.util.spec_from_file_location('code2vec_amd.ops._c2v_hip', path)
+            mod = importlib.util.module_from_spec(spec)
+            spec.loader.exec_module(mod)
+            _EXT = mod
+            return _EXT
+        except Exception as e:  # noqa: BLE001
+            last_err = e
+    _EXT_ERR = last_err or FileNotFoundError(
+        'no _c2v_hip.so found in %s' % (candidates,))
+    return None
 
 
 def hip_ext(required: bool = False):

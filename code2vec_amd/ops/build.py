@@ -1,0 +1,44 @@
+"""In-tree build of the CDNA4 HIP extension.
+
+Builds `code2vec_amd/ops/_build/_c2v_hip.so` for gfx950 via
+torch.utils.cpp_extension (which drives hipcc for the .hip source) and copies
+it next to this file so the loader (ops/__init__.py) and the gpurun snapshot
+both see it. hipcc cross-compiles without a GPU, so this runs on the CPU-only
+build host too.
+
+Usage: python -m code2vec_amd.ops.build
+"""
+
+import os
+import shutil
+import sys
+
+
+def build(verbose: bool = True) -> str:
+    os.environ.setdefault('PYTORCH_ROCM_ARCH', 'gfx950')
+    here = os.path.dirname(os.path.abspath(__file__))
+    build_dir = os.path.join(here, '_build')
+    os.makedirs(build_dir, exist_ok=True)
+    src = os.path.join(here, 'csrc', 'c2v_kernels.hip')
+
+    from torch.utils import cpp_extension
+    cpp_extension.load(
+        name='_c2v_hip',
+        sources=[src],
+        build_directory=build_dir,
+        extra_cflags=['-O3'],
+        extra_cuda_cflags=['-O3', '-std=c++17'],
+        verbose=verbose,
+        is_python_module=False,
+        with_cuda=True,
+    )
+    built = os.path.join(build_dir, '_c2v_hip.so')
+    dest = os.path.join(here, '_c2v_hip.so')
+    shutil.copy2(built, dest)
+    return dest
+
+
+if __name__ == '__main__':
+    path = build()
+    print('built:', path)
+    sys.exit(0)

@@ -1,0 +1,836 @@
+// code2vec_amd — hand-written CDNA4 (gfx950 / MI355X) kernels.
+//
+// Covers the reference hot-op inventory:
+//   K1-K3  fused embedding gather + concat + dropout  (k_gather_concat)
+//   K4     MFMA bf16 128x128-tile GEMM w/ tanh epilogue (k_gemm_bt) — the
+//          m97 structure from the CDNA4 guide: global_load_lds width-16
+//          staging, 16x16x32 bf16 MFMA, 4 waves x (64x64) per block,
+//          XCD-aware bijective block swizzle
+//   K5-K7  fused masked-softmax attention reduce, fwd+bwd (k_attn_*) —
+//          the whole (C,D) context tile for one example stays in LDS
+//          (200x384 bf16 = 150 KiB of the 160 KiB/CU)
+//   K9     fused large-vocab log-softmax + NLL, fwd+bwd (k_ce_*)
+//   K10    dense Adam (TF formulation) + lazy sparse-row Adam for the
+//          embedding tables (k_adam_*)
+//
+// Compute dtype is bf16 with fp32 accumulation; master weights fp32.
+// Wavefront size is 64 everywhere (CDNA4), blocks are multiples of 64.
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include <cstdint>
+
+#define DEVI __device__ __forceinline__
+
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+using u32 = uint32_t;
+using u64 = uint64_t;
+
+// ---------------------------------------------------------------------------
+// small helpers
+// ---------------------------------------------------------------------------
+
+DEVI float bf2f(ushort u) {
+  union { u32 u; float f; } cvt;
+  cvt.u = ((u32)u) << 16;
+  return cvt.f;
+}
+
+DEVI ushort f2bf(float f) {
+  // round-to-nearest-even, NaN-safe enough for our ranges
+  union { float f; u32 u; } cvt;
+  cvt.f = f;
+  u32 u = cvt.u;
+  u32 rounded = (u + 0x7FFFu + ((u >> 16) & 1u)) >> 16;
+  return (ushort)rounded;
+}
+
+// splitmix64 — must match code2vec_amd/ops/reference.py::_splitmix64 bit-for-bit
+DEVI u64 splitmix64(u64 x) {
+  u64 z = x * 0x9E3779B97F4A7C15ull;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+  return z ^ (z >> 31);
+}
+
+DEVI float hash_uniform(u64 seed, u64 idx) {
+  u64 h = splitmix64(seed + idx);
+  return (float)((h >> 40) & 0xFFFFFFull) * (1.0f / 16777216.0f);
+}
+
+DEVI float wave_reduce_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+  return v;
+}
+
+DEVI float wave_reduce_max(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_down(v, off, 64));
+  return v;
+}
+
+// ---------------------------------------------------------------------------
+// K1-K3: fused gather + concat + dropout
+// out (Nrows, 3d) bf16 from two fp32 tables; element (row, off):
+//   off <  d   -> tok_table[src[row]][off]
+//   off < 2d   -> path_table[path[row]][off-d]
+//   else       -> tok_table[tgt[row]][off-2d]
+// dropout keep-mask is the splitmix64 counter hash over the flat element
+// index (identical to the CPU reference), scaled by 1/keep_prob.
+// ---------------------------------------------------------------------------
+
+__global__ void k_gather_concat(
+    const float* __restrict__ tok, const float* __restrict__ path,
+    const int* __restrict__ src_ids, const int* __restrict__ path_ids,
+    const int* __restrict__ tgt_ids, ushort* __restrict__ out,
+    int n_rows, int d, float keep_prob, u64 seed, int apply_dropout) {
+  const int slots_per_row = (3 * d) / 8;
+  const long total = (long)n_rows * slots_per_row;
+  const float inv_keep = 1.0f / keep_prob;
+  for (long slot = blockIdx.x * blockDim.x + threadIdx.x; slot < total;
+       slot += (long)gridDim.x * blockDim.x) {
+    const int row = (int)(slot / slots_per_row);
+    const int off = (int)(slot % slots_per_row) * 8;
+    const float* base;
+    int col;
+    if (off < d) {
+      base = tok + (long)src_ids[row] * d;
+      col = off;
+    } else if (off < 2 * d) {
+      base = path + (long)path_ids[row] * d;
+      col = off - d;
+    } else {
+      base = tok + (long)tgt_ids[row] * d;
+      col = off - 2 * d;
+    }
+    float4 lo = *reinterpret_cast<const float4*>(base + col);
+    float4 hi = *reinterpret_cast<const float4*>(base + col + 4);
+    float vals[8] = {lo.x, lo.y, lo.z, lo.w, hi.x, hi.y, hi.z, hi.w};
+    if (apply_dropout) {
+      const u64 eidx = (u64)row * (3 * d) + off;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const bool keep = hash_uniform(seed, eidx + j) < keep_prob;
+        vals[j] = keep ? vals[j] * inv_keep : 0.0f;
+      }
+    }
+    ushort outv[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) outv[j] = f2bf(vals[j]);
+    *reinterpret_cast<ulonglong2*>(out + (long)row * (3 * d) + off) =
+        *reinterpret_cast<ulonglong2*>(outv);
+  }
+}
+
+// dropout backward: scale d_ctx by the same keep mask (bf16 in/out)
+__global__ void k_dropout_bwd(const ushort* __restrict__ g_in,
+                              ushort* __restrict__ g_out, long total,
+                              float keep_prob, u64 seed) {
+  const float inv_keep = 1.0f / keep_prob;
+  for (long i8 = blockIdx.x * blockDim.x + threadIdx.x; i8 * 8 < total;
+       i8 += (long)gridDim.x * blockDim.x) {
+    const long e = i8 * 8;
+    ulonglong2 packed = *reinterpret_cast<const ulonglong2*>(g_in + e);
+    ushort* u = reinterpret_cast<ushort*>(&packed);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const bool keep = hash_uniform(seed, e + j) < keep_prob;
+      u[j] = keep ? f2bf(bf2f(u[j]) * inv_keep) : (ushort)0;
+    }
+    *reinterpret_cast<ulonglong2*>(g_out + e) = packed;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K4: bf16 MFMA GEMM, C(N,M) = A(N,K) @ Bt(M,K)^T, optional tanh epilogue.
+// m97 structure (CDNA4 guide §5): 128x128 tile, BK=32, 256 threads (4 waves,
+// 2x2 of 64x64), LDS staging via __builtin_amdgcn_global_load_lds width 16,
+// v_mfma_f32_16x16x32_bf16, XCD-aware bijective grid swizzle (m204).
+// Ragged N/M handled by clamping staged rows and masking the C-store.
+// ---------------------------------------------------------------------------
+
+#define GEMM_BM 128
+#define GEMM_BN 128
+#define GEMM_BK 32
+
+template <bool TANH>
+__launch_bounds__(256)
+__global__ void k_gemm_bt(const ushort* __restrict__ A,
+                          const ushort* __restrict__ Bt,
+                          ushort* __restrict__ C, int N, int M, int K) {
+  __shared__ ushort lds_a[GEMM_BM * GEMM_BK];
+  __shared__ ushort lds_b[GEMM_BN * GEMM_BK];
+
+  // XCD-aware bijective swizzle of the (row-tile, col-tile) space
+  const int n_tiles = (N + GEMM_BM - 1) / GEMM_BM;
+  const int m_tiles = (M + GEMM_BN - 1) / GEMM_BN;
+  const int nwg = n_tiles * m_tiles;
+  int wg = blockIdx.x;
+  {
+    const int nxcd = 8;
+    const int q = nwg / nxcd, r = nwg % nxcd;
+    const int xcd = wg % nxcd, orig = wg / nxcd;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig;
+  }
+  const int tile_n = wg / m_tiles;   // row-tile of A
+  const int tile_m = wg % m_tiles;   // row-tile of Bt (= col-tile of C)
+  const int row0 = tile_n * GEMM_BM;
+  const int col0 = tile_m * GEMM_BN;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;           // 4 waves: 2x2 over (64,64) subtiles
+  const int wrow = (wid >> 1) * 64;   // wave's row offset in the tile
+  const int wcol = (wid & 1) * 64;    // wave's col offset
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int k_steps = K / GEMM_BK;
+  for (int ks = 0; ks < k_steps; ++ks) {
+    const int k0 = ks * GEMM_BK;
+    // ---- stage A and B tiles: each wave issues 2+2 width-16 direct loads.
+    // One wave-load covers 1 KiB of LDS = 16 rows x 32 cols bf16; lane i
+    // sources row (i>>2), byte-col (i&3)*16 of the 16x64B sub-tile.
+    {
+      const int sub = wid;                       // wave's 16-row chunk pair
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        const int r16 = sub * 2 + half;          // 16-row group index (0..7)
+        const int lrow = r16 * 16 + (lane >> 2);
+        const int kb = (lane & 3) * 8;           // 8 bf16 = 16 B
+        {
+          const int grow = min(row0 + lrow, N - 1);
+          const ushort* gp = A + (long)grow * K + k0 + kb;
+          __builtin_amdgcn_global_load_lds(
+              (const __attribute__((address_space(1))) u32*)gp,
+              (__attribute__((address_space(3))) u32*)(lds_a + r16 * 16 * GEMM_BK),
+              16, 0, 0);
+        }
+        {
+          const int grow = min(col0 + lrow, M - 1);
+          const ushort* gp = Bt + (long)grow * K + k0 + kb;
+          __builtin_amdgcn_global_load_lds(
+              (const __attribute__((address_space(1))) u32*)gp,
+              (__attribute__((address_space(3))) u32*)(lds_b + r16 * 16 * GEMM_BK),
+              16, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- fragments + MFMA. a-frag: row (lane&15) of the 16-row block,
+    // k = (lane>>4)*8..+8 (contiguous 16 B -> ds_read_b128).
+    bf16x8 afrag[4], bfrag[4];
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+      const int r = wrow + mi * 16 + (lane & 15);
+      afrag[mi] = *reinterpret_cast<const bf16x8*>(
+          lds_a + r * GEMM_BK + (lane >> 4) * 8);
+    }
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      const int r = wcol + ni * 16 + (lane & 15);
+      bfrag[ni] = *reinterpret_cast<const bf16x8*>(
+          lds_b + r * GEMM_BK + (lane >> 4) * 8);
+    }
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
+    __syncthreads();
+  }
+
+  // ---- epilogue: C/D layout col=lane&15, row=(lane>>4)*4+reg (guide §3)
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      const int col = col0 + wcol + ni * 16 + (lane & 15);
+      if (col >= M) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = row0 + wrow + mi * 16 + (lane >> 4) * 4 + r;
+        if (row >= N) continue;
+        float v = acc[mi][ni][r];
+        if (TANH) v = tanhf(v);
+        C[(long)row * M + col] = f2bf(v);
+      }
+    }
+  }
+}
+
+// elementwise tanh backward: d_z = d_y * (1 - y^2), bf16
+__global__ void k_tanh_bwd_mul(const ushort* __restrict__ dy,
+                               const ushort* __restrict__ y,
+                               ushort* __restrict__ dz, long total) {
+  for (long i8 = blockIdx.x * blockDim.x + threadIdx.x; i8 * 8 < total;
+       i8 += (long)gridDim.x * blockDim.x) {
+    const long e = i8 * 8;
+    ulonglong2 pd = *reinterpret_cast<const ulonglong2*>(dy + e);
+    ulonglong2 py = *reinterpret_cast<const ulonglong2*>(y + e);
+    ushort* ud = reinterpret_cast<ushort*>(&pd);
+    ushort* uy = reinterpret_cast<ushort*>(&py);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float yv = bf2f(uy[j]);
+      ud[j] = f2bf(bf2f(ud[j]) * (1.0f - yv * yv));
+    }
+    *reinterpret_cast<ulonglong2*>(dz + e) = pd;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K5-K7: fused masked-softmax attention (one workgroup per example).
+// LDS holds the example's whole (C,D) bf16 tile + the attention vector +
+// per-context scores (java14m: 200x384x2B = 150 KiB).
+//   scores_c = comb_c . a + log(mask_c);  alpha = softmax_c; code = sum alpha*comb
+// All-masked rows produce zeros (reference would NaN).
+// ---------------------------------------------------------------------------
+
+extern __shared__ unsigned char smem[];
+
+__launch_bounds__(256)
+__global__ void k_attn_fwd(const ushort* __restrict__ comb,
+                           const float* __restrict__ a,
+                           const float* __restrict__ mask,
+                           float* __restrict__ code,
+                           float* __restrict__ alpha_out, int B, int C, int D) {
+  ushort* tile = reinterpret_cast<ushort*>(smem);                 // C*D bf16
+  float* sc = reinterpret_cast<float*>(smem + (size_t)C * D * 2); // C f32
+  float* av = sc + C;                                             // D f32
+  __shared__ float red[4];
+
+  const int b = blockIdx.x;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const ushort* src = comb + (long)b * C * D;
+
+  // stage tile (vectorized 8-wide) + attention vector
+  const int total8 = (C * D) / 8;
+  for (int s = tid; s < total8; s += blockDim.x)
+    *reinterpret_cast<ulonglong2*>(tile + (long)s * 8) =
+        *reinterpret_cast<const ulonglong2*>(src + (long)s * 8);
+  for (int i = tid; i < D; i += blockDim.x) av[i] = a[i];
+  __syncthreads();
+
+  // scores: each wave takes contexts strided by 4; lane covers D/64 elems
+  const int per_lane = D / 64;
+  for (int c = wid; c < C; c += 4) {
+    float dot = 0.f;
+    const ushort* rowp = tile + (long)c * D + lane * per_lane;
+#pragma unroll 8
+    for (int j = 0; j < per_lane; ++j) dot += bf2f(rowp[j]) * av[lane * per_lane + j];
+    dot = wave_reduce_sum(dot);
+    if (lane == 0)
+      sc[c] = (mask[(long)b * C + c] > 0.f) ? dot : -3.0e38f;
+  }
+  __syncthreads();
+
+  // softmax over C (one wave handles the reduction; C <= a few thousand)
+  if (wid == 0) {
+    float m = -3.0e38f;
+    for (int c = lane; c < C; c += 64) m = fmaxf(m, sc[c]);
+    m = wave_reduce_max(m);
+    m = __shfl(m, 0, 64);
+    const bool any_valid = m > -1.0e38f;
+    float s = 0.f;
+    for (int c = lane; c < C; c += 64) {
+      const float e = any_valid ? __expf(sc[c] - m) : 0.f;
+      const float e2 = (sc[c] > -1.0e38f) ? e : 0.f;
+      sc[c] = e2;
+      s += e2;
+    }
+    s = wave_reduce_sum(s);
+    s = __shfl(s, 0, 64);
+    const float inv = (s > 0.f) ? 1.0f / s : 0.f;
+    for (int c = lane; c < C; c += 64) {
+      sc[c] *= inv;
+      alpha_out[(long)b * C + c] = sc[c];
+    }
+  }
+  __syncthreads();
+
+  // weighted sum: thread t accumulates output columns t, t+256, ...
+  for (int col = tid; col < D; col += blockDim.x) {
+    float acc = 0.f;
+    for (int c = 0; c < C; ++c) acc += sc[c] * bf2f(tile[(long)c * D + col]);
+    code[(long)b * D + col] = acc;
+  }
+  (void)red;
+}
+
+__launch_bounds__(256)
+__global__ void k_attn_bwd(const ushort* __restrict__ comb,
+                           const float* __restrict__ a,
+                           const float* __restrict__ alpha,
+                           const float* __restrict__ d_code,
+                           ushort* __restrict__ d_comb,
+                           float* __restrict__ d_a_partial,  // (grid,D)
+                           int B, int C, int D) {
+  ushort* tile = reinterpret_cast<ushort*>(smem);                 // C*D bf16
+  float* dal = reinterpret_cast<float*>(smem + (size_t)C * D * 2);// C f32: d_alpha then d_e
+  float* av = dal + C;                                            // D f32
+  float* dv = av + D;                                             // D f32
+  __shared__ float s_inner;
+
+  const int b = blockIdx.x;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const ushort* src = comb + (long)b * C * D;
+
+  const int total8 = (C * D) / 8;
+  for (int s = tid; s < total8; s += blockDim.x)
+    *reinterpret_cast<ulonglong2*>(tile + (long)s * 8) =
+        *reinterpret_cast<const ulonglong2*>(src + (long)s * 8);
+  for (int i = tid; i < D; i += blockDim.x) {
+    av[i] = a[i];
+    dv[i] = d_code[(long)b * D + i];
+  }
+  __syncthreads();
+
+  // d_alpha_c = comb_c . dv
+  const int per_lane = D / 64;
+  for (int c = wid; c < C; c += 4) {
+    float dot = 0.f;
+    const ushort* rowp = tile + (long)c * D + lane * per_lane;
+#pragma unroll 8
+    for (int j = 0; j < per_lane; ++j) dot += bf2f(rowp[j]) * dv[lane * per_lane + j];
+    dot = wave_reduce_sum(dot);
+    if (lane == 0) dal[c] = dot;
+  }
+  __syncthreads();
+
+  // inner = sum alpha*d_alpha ; d_e_c = alpha_c (d_alpha_c - inner)
+  if (wid == 0) {
+    float inner = 0.f;
+    for (int c = lane; c < C; c += 64)
+      inner += alpha[(long)b * C + c] * dal[c];
+    inner = wave_reduce_sum(inner);
+    if (lane == 0) s_inner = inner;
+  }
+  __syncthreads();
+  const float inner = s_inner;
+  for (int c = tid; c < C; c += blockDim.x)
+    dal[c] = alpha[(long)b * C + c] * (dal[c] - inner);
+  __syncthreads();
+
+  // d_comb_c = alpha_c * dv + d_e_c * a   (write bf16)
+  for (int s = tid; s < C * D; s += blockDim.x) {
+    const int c = s / D, col = s % D;
+    const float al = alpha[(long)b * C + c];
+    d_comb[(long)b * C * D + s] = f2bf(al * dv[col] + dal[c] * av[col]);
+  }
+
+  // d_a partial: per-block row in d_a_partial
+  for (int col = tid; col < D; col += blockDim.x) {
+    float acc = 0.f;
+    for (int c = 0; c < C; ++c) acc += dal[c] * bf2f(tile[(long)c * D + col]);
+    d_a_partial[(long)blockIdx.x * D + col] = acc;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K9: fused CE over the target vocab. One workgroup per row.
+// fwd: online max/sum-exp -> (loss, lse). bwd: d = scale*(softmax - onehot).
+// ---------------------------------------------------------------------------
+
+__launch_bounds__(256)
+__global__ void k_ce_fwd(const ushort* __restrict__ logits,
+                         const long* __restrict__ labels,
+                         float* __restrict__ loss, float* __restrict__ lse,
+                         int B, int V) {
+  __shared__ float sm[4], ss[4];
+  const int b = blockIdx.x;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const ushort* row = logits + (long)b * V;
+
+  float m = -3.0e38f, s = 0.f;
+  const int v8 = V / 8;
+  for (int i = tid; i < v8; i += blockDim.x) {
+    ulonglong2 packed = *reinterpret_cast<const ulonglong2*>(row + (long)i * 8);
+    const ushort* u = reinterpret_cast<const ushort*>(&packed);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float x = bf2f(u[j]);
+      if (x > m) { s = s * __expf(m - x) + 1.f; m = x; }
+      else s += __expf(x - m);
+    }
+  }
+  for (int i = v8 * 8 + tid; i < V; i += blockDim.x) {
+    const float x = bf2f(row[i]);
+    if (x > m) { s = s * __expf(m - x) + 1.f; m = x; }
+    else s += __expf(x - m);
+  }
+  // reduce (m, s) across lanes then waves
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    const float mo = __shfl_down(m, off, 64);
+    const float so = __shfl_down(s, off, 64);
+    if (mo > m) { s = s * __expf(m - mo) + so; m = mo; }
+    else s += so * __expf(mo - m);
+  }
+  if (lane == 0) { sm[wid] = m; ss[wid] = s; }
+  __syncthreads();
+  if (tid == 0) {
+    float M = sm[0], S = ss[0];
+#pragma unroll
+    for (int w = 1; w < 4; ++w) {
+      if (sm[w] > M) { S = S * __expf(M - sm[w]) + ss[w]; M = sm[w]; }
+      else S += ss[w] * __expf(sm[w] - M);
+    }
+    const float l = M + __logf(S);
+    lse[b] = l;
+    loss[b] = l - bf2f(row[labels[b]]);
+  }
+}
+
+__global__ void k_ce_bwd(const ushort* __restrict__ logits,
+                         const float* __restrict__ lse,
+                         const long* __restrict__ labels,
+                         ushort* __restrict__ d_logits, float scale, int B,
+                         int V) {
+  const int b = blockIdx.y;
+  const float l = lse[b];
+  const long lab = labels[b];
+  const ushort* row = logits + (long)b * V;
+  ushort* drow = d_logits + (long)b * V;
+  const int v8 = V / 8;
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < v8;
+       i += gridDim.x * blockDim.x) {
+    ulonglong2 packed = *reinterpret_cast<const ulonglong2*>(row + (long)i * 8);
+    ushort* u = reinterpret_cast<ushort*>(&packed);
+    const long e0 = (long)i * 8;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float p = __expf(bf2f(u[j]) - l);
+      if (e0 + j == lab) p -= 1.f;
+      u[j] = f2bf(p * scale);
+    }
+    *reinterpret_cast<ulonglong2*>(drow + e0) = packed;
+  }
+  if (blockIdx.x == 0 && threadIdx.x < (V - v8 * 8)) {
+    const long e = (long)v8 * 8 + threadIdx.x;
+    float p = __expf(bf2f(row[e]) - l);
+    if (e == lab) p -= 1.f;
+    drow[e] = f2bf(p * scale);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K10: Adam, TF AdamOptimizer formulation:
+//   m <- b1*m + (1-b1)*g ; v <- b2*v + (1-b2)*g^2
+//   lr_t = lr*sqrt(1-b2^t)/(1-b1^t) ; p -= lr_t * m/(sqrt(v)+eps)
+// Dense kernel optionally refreshes a bf16 shadow. Sparse path: atomic
+// row-accumulate into a compact fp32 buffer (dedup'd ids), then per-row
+// lazy update of p/m/v.
+// ---------------------------------------------------------------------------
+
+template <typename G>
+__global__ void k_adam_dense(float* __restrict__ p, const G* __restrict__ g,
+                             float* __restrict__ m, float* __restrict__ v,
+                             ushort* __restrict__ shadow, long n, float lr_t,
+                             float b1, float b2, float eps) {
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    float gv;
+    if constexpr (sizeof(G) == 2) gv = bf2f(((const ushort*)g)[i]);
+    else gv = ((const float*)g)[i];
+    const float mi = b1 * m[i] + (1.f - b1) * gv;
+    const float vi = b2 * v[i] + (1.f - b2) * gv * gv;
+    m[i] = mi;
+    v[i] = vi;
+    const float pv = p[i] - lr_t * mi / (sqrtf(vi) + eps);
+    p[i] = pv;
+    if (shadow != nullptr) shadow[i] = f2bf(pv);
+  }
+}
+
+template <typename G>
+__global__ void k_rows_accum(const G* __restrict__ rows,
+                             const int* __restrict__ inverse,
+                             float* __restrict__ acc, long n_rows, int d) {
+  for (long s = blockIdx.x * blockDim.x + threadIdx.x; s < n_rows * d;
+       s += (long)gridDim.x * blockDim.x) {
+    const long r = s / d;
+    const int col = (int)(s % d);
+    float gv;
+    if constexpr (sizeof(G) == 2) gv = bf2f(((const ushort*)rows)[s]);
+    else gv = ((const float*)rows)[s];
+    if (gv != 0.f)
+      atomicAdd(acc + (long)inverse[r] * d + col, gv);
+  }
+}
+
+__global__ void k_adam_rows(float* __restrict__ p, const long* __restrict__ ids,
+                            const float* __restrict__ acc, float* __restrict__ m,
+                            float* __restrict__ v, long n_uniq, int d,
+                            float lr_t, float b1, float b2, float eps) {
+  for (long s = blockIdx.x * blockDim.x + threadIdx.x; s < n_uniq * d;
+       s += (long)gridDim.x * blockDim.x) {
+    const long u = s / d;
+    const int col = (int)(s % d);
+    const long off = ids[u] * d + col;
+    const float gv = acc[s];
+    const float mi = b1 * m[off] + (1.f - b1) * gv;
+    const float vi = b2 * v[off] + (1.f - b2) * gv * gv;
+    m[off] = mi;
+    v[off] = vi;
+    p[off] -= lr_t * mi / (sqrtf(vi) + eps);
+  }
+}
+
+// ===========================================================================
+// host wrappers
+// ===========================================================================
+
+namespace {
+
+inline hipStream_t cur_stream() {
+  return at::cuda::getCurrentCUDAStream().stream();
+}
+
+inline int grid_1d(long total, int block, int cap = 2048) {
+  long g = (total + block - 1) / block;
+  return (int)std::min<long>(g, cap);
+}
+
+#define CHECK_DEV(t) TORCH_CHECK((t).is_cuda(), #t " must be on GPU")
+#define CHECK_CONT(t) TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+
+const ushort* bf_ptr(const torch::Tensor& t) {
+  return reinterpret_cast<const ushort*>(t.data_ptr<at::BFloat16>());
+}
+ushort* bf_ptr_mut(torch::Tensor& t) {
+  return reinterpret_cast<ushort*>(t.data_ptr<at::BFloat16>());
+}
+
+torch::Tensor gather_concat_fwd(torch::Tensor tok, torch::Tensor path,
+                                torch::Tensor src, torch::Tensor pth,
+                                torch::Tensor tgt, double keep_prob,
+                                int64_t seed, bool training) {
+  CHECK_DEV(tok); CHECK_CONT(tok); CHECK_DEV(path); CHECK_CONT(path);
+  CHECK_DEV(src); CHECK_CONT(src);
+  TORCH_CHECK(src.scalar_type() == torch::kInt32, "ids must be int32");
+  const int d = tok.size(1);
+  TORCH_CHECK(d % 8 == 0, "embedding dim must be a multiple of 8");
+  const long n_rows = src.numel();
+  auto out = torch::empty({n_rows, 3 * (long)d},
+                          tok.options().dtype(torch::kBFloat16));
+  const bool drop = training && keep_prob < 1.0;
+  const long slots = n_rows * (3 * d / 8);
+  k_gather_concat<<<grid_1d(slots, 256), 256, 0, cur_stream()>>>(
+      tok.data_ptr<float>(), path.data_ptr<float>(), src.data_ptr<int>(),
+      pth.data_ptr<int>(), tgt.data_ptr<int>(), bf_ptr_mut(out), (int)n_rows,
+      d, (float)keep_prob, (u64)seed, drop ? 1 : 0);
+  return out;
+}
+
+torch::Tensor gather_concat_bwd(torch::Tensor d_ctx, double keep_prob,
+                                int64_t seed, bool training) {
+  CHECK_DEV(d_ctx); CHECK_CONT(d_ctx);
+  if (!(training && keep_prob < 1.0)) return d_ctx;
+  auto out = torch::empty_like(d_ctx);
+  const long total = d_ctx.numel();
+  TORCH_CHECK(total % 8 == 0);
+  k_dropout_bwd<<<grid_1d(total / 8, 256), 256, 0, cur_stream()>>>(
+      bf_ptr(d_ctx), bf_ptr_mut(out), total, (float)keep_prob, (u64)seed);
+  return out;
+}
+
+torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor Bt, bool tanh_epilogue) {
+  CHECK_DEV(A); CHECK_CONT(A); CHECK_DEV(Bt); CHECK_CONT(Bt);
+  TORCH_CHECK(A.scalar_type() == torch::kBFloat16 &&
+              Bt.scalar_type() == torch::kBFloat16);
+  const int N = A.size(0), K = A.size(1), M = Bt.size(0);
+  TORCH_CHECK(Bt.size(1) == K, "K mismatch");
+  TORCH_CHECK(K % GEMM_BK == 0, "K must be a multiple of 32");
+  auto C = torch::empty({N, M}, A.options());
+  const int n_tiles = (N + GEMM_BM - 1) / GEMM_BM;
+  const int m_tiles = (M + GEMM_BN - 1) / GEMM_BN;
+  dim3 grid(n_tiles * m_tiles);
+  if (tanh_epilogue)
+    k_gemm_bt<true><<<grid, 256, 0, cur_stream()>>>(
+        bf_ptr(A), bf_ptr(Bt), bf_ptr_mut(C), N, M, K);
+  else
+    k_gemm_bt<false><<<grid, 256, 0, cur_stream()>>>(
+        bf_ptr(A), bf_ptr(Bt), bf_ptr_mut(C), N, M, K);
+  return C;
+}
+
+torch::Tensor transform_tanh_fwd(torch::Tensor ctx, torch::Tensor w_oi) {
+  return gemm_bt(ctx, w_oi, true);
+}
+
+torch::Tensor gemm_bt_bf16(torch::Tensor A, torch::Tensor Bt) {
+  return gemm_bt(A, Bt, false);
+}
+
+torch::Tensor tanh_bwd_mul(torch::Tensor dy, torch::Tensor y) {
+  CHECK_DEV(dy); CHECK_CONT(dy); CHECK_DEV(y); CHECK_CONT(y);
+  auto dz = torch::empty_like(dy);
+  const long total = dy.numel();
+  TORCH_CHECK(total % 8 == 0);
+  k_tanh_bwd_mul<<<grid_1d(total / 8, 256), 256, 0, cur_stream()>>>(
+      bf_ptr(dy), bf_ptr(y), bf_ptr_mut(dz), total);
+  return dz;
+}
+
+static void ensure_lds_limit(const void* fn, size_t bytes) {
+  static size_t configured = 0;
+  if (bytes > 65536 && bytes > configured) {
+    hipFuncSetAttribute(fn, hipFuncAttributeMaxDynamicSharedMemorySize,
+                        (int)bytes);
+    configured = bytes;
+  }
+}
+
+std::vector<torch::Tensor> attention_fwd(torch::Tensor comb, torch::Tensor a,
+                                         torch::Tensor mask) {
+  CHECK_DEV(comb); CHECK_CONT(comb); CHECK_DEV(a); CHECK_DEV(mask);
+  TORCH_CHECK(comb.dim() == 3);
+  const int B = comb.size(0), C = comb.size(1), D = comb.size(2);
+  TORCH_CHECK(D % 64 == 0, "code vector size must be a multiple of 64");
+  auto a32 = a.to(torch::kFloat32).contiguous();
+  auto mask32 = mask.to(torch::kFloat32).contiguous();
+  auto code = torch::empty({B, D}, comb.options().dtype(torch::kFloat32));
+  auto alpha = torch::empty({B, C}, comb.options().dtype(torch::kFloat32));
+  const size_t lds = (size_t)C * D * 2 + (size_t)C * 4 + (size_t)D * 4;
+  TORCH_CHECK(lds <= 160 * 1024, "attention tile exceeds LDS capacity");
+  ensure_lds_limit((const void*)k_attn_fwd, lds);
+  k_attn_fwd<<<B, 256, lds, cur_stream()>>>(
+      bf_ptr(comb), a32.data_ptr<float>(), mask32.data_ptr<float>(),
+      code.data_ptr<float>(), alpha.data_ptr<float>(), B, C, D);
+  return {code, alpha};
+}
+
+std::vector<torch::Tensor> attention_bwd(torch::Tensor comb, torch::Tensor a,
+                                         torch::Tensor alpha,
+                                         torch::Tensor d_code) {
+  CHECK_DEV(comb); CHECK_CONT(comb);
+  const int B = comb.size(0), C = comb.size(1), D = comb.size(2);
+  TORCH_CHECK(D % 64 == 0);
+  auto a32 = a.to(torch::kFloat32).contiguous();
+  auto alpha32 = alpha.contiguous();
+  auto dcode32 = d_code.to(torch::kFloat32).contiguous();
+  auto d_comb = torch::empty_like(comb);
+  auto d_a_partial = torch::empty({B, D}, comb.options().dtype(torch::kFloat32));
+  const size_t lds = (size_t)C * D * 2 + (size_t)C * 4 + (size_t)D * 8;
+  TORCH_CHECK(lds <= 160 * 1024, "attention tile exceeds LDS capacity");
+  ensure_lds_limit((const void*)k_attn_bwd, lds);
+  k_attn_bwd<<<B, 256, lds, cur_stream()>>>(
+      bf_ptr(comb), a32.data_ptr<float>(), alpha32.data_ptr<float>(),
+      dcode32.data_ptr<float>(), bf_ptr_mut(d_comb),
+      d_a_partial.data_ptr<float>(), B, C, D);
+  auto d_a = d_a_partial.sum(0);
+  return {d_comb, d_a};
+}
+
+std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor labels) {
+  CHECK_DEV(logits); CHECK_CONT(logits);
+  TORCH_CHECK(labels.scalar_type() == torch::kInt64);
+  const int B = logits.size(0), V = logits.size(1);
+  auto labels_c = labels.contiguous();
+  auto loss = torch::empty({B}, logits.options().dtype(torch::kFloat32));
+  auto lse = torch::empty({B}, logits.options().dtype(torch::kFloat32));
+  k_ce_fwd<<<B, 256, 0, cur_stream()>>>(
+      bf_ptr(logits), labels_c.data_ptr<long>(), loss.data_ptr<float>(),
+      lse.data_ptr<float>(), B, V);
+  return {loss, lse};
+}
+
+torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor lse,
+                     torch::Tensor labels, double scale) {
+  CHECK_DEV(logits); CHECK_CONT(logits);
+  const int B = logits.size(0), V = logits.size(1);
+  auto labels_c = labels.contiguous();
+  auto d = torch::empty_like(logits);
+  dim3 grid(grid_1d(V / 8, 256, 64), B);
+  k_ce_bwd<<<grid, 256, 0, cur_stream()>>>(
+      bf_ptr(logits), lse.data_ptr<float>(), labels_c.data_ptr<long>(),
+      bf_ptr_mut(d), (float)scale, B, V);
+  return d;
+}
+
+void adam_dense_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                     torch::Tensor v, int64_t step, double lr, double beta1,
+                     double beta2, double eps, torch::Tensor shadow) {
+  CHECK_DEV(p); CHECK_CONT(p);
+  auto g_c = g.contiguous();
+  const long n = p.numel();
+  TORCH_CHECK(g_c.numel() == n && m.numel() == n && v.numel() == n);
+  const float lr_t = (float)(lr * std::sqrt(1.0 - std::pow(beta2, (double)step)) /
+                             (1.0 - std::pow(beta1, (double)step)));
+  ushort* shadow_ptr = nullptr;
+  if (shadow.defined() && shadow.numel() == n)
+    shadow_ptr = reinterpret_cast<ushort*>(shadow.data_ptr<at::BFloat16>());
+  const int grid = grid_1d(n, 256);
+  if (g_c.scalar_type() == torch::kBFloat16)
+    k_adam_dense<ushort><<<grid, 256, 0, cur_stream()>>>(
+        p.data_ptr<float>(), bf_ptr(g_c), m.data_ptr<float>(),
+        v.data_ptr<float>(), shadow_ptr, n, lr_t, (float)beta1, (float)beta2,
+        (float)eps);
+  else
+    k_adam_dense<float><<<grid, 256, 0, cur_stream()>>>(
+        p.data_ptr<float>(), g_c.data_ptr<float>(), m.data_ptr<float>(),
+        v.data_ptr<float>(), shadow_ptr, n, lr_t, (float)beta1, (float)beta2,
+        (float)eps);
+}
+
+void adam_sparse_rows_step(torch::Tensor p, torch::Tensor uniq_ids,
+                           torch::Tensor inverse, torch::Tensor grad_rows,
+                           torch::Tensor m, torch::Tensor v, int64_t step,
+                           double lr, double beta1, double beta2, double eps) {
+  CHECK_DEV(p); CHECK_CONT(p);
+  auto rows_c = grad_rows.contiguous();
+  auto inv_c = inverse.contiguous();
+  auto ids_c = uniq_ids.contiguous();
+  const long n_rows = rows_c.size(0);
+  const int d = rows_c.size(1);
+  const long n_uniq = ids_c.numel();
+  auto acc = torch::zeros({n_uniq, (long)d},
+                          p.options().dtype(torch::kFloat32));
+  const float lr_t = (float)(lr * std::sqrt(1.0 - std::pow(beta2, (double)step)) /
+                             (1.0 - std::pow(beta1, (double)step)));
+  if (rows_c.scalar_type() == torch::kBFloat16)
+    k_rows_accum<ushort><<<grid_1d(n_rows * d, 256), 256, 0, cur_stream()>>>(
+        bf_ptr(rows_c), inv_c.data_ptr<int>(), acc.data_ptr<float>(), n_rows, d);
+  else
+    k_rows_accum<float><<<grid_1d(n_rows * d, 256), 256, 0, cur_stream()>>>(
+        rows_c.data_ptr<float>(), inv_c.data_ptr<int>(), acc.data_ptr<float>(),
+        n_rows, d);
+  k_adam_rows<<<grid_1d(n_uniq * d, 256), 256, 0, cur_stream()>>>(
+      p.data_ptr<float>(), ids_c.data_ptr<long>(), acc.data_ptr<float>(),
+      m.data_ptr<float>(), v.data_ptr<float>(), n_uniq, d, lr_t, (float)beta1,
+      (float)beta2, (float)eps);
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("gather_concat_fwd", &gather_concat_fwd);
+  mod.def("gather_concat_bwd", &gather_concat_bwd);
+  mod.def("transform_tanh_fwd", &transform_tanh_fwd);
+  mod.def("gemm_bt_bf16", &gemm_bt_bf16);
+  mod.def("tanh_bwd_mul", &tanh_bwd_mul);
+  mod.def("attention_fwd", &attention_fwd);
+  mod.def("attention_bwd", &attention_bwd);
+  mod.def("ce_fwd", &ce_fwd);
+  mod.def("ce_bwd", &ce_bwd);
+  mod.def("adam_dense_step", &adam_dense_step);
+  mod.def("adam_sparse_rows_step", &adam_sparse_rows_step);
+}
