@@ -1,0 +1,147 @@
+// Native .c2v line parser — the C++ core of the data runtime.
+//
+// The reference's input pipeline is tf.data's C++ CsvDataset + string_split +
+// static hash tables (path_context_reader.py:119-228); this is the
+// MI355X-native equivalent: one C++ parser object holding the three
+// string->index hash maps, parsing batches of lines into int32/float32
+// torch tensors with a std::thread pool. Semantics identical to the Python
+// reader (code2vec_amd/data/reader.py), which remains the fallback and the
+// test oracle.
+
+#include <torch/extension.h>
+
+#include <atomic>
+#include <cstring>
+#include <string>
+#include <string_view>
+#include <thread>
+#include <unordered_map>
+#include <vector>
+
+namespace {
+
+struct Parser {
+  std::unordered_map<std::string, int> tok, path, tgt;
+  int tok_pad = 0, tok_oov = 0, path_pad = 0, path_oov = 0, tgt_oov = 0;
+  int max_contexts = 200;
+  int n_threads = 6;  // reference READER_NUM_PARALLEL_BATCHES default
+
+  Parser(const std::unordered_map<std::string, int>& tok_map,
+         const std::unordered_map<std::string, int>& path_map,
+         const std::unordered_map<std::string, int>& tgt_map, int tok_pad_,
+         int tok_oov_, int path_pad_, int path_oov_, int tgt_oov_,
+         int max_contexts_, int n_threads_)
+      : tok(tok_map), path(path_map), tgt(tgt_map), tok_pad(tok_pad_),
+        tok_oov(tok_oov_), path_pad(path_pad_), path_oov(path_oov_),
+        tgt_oov(tgt_oov_), max_contexts(max_contexts_),
+        n_threads(std::max(1, n_threads_)) {}
+
+  inline int look(const std::unordered_map<std::string, int>& m,
+                  std::string_view s, int oov) const {
+    auto it = m.find(std::string(s));
+    return it == m.end() ? oov : it->second;
+  }
+
+  // Parse one line into the row buffers. Returns target index.
+  int parse_line(std::string_view line, int* src_row, int* path_row,
+                 int* tgt_row, float* mask_row) const {
+    const int C = max_contexts;
+    for (int c = 0; c < C; ++c) {
+      src_row[c] = tok_pad;
+      path_row[c] = path_pad;
+      tgt_row[c] = tok_pad;
+      mask_row[c] = 0.f;
+    }
+    // strip trailing newline
+    while (!line.empty() && (line.back() == '\n' || line.back() == '\r'))
+      line.remove_suffix(1);
+
+    size_t pos = line.find(' ');
+    std::string_view target = line.substr(0, pos == std::string_view::npos
+                                                 ? line.size() : pos);
+    int target_idx = target.empty() ? tgt_oov : look(tgt, target, tgt_oov);
+
+    int c = 0;
+    while (pos != std::string_view::npos && c < C) {
+      size_t start = pos + 1;
+      pos = line.find(' ', start);
+      std::string_view field = line.substr(
+          start, (pos == std::string_view::npos ? line.size() : pos) - start);
+      if (field.empty()) { ++c; continue; }
+      // split on ',' into up to 3 parts; missing parts stay PAD
+      size_t c1 = field.find(',');
+      size_t c2 = c1 == std::string_view::npos ? std::string_view::npos
+                                               : field.find(',', c1 + 1);
+      std::string_view p0 = field.substr(0, c1);
+      std::string_view p1 = c1 == std::string_view::npos
+                                ? std::string_view()
+                                : field.substr(c1 + 1,
+                                               (c2 == std::string_view::npos
+                                                    ? field.size()
+                                                    : c2) - c1 - 1);
+      std::string_view p2 = c2 == std::string_view::npos
+                                ? std::string_view()
+                                : field.substr(c2 + 1);
+      int si = p0.empty() ? tok_pad : look(tok, p0, tok_oov);
+      int pi = p1.empty() ? path_pad : look(path, p1, path_oov);
+      int ti = p2.empty() ? tok_pad : look(tok, p2, tok_oov);
+      src_row[c] = si;
+      path_row[c] = pi;
+      tgt_row[c] = ti;
+      mask_row[c] = (si != tok_pad || ti != tok_pad || pi != path_pad) ? 1.f : 0.f;
+      ++c;
+    }
+    return target_idx;
+  }
+
+  // Parse a batch of lines (thread-parallel across rows).
+  py::tuple parse_batch(const std::vector<std::string>& lines) const {
+    const int64_t B = (int64_t)lines.size();
+    const int64_t C = max_contexts;
+    auto opts = torch::TensorOptions().dtype(torch::kInt32);
+    auto src = torch::empty({B, C}, opts);
+    auto pth = torch::empty({B, C}, opts);
+    auto tgt = torch::empty({B, C}, opts);
+    auto mask = torch::empty({B, C}, torch::TensorOptions().dtype(torch::kFloat32));
+    auto tidx = torch::empty({B}, torch::TensorOptions().dtype(torch::kInt64));
+
+    int* src_p = src.data_ptr<int>();
+    int* pth_p = pth.data_ptr<int>();
+    int* tgt_p = tgt.data_ptr<int>();
+    float* mask_p = mask.data_ptr<float>();
+    int64_t* tidx_p = tidx.data_ptr<int64_t>();
+
+    {
+      py::gil_scoped_release release;  // parsing is pure C++
+      const int nt = (int)std::min<int64_t>(n_threads, std::max<int64_t>(1, B));
+      std::atomic<int64_t> next(0);
+      auto work = [&]() {
+        int64_t i;
+        while ((i = next.fetch_add(1)) < B) {
+          tidx_p[i] = parse_line(lines[i], src_p + i * C, pth_p + i * C,
+                                 tgt_p + i * C, mask_p + i * C);
+        }
+      };
+      if (nt <= 1) {
+        work();
+      } else {
+        std::vector<std::thread> pool;
+        pool.reserve(nt);
+        for (int t = 0; t < nt; ++t) pool.emplace_back(work);
+        for (auto& th : pool) th.join();
+      }
+    }
+    return py::make_tuple(src, pth, tgt, mask, tidx);
+  }
+};
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  py::class_<Parser>(m, "Parser")
+      .def(py::init<const std::unordered_map<std::string, int>&,
+                    const std::unordered_map<std::string, int>&,
+                    const std::unordered_map<std::string, int>&, int, int, int,
+                    int, int, int, int>())
+      .def("parse_batch", &Parser::parse_batch);
+}

@@ -79,17 +79,48 @@ class ReaderBatch(NamedTuple):
             else self.target_index.to(device, non_blocking=non_blocking))
 
 
+_NATIVE_READER_MOD = None
+_NATIVE_READER_TRIED = False
+
+
+def _load_native_reader_module():
+    """Load data/_c2v_reader.so exactly once per process (re-executing a
+    pybind11 extension init in the same process crashes)."""
+    global _NATIVE_READER_MOD, _NATIVE_READER_TRIED
+    if _NATIVE_READER_TRIED:
+        return _NATIVE_READER_MOD
+    _NATIVE_READER_TRIED = True
+    try:
+        import importlib.util
+        here = os.path.dirname(os.path.abspath(__file__))
+        for cand in (os.path.join(here, '_c2v_reader.so'),
+                     os.path.join(here, '_build', '_c2v_reader.so')):
+            if os.path.isfile(cand):
+                spec = importlib.util.spec_from_file_location(
+                    'code2vec_amd.data._c2v_reader', cand)
+                mod = importlib.util.module_from_spec(spec)
+                spec.loader.exec_module(mod)
+                _NATIVE_READER_MOD = mod
+                break
+    except Exception:  # noqa: BLE001
+        _NATIVE_READER_MOD = None
+    return _NATIVE_READER_MOD
+
+
 class PathContextReader:
     def __init__(self, vocabs: Code2VecVocabs, config: Config,
                  estimator_action: EstimatorAction,
                  repeat_endlessly: bool = False, keep_strings: bool = False,
-                 world_size: int = 1, rank: int = 0):
+                 world_size: int = 1, rank: int = 0, use_native: bool = True):
         self.vocabs = vocabs
         self.config = config
         self.estimator_action = estimator_action
         self.repeat_endlessly = repeat_endlessly
-        # strings are needed for evaluate (metrics/log) and predict (attention display)
+        # context strings are needed for predict (attention display); evaluate
+        # needs only the target strings (metrics + log.txt), which the native
+        # path extracts in Python
         self.keep_strings = keep_strings or estimator_action.is_evaluate_or_predict
+        self.need_target_strings = estimator_action.is_evaluate_or_predict
         self.world_size = world_size
         self.rank = rank
 
@@ -106,6 +137,26 @@ class PathContextReader:
         self._tgt_oov = tgt.oov_index
         self._tgt_oov_word = tgt.special_words.OOV
         self._tok_pad_word = tok.special_words.PAD
+        self._native = None
+        native_ok = use_native and (not keep_strings) and \
+            not estimator_action.is_predict
+        if native_ok:
+            self._native = self._try_native_parser()
+
+    def _try_native_parser(self):
+        """Multithreaded C++ parser (data/csrc/c2v_reader.cpp) — the native
+        core of the pipeline; the Python path below is the fallback oracle."""
+        mod = _load_native_reader_module()
+        if mod is None:
+            return None
+        try:
+            return mod.Parser(
+                self._tok_w2i, self._pth_w2i, self._tgt_w2i,
+                self._tok_pad, self._tok_oov, self._pth_pad,
+                self._pth_oov, self._tgt_oov, self.config.MAX_CONTEXTS,
+                self.config.READER_NUM_PARALLEL_BATCHES)
+        except Exception:  # noqa: BLE001  (fallback to the Python parser)
+            return None
 
     # ---- per-row parsing ----
 
@@ -205,6 +256,10 @@ class PathContextReader:
         batch_size = 1 if self.estimator_action.is_predict else \
             self.config.batch_size(is_evaluating=self.estimator_action.is_evaluate)
 
+        if self._native is not None and not self.estimator_action.is_predict:
+            yield from self._iter_batches_native(lines, batch_size)
+            return
+
         rows = []
         for i, line in enumerate(lines):
             if self.world_size > 1 and (i % self.world_size) != self.rank:
@@ -220,6 +275,84 @@ class PathContextReader:
                 rows = []
         if rows:
             yield self._collate(rows)
+
+    def _iter_batches_native(self, lines: Iterable[str],
+                             batch_size: int) -> Iterator[ReaderBatch]:
+        """C++ parser path: parse line chunks in parallel, filter rows with
+        tensor ops, accumulate and emit exact-size batches."""
+        pin = torch.cuda.is_available()
+        want_targets = self.need_target_strings
+        pending = []            # filtered (src,pth,tgt,mask,tidx) chunks
+        pending_targets = []    # filtered target strings (when wanted)
+        n_pending = 0
+        chunk: List[str] = []
+        chunk_size = max(batch_size, 512)
+
+        def flush_chunk():
+            nonlocal n_pending
+            if not chunk:
+                return
+            src, pth, tgt, mask, tidx = self._native.parse_batch(chunk)
+            keep = mask.any(dim=1)
+            if self.estimator_action.is_train:
+                keep &= tidx > self._tgt_oov
+            targets = None
+            if want_targets:
+                targets = [(l.split(' ', 1)[0].rstrip('\n') or self._tgt_oov_word)
+                           for l in chunk]
+            if not bool(keep.all()):
+                idx = keep.nonzero(as_tuple=True)[0]
+                src, pth, tgt = src[idx], pth[idx], tgt[idx]
+                mask, tidx = mask[idx], tidx[idx]
+                if targets is not None:
+                    targets = [targets[i] for i in idx.tolist()]
+            if src.shape[0]:
+                pending.append((src, pth, tgt, mask, tidx))
+                if targets is not None:
+                    pending_targets.extend(targets)
+                n_pending += src.shape[0]
+            chunk.clear()
+
+        def make_batch(tensors, targets):
+            if pin:
+                tensors = [t.contiguous().pin_memory() for t in tensors]
+            return ReaderBatch(source_token_indices=tensors[0],
+                               path_indices=tensors[1],
+                               target_token_indices=tensors[2],
+                               context_valid_mask=tensors[3],
+                               target_index=tensors[4],
+                               target_string=targets)
+
+        def emit(final=False):
+            nonlocal n_pending, pending_targets
+            cat = [torch.cat([p[i] for p in pending]) for i in range(5)]
+            targets = pending_targets
+            pending.clear()
+            pending_targets = []
+            start = 0
+            total = cat[0].shape[0]
+            while total - start >= batch_size or (final and start < total):
+                end = min(start + batch_size, total)
+                yield make_batch([t[start:end] for t in cat],
+                                 targets[start:end] if want_targets else None)
+                start = end
+            if start < total:
+                pending.append(tuple(t[start:] for t in cat))
+                if want_targets:
+                    pending_targets = targets[start:]
+            n_pending = total - start
+
+        for i, line in enumerate(lines):
+            if self.world_size > 1 and (i % self.world_size) != self.rank:
+                continue
+            chunk.append(line)
+            if len(chunk) >= chunk_size:
+                flush_chunk()
+                if n_pending >= batch_size:
+                    yield from emit()
+        flush_chunk()
+        if n_pending:
+            yield from emit(final=True)
 
     def _collate(self, rows) -> ReaderBatch:
         pin = torch.cuda.is_available()

@@ -14,12 +14,26 @@ import shutil
 import sys
 
 
+def _is_fresh(dest: str, *sources: str) -> bool:
+    """True if dest exists and is newer than every source. Used to avoid
+    re-invoking cpp_extension.load in a process that may already have the
+    module loaded (double pybind init in one process segfaults)."""
+    if not os.path.isfile(dest):
+        return False
+    dest_m = os.path.getmtime(dest)
+    return all(os.path.getmtime(s) < dest_m for s in sources if os.path.isfile(s))
+
+
 def build(verbose: bool = True) -> str:
     os.environ.setdefault('PYTORCH_ROCM_ARCH', 'gfx950')
     here = os.path.dirname(os.path.abspath(__file__))
     build_dir = os.path.join(here, '_build')
     os.makedirs(build_dir, exist_ok=True)
     src = os.path.join(here, 'csrc', 'c2v_kernels.hip')
+    dest = os.path.join(here, '_c2v_hip.so')
+    if _is_fresh(dest, src):
+        build_reader(verbose=verbose)
+        return dest
 
     from torch.utils import cpp_extension
     cpp_extension.load(
@@ -33,7 +47,33 @@ def build(verbose: bool = True) -> str:
         with_cuda=True,
     )
     built = os.path.join(build_dir, '_c2v_hip.so')
-    dest = os.path.join(here, '_c2v_hip.so')
+    shutil.copy2(built, dest)
+    build_reader(verbose=verbose)
+    return dest
+
+
+def build_reader(verbose: bool = True) -> str:
+    """CPU-only native reader extension (no HIP): built into
+    code2vec_amd/data/_c2v_reader.so."""
+    here = os.path.dirname(os.path.abspath(__file__))
+    data_dir = os.path.join(os.path.dirname(here), 'data')
+    build_dir = os.path.join(data_dir, '_build')
+    os.makedirs(build_dir, exist_ok=True)
+    src = os.path.join(data_dir, 'csrc', 'c2v_reader.cpp')
+    dest = os.path.join(data_dir, '_c2v_reader.so')
+    if _is_fresh(dest, src):
+        return dest
+    from torch.utils import cpp_extension
+    cpp_extension.load(
+        name='_c2v_reader',
+        sources=[src],
+        build_directory=build_dir,
+        extra_cflags=['-O3', '-std=c++17'],
+        verbose=verbose,
+        is_python_module=False,
+        with_cuda=False,
+    )
+    built = os.path.join(build_dir, '_c2v_reader.so')
     shutil.copy2(built, dest)
     return dest
 

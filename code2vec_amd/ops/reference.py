@@ -67,7 +67,10 @@ def gather_concat_bwd(d_ctx: torch.Tensor, keep_prob: float, seed: int,
     which consumes (ids, d_rows) directly."""
     if training and keep_prob < 1.0:
         mask = dropout_keep_mask(seed, d_ctx.numel(), keep_prob, d_ctx.device)
-        d_ctx = d_ctx * (mask.reshape(d_ctx.shape).to(d_ctx.dtype) / keep_prob)
+        # fp32 scale then one rounding to the storage dtype — matches the
+        # HIP kernel's bf2f -> *(1/keep) -> f2bf path exactly
+        d_ctx = (d_ctx.float() * (mask.reshape(d_ctx.shape).float() / keep_prob)
+                 ).to(d_ctx.dtype)
     return d_ctx
 
 

@@ -1,0 +1,92 @@
+"""C++ parser (data/csrc/c2v_reader.cpp) vs the Python reader oracle —
+batch-for-batch identical tensors on the same input file (runs on CPU; the
+native .so is built by __graft_entry__.build / ops.build.build_reader)."""
+
+import pickle
+import random
+
+import pytest
+import torch
+
+from code2vec_amd.config import Config
+from code2vec_amd.data.reader import EstimatorAction, PathContextReader
+from code2vec_amd.vocabularies import Code2VecVocabs
+
+
+@pytest.fixture()
+def setup(tmp_path):
+    try:
+        from code2vec_amd.ops.build import build_reader
+        build_reader(verbose=False)
+    except Exception as e:  # noqa: BLE001
+        pytest.skip('native reader unavailable: %r' % (e,))
+    prefix = str(tmp_path / 'ds')
+    tokens = ['tk%d' % i for i in range(30)]
+    paths = ['ph%d' % i for i in range(20)]
+    targets = ['tg|%d' % i for i in range(10)]
+    with open(prefix + '.dict.c2v', 'wb') as f:
+        pickle.dump({t: 5 for t in tokens}, f)
+        pickle.dump({p: 5 for p in paths}, f)
+        pickle.dump({t: 5 for t in targets}, f)
+    cfg = Config(set_defaults=True)
+    cfg.TRAIN_DATA_PATH_PREFIX = prefix
+    cfg.MAX_CONTEXTS = 5
+    cfg.TRAIN_BATCH_SIZE = cfg.TEST_BATCH_SIZE = 4
+    cfg.NUM_TRAIN_EPOCHS = 1
+    cfg.SHUFFLE_BUFFER_SIZE = 0
+    vocabs = Code2VecVocabs(cfg)
+
+    rng = random.Random(3)
+    data = tmp_path / 'mix.c2v'
+    lines = []
+    for i in range(57):
+        kind = rng.random()
+        if kind < 0.1:
+            lines.append('unknown_target tk0,ph0,tk1\n')   # OOV target
+        elif kind < 0.15:
+            lines.append('tg|1\n')                          # no contexts
+        elif kind < 0.2:
+            lines.append('tg|2 zz,yy,xx\n')                 # all-OOV context (mask 0)
+        else:
+            n = rng.randint(1, 5)
+            ctxs = ' '.join('%s,%s,%s' % (rng.choice(tokens), rng.choice(paths),
+                                          rng.choice(tokens)) for _ in range(n))
+            pad = ' ' * (5 - n)
+            lines.append('%s %s%s\n' % (rng.choice(targets), ctxs, pad))
+    data.write_text(''.join(lines))
+    return cfg, vocabs, str(data)
+
+
+@pytest.mark.parametrize('action', [EstimatorAction.Train, EstimatorAction.Evaluate])
+def test_native_matches_python(setup, action):
+    cfg, vocabs, data = setup
+    nat = PathContextReader(vocabs, cfg, action, keep_strings=False, use_native=True)
+    if nat._native is None:
+        pytest.skip('native parser not loaded')
+    py = PathContextReader(vocabs, cfg, action, keep_strings=False, use_native=False)
+    nb = list(nat.iter_batches(data))
+    pb = list(py.iter_batches(data))
+    assert len(nb) == len(pb) and len(nb) > 0
+    for a, b in zip(nb, pb):
+        assert torch.equal(a.source_token_indices, b.source_token_indices)
+        assert torch.equal(a.path_indices, b.path_indices)
+        assert torch.equal(a.target_token_indices, b.target_token_indices)
+        assert torch.equal(a.context_valid_mask, b.context_valid_mask)
+        assert torch.equal(a.target_index, b.target_index)
+        if action is EstimatorAction.Evaluate:
+            assert a.target_string == b.target_string
+
+
+def test_native_world_sharding(setup):
+    cfg, vocabs, data = setup
+    r0 = PathContextReader(vocabs, cfg, EstimatorAction.Train,
+                           world_size=2, rank=0)
+    r1 = PathContextReader(vocabs, cfg, EstimatorAction.Train,
+                           world_size=2, rank=1)
+    if r0._native is None:
+        pytest.skip('native parser not loaded')
+    full = PathContextReader(vocabs, cfg, EstimatorAction.Train, use_native=False)
+    n0 = sum(b.source_token_indices.shape[0] for b in r0.iter_batches(data))
+    n1 = sum(b.source_token_indices.shape[0] for b in r1.iter_batches(data))
+    nf = sum(b.source_token_indices.shape[0] for b in full.iter_batches(data))
+    assert n0 + n1 == nf
