@@ -10,6 +10,7 @@ Dropout uses a counter-based splitmix64 hash so CPU and HIP backends generate
 the identical mask from (seed, element-index).
 """
 
+import math
 from typing import Tuple
 
 import torch
@@ -154,6 +155,68 @@ def ce_bwd(logits: torch.Tensor, lse: torch.Tensor, labels: torch.Tensor,
     p.scatter_add_(1, labels.reshape(-1, 1),
                    torch.full((logits.shape[0], 1), -1.0, device=logits.device))
     return (p * scale).to(logits.dtype)
+
+
+# ---------------------------------------------------------------------------
+# sampled softmax (training-time approximation of the full-vocab CE)
+# ---------------------------------------------------------------------------
+
+def log_uniform_probs(ids: torch.Tensor, vocab_size: int) -> torch.Tensor:
+    """Zipf/log-uniform sampling probability of index k:
+    q(k) = log((k+2)/(k+1)) / log(V+1). The target vocabulary is sorted by
+    frequency (vocabularies top-N by count), so index order ~ rank order."""
+    idf = ids.float()
+    return torch.log((idf + 2.0) / (idf + 1.0)) / math.log(vocab_size + 1.0)
+
+
+def sample_log_uniform(n: int, vocab_size: int, device,
+                       generator=None) -> torch.Tensor:
+    """Draw n ids (with replacement) from the log-uniform distribution over
+    [0, vocab_size)."""
+    u = torch.rand(n, device=device, generator=generator)
+    ids = (torch.exp(u * math.log(vocab_size + 1.0)) - 1.0).long()
+    return ids.clamp_(0, vocab_size - 1)
+
+
+def sampled_ce_fwd(logits_cand: torch.Tensor, labels: torch.Tensor,
+                   sampled: torch.Tensor, corr_true: torch.Tensor,
+                   corr_samp: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+    """logits_cand (B, B+S): candidate logits where column b is row b's true
+    class and columns B.. are the shared sampled negatives.
+    Row b's effective logit set: z0 = logits[b,b] - corr_true[b];
+    z_j = logits[b,B+j] - corr_samp[j], with accidental hits
+    (sampled[j] == labels[b]) masked to -inf.
+    Returns (per-row loss, per-row lse over the effective set)."""
+    B = logits_cand.shape[0]
+    lf = logits_cand.float()
+    z0 = lf.diagonal() - corr_true                          # (B,)
+    zs = lf[:, B:] - corr_samp.reshape(1, -1)               # (B,S)
+    hit = sampled.reshape(1, -1) == labels.reshape(-1, 1)   # (B,S)
+    zs = torch.where(hit, torch.full_like(zs, -3.0e38), zs)
+    z = torch.cat([z0.reshape(-1, 1), zs], dim=1)           # (B,1+S)
+    lse = torch.logsumexp(z, dim=1)
+    return lse - z0, lse
+
+
+def sampled_ce_bwd(logits_cand: torch.Tensor, labels: torch.Tensor,
+                   sampled: torch.Tensor, corr_true: torch.Tensor,
+                   corr_samp: torch.Tensor, lse: torch.Tensor,
+                   scale: float) -> torch.Tensor:
+    """d_logits_cand (B, B+S) in logits_cand.dtype: softmax over the effective
+    set minus the one-hot at the true class, times scale. Columns 0..B-1
+    other than the diagonal get zero."""
+    B, T = logits_cand.shape
+    lf = logits_cand.float()
+    out = torch.zeros_like(lf)
+    z0 = lf.diagonal() - corr_true
+    p0 = torch.exp(z0 - lse) - 1.0
+    out.diagonal().copy_(p0 * scale)
+    zs = lf[:, B:] - corr_samp.reshape(1, -1)
+    hit = sampled.reshape(1, -1) == labels.reshape(-1, 1)
+    ps = torch.exp(zs - lse.reshape(-1, 1))
+    ps = torch.where(hit, torch.zeros_like(ps), ps)
+    out[:, B:] = ps * scale
+    return out.to(logits_cand.dtype)
 
 
 # ---------------------------------------------------------------------------
