@@ -174,17 +174,41 @@ class Code2VecNetwork:
 
         st = self.forward(src_ids, path_ids, tgt_ids, valid_mask, training=True)
         code_c = st.code.to(self.compute_dtype)
-        logits = code_c @ self.target_shadow.t()
-        loss_rows, lse = F.ce_fwd(logits, labels)
-        loss = loss_rows.float().mean()
+        S = int(cfg.SAMPLED_SOFTMAX_SIZE)
+        V = self.target_table.shape[0]
+        use_sampled = 0 < S < V
 
-        # ---- backward ----
-        d_logits = F.ce_bwd(logits, lse, labels, 1.0 / B)
-        # Dense target-table grad first: it is the big all-reduce (≈200 MB
-        # bf16 on java14m), launched async so it overlaps the rest of backward.
-        d_target = (d_logits.t() @ code_c)                      # (V,D) compute dtype
-        reducer.allreduce_dense('target_table', d_target)
-        d_code = (d_logits @ self.target_shadow).float()        # (B,D)
+        if use_sampled:
+            # Sampled-softmax path (BASELINE config 4): candidate set =
+            # [B local labels | S shared log-uniform negatives]; the target
+            # table gets SPARSE row grads (all-gathered under DP like the
+            # embedding tables) instead of the dense 200 MB all-reduce.
+            from ..ops.reference import log_uniform_probs, sample_log_uniform
+            sampled = sample_log_uniform(S, V, code_c.device)
+            cand = torch.cat([labels, sampled])                  # (B+S,)
+            w_cand = self.target_shadow.index_select(0, cand)    # (B+S, D)
+            logits_cand = code_c @ w_cand.t()                    # (B, B+S)
+            corr_true = torch.log(log_uniform_probs(labels, V) * S)
+            corr_samp = torch.log(log_uniform_probs(sampled, V) * S)
+            loss_rows, lse = F.sampled_ce_fwd(logits_cand, labels, sampled,
+                                              corr_true, corr_samp)
+            loss = loss_rows.float().mean()
+            d_cand = F.sampled_ce_bwd(logits_cand, labels, sampled, corr_true,
+                                      corr_samp, lse, 1.0 / B)
+            d_target_rows = d_cand.t() @ code_c                  # (B+S, D)
+            cand_g, target_rows_g = reducer.allgather_sparse(cand, d_target_rows)
+            d_code = (d_cand @ w_cand).float()                   # (B,D)
+            d_target = None
+        else:
+            logits = code_c @ self.target_shadow.t()
+            loss_rows, lse = F.ce_fwd(logits, labels)
+            loss = loss_rows.float().mean()
+            d_logits = F.ce_bwd(logits, lse, labels, 1.0 / B)
+            # Dense target-table grad first: it is the big all-reduce (≈200 MB
+            # bf16 on java14m), launched async to overlap the rest of backward.
+            d_target = (d_logits.t() @ code_c)                   # (V,D)
+            reducer.allreduce_dense('target_table', d_target)
+            d_code = (d_logits @ self.target_shadow).float()     # (B,D)
 
         d_comb3, d_a = F.attention_bwd(st.comb.reshape(B, C, D), self.a_c,
                                        st.alpha, d_code)
@@ -222,10 +246,18 @@ class Code2VecNetwork:
         F.adam_dense_step(self.a, d_a, self._adam_m['a'], self._adam_v['a'],
                           t, lr, b1, b2, eps)
         self._refresh_shadows(only_w=True)
-        reducer.wait('target_table')
-        F.adam_dense_step(self.target_table, d_target,
-                          self._adam_m['target_table'], self._adam_v['target_table'],
-                          t, lr, b1, b2, eps, shadow=self.target_shadow)
+        if use_sampled:
+            F.adam_sparse_rows_step(self.target_table, cand_g, target_rows_g,
+                                    self._adam_m['target_table'],
+                                    self._adam_v['target_table'],
+                                    t, lr, b1, b2, eps,
+                                    shadow=self.target_shadow)
+        else:
+            reducer.wait('target_table')
+            F.adam_dense_step(self.target_table, d_target,
+                              self._adam_m['target_table'],
+                              self._adam_v['target_table'],
+                              t, lr, b1, b2, eps, shadow=self.target_shadow)
         return loss
 
     # ---- evaluation / prediction forward ----

@@ -532,6 +532,88 @@ __global__ void k_ce_bwd(const ushort* __restrict__ logits,
 }
 
 // ---------------------------------------------------------------------------
+// Sampled softmax CE (training-time; BASELINE config 4). Candidate logits
+// L (B, T=B+S): column b is row b's true class, columns B.. are S shared
+// log-uniform negatives. Effective per-row set: {L[b,b]-corr_true[b]} ∪
+// {L[b,B+j]-corr_samp[j] | sampled[j] != labels[b]} (accidental hits masked).
+// ---------------------------------------------------------------------------
+
+__launch_bounds__(256)
+__global__ void k_sampled_ce_fwd(const ushort* __restrict__ L,
+                                 const long* __restrict__ labels,
+                                 const long* __restrict__ sampled,
+                                 const float* __restrict__ corr_true,
+                                 const float* __restrict__ corr_samp,
+                                 float* __restrict__ loss,
+                                 float* __restrict__ lse_out, int B, int S) {
+  __shared__ float sm[4], ss[4];
+  const int b = blockIdx.x;
+  const int T = B + S;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const ushort* row = L + (long)b * T;
+  const long lab = labels[b];
+  const float z0 = bf2f(row[b]) - corr_true[b];
+
+  float m = z0, s = 1.f;  // thread 0's stream starts with z0; others -inf
+  if (tid != 0) { m = -3.0e38f; s = 0.f; }
+  for (int j = tid; j < S; j += blockDim.x) {
+    if (sampled[j] == lab) continue;
+    const float x = bf2f(row[B + j]) - corr_samp[j];
+    if (x > m) { s = s * __expf(m - x) + 1.f; m = x; }
+    else s += __expf(x - m);
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    const float mo = __shfl_down(m, off, 64);
+    const float so = __shfl_down(s, off, 64);
+    if (mo > m) { s = s * __expf(m - mo) + so; m = mo; }
+    else s += so * __expf(mo - m);
+  }
+  if (lane == 0) { sm[wid] = m; ss[wid] = s; }
+  __syncthreads();
+  if (tid == 0) {
+    float M = sm[0], Ssum = ss[0];
+#pragma unroll
+    for (int w = 1; w < 4; ++w) {
+      if (sm[w] > M) { Ssum = Ssum * __expf(M - sm[w]) + ss[w]; M = sm[w]; }
+      else Ssum += ss[w] * __expf(sm[w] - M);
+    }
+    const float l = M + __logf(Ssum);
+    lse_out[b] = l;
+    loss[b] = l - z0;
+  }
+}
+
+__global__ void k_sampled_ce_bwd(const ushort* __restrict__ L,
+                                 const long* __restrict__ labels,
+                                 const long* __restrict__ sampled,
+                                 const float* __restrict__ corr_true,
+                                 const float* __restrict__ corr_samp,
+                                 const float* __restrict__ lse,
+                                 ushort* __restrict__ dL, float scale, int B,
+                                 int S) {
+  const int b = blockIdx.y;
+  const int T = B + S;
+  const long lab = labels[b];
+  const float l = lse[b];
+  const ushort* row = L + (long)b * T;
+  ushort* drow = dL + (long)b * T;
+  for (int j = blockIdx.x * blockDim.x + threadIdx.x; j < S;
+       j += gridDim.x * blockDim.x) {
+    float p = 0.f;
+    if (sampled[j] != lab)
+      p = __expf(bf2f(row[B + j]) - corr_samp[j] - l);
+    drow[B + j] = f2bf(p * scale);
+  }
+  if (blockIdx.x == 0 && threadIdx.x == 0) {
+    const float z0 = bf2f(row[b]) - corr_true[b];
+    drow[b] = f2bf((__expf(z0 - l) - 1.f) * scale);
+  }
+}
+
+// ---------------------------------------------------------------------------
 // K10: Adam, TF AdamOptimizer formulation:
 //   m <- b1*m + (1-b1)*g ; v <- b2*v + (1-b2)*g^2
 //   lr_t = lr*sqrt(1-b2^t)/(1-b1^t) ; p -= lr_t * m/(sqrt(v)+eps)
@@ -578,8 +660,9 @@ __global__ void k_rows_accum(const G* __restrict__ rows,
 
 __global__ void k_adam_rows(float* __restrict__ p, const long* __restrict__ ids,
                             const float* __restrict__ acc, float* __restrict__ m,
-                            float* __restrict__ v, long n_uniq, int d,
-                            float lr_t, float b1, float b2, float eps) {
+                            float* __restrict__ v, ushort* __restrict__ shadow,
+                            long n_uniq, int d, float lr_t, float b1, float b2,
+                            float eps) {
   for (long s = blockIdx.x * blockDim.x + threadIdx.x; s < n_uniq * d;
        s += (long)gridDim.x * blockDim.x) {
     const long u = s / d;
@@ -590,7 +673,9 @@ __global__ void k_adam_rows(float* __restrict__ p, const long* __restrict__ ids,
     const float vi = b2 * v[off] + (1.f - b2) * gv * gv;
     m[off] = mi;
     v[off] = vi;
-    p[off] -= lr_t * mi / (sqrtf(vi) + eps);
+    const float pv = p[off] - lr_t * mi / (sqrtf(vi) + eps);
+    p[off] = pv;
+    if (shadow != nullptr) shadow[off] = f2bf(pv);
   }
 }
 
@@ -794,7 +879,8 @@ void adam_dense_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
 void adam_sparse_rows_step(torch::Tensor p, torch::Tensor uniq_ids,
                            torch::Tensor inverse, torch::Tensor grad_rows,
                            torch::Tensor m, torch::Tensor v, int64_t step,
-                           double lr, double beta1, double beta2, double eps) {
+                           double lr, double beta1, double beta2, double eps,
+                           torch::Tensor shadow) {
   CHECK_DEV(p); CHECK_CONT(p);
   auto rows_c = grad_rows.contiguous();
   auto inv_c = inverse.contiguous();
@@ -813,10 +899,55 @@ void adam_sparse_rows_step(torch::Tensor p, torch::Tensor uniq_ids,
     k_rows_accum<float><<<grid_1d(n_rows * d, 256), 256, 0, cur_stream()>>>(
         rows_c.data_ptr<float>(), inv_c.data_ptr<int>(), acc.data_ptr<float>(),
         n_rows, d);
+  ushort* shadow_ptr = nullptr;
+  if (shadow.defined() && shadow.numel() == p.numel())
+    shadow_ptr = reinterpret_cast<ushort*>(shadow.data_ptr<at::BFloat16>());
   k_adam_rows<<<grid_1d(n_uniq * d, 256), 256, 0, cur_stream()>>>(
       p.data_ptr<float>(), ids_c.data_ptr<long>(), acc.data_ptr<float>(),
-      m.data_ptr<float>(), v.data_ptr<float>(), n_uniq, d, lr_t, (float)beta1,
-      (float)beta2, (float)eps);
+      m.data_ptr<float>(), v.data_ptr<float>(), shadow_ptr, n_uniq, d, lr_t,
+      (float)beta1, (float)beta2, (float)eps);
+}
+
+std::vector<torch::Tensor> sampled_ce_fwd(torch::Tensor logits_cand,
+                                          torch::Tensor labels,
+                                          torch::Tensor sampled,
+                                          torch::Tensor corr_true,
+                                          torch::Tensor corr_samp) {
+  CHECK_DEV(logits_cand); CHECK_CONT(logits_cand);
+  const int B = logits_cand.size(0);
+  const int S = (int)sampled.numel();
+  TORCH_CHECK(logits_cand.size(1) == B + S, "candidate layout mismatch");
+  auto labels_c = labels.contiguous();
+  auto sampled_c = sampled.contiguous();
+  auto ct = corr_true.contiguous();
+  auto cs = corr_samp.contiguous();
+  auto loss = torch::empty({B}, logits_cand.options().dtype(torch::kFloat32));
+  auto lse = torch::empty({B}, logits_cand.options().dtype(torch::kFloat32));
+  k_sampled_ce_fwd<<<B, 256, 0, cur_stream()>>>(
+      bf_ptr(logits_cand), labels_c.data_ptr<long>(),
+      sampled_c.data_ptr<long>(), ct.data_ptr<float>(), cs.data_ptr<float>(),
+      loss.data_ptr<float>(), lse.data_ptr<float>(), B, S);
+  return {loss, lse};
+}
+
+torch::Tensor sampled_ce_bwd(torch::Tensor logits_cand, torch::Tensor labels,
+                             torch::Tensor sampled, torch::Tensor corr_true,
+                             torch::Tensor corr_samp, torch::Tensor lse,
+                             double scale) {
+  CHECK_DEV(logits_cand); CHECK_CONT(logits_cand);
+  const int B = logits_cand.size(0);
+  const int S = (int)sampled.numel();
+  auto labels_c = labels.contiguous();
+  auto sampled_c = sampled.contiguous();
+  auto ct = corr_true.contiguous();
+  auto cs = corr_samp.contiguous();
+  auto d = torch::zeros_like(logits_cand);
+  dim3 grid(grid_1d(S, 256, 32), B);
+  k_sampled_ce_bwd<<<grid, 256, 0, cur_stream()>>>(
+      bf_ptr(logits_cand), labels_c.data_ptr<long>(),
+      sampled_c.data_ptr<long>(), ct.data_ptr<float>(), cs.data_ptr<float>(),
+      lse.data_ptr<float>(), bf_ptr_mut(d), (float)scale, B, S);
+  return d;
 }
 
 }  // namespace
@@ -833,4 +964,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("ce_bwd", &ce_bwd);
   mod.def("adam_dense_step", &adam_dense_step);
   mod.def("adam_sparse_rows_step", &adam_sparse_rows_step);
+  mod.def("sampled_ce_fwd", &sampled_ce_fwd);
+  mod.def("sampled_ce_bwd", &sampled_ce_bwd);
 }

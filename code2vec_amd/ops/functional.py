@@ -90,11 +90,30 @@ def adam_dense_step(p, g, m, v, step: int, lr: float, beta1: float,
 
 
 def adam_sparse_rows_step(p, ids, grad_rows, m, v, step: int, lr: float,
-                          beta1: float, beta2: float, eps: float):
+                          beta1: float, beta2: float, eps: float, shadow=None):
     if backend_for(p) == 'hip':
         uniq, inverse = torch.unique(ids.long(), return_inverse=True)
         hip_ext(True).adam_sparse_rows_step(
             p, uniq.to(torch.int64), inverse.to(torch.int32), grad_rows, m, v,
-            int(step), float(lr), float(beta1), float(beta2), float(eps))
+            int(step), float(lr), float(beta1), float(beta2), float(eps),
+            shadow if shadow is not None else torch.empty(0))
         return
-    ref.adam_sparse_rows_step(p, ids, grad_rows, m, v, step, lr, beta1, beta2, eps)
+    ref.adam_sparse_rows_step(p, ids, grad_rows, m, v, step, lr, beta1, beta2,
+                              eps, shadow)
+
+
+def sampled_ce_fwd(logits_cand, labels, sampled, corr_true, corr_samp):
+    if backend_for(logits_cand) == 'hip':
+        return hip_ext(True).sampled_ce_fwd(logits_cand, labels, sampled,
+                                            corr_true, corr_samp)
+    return ref.sampled_ce_fwd(logits_cand, labels, sampled, corr_true, corr_samp)
+
+
+def sampled_ce_bwd(logits_cand, labels, sampled, corr_true, corr_samp, lse,
+                   scale: float):
+    if backend_for(logits_cand) == 'hip':
+        return hip_ext(True).sampled_ce_bwd(logits_cand, labels, sampled,
+                                            corr_true, corr_samp, lse,
+                                            float(scale))
+    return ref.sampled_ce_bwd(logits_cand, labels, sampled, corr_true,
+                              corr_samp, lse, scale)

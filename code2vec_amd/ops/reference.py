@@ -241,7 +241,7 @@ def adam_dense_step(p: torch.Tensor, g: torch.Tensor, m: torch.Tensor,
 def adam_sparse_rows_step(p: torch.Tensor, ids: torch.Tensor,
                           grad_rows: torch.Tensor, m: torch.Tensor,
                           v: torch.Tensor, step: int, lr: float, beta1: float,
-                          beta2: float, eps: float):
+                          beta2: float, eps: float, shadow: torch.Tensor = None):
     """Lazy sparse-row Adam (TF AdamOptimizer._apply_sparse semantics: only
     touched rows update their moments). `ids` may contain duplicates; grads
     for duplicate rows are summed first."""
@@ -259,3 +259,5 @@ def adam_sparse_rows_step(p: torch.Tensor, ids: torch.Tensor,
     m.index_copy_(0, uniq, m_rows)
     v.index_copy_(0, uniq, v_rows)
     p.index_copy_(0, uniq, p_rows)
+    if shadow is not None:
+        shadow.index_copy_(0, uniq, p_rows.to(shadow.dtype))

@@ -157,6 +157,43 @@ def test_ce_fwd_bwd(B, V):
 
 # ---------------------------------------------------------------------------
 
+@pytest.mark.parametrize("B,S", [(8, 32), (1024, 8192)])
+def test_sampled_ce_fwd_bwd(B, S):
+    V = 261246
+    logits = randn(B, B + S, dtype=torch.bfloat16, scale=2.0, seed=B)
+    labels = torch.randint(0, V, (B,)).cuda()
+    sampled = torch.randint(0, V, (S,)).cuda()
+    sampled[0] = labels[0]  # accidental hit
+    ct = torch.log(R.log_uniform_probs(labels, V) * S)
+    cs = torch.log(R.log_uniform_probs(sampled, V) * S)
+    loss, lse = ext().sampled_ce_fwd(logits, labels, sampled, ct, cs)
+    loss_ref, lse_ref = R.sampled_ce_fwd(logits, labels, sampled, ct, cs)
+    assert (lse - lse_ref).abs().max().item() < 2e-3
+    assert (loss - loss_ref).abs().max().item() < 4e-3
+    d = ext().sampled_ce_bwd(logits, labels, sampled, ct, cs, lse, 1.0 / B)
+    d_ref = R.sampled_ce_bwd(logits, labels, sampled, ct, cs, lse_ref, 1.0 / B)
+    assert (d.float() - d_ref.float()).abs().max().item() < 2e-3
+
+
+def test_adam_sparse_rows_with_shadow():
+    torch.manual_seed(13)
+    Vr, d, n = 32, 16, 100
+    p = randn(Vr, d); p_ref = p.clone()
+    m = torch.zeros_like(p); v = torch.zeros_like(p)
+    m_ref = m.clone(); v_ref = v.clone()
+    shadow = p.to(torch.bfloat16)
+    shadow_ref = p_ref.to(torch.bfloat16)
+    ids = torch.randint(0, Vr, (n,), dtype=torch.int64).cuda()
+    rows = randn(n, d, dtype=torch.bfloat16)
+    from code2vec_amd.ops import functional as F
+    F.adam_sparse_rows_step(p, ids, rows, m, v, 1, 1e-3, 0.9, 0.999, 1e-8,
+                            shadow=shadow)
+    R.adam_sparse_rows_step(p_ref, ids, rows, m_ref, v_ref, 1, 1e-3, 0.9,
+                            0.999, 1e-8, shadow=shadow_ref)
+    assert (p - p_ref).abs().max().item() < 1e-5
+    assert torch.equal(shadow, shadow_ref)
+
+
 def test_adam_dense():
     torch.manual_seed(11)
     n = 10000
