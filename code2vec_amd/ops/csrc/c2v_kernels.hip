@@ -298,47 +298,56 @@ __global__ void k_tanh_bwd_mul(const ushort* __restrict__ dy,
 // All-masked rows produce zeros (reference would NaN).
 // ---------------------------------------------------------------------------
 
+// Streaming design (v2): NO whole-tile LDS staging — the 150 KiB tile capped
+// occupancy at 1 block/CU and left the kernels 5-7x off the BW floor. LDS
+// holds only the per-context scores (C f32) and the D-sized vectors; comb is
+// streamed from HBM/L2 with coalesced accesses (contexts row-major: the dot
+// pass reads 16 B/lane per context row; the column passes read/write
+// thread-per-column so consecutive threads touch consecutive addresses).
+
 extern __shared__ unsigned char smem[];
 
-__launch_bounds__(256)
+#define ATTN_THREADS 256
+
+__launch_bounds__(ATTN_THREADS)
 __global__ void k_attn_fwd(const ushort* __restrict__ comb,
                            const float* __restrict__ a,
                            const float* __restrict__ mask,
                            float* __restrict__ code,
                            float* __restrict__ alpha_out, int B, int C, int D) {
-  ushort* tile = reinterpret_cast<ushort*>(smem);                 // C*D bf16
-  float* sc = reinterpret_cast<float*>(smem + (size_t)C * D * 2); // C f32
-  float* av = sc + C;                                             // D f32
-  __shared__ float red[4];
-
+  float* sc = reinterpret_cast<float*>(smem);  // C f32 scores -> alpha
   const int b = blockIdx.x;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
-  const ushort* src = comb + (long)b * C * D;
+  const int n_waves = blockDim.x >> 6;
+  const ushort* base = comb + (long)b * C * D;
 
-  // stage tile (vectorized 8-wide) + attention vector
-  const int total8 = (C * D) / 8;
-  for (int s = tid; s < total8; s += blockDim.x)
-    *reinterpret_cast<ulonglong2*>(tile + (long)s * 8) =
-        *reinterpret_cast<const ulonglong2*>(src + (long)s * 8);
-  for (int i = tid; i < D; i += blockDim.x) av[i] = a[i];
-  __syncthreads();
+  // per-lane slice of `a` for the dot pass: lane covers [lane*8, lane*8+8)
+  const int d8 = D / 8;  // D % 64 == 0 is checked host-side (so D%8==0)
+  float areg[8];
+  const bool lane_active = lane < d8;
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    areg[j] = lane_active ? a[lane * 8 + j] : 0.f;
 
-  // scores: each wave takes contexts strided by 4; lane covers D/64 elems
-  const int per_lane = D / 64;
-  for (int c = wid; c < C; c += 4) {
+  // ---- scores: one context per wave per iteration
+  for (int c = wid; c < C; c += n_waves) {
     float dot = 0.f;
-    const ushort* rowp = tile + (long)c * D + lane * per_lane;
-#pragma unroll 8
-    for (int j = 0; j < per_lane; ++j) dot += bf2f(rowp[j]) * av[lane * per_lane + j];
+    if (lane_active) {
+      ulonglong2 packed = *reinterpret_cast<const ulonglong2*>(
+          base + (long)c * D + lane * 8);
+      const ushort* u = reinterpret_cast<const ushort*>(&packed);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dot += bf2f(u[j]) * areg[j];
+    }
     dot = wave_reduce_sum(dot);
     if (lane == 0)
       sc[c] = (mask[(long)b * C + c] > 0.f) ? dot : -3.0e38f;
   }
   __syncthreads();
 
-  // softmax over C (one wave handles the reduction; C <= a few thousand)
+  // ---- softmax over C (single wave; C is at most a few thousand)
   if (wid == 0) {
     float m = -3.0e38f;
     for (int c = lane; c < C; c += 64) m = fmaxf(m, sc[c]);
@@ -347,10 +356,9 @@ __global__ void k_attn_fwd(const ushort* __restrict__ comb,
     const bool any_valid = m > -1.0e38f;
     float s = 0.f;
     for (int c = lane; c < C; c += 64) {
-      const float e = any_valid ? __expf(sc[c] - m) : 0.f;
-      const float e2 = (sc[c] > -1.0e38f) ? e : 0.f;
-      sc[c] = e2;
-      s += e2;
+      const float e = (any_valid && sc[c] > -1.0e38f) ? __expf(sc[c] - m) : 0.f;
+      sc[c] = e;
+      s += e;
     }
     s = wave_reduce_sum(s);
     s = __shfl(s, 0, 64);
@@ -362,16 +370,16 @@ __global__ void k_attn_fwd(const ushort* __restrict__ comb,
   }
   __syncthreads();
 
-  // weighted sum: thread t accumulates output columns t, t+256, ...
+  // ---- weighted sum: thread t owns output columns t, t+T, ...
   for (int col = tid; col < D; col += blockDim.x) {
     float acc = 0.f;
-    for (int c = 0; c < C; ++c) acc += sc[c] * bf2f(tile[(long)c * D + col]);
+    const ushort* p = base + col;
+    for (int c = 0; c < C; ++c) acc += sc[c] * bf2f(p[(long)c * D]);
     code[(long)b * D + col] = acc;
   }
-  (void)red;
 }
 
-__launch_bounds__(256)
+__launch_bounds__(ATTN_THREADS)
 __global__ void k_attn_bwd(const ushort* __restrict__ comb,
                            const float* __restrict__ a,
                            const float* __restrict__ alpha,
@@ -379,66 +387,66 @@ __global__ void k_attn_bwd(const ushort* __restrict__ comb,
                            ushort* __restrict__ d_comb,
                            float* __restrict__ d_a_partial,  // (grid,D)
                            int B, int C, int D) {
-  ushort* tile = reinterpret_cast<ushort*>(smem);                 // C*D bf16
-  float* dal = reinterpret_cast<float*>(smem + (size_t)C * D * 2);// C f32: d_alpha then d_e
-  float* av = dal + C;                                            // D f32
-  float* dv = av + D;                                             // D f32
+  float* dal = reinterpret_cast<float*>(smem);  // C f32: d_alpha then d_e
   __shared__ float s_inner;
-
   const int b = blockIdx.x;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
-  const ushort* src = comb + (long)b * C * D;
+  const int n_waves = blockDim.x >> 6;
+  const ushort* base = comb + (long)b * C * D;
+  const float* alpha_row = alpha + (long)b * C;
 
-  const int total8 = (C * D) / 8;
-  for (int s = tid; s < total8; s += blockDim.x)
-    *reinterpret_cast<ulonglong2*>(tile + (long)s * 8) =
-        *reinterpret_cast<const ulonglong2*>(src + (long)s * 8);
-  for (int i = tid; i < D; i += blockDim.x) {
-    av[i] = a[i];
-    dv[i] = d_code[(long)b * D + i];
-  }
-  __syncthreads();
+  const int d8 = D / 8;
+  const bool lane_active = lane < d8;
+  float dvreg[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    dvreg[j] = lane_active ? d_code[(long)b * D + lane * 8 + j] : 0.f;
 
-  // d_alpha_c = comb_c . dv
-  const int per_lane = D / 64;
-  for (int c = wid; c < C; c += 4) {
+  // ---- d_alpha_c = comb_c . dv
+  for (int c = wid; c < C; c += n_waves) {
     float dot = 0.f;
-    const ushort* rowp = tile + (long)c * D + lane * per_lane;
-#pragma unroll 8
-    for (int j = 0; j < per_lane; ++j) dot += bf2f(rowp[j]) * dv[lane * per_lane + j];
+    if (lane_active) {
+      ulonglong2 packed = *reinterpret_cast<const ulonglong2*>(
+          base + (long)c * D + lane * 8);
+      const ushort* u = reinterpret_cast<const ushort*>(&packed);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dot += bf2f(u[j]) * dvreg[j];
+    }
     dot = wave_reduce_sum(dot);
     if (lane == 0) dal[c] = dot;
   }
   __syncthreads();
 
-  // inner = sum alpha*d_alpha ; d_e_c = alpha_c (d_alpha_c - inner)
+  // ---- inner = sum alpha*d_alpha ; d_e_c = alpha_c (d_alpha_c - inner)
   if (wid == 0) {
     float inner = 0.f;
-    for (int c = lane; c < C; c += 64)
-      inner += alpha[(long)b * C + c] * dal[c];
+    for (int c = lane; c < C; c += 64) inner += alpha_row[c] * dal[c];
     inner = wave_reduce_sum(inner);
     if (lane == 0) s_inner = inner;
   }
   __syncthreads();
   const float inner = s_inner;
   for (int c = tid; c < C; c += blockDim.x)
-    dal[c] = alpha[(long)b * C + c] * (dal[c] - inner);
+    dal[c] = alpha_row[c] * (dal[c] - inner);
   __syncthreads();
 
-  // d_comb_c = alpha_c * dv + d_e_c * a   (write bf16)
-  for (int s = tid; s < C * D; s += blockDim.x) {
-    const int c = s / D, col = s % D;
-    const float al = alpha[(long)b * C + c];
-    d_comb[(long)b * C * D + s] = f2bf(al * dv[col] + dal[c] * av[col]);
-  }
-
-  // d_a partial: per-block row in d_a_partial
+  // ---- column passes: thread t owns cols t, t+T, ...
+  // d_comb[c,col] = alpha_c*dv[col] + d_e_c*a[col]  (coalesced writes per c)
+  // d_a[col]     += sum_c d_e_c * comb[c,col]       (coalesced reads per c)
   for (int col = tid; col < D; col += blockDim.x) {
-    float acc = 0.f;
-    for (int c = 0; c < C; ++c) acc += dal[c] * bf2f(tile[(long)c * D + col]);
-    d_a_partial[(long)blockIdx.x * D + col] = acc;
+    const float av = a[col];
+    const float dvv = d_code[(long)b * D + col];
+    const ushort* rp = base + col;
+    ushort* wp = d_comb + (long)b * C * D + col;
+    float da = 0.f;
+    for (int c = 0; c < C; ++c) {
+      const float de = dal[c];
+      da += de * bf2f(rp[(long)c * D]);
+      wp[(long)c * D] = f2bf(alpha_row[c] * dvv + de * av);
+    }
+    d_a_partial[(long)blockIdx.x * D + col] = da;
   }
 }
 
@@ -627,7 +635,44 @@ __global__ void k_adam_dense(float* __restrict__ p, const G* __restrict__ g,
                              float* __restrict__ m, float* __restrict__ v,
                              ushort* __restrict__ shadow, long n, float lr_t,
                              float b1, float b2, float eps) {
-  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+  // vectorized x4 main body (p/m/v as float4, g as bf16x4 or float4)
+  const long n4 = n / 4;
+  for (long i4 = blockIdx.x * blockDim.x + threadIdx.x; i4 < n4;
+       i4 += (long)gridDim.x * blockDim.x) {
+    const long i = i4 * 4;
+    float4 pv = *reinterpret_cast<float4*>(p + i);
+    float4 mv = *reinterpret_cast<float4*>(m + i);
+    float4 vv = *reinterpret_cast<float4*>(v + i);
+    float gv[4];
+    if constexpr (sizeof(G) == 2) {
+      ulonglong1 packed = *reinterpret_cast<const ulonglong1*>((const ushort*)g + i);
+      const ushort* u = reinterpret_cast<const ushort*>(&packed);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) gv[j] = bf2f(u[j]);
+    } else {
+      float4 gf = *reinterpret_cast<const float4*>((const float*)g + i);
+      gv[0] = gf.x; gv[1] = gf.y; gv[2] = gf.z; gv[3] = gf.w;
+    }
+    float* pp = &pv.x; float* mp = &mv.x; float* vp = &vv.x;
+    ushort sh[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float mi = b1 * mp[j] + (1.f - b1) * gv[j];
+      const float vi = b2 * vp[j] + (1.f - b2) * gv[j] * gv[j];
+      mp[j] = mi; vp[j] = vi;
+      const float pn = pp[j] - lr_t * mi / (sqrtf(vi) + eps);
+      pp[j] = pn;
+      sh[j] = f2bf(pn);
+    }
+    *reinterpret_cast<float4*>(p + i) = pv;
+    *reinterpret_cast<float4*>(m + i) = mv;
+    *reinterpret_cast<float4*>(v + i) = vv;
+    if (shadow != nullptr)
+      *reinterpret_cast<ulonglong1*>(shadow + i) =
+          *reinterpret_cast<ulonglong1*>(sh);
+  }
+  // scalar tail
+  for (long i = n4 * 4 + blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (long)gridDim.x * blockDim.x) {
     float gv;
     if constexpr (sizeof(G) == 2) gv = bf2f(((const ushort*)g)[i]);
@@ -775,28 +820,18 @@ torch::Tensor tanh_bwd_mul(torch::Tensor dy, torch::Tensor y) {
   return dz;
 }
 
-static void ensure_lds_limit(const void* fn, size_t bytes) {
-  static size_t configured = 0;
-  if (bytes > 65536 && bytes > configured) {
-    hipFuncSetAttribute(fn, hipFuncAttributeMaxDynamicSharedMemorySize,
-                        (int)bytes);
-    configured = bytes;
-  }
-}
-
 std::vector<torch::Tensor> attention_fwd(torch::Tensor comb, torch::Tensor a,
                                          torch::Tensor mask) {
   CHECK_DEV(comb); CHECK_CONT(comb); CHECK_DEV(a); CHECK_DEV(mask);
   TORCH_CHECK(comb.dim() == 3);
   const int B = comb.size(0), C = comb.size(1), D = comb.size(2);
-  TORCH_CHECK(D % 64 == 0, "code vector size must be a multiple of 64");
+  TORCH_CHECK(D % 64 == 0 && D <= 512,
+              "code vector size must be a multiple of 64, <= 512");
   auto a32 = a.to(torch::kFloat32).contiguous();
   auto mask32 = mask.to(torch::kFloat32).contiguous();
   auto code = torch::empty({B, D}, comb.options().dtype(torch::kFloat32));
   auto alpha = torch::empty({B, C}, comb.options().dtype(torch::kFloat32));
-  const size_t lds = (size_t)C * D * 2 + (size_t)C * 4 + (size_t)D * 4;
-  TORCH_CHECK(lds <= 160 * 1024, "attention tile exceeds LDS capacity");
-  ensure_lds_limit((const void*)k_attn_fwd, lds);
+  const size_t lds = (size_t)C * 4;
   k_attn_fwd<<<B, 256, lds, cur_stream()>>>(
       bf_ptr(comb), a32.data_ptr<float>(), mask32.data_ptr<float>(),
       code.data_ptr<float>(), alpha.data_ptr<float>(), B, C, D);
@@ -808,15 +843,13 @@ std::vector<torch::Tensor> attention_bwd(torch::Tensor comb, torch::Tensor a,
                                          torch::Tensor d_code) {
   CHECK_DEV(comb); CHECK_CONT(comb);
   const int B = comb.size(0), C = comb.size(1), D = comb.size(2);
-  TORCH_CHECK(D % 64 == 0);
+  TORCH_CHECK(D % 64 == 0 && D <= 512);
   auto a32 = a.to(torch::kFloat32).contiguous();
   auto alpha32 = alpha.contiguous();
   auto dcode32 = d_code.to(torch::kFloat32).contiguous();
   auto d_comb = torch::empty_like(comb);
   auto d_a_partial = torch::empty({B, D}, comb.options().dtype(torch::kFloat32));
-  const size_t lds = (size_t)C * D * 2 + (size_t)C * 4 + (size_t)D * 8;
-  TORCH_CHECK(lds <= 160 * 1024, "attention tile exceeds LDS capacity");
-  ensure_lds_limit((const void*)k_attn_bwd, lds);
+  const size_t lds = (size_t)C * 4;
   k_attn_bwd<<<B, 256, lds, cur_stream()>>>(
       bf_ptr(comb), a32.data_ptr<float>(), alpha32.data_ptr<float>(),
       dcode32.data_ptr<float>(), bf_ptr_mut(d_comb),
@@ -863,7 +896,7 @@ void adam_dense_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
   ushort* shadow_ptr = nullptr;
   if (shadow.defined() && shadow.numel() == n)
     shadow_ptr = reinterpret_cast<ushort*>(shadow.data_ptr<at::BFloat16>());
-  const int grid = grid_1d(n, 256);
+  const int grid = grid_1d(std::max<long>(n / 4, 1), 256);
   if (g_c.scalar_type() == torch::kBFloat16)
     k_adam_dense<ushort><<<grid, 256, 0, cur_stream()>>>(
         p.data_ptr<float>(), bf_ptr(g_c), m.data_ptr<float>(),
