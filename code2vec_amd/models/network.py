@@ -268,9 +268,9 @@ class Code2VecNetwork:
         """Returns (topk_indices (B,k) int64, topk_scores (B,k) fp32,
         code (B,D) fp32, alpha (B,C) fp32)."""
         st = self.forward(src_ids, path_ids, tgt_ids, valid_mask, training=False)
-        logits = self.logits(st.code).float()
+        logits = self.logits(st.code)
         # k is clamped to the vocab size (reference: tensorflow_model.py:299-301)
-        scores, indices = torch.topk(logits, k=min(top_k, logits.shape[1]), dim=1)
+        scores, indices = F.topk(logits, k=min(top_k, logits.shape[1]))
         if normalize_scores:
             scores = torch.softmax(scores, dim=1)
         return indices, scores, st.code, st.alpha

@@ -724,6 +724,90 @@ __global__ void k_adam_rows(float* __restrict__ p, const long* __restrict__ ids,
   }
 }
 
+// ---------------------------------------------------------------------------
+// K11: per-row top-k over the target vocabulary (k <= 32).
+// One workgroup per row: each thread keeps a sorted local top-k of its
+// strided slice in registers, candidates go to LDS, then wave 0 does k
+// max-scan passes over the 256*k candidates. Ties resolve to the lower index.
+// ---------------------------------------------------------------------------
+
+__launch_bounds__(256)
+__global__ void k_topk(const ushort* __restrict__ logits,
+                       float* __restrict__ out_vals,
+                       long* __restrict__ out_idx, int V, int K) {
+  extern __shared__ unsigned char smem_raw[];
+  float* cand_v = reinterpret_cast<float*>(smem_raw);          // 256*K
+  int* cand_i = reinterpret_cast<int*>(cand_v + blockDim.x * K);
+
+  const int b = blockIdx.x;
+  const int tid = threadIdx.x;
+  const ushort* row = logits + (long)b * V;
+
+  float lv[32];
+  int li[32];
+#pragma unroll
+  for (int j = 0; j < 32; ++j) { lv[j] = -3.0e38f; li[j] = 0x7FFFFFFF; }
+
+  for (int c = tid; c < V; c += blockDim.x) {
+    const float x = bf2f(row[c]);
+    if (x > lv[K - 1] || (x == lv[K - 1] && c < li[K - 1])) {
+      // insertion into the sorted (desc, then idx asc) local list
+      int pos = K - 1;
+      while (pos > 0 && (x > lv[pos - 1] ||
+                         (x == lv[pos - 1] && c < li[pos - 1]))) {
+        lv[pos] = lv[pos - 1];
+        li[pos] = li[pos - 1];
+        --pos;
+      }
+      lv[pos] = x;
+      li[pos] = c;
+    }
+  }
+  for (int j = 0; j < K; ++j) {
+    cand_v[tid * K + j] = lv[j];
+    cand_i[tid * K + j] = li[j];
+  }
+  __syncthreads();
+
+  // k selection passes; all threads participate so __syncthreads is legal.
+  __shared__ float red_v[4];
+  __shared__ int red_i[4], red_s[4];
+  const int n_cand = blockDim.x * K;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  for (int sel = 0; sel < K; ++sel) {
+    float m = -3.0e38f;
+    int mi = 0x7FFFFFFF;
+    int mslot = -1;
+    for (int s = tid; s < n_cand; s += blockDim.x) {
+      const float x = cand_v[s];
+      const int ci = cand_i[s];
+      if (x > m || (x == m && ci < mi)) { m = x; mi = ci; mslot = s; }
+    }
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      const float mo = __shfl_down(m, off, 64);
+      const int io = __shfl_down(mi, off, 64);
+      const int so = __shfl_down(mslot, off, 64);
+      if (mo > m || (mo == m && io < mi)) { m = mo; mi = io; mslot = so; }
+    }
+    if (lane == 0) { red_v[wid] = m; red_i[wid] = mi; red_s[wid] = mslot; }
+    __syncthreads();
+    if (tid == 0) {
+      float M = red_v[0]; int Mi = red_i[0]; int Ms = red_s[0];
+#pragma unroll
+      for (int w = 1; w < 4; ++w)
+        if (red_v[w] > M || (red_v[w] == M && red_i[w] < Mi)) {
+          M = red_v[w]; Mi = red_i[w]; Ms = red_s[w];
+        }
+      out_vals[(long)b * K + sel] = M;
+      out_idx[(long)b * K + sel] = Mi;
+      if (Ms >= 0) cand_v[Ms] = -3.0e38f;
+    }
+    __syncthreads();
+  }
+}
+
 // ===========================================================================
 // host wrappers
 // ===========================================================================
@@ -856,6 +940,20 @@ std::vector<torch::Tensor> attention_bwd(torch::Tensor comb, torch::Tensor a,
       d_a_partial.data_ptr<float>(), B, C, D);
   auto d_a = d_a_partial.sum(0);
   return {d_comb, d_a};
+}
+
+std::vector<torch::Tensor> topk(torch::Tensor logits, int64_t k) {
+  CHECK_DEV(logits); CHECK_CONT(logits);
+  TORCH_CHECK(logits.scalar_type() == torch::kBFloat16);
+  const int B = logits.size(0), V = logits.size(1);
+  TORCH_CHECK(k >= 1 && k <= 32 && k <= V);
+  auto vals = torch::empty({B, k}, logits.options().dtype(torch::kFloat32));
+  auto idx = torch::empty({B, k}, logits.options().dtype(torch::kInt64));
+  const size_t lds = 256 * (size_t)k * 8;
+  k_topk<<<B, 256, lds, cur_stream()>>>(bf_ptr(logits),
+                                        vals.data_ptr<float>(),
+                                        idx.data_ptr<long>(), V, (int)k);
+  return {vals, idx};
 }
 
 std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor labels) {
@@ -999,4 +1097,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("adam_sparse_rows_step", &adam_sparse_rows_step);
   mod.def("sampled_ce_fwd", &sampled_ce_fwd);
   mod.def("sampled_ce_bwd", &sampled_ce_bwd);
+  mod.def("topk", &topk);
 }

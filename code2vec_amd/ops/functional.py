@@ -79,6 +79,15 @@ def ce_bwd(logits, lse, labels, scale: float) -> torch.Tensor:
     return ref.ce_bwd(logits, lse, labels, scale)
 
 
+def topk(logits, k: int):
+    """Per-row top-k (values fp32 desc, tie → lower index; int64 indices)."""
+    if backend_for(logits) == 'hip' and logits.dtype == torch.bfloat16 and k <= 32:
+        vals, idx = hip_ext(True).topk(logits, int(k))
+        return vals, idx
+    vals, idx = torch.topk(logits.float(), k=k, dim=1)
+    return vals, idx
+
+
 def adam_dense_step(p, g, m, v, step: int, lr: float, beta1: float,
                     beta2: float, eps: float, shadow=None):
     if backend_for(p) == 'hip':

@@ -157,6 +157,23 @@ def test_ce_fwd_bwd(B, V):
 
 # ---------------------------------------------------------------------------
 
+@pytest.mark.parametrize("B,V,k", [(4, 1000, 10), (32, 261246, 10), (2, 50, 32)])
+def test_topk_vs_torch(B, V, k):
+    logits = randn(B, V, dtype=torch.bfloat16, scale=3.0, seed=B + V)
+    vals, idx = ext().topk(logits, k)
+    ref_vals, ref_idx = torch.topk(logits.float(), k=k, dim=1)
+    assert torch.equal(vals, ref_vals)
+    # indices must point at the same values (tie order may differ from torch)
+    picked = logits.float().gather(1, idx)
+    assert torch.equal(picked, ref_vals)
+    # our tie rule: strictly non-increasing values, index ascending on ties
+    for b in range(B):
+        for j in range(1, k):
+            assert vals[b, j] < vals[b, j - 1] or idx[b, j] > idx[b, j - 1]
+
+
+# ---------------------------------------------------------------------------
+
 @pytest.mark.parametrize("B,S", [(8, 32), (1024, 8192)])
 def test_sampled_ce_fwd_bwd(B, S):
     V = 261246
