@@ -105,14 +105,16 @@ class Code2VecModel(Code2VecModelBase):
         window_examples = 0
         start = time.time()
         multi_batch_start = start
-        for batch in reader.iter_batches():
-            b = batch.to(device) if device.type != 'cpu' else batch
+        from ..data.prefetcher import BatchPrefetcher
+        prefetcher = BatchPrefetcher(reader.iter_batches(), device,
+                                     depth=cfg.READER_QUEUE_DEPTH)
+        for b in prefetcher:
             loss = self.network.train_step(
                 b.source_token_indices, b.path_indices, b.target_token_indices,
                 b.context_valid_mask, b.target_index, reducer=self.reducer)
             batch_num += 1
             sum_loss += float(loss)
-            window_examples += batch.source_token_indices.shape[0] * self.world_size
+            window_examples += b.source_token_indices.shape[0] * self.world_size
 
             if batch_num % cfg.NUM_BATCHES_TO_LOG_PROGRESS == 0:
                 elapsed = time.time() - multi_batch_start
