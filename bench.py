@@ -85,6 +85,8 @@ def main():
     ap.add_argument('--steps', type=int, default=30)
     ap.add_argument('--warmup', type=int, default=10)
     ap.add_argument('--batch', type=int, default=1024, help='per-GPU batch size')
+    ap.add_argument('--no-graph', action='store_true',
+                    help='disable hipGraph-captured stepping (single-GPU only)')
     args = ap.parse_args()
 
     world_size = int(os.environ.get('WORLD_SIZE', '1'))
@@ -115,6 +117,20 @@ def main():
                           device=device)
     batches = synth_batches(cfg, device, args.batch, seed=1234 + rank)
 
+    # hipGraph-captured stepping: the whole train step (fwd+bwd+Adam) replays
+    # as one graph launch. Single-process path only; DP keeps eager launches
+    # (they overlap the RCCL collectives).
+    use_graph = (device.startswith('cuda') and not distributed
+                 and not args.no_graph)
+    if use_graph:
+        graph_step = net.make_graph_step(args.batch)
+
+        def do_step(b):
+            graph_step.step(*b)
+    else:
+        def do_step(b):
+            net.train_step(*b, reducer=reducer)
+
     def barrier_sync():
         if distributed:
             import torch.distributed as dist
@@ -123,12 +139,12 @@ def main():
             torch.cuda.synchronize()
 
     for i in range(args.warmup):
-        net.train_step(*batches[i % len(batches)], reducer=reducer)
+        do_step(batches[i % len(batches)])
     barrier_sync()
 
     t0 = time.perf_counter()
     for i in range(args.steps):
-        net.train_step(*batches[i % len(batches)], reducer=reducer)
+        do_step(batches[i % len(batches)])
     barrier_sync()
     elapsed = time.perf_counter() - t0
 
@@ -161,6 +177,7 @@ def main():
                 'seq_len': cfg.MAX_CONTEXTS,
                 'parallelism': 'dp%d' % n_gpus,
                 'softmax': 'full',
+                'stepping': 'hipgraph' if use_graph else 'eager',
                 'token_vocab': cfg.MAX_TOKEN_VOCAB_SIZE,
                 'path_vocab': cfg.MAX_PATH_VOCAB_SIZE,
                 'target_vocab': cfg.MAX_TARGET_VOCAB_SIZE,

@@ -33,6 +33,7 @@ class ForwardState(NamedTuple):
     code: torch.Tensor       # (B, D) fp32
     alpha: torch.Tensor      # (B, C) fp32
     seed: int
+    seed_t: Optional[torch.Tensor] = None
 
 
 class NullReducer:
@@ -95,6 +96,14 @@ class Code2VecNetwork:
         self._refresh_shadows()
         self._seed_counter = torch.Generator(device='cpu').manual_seed(4242).initial_seed()
         self._step_ctr = 0
+        # Device-resident RNG seed and Adam step counters: under hipGraph
+        # capture these are advanced by in-graph tensor ops so every replay
+        # sees fresh values (a host scalar would be baked into the capture).
+        self._seed_t = None
+        self._step_t = None
+        if self.device.type == 'cuda':
+            self._seed_t = torch.zeros(1, dtype=torch.int64, device=self.device)
+            self._step_t = torch.zeros(1, dtype=torch.int32, device=self.device)
 
     # ---- parameters ----
 
@@ -139,6 +148,8 @@ class Code2VecNetwork:
                     self._adam_m[n].copy_(sd['adam_m.' + n].to(self.device))
                     self._adam_v[n].copy_(sd['adam_v.' + n].to(self.device))
         self._refresh_shadows()
+        if self._step_t is not None:
+            self._step_t.fill_(self.adam_step)
 
     # ---- forward ----
 
@@ -148,13 +159,19 @@ class Code2VecNetwork:
         D = self.config.CODE_VECTOR_SIZE
         self._step_ctr += 1
         seed = (self._seed_counter + self._step_ctr * 2654435761) & 0x7FFFFFFFFFFFFFFF
+        seed_t = None
+        if training and self._seed_t is not None:
+            self._seed_t.add_(2654435761)   # in-graph advance (capture-safe)
+            seed_t = self._seed_t
+            seed = 0
         keep = self.config.DROPOUT_KEEP_RATE if training else 1.0
         ctx = F.gather_concat_fwd(self.tok_table, self.path_table, src_ids,
                                   path_ids, tgt_ids, keep, seed, training,
-                                  out_dtype=self.compute_dtype)
+                                  out_dtype=self.compute_dtype, seed_t=seed_t)
         comb = F.transform_tanh_fwd(ctx, self.w_oi)                  # (B*C, D)
         code, alpha = F.attention_fwd(comb.reshape(B, C, D), self.a_c, valid_mask)
-        return ForwardState(ctx=ctx, comb=comb, code=code, alpha=alpha, seed=seed)
+        return ForwardState(ctx=ctx, comb=comb, code=code, alpha=alpha,
+                            seed=seed, seed_t=seed_t)
 
     def logits(self, code: torch.Tensor) -> torch.Tensor:
         """code (B,D) fp32 → (B, V_tgt) compute dtype via hipBLASLt
@@ -216,7 +233,8 @@ class Code2VecNetwork:
         d_ctx, d_w = F.transform_tanh_bwd(st.ctx, self.w_io, st.comb, d_comb)
         reducer.allreduce_dense('w', d_w)
         reducer.allreduce_dense('a', d_a)
-        d_ctx = F.gather_concat_bwd(d_ctx, cfg.DROPOUT_KEEP_RATE, st.seed, True)
+        d_ctx = F.gather_concat_bwd(d_ctx, cfg.DROPOUT_KEEP_RATE, st.seed, True,
+                                    seed_t=st.seed_t)
 
         # Sparse embedding grads: (ids, rows) pairs; under DP these are
         # all-gathered (not dense-all-reduced) — SURVEY §2.4.
@@ -233,32 +251,46 @@ class Code2VecNetwork:
         t, lr = self.adam_step, cfg.ADAM_LR
         b1, b2, eps = cfg.ADAM_BETA1, cfg.ADAM_BETA2, cfg.ADAM_EPS
 
+        if self._step_t is not None:
+            self._step_t.add_(1)            # in-graph advance (capture-safe)
+        st_t = self._step_t
         F.adam_sparse_rows_step(self.tok_table, tok_ids, tok_rows,
                                 self._adam_m['tok_table'], self._adam_v['tok_table'],
-                                t, lr, b1, b2, eps)
+                                t, lr, b1, b2, eps, step_t=st_t)
         F.adam_sparse_rows_step(self.path_table, path_ids_flat, path_rows,
                                 self._adam_m['path_table'], self._adam_v['path_table'],
-                                t, lr, b1, b2, eps)
+                                t, lr, b1, b2, eps, step_t=st_t)
         reducer.wait('w')
         F.adam_dense_step(self.w, d_w, self._adam_m['w'], self._adam_v['w'],
-                          t, lr, b1, b2, eps)
+                          t, lr, b1, b2, eps, step_t=st_t)
         reducer.wait('a')
         F.adam_dense_step(self.a, d_a, self._adam_m['a'], self._adam_v['a'],
-                          t, lr, b1, b2, eps)
+                          t, lr, b1, b2, eps, step_t=st_t)
         self._refresh_shadows(only_w=True)
         if use_sampled:
             F.adam_sparse_rows_step(self.target_table, cand_g, target_rows_g,
                                     self._adam_m['target_table'],
                                     self._adam_v['target_table'],
                                     t, lr, b1, b2, eps,
-                                    shadow=self.target_shadow)
+                                    shadow=self.target_shadow, step_t=st_t)
         else:
             reducer.wait('target_table')
             F.adam_dense_step(self.target_table, d_target,
                               self._adam_m['target_table'],
                               self._adam_v['target_table'],
-                              t, lr, b1, b2, eps, shadow=self.target_shadow)
+                              t, lr, b1, b2, eps, shadow=self.target_shadow,
+                              step_t=st_t)
         return loss
+
+    def adam_step_host_sync(self, delta: int = 0):
+        """Keep the HOST adam_step mirror in sync when the device counter is
+        advanced by a graph replay (the python train_step body didn't run)."""
+        self.adam_step += delta
+
+    # ---- hipGraph-captured training step ----
+
+    def make_graph_step(self, batch_size: int):
+        return GraphTrainStep(self, batch_size)
 
     # ---- evaluation / prediction forward ----
 
@@ -274,3 +306,51 @@ class Code2VecNetwork:
         if normalize_scores:
             scores = torch.softmax(scores, dim=1)
         return indices, scores, st.code, st.alpha
+
+
+class GraphTrainStep:
+    """hipGraph-captured training step (single-process path).
+
+    Captures the ENTIRE train_step — forward, backward, Adam — into one HIP
+    graph with static input buffers. All step-varying state (dropout seed,
+    Adam bias-correction step) lives in device memory and is advanced by
+    captured tensor ops, so each replay is a fresh, correct optimizer step
+    with one graph launch instead of ~40 Python-driven kernel launches."""
+
+    def __init__(self, net: Code2VecNetwork, batch_size: int, warmup: int = 2):
+        assert net.device.type == 'cuda', 'graph capture needs a GPU'
+        self.net = net
+        C = net.config.MAX_CONTEXTS
+        dev = net.device
+        self.src = torch.zeros(batch_size, C, dtype=torch.int32, device=dev)
+        self.pth = torch.zeros(batch_size, C, dtype=torch.int32, device=dev)
+        self.tgt = torch.zeros(batch_size, C, dtype=torch.int32, device=dev)
+        self.mask = torch.ones(batch_size, C, dtype=torch.float32, device=dev)
+        self.labels = torch.ones(batch_size, dtype=torch.int64, device=dev)
+
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(warmup):  # real steps (python body runs: host and
+                net.train_step(self.src, self.pth, self.tgt, self.mask,
+                               self.labels)   # device counters both advance)
+        torch.cuda.current_stream().wait_stream(side)
+
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self.loss = net.train_step(self.src, self.pth, self.tgt,
+                                       self.mask, self.labels)
+        # capture RECORDS the device ops without executing them, but the
+        # python body still bumped the host counter — undo that: the capture
+        # itself is not a training step.
+        net.adam_step_host_sync(-1)
+
+    def step(self, src, pth, tgt, mask, labels) -> torch.Tensor:
+        self.src.copy_(src)
+        self.pth.copy_(pth)
+        self.tgt.copy_(tgt)
+        self.mask.copy_(mask)
+        self.labels.copy_(labels)
+        self.graph.replay()
+        self.net.adam_step_host_sync(+1)
+        return self.loss

@@ -88,7 +88,9 @@ __global__ void k_gather_concat(
     const float* __restrict__ tok, const float* __restrict__ path,
     const int* __restrict__ src_ids, const int* __restrict__ path_ids,
     const int* __restrict__ tgt_ids, ushort* __restrict__ out,
-    int n_rows, int d, float keep_prob, u64 seed, int apply_dropout) {
+    int n_rows, int d, float keep_prob, u64 seed_scalar,
+    const long* __restrict__ seed_ptr, int apply_dropout) {
+  const u64 seed = seed_ptr ? (u64)*seed_ptr : seed_scalar;
   const int slots_per_row = (3 * d) / 8;
   const long total = (long)n_rows * slots_per_row;
   const float inv_keep = 1.0f / keep_prob;
@@ -130,7 +132,9 @@ __global__ void k_gather_concat(
 // dropout backward: scale d_ctx by the same keep mask (bf16 in/out)
 __global__ void k_dropout_bwd(const ushort* __restrict__ g_in,
                               ushort* __restrict__ g_out, long total,
-                              float keep_prob, u64 seed) {
+                              float keep_prob, u64 seed_scalar,
+                              const long* __restrict__ seed_ptr) {
+  const u64 seed = seed_ptr ? (u64)*seed_ptr : seed_scalar;
   const float inv_keep = 1.0f / keep_prob;
   for (long i8 = blockIdx.x * blockDim.x + threadIdx.x; i8 * 8 < total;
        i8 += (long)gridDim.x * blockDim.x) {
@@ -634,7 +638,12 @@ template <typename G>
 __global__ void k_adam_dense(float* __restrict__ p, const G* __restrict__ g,
                              float* __restrict__ m, float* __restrict__ v,
                              ushort* __restrict__ shadow, long n, float lr_t,
-                             float b1, float b2, float eps) {
+                             float b1, float b2, float eps, float lr,
+                             const int* __restrict__ step_ptr) {
+  if (step_ptr) {
+    const float t = (float)*step_ptr;
+    lr_t = lr * sqrtf(1.f - powf(b2, t)) / (1.f - powf(b1, t));
+  }
   // vectorized x4 main body (p/m/v as float4, g as bf16x4 or float4)
   const long n4 = n / 4;
   for (long i4 = blockIdx.x * blockDim.x + threadIdx.x; i4 < n4;
@@ -684,6 +693,91 @@ __global__ void k_adam_dense(float* __restrict__ p, const G* __restrict__ g,
     const float pv = p[i] - lr_t * mi / (sqrtf(vi) + eps);
     p[i] = pv;
     if (shadow != nullptr) shadow[i] = f2bf(pv);
+  }
+}
+
+// Hash-based id dedup (replaces sort-based torch.unique on the sparse-Adam
+// path): open-addressing table with atomicCAS claims; a claiming thread takes
+// a compact index from a global counter, so inverse mapping and unique list
+// come out of one pass with no sort and no host sync.
+// Three spin-free passes (an intra-wave publish/spin pair can deadlock under
+// the exec-mask divergence model, so claim / compact / lookup are separate
+// kernel launches with the stream as the barrier):
+//   1. claim: CAS the id into its probe slot
+//   2. compact: every occupied slot takes a compact index
+//   3. lookup: every input id re-probes and reads its compact index
+template <typename I>
+__global__ void k_hash_claim(const I* __restrict__ ids, long n,
+                             int* __restrict__ tbl_id, u32 mask_) {
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    const int id = (int)ids[i];
+    u32 slot = ((u32)id * 2654435761u) & mask_;
+    for (;;) {
+      const int seen = atomicCAS(&tbl_id[slot], -1, id);
+      if (seen == -1 || seen == id) break;
+      slot = (slot + 1) & mask_;
+    }
+  }
+}
+
+__global__ void k_hash_compact(const int* __restrict__ tbl_id,
+                               int* __restrict__ tbl_cidx,
+                               long* __restrict__ uniq_out,
+                               int* __restrict__ n_uniq, u32 cap) {
+  for (u32 s = blockIdx.x * blockDim.x + threadIdx.x; s < cap;
+       s += gridDim.x * blockDim.x) {
+    const int id = tbl_id[s];
+    if (id != -1) {
+      const int cidx = atomicAdd(n_uniq, 1);
+      tbl_cidx[s] = cidx;
+      uniq_out[cidx] = id;
+    }
+  }
+}
+
+template <typename I>
+__global__ void k_hash_lookup(const I* __restrict__ ids, long n,
+                              const int* __restrict__ tbl_id,
+                              const int* __restrict__ tbl_cidx,
+                              int* __restrict__ inverse_out, u32 mask_) {
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    const int id = (int)ids[i];
+    u32 slot = ((u32)id * 2654435761u) & mask_;
+    while (tbl_id[slot] != id) slot = (slot + 1) & mask_;
+    inverse_out[i] = tbl_cidx[slot];
+  }
+}
+
+// dynamic-count variant of k_adam_rows: row count read from device memory
+// (no host sync on the dedup result)
+__global__ void k_adam_rows_dyn(float* __restrict__ p,
+                                const long* __restrict__ ids,
+                                const float* __restrict__ acc,
+                                float* __restrict__ m, float* __restrict__ v,
+                                ushort* __restrict__ shadow,
+                                const int* __restrict__ n_uniq_ptr, int d,
+                                float lr_t, float b1, float b2, float eps,
+                                float lr, const int* __restrict__ step_ptr) {
+  const long n_uniq = *n_uniq_ptr;
+  if (step_ptr) {
+    const float t = (float)*step_ptr;
+    lr_t = lr * sqrtf(1.f - powf(b2, t)) / (1.f - powf(b1, t));
+  }
+  for (long s = blockIdx.x * blockDim.x + threadIdx.x; s < n_uniq * d;
+       s += (long)gridDim.x * blockDim.x) {
+    const long u = s / d;
+    const int col = (int)(s % d);
+    const long off = ids[u] * d + col;
+    const float gv = acc[s];
+    const float mi = b1 * m[off] + (1.f - b1) * gv;
+    const float vi = b2 * v[off] + (1.f - b2) * gv * gv;
+    m[off] = mi;
+    v[off] = vi;
+    const float pv = p[off] - lr_t * mi / (sqrtf(vi) + eps);
+    p[off] = pv;
+    if (shadow != nullptr) shadow[off] = f2bf(pv);
   }
 }
 
@@ -836,7 +930,10 @@ ushort* bf_ptr_mut(torch::Tensor& t) {
 torch::Tensor gather_concat_fwd(torch::Tensor tok, torch::Tensor path,
                                 torch::Tensor src, torch::Tensor pth,
                                 torch::Tensor tgt, double keep_prob,
-                                int64_t seed, bool training) {
+                                int64_t seed, bool training,
+                                torch::Tensor seed_t) {
+  const long* seed_ptr = (seed_t.defined() && seed_t.numel() == 1)
+                             ? seed_t.data_ptr<long>() : nullptr;
   CHECK_DEV(tok); CHECK_CONT(tok); CHECK_DEV(path); CHECK_CONT(path);
   CHECK_DEV(src); CHECK_CONT(src);
   TORCH_CHECK(src.scalar_type() == torch::kInt32, "ids must be int32");
@@ -850,19 +947,23 @@ torch::Tensor gather_concat_fwd(torch::Tensor tok, torch::Tensor path,
   k_gather_concat<<<grid_1d(slots, 256), 256, 0, cur_stream()>>>(
       tok.data_ptr<float>(), path.data_ptr<float>(), src.data_ptr<int>(),
       pth.data_ptr<int>(), tgt.data_ptr<int>(), bf_ptr_mut(out), (int)n_rows,
-      d, (float)keep_prob, (u64)seed, drop ? 1 : 0);
+      d, (float)keep_prob, (u64)seed, seed_ptr, drop ? 1 : 0);
   return out;
 }
 
 torch::Tensor gather_concat_bwd(torch::Tensor d_ctx, double keep_prob,
-                                int64_t seed, bool training) {
+                                int64_t seed, bool training,
+                                torch::Tensor seed_t) {
   CHECK_DEV(d_ctx); CHECK_CONT(d_ctx);
   if (!(training && keep_prob < 1.0)) return d_ctx;
+  const long* seed_ptr = (seed_t.defined() && seed_t.numel() == 1)
+                             ? seed_t.data_ptr<long>() : nullptr;
   auto out = torch::empty_like(d_ctx);
   const long total = d_ctx.numel();
   TORCH_CHECK(total % 8 == 0);
   k_dropout_bwd<<<grid_1d(total / 8, 256), 256, 0, cur_stream()>>>(
-      bf_ptr(d_ctx), bf_ptr_mut(out), total, (float)keep_prob, (u64)seed);
+      bf_ptr(d_ctx), bf_ptr_mut(out), total, (float)keep_prob, (u64)seed,
+      seed_ptr);
   return out;
 }
 
@@ -984,7 +1085,10 @@ torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor lse,
 
 void adam_dense_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                      torch::Tensor v, int64_t step, double lr, double beta1,
-                     double beta2, double eps, torch::Tensor shadow) {
+                     double beta2, double eps, torch::Tensor shadow,
+                     torch::Tensor step_t) {
+  const int* step_ptr = (step_t.defined() && step_t.numel() == 1)
+                            ? step_t.data_ptr<int>() : nullptr;
   CHECK_DEV(p); CHECK_CONT(p);
   auto g_c = g.contiguous();
   const long n = p.numel();
@@ -999,12 +1103,12 @@ void adam_dense_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
     k_adam_dense<ushort><<<grid, 256, 0, cur_stream()>>>(
         p.data_ptr<float>(), bf_ptr(g_c), m.data_ptr<float>(),
         v.data_ptr<float>(), shadow_ptr, n, lr_t, (float)beta1, (float)beta2,
-        (float)eps);
+        (float)eps, (float)lr, step_ptr);
   else
     k_adam_dense<float><<<grid, 256, 0, cur_stream()>>>(
         p.data_ptr<float>(), g_c.data_ptr<float>(), m.data_ptr<float>(),
         v.data_ptr<float>(), shadow_ptr, n, lr_t, (float)beta1, (float)beta2,
-        (float)eps);
+        (float)eps, (float)lr, step_ptr);
 }
 
 void adam_sparse_rows_step(torch::Tensor p, torch::Tensor uniq_ids,
@@ -1037,6 +1141,72 @@ void adam_sparse_rows_step(torch::Tensor p, torch::Tensor uniq_ids,
       p.data_ptr<float>(), ids_c.data_ptr<long>(), acc.data_ptr<float>(),
       m.data_ptr<float>(), v.data_ptr<float>(), shadow_ptr, n_uniq, d, lr_t,
       (float)beta1, (float)beta2, (float)eps);
+}
+
+void adam_sparse_rows_hash(torch::Tensor p, torch::Tensor ids,
+                           torch::Tensor grad_rows, torch::Tensor m,
+                           torch::Tensor v, int64_t step, double lr,
+                           double beta1, double beta2, double eps,
+                           torch::Tensor shadow, torch::Tensor step_t) {
+  const int* step_ptr = (step_t.defined() && step_t.numel() == 1)
+                            ? step_t.data_ptr<int>() : nullptr;
+  CHECK_DEV(p); CHECK_CONT(p);
+  auto ids_c = ids.contiguous();
+  auto rows_c = grad_rows.contiguous();
+  const long n = ids_c.numel();
+  const int d = rows_c.size(1);
+  const long V = p.size(0);
+  // table capacity: power of two >= 2 * min(n, V)
+  long want = 2 * std::min(n, V);
+  u32 cap = 1;
+  while (cap < (u32)want) cap <<= 1;
+  auto opts_i32 = p.options().dtype(torch::kInt32);
+  auto tbl_id = torch::full({(long)cap}, -1, opts_i32);
+  auto tbl_cidx = torch::full({(long)cap}, -1, opts_i32);
+  auto uniq = torch::empty({n}, p.options().dtype(torch::kInt64));
+  auto n_uniq = torch::zeros({1}, opts_i32);
+  auto inverse = torch::empty({n}, opts_i32);
+  auto acc = torch::zeros({n, (long)d}, p.options().dtype(torch::kFloat32));
+
+  const u32 mask_ = cap - 1;
+  if (ids_c.scalar_type() == torch::kInt32) {
+    k_hash_claim<int><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
+        ids_c.data_ptr<int>(), n, tbl_id.data_ptr<int>(), mask_);
+    k_hash_compact<<<grid_1d(cap, 256), 256, 0, cur_stream()>>>(
+        tbl_id.data_ptr<int>(), tbl_cidx.data_ptr<int>(),
+        uniq.data_ptr<long>(), n_uniq.data_ptr<int>(), cap);
+    k_hash_lookup<int><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
+        ids_c.data_ptr<int>(), n, tbl_id.data_ptr<int>(),
+        tbl_cidx.data_ptr<int>(), inverse.data_ptr<int>(), mask_);
+  } else {
+    k_hash_claim<long><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
+        ids_c.data_ptr<long>(), n, tbl_id.data_ptr<int>(), mask_);
+    k_hash_compact<<<grid_1d(cap, 256), 256, 0, cur_stream()>>>(
+        tbl_id.data_ptr<int>(), tbl_cidx.data_ptr<int>(),
+        uniq.data_ptr<long>(), n_uniq.data_ptr<int>(), cap);
+    k_hash_lookup<long><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
+        ids_c.data_ptr<long>(), n, tbl_id.data_ptr<int>(),
+        tbl_cidx.data_ptr<int>(), inverse.data_ptr<int>(), mask_);
+  }
+
+  if (rows_c.scalar_type() == torch::kBFloat16)
+    k_rows_accum<ushort><<<grid_1d(n * d, 256), 256, 0, cur_stream()>>>(
+        bf_ptr(rows_c), inverse.data_ptr<int>(), acc.data_ptr<float>(), n, d);
+  else
+    k_rows_accum<float><<<grid_1d(n * d, 256), 256, 0, cur_stream()>>>(
+        rows_c.data_ptr<float>(), inverse.data_ptr<int>(), acc.data_ptr<float>(),
+        n, d);
+
+  const float lr_t = (float)(lr * std::sqrt(1.0 - std::pow(beta2, (double)step)) /
+                             (1.0 - std::pow(beta1, (double)step)));
+  ushort* shadow_ptr = nullptr;
+  if (shadow.defined() && shadow.numel() == p.numel())
+    shadow_ptr = reinterpret_cast<ushort*>(shadow.data_ptr<at::BFloat16>());
+  k_adam_rows_dyn<<<grid_1d(n * d, 256), 256, 0, cur_stream()>>>(
+      p.data_ptr<float>(), uniq.data_ptr<long>(), acc.data_ptr<float>(),
+      m.data_ptr<float>(), v.data_ptr<float>(), shadow_ptr,
+      n_uniq.data_ptr<int>(), d, lr_t, (float)beta1, (float)beta2, (float)eps,
+      (float)lr, step_ptr);
 }
 
 std::vector<torch::Tensor> sampled_ce_fwd(torch::Tensor logits_cand,
@@ -1095,6 +1265,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("ce_bwd", &ce_bwd);
   mod.def("adam_dense_step", &adam_dense_step);
   mod.def("adam_sparse_rows_step", &adam_sparse_rows_step);
+  mod.def("adam_sparse_rows_hash", &adam_sparse_rows_hash);
   mod.def("sampled_ce_fwd", &sampled_ce_fwd);
   mod.def("sampled_ce_bwd", &sampled_ce_bwd);
   mod.def("topk", &topk);

@@ -12,21 +12,30 @@ from . import reference as ref
 
 def gather_concat_fwd(tok_table, path_table, src_ids, path_ids, tgt_ids,
                       keep_prob: float, seed: int, training: bool,
-                      out_dtype=torch.bfloat16) -> torch.Tensor:
+                      out_dtype=torch.bfloat16, seed_t=None) -> torch.Tensor:
+    """seed_t: optional 1-elem int64 device tensor overriding `seed` — used
+    under hipGraph capture so each replay reads a fresh seed."""
     if backend_for(tok_table) == 'hip':
         assert out_dtype == torch.bfloat16, "HIP gather_concat emits bf16"
         return hip_ext(True).gather_concat_fwd(
             tok_table, path_table, src_ids, path_ids, tgt_ids,
-            float(keep_prob), int(seed), bool(training))
+            float(keep_prob), int(seed), bool(training),
+            seed_t if seed_t is not None else torch.empty(0))
+    if seed_t is not None:
+        seed = int(seed_t.item())
     return ref.gather_concat_fwd(tok_table, path_table, src_ids, path_ids,
                                  tgt_ids, keep_prob, seed, training,
                                  out_dtype=out_dtype)
 
 
-def gather_concat_bwd(d_ctx, keep_prob: float, seed: int, training: bool) -> torch.Tensor:
+def gather_concat_bwd(d_ctx, keep_prob: float, seed: int, training: bool,
+                      seed_t=None) -> torch.Tensor:
     if backend_for(d_ctx) == 'hip':
-        return hip_ext(True).gather_concat_bwd(d_ctx, float(keep_prob),
-                                               int(seed), bool(training))
+        return hip_ext(True).gather_concat_bwd(
+            d_ctx, float(keep_prob), int(seed), bool(training),
+            seed_t if seed_t is not None else torch.empty(0))
+    if seed_t is not None:
+        seed = int(seed_t.item())
     return ref.gather_concat_bwd(d_ctx, keep_prob, seed, training)
 
 
@@ -89,24 +98,33 @@ def topk(logits, k: int):
 
 
 def adam_dense_step(p, g, m, v, step: int, lr: float, beta1: float,
-                    beta2: float, eps: float, shadow=None):
+                    beta2: float, eps: float, shadow=None, step_t=None):
+    """step_t: optional 1-elem int32 device tensor overriding `step` (the
+    bias-correction term is then computed in-kernel per replay)."""
     if backend_for(p) == 'hip':
         hip_ext(True).adam_dense_step(
             p, g, m, v, int(step), float(lr), float(beta1), float(beta2),
-            float(eps), shadow if shadow is not None else torch.empty(0))
+            float(eps), shadow if shadow is not None else torch.empty(0),
+            step_t if step_t is not None else torch.empty(0))
         return
+    if step_t is not None:
+        step = int(step_t.item())
     ref.adam_dense_step(p, g, m, v, step, lr, beta1, beta2, eps, shadow)
 
 
 def adam_sparse_rows_step(p, ids, grad_rows, m, v, step: int, lr: float,
-                          beta1: float, beta2: float, eps: float, shadow=None):
+                          beta1: float, beta2: float, eps: float, shadow=None,
+                          step_t=None):
     if backend_for(p) == 'hip':
-        uniq, inverse = torch.unique(ids.long(), return_inverse=True)
-        hip_ext(True).adam_sparse_rows_step(
-            p, uniq.to(torch.int64), inverse.to(torch.int32), grad_rows, m, v,
-            int(step), float(lr), float(beta1), float(beta2), float(eps),
-            shadow if shadow is not None else torch.empty(0))
+        # hash-based dedup + accumulate + lazy row update, no sort / no sync
+        hip_ext(True).adam_sparse_rows_hash(
+            p, ids, grad_rows, m, v, int(step), float(lr), float(beta1),
+            float(beta2), float(eps),
+            shadow if shadow is not None else torch.empty(0),
+            step_t if step_t is not None else torch.empty(0))
         return
+    if step_t is not None:
+        step = int(step_t.item())
     ref.adam_sparse_rows_step(p, ids, grad_rows, m, v, step, lr, beta1, beta2,
                               eps, shadow)
 

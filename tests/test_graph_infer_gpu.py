@@ -37,3 +37,59 @@ def test_graph_capture_matches_eager():
     assert torch.equal(idx_e2, idx_g2)
     assert torch.allclose(code_e2, code_g2, atol=1e-5)
     assert not torch.equal(idx_e, idx_e2)
+
+
+def test_graph_train_step_matches_eager():
+    """hipGraph-captured training replays must produce the same parameters
+    and losses as eager train_step (device-side seed/step counters)."""
+    from code2vec_amd.config import Config
+    from code2vec_amd.models.network import Code2VecNetwork
+
+    def cfg():
+        c = Config(set_defaults=True)
+        c.TRAIN_DATA_PATH_PREFIX = 'unused'
+        c.MAX_CONTEXTS = 16
+        c.TOKEN_EMBEDDINGS_SIZE = 64
+        c.PATH_EMBEDDINGS_SIZE = 64
+        c.CODE_VECTOR_SIZE = 192
+        c.TARGET_EMBEDDINGS_SIZE = 192
+        c.DROPOUT_KEEP_RATE = 0.75  # dropout ON: device seed must line up
+        return c
+
+    B, C = 32, 16
+    g = torch.Generator().manual_seed(21)
+    real = []
+    for _ in range(4):
+        src = torch.randint(0, 500, (B, C), generator=g, dtype=torch.int32).cuda()
+        pth = torch.randint(0, 300, (B, C), generator=g, dtype=torch.int32).cuda()
+        tgt = torch.randint(0, 500, (B, C), generator=g, dtype=torch.int32).cuda()
+        mask = torch.ones(B, C, device='cuda')
+        labels = torch.randint(1, 200, (B,), generator=g).cuda()
+        real.append((src, pth, tgt, mask, labels))
+    zero = (torch.zeros(B, C, dtype=torch.int32, device='cuda'),
+            torch.zeros(B, C, dtype=torch.int32, device='cuda'),
+            torch.zeros(B, C, dtype=torch.int32, device='cuda'),
+            torch.ones(B, C, device='cuda'),
+            torch.ones(B, dtype=torch.int64, device='cuda'))
+
+    torch.manual_seed(5)
+    eager = Code2VecNetwork(cfg(), 500, 300, 200, device='cuda:0')
+    torch.manual_seed(5)
+    graphed = Code2VecNetwork(cfg(), 500, 300, 200, device='cuda:0')
+
+    # eager: 2 zero-steps (mirror of graph warmup) then the real batches
+    eager_losses = []
+    for _ in range(2):
+        eager.train_step(*zero)
+    for b in real:
+        eager_losses.append(float(eager.train_step(*b)))
+
+    gts = graphed.make_graph_step(B)  # internally: 2 zero-warmups + capture
+    graph_losses = [float(gts.step(*b)) for b in real]
+
+    for le, lg in zip(eager_losses, graph_losses):
+        assert abs(le - lg) < 1e-4, (eager_losses, graph_losses)
+    for n in eager.param_names():
+        d = (eager.get_param(n) - graphed.get_param(n)).abs().max().item()
+        assert d < 1e-5, (n, d)
+    assert eager.adam_step == graphed.adam_step

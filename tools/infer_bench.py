@@ -2,6 +2,9 @@
 """Inference benchmark (BASELINE.json config 5): top-k predict + code-vector
 export at batch 4096 on 1 MI355X, eager launches vs hipGraph replay."""
 
+import os
+import sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import argparse
 import json
 import time

@@ -2,6 +2,9 @@
 """Micro-benchmarks for individual engine ops on MI355X (run via gpurun).
 Prints per-op times so kernel changes can be A/B'd without full bench runs."""
 
+import os
+import sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import time
 
 import torch
