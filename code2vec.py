@@ -9,7 +9,13 @@ from code2vec_amd.vocabularies import VocabType
 
 
 def load_model(config: Config) -> Code2VecModel:
-    return Code2VecModel(config)
+    """Builds the model; under torchrun (WORLD_SIZE>1) the data-parallel
+    reducer (RCCL over xGMI on GPU, gloo on CPU) is wired automatically."""
+    from code2vec_amd.parallel.ddp import Reducer, init_distributed_from_env
+    rank, world_size = init_distributed_from_env()
+    reducer = Reducer() if world_size > 1 else None
+    return Code2VecModel(config, reducer=reducer, world_size=world_size,
+                         rank=rank)
 
 
 if __name__ == '__main__':
