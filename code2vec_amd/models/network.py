@@ -251,35 +251,41 @@ class Code2VecNetwork:
         t, lr = self.adam_step, cfg.ADAM_LR
         b1, b2, eps = cfg.ADAM_BETA1, cfg.ADAM_BETA2, cfg.ADAM_EPS
 
+        st_t = None
         if self._step_t is not None:
             self._step_t.add_(1)            # in-graph advance (capture-safe)
-        st_t = self._step_t
+            # precompute the bias-corrected lr_t on device (capture-safe)
+            tf32 = self._step_t.to(torch.float32)
+            st_t = (lr * torch.sqrt(1.0 - torch.pow(torch.tensor(
+                b2, device=self.device), tf32))
+                    / (1.0 - torch.pow(torch.tensor(b1, device=self.device),
+                                       tf32))).reshape(1)
         F.adam_sparse_rows_step(self.tok_table, tok_ids, tok_rows,
                                 self._adam_m['tok_table'], self._adam_v['tok_table'],
-                                t, lr, b1, b2, eps, step_t=st_t)
+                                t, lr, b1, b2, eps, lrt_t=st_t)
         F.adam_sparse_rows_step(self.path_table, path_ids_flat, path_rows,
                                 self._adam_m['path_table'], self._adam_v['path_table'],
-                                t, lr, b1, b2, eps, step_t=st_t)
+                                t, lr, b1, b2, eps, lrt_t=st_t)
         reducer.wait('w')
         F.adam_dense_step(self.w, d_w, self._adam_m['w'], self._adam_v['w'],
-                          t, lr, b1, b2, eps, step_t=st_t)
+                          t, lr, b1, b2, eps, lrt_t=st_t)
         reducer.wait('a')
         F.adam_dense_step(self.a, d_a, self._adam_m['a'], self._adam_v['a'],
-                          t, lr, b1, b2, eps, step_t=st_t)
+                          t, lr, b1, b2, eps, lrt_t=st_t)
         self._refresh_shadows(only_w=True)
         if use_sampled:
             F.adam_sparse_rows_step(self.target_table, cand_g, target_rows_g,
                                     self._adam_m['target_table'],
                                     self._adam_v['target_table'],
                                     t, lr, b1, b2, eps,
-                                    shadow=self.target_shadow, step_t=st_t)
+                                    shadow=self.target_shadow, lrt_t=st_t)
         else:
             reducer.wait('target_table')
             F.adam_dense_step(self.target_table, d_target,
                               self._adam_m['target_table'],
                               self._adam_v['target_table'],
                               t, lr, b1, b2, eps, shadow=self.target_shadow,
-                              step_t=st_t)
+                              lrt_t=st_t)
         return loss
 
     def adam_step_host_sync(self, delta: int = 0):

@@ -638,12 +638,9 @@ template <typename G>
 __global__ void k_adam_dense(float* __restrict__ p, const G* __restrict__ g,
                              float* __restrict__ m, float* __restrict__ v,
                              ushort* __restrict__ shadow, long n, float lr_t,
-                             float b1, float b2, float eps, float lr,
-                             const int* __restrict__ step_ptr) {
-  if (step_ptr) {
-    const float t = (float)*step_ptr;
-    lr_t = lr * sqrtf(1.f - powf(b2, t)) / (1.f - powf(b1, t));
-  }
+                             float b1, float b2, float eps,
+                             const float* __restrict__ lrt_ptr) {
+  if (lrt_ptr) lr_t = *lrt_ptr;
   // vectorized x4 main body (p/m/v as float4, g as bf16x4 or float4)
   const long n4 = n / 4;
   for (long i4 = blockIdx.x * blockDim.x + threadIdx.x; i4 < n4;
@@ -714,8 +711,15 @@ __global__ void k_hash_claim(const I* __restrict__ ids, long n,
     const int id = (int)ids[i];
     u32 slot = ((u32)id * 2654435761u) & mask_;
     for (;;) {
-      const int seen = atomicCAS(&tbl_id[slot], -1, id);
-      if (seen == -1 || seen == id) break;
+      // plain read first: duplicate ids (notably the PAD id, ~30% of a
+      // batch) would otherwise serialize on same-address atomicCAS
+      const int cur = tbl_id[slot];
+      if (cur == id) break;
+      if (cur == -1) {
+        const int seen = atomicCAS(&tbl_id[slot], -1, id);
+        if (seen == -1 || seen == id) break;
+        // claimed by a different id meanwhile: fall through to next slot
+      }
       slot = (slot + 1) & mask_;
     }
   }
@@ -759,12 +763,9 @@ __global__ void k_adam_rows_dyn(float* __restrict__ p,
                                 ushort* __restrict__ shadow,
                                 const int* __restrict__ n_uniq_ptr, int d,
                                 float lr_t, float b1, float b2, float eps,
-                                float lr, const int* __restrict__ step_ptr) {
+                                const float* __restrict__ lrt_ptr) {
   const long n_uniq = *n_uniq_ptr;
-  if (step_ptr) {
-    const float t = (float)*step_ptr;
-    lr_t = lr * sqrtf(1.f - powf(b2, t)) / (1.f - powf(b1, t));
-  }
+  if (lrt_ptr) lr_t = *lrt_ptr;
   for (long s = blockIdx.x * blockDim.x + threadIdx.x; s < n_uniq * d;
        s += (long)gridDim.x * blockDim.x) {
     const long u = s / d;
@@ -1086,9 +1087,9 @@ torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor lse,
 void adam_dense_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                      torch::Tensor v, int64_t step, double lr, double beta1,
                      double beta2, double eps, torch::Tensor shadow,
-                     torch::Tensor step_t) {
-  const int* step_ptr = (step_t.defined() && step_t.numel() == 1)
-                            ? step_t.data_ptr<int>() : nullptr;
+                     torch::Tensor lrt_t) {
+  const float* lrt_ptr = (lrt_t.defined() && lrt_t.numel() == 1)
+                             ? lrt_t.data_ptr<float>() : nullptr;
   CHECK_DEV(p); CHECK_CONT(p);
   auto g_c = g.contiguous();
   const long n = p.numel();
@@ -1103,12 +1104,12 @@ void adam_dense_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
     k_adam_dense<ushort><<<grid, 256, 0, cur_stream()>>>(
         p.data_ptr<float>(), bf_ptr(g_c), m.data_ptr<float>(),
         v.data_ptr<float>(), shadow_ptr, n, lr_t, (float)beta1, (float)beta2,
-        (float)eps, (float)lr, step_ptr);
+        (float)eps, lrt_ptr);
   else
     k_adam_dense<float><<<grid, 256, 0, cur_stream()>>>(
         p.data_ptr<float>(), g_c.data_ptr<float>(), m.data_ptr<float>(),
         v.data_ptr<float>(), shadow_ptr, n, lr_t, (float)beta1, (float)beta2,
-        (float)eps, (float)lr, step_ptr);
+        (float)eps, lrt_ptr);
 }
 
 void adam_sparse_rows_step(torch::Tensor p, torch::Tensor uniq_ids,
@@ -1147,9 +1148,9 @@ void adam_sparse_rows_hash(torch::Tensor p, torch::Tensor ids,
                            torch::Tensor grad_rows, torch::Tensor m,
                            torch::Tensor v, int64_t step, double lr,
                            double beta1, double beta2, double eps,
-                           torch::Tensor shadow, torch::Tensor step_t) {
-  const int* step_ptr = (step_t.defined() && step_t.numel() == 1)
-                            ? step_t.data_ptr<int>() : nullptr;
+                           torch::Tensor shadow, torch::Tensor lrt_t) {
+  const float* lrt_ptr = (lrt_t.defined() && lrt_t.numel() == 1)
+                             ? lrt_t.data_ptr<float>() : nullptr;
   CHECK_DEV(p); CHECK_CONT(p);
   auto ids_c = ids.contiguous();
   auto rows_c = grad_rows.contiguous();
@@ -1206,7 +1207,7 @@ void adam_sparse_rows_hash(torch::Tensor p, torch::Tensor ids,
       p.data_ptr<float>(), uniq.data_ptr<long>(), acc.data_ptr<float>(),
       m.data_ptr<float>(), v.data_ptr<float>(), shadow_ptr,
       n_uniq.data_ptr<int>(), d, lr_t, (float)beta1, (float)beta2, (float)eps,
-      (float)lr, step_ptr);
+      lrt_ptr);
 }
 
 std::vector<torch::Tensor> sampled_ce_fwd(torch::Tensor logits_cand,

@@ -98,33 +98,29 @@ def topk(logits, k: int):
 
 
 def adam_dense_step(p, g, m, v, step: int, lr: float, beta1: float,
-                    beta2: float, eps: float, shadow=None, step_t=None):
-    """step_t: optional 1-elem int32 device tensor overriding `step` (the
-    bias-correction term is then computed in-kernel per replay)."""
+                    beta2: float, eps: float, shadow=None, lrt_t=None):
+    """lrt_t: optional 1-elem fp32 device tensor with the precomputed
+    bias-corrected lr_t (used under hipGraph capture)."""
     if backend_for(p) == 'hip':
         hip_ext(True).adam_dense_step(
             p, g, m, v, int(step), float(lr), float(beta1), float(beta2),
             float(eps), shadow if shadow is not None else torch.empty(0),
-            step_t if step_t is not None else torch.empty(0))
+            lrt_t if lrt_t is not None else torch.empty(0))
         return
-    if step_t is not None:
-        step = int(step_t.item())
     ref.adam_dense_step(p, g, m, v, step, lr, beta1, beta2, eps, shadow)
 
 
 def adam_sparse_rows_step(p, ids, grad_rows, m, v, step: int, lr: float,
                           beta1: float, beta2: float, eps: float, shadow=None,
-                          step_t=None):
+                          lrt_t=None):
     if backend_for(p) == 'hip':
         # hash-based dedup + accumulate + lazy row update, no sort / no sync
         hip_ext(True).adam_sparse_rows_hash(
             p, ids, grad_rows, m, v, int(step), float(lr), float(beta1),
             float(beta2), float(eps),
             shadow if shadow is not None else torch.empty(0),
-            step_t if step_t is not None else torch.empty(0))
+            lrt_t if lrt_t is not None else torch.empty(0))
         return
-    if step_t is not None:
-        step = int(step_t.item())
     ref.adam_sparse_rows_step(p, ids, grad_rows, m, v, step, lr, beta1, beta2,
                               eps, shadow)
 

@@ -38,7 +38,7 @@ def test_gather_concat_no_dropout():
     src = torch.randint(0, Vt, (B, C), dtype=torch.int32).cuda()
     pth = torch.randint(0, Vp, (B, C), dtype=torch.int32).cuda()
     tgt = torch.randint(0, Vt, (B, C), dtype=torch.int32).cuda()
-    out = ext().gather_concat_fwd(tok, path, src, pth, tgt, 1.0, 7, True)
+    out = ext().gather_concat_fwd(tok, path, src, pth, tgt, 1.0, 7, True, torch.empty(0))
     ref = R.gather_concat_fwd(tok, path, src, pth, tgt, 1.0, 7, True)
     assert out.dtype == torch.bfloat16
     assert torch.equal(out, ref)
@@ -52,18 +52,18 @@ def test_gather_concat_dropout_matches_reference_hash():
     pth = torch.randint(0, Vp, (B, C), dtype=torch.int32).cuda()
     tgt = torch.randint(0, Vt, (B, C), dtype=torch.int32).cuda()
     seed = 987654321
-    out = ext().gather_concat_fwd(tok, path, src, pth, tgt, 0.75, seed, True)
+    out = ext().gather_concat_fwd(tok, path, src, pth, tgt, 0.75, seed, True, torch.empty(0))
     ref = R.gather_concat_fwd(tok, path, src, pth, tgt, 0.75, seed, True)
     assert torch.equal(out, ref)
     # eval: no dropout even with keep<1
-    out_eval = ext().gather_concat_fwd(tok, path, src, pth, tgt, 0.75, seed, False)
+    out_eval = ext().gather_concat_fwd(tok, path, src, pth, tgt, 0.75, seed, False, torch.empty(0))
     ref_eval = R.gather_concat_fwd(tok, path, src, pth, tgt, 0.75, seed, False)
     assert torch.equal(out_eval, ref_eval)
 
 
 def test_dropout_bwd_mask():
     g = randn(64, 24, dtype=torch.bfloat16, seed=2)
-    out = ext().gather_concat_bwd(g, 0.75, 555, True)
+    out = ext().gather_concat_bwd(g, 0.75, 555, True, torch.empty(0))
     ref = R.gather_concat_bwd(g, 0.75, 555, True)
     assert torch.equal(out, ref)
 
@@ -221,7 +221,7 @@ def test_adam_dense():
     m_ref = m.clone(); v_ref = v.clone()
     shadow = torch.zeros(n, dtype=torch.bfloat16).cuda()
     for t in (1, 2, 3):
-        ext().adam_dense_step(p, g, m, v, t, 1e-3, 0.9, 0.999, 1e-8, shadow)
+        ext().adam_dense_step(p, g, m, v, t, 1e-3, 0.9, 0.999, 1e-8, shadow, torch.empty(0))
         R.adam_dense_step(p_ref, g, m_ref, v_ref, t, 1e-3, 0.9, 0.999, 1e-8)
     assert (p - p_ref).abs().max().item() < 1e-6
     assert (m - m_ref).abs().max().item() < 1e-6

@@ -69,7 +69,7 @@ def main():
     m = torch.zeros_like(p)
     v = torch.zeros_like(p)
     timeit('adam_dense target (w/ shadow)',
-           lambda: ext.adam_dense_step(p, g, m, v, 1, 1e-3, 0.9, 0.999, 1e-8, shadow))
+           lambda: ext.adam_dense_step(p, g, m, v, 1, 1e-3, 0.9, 0.999, 1e-8, shadow, torch.empty(0)))
 
     ids = torch.randint(0, 1301137, (2 * N,), device=dev)
     rows = torch.randn(2 * N, d, device=dev).to(torch.bfloat16)
@@ -78,11 +78,14 @@ def main():
     vv = torch.zeros_like(tok)
 
     def sparse_step():
-        uniq, inverse = torch.unique(ids, return_inverse=True)
-        ext.adam_sparse_rows_step(tok, uniq, inverse.to(torch.int32), rows,
-                                  mm, vv, 1, 1e-3, 0.9, 0.999, 1e-8,
-                                  torch.empty(0))
-    timeit('sparse adam incl. unique (tok)', sparse_step)
+        ext.adam_sparse_rows_hash(tok, ids, rows, mm, vv, 1, 1e-3, 0.9,
+                                  0.999, 1e-8, torch.empty(0), torch.empty(0))
+    timeit('sparse adam (hash dedup, tok)', sparse_step)
+    ids_pad = ids.clone(); ids_pad[::3] = 0   # heavy PAD duplication
+    def sparse_step_pad():
+        ext.adam_sparse_rows_hash(tok, ids_pad, rows, mm, vv, 1, 1e-3, 0.9,
+                                  0.999, 1e-8, torch.empty(0), torch.empty(0))
+    timeit('sparse adam (hash, 33% PAD ids)', sparse_step_pad)
     timeit('torch.unique only', lambda: torch.unique(ids, return_inverse=True))
 
 
