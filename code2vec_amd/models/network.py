@@ -254,12 +254,12 @@ class Code2VecNetwork:
         st_t = None
         if self._step_t is not None:
             self._step_t.add_(1)            # in-graph advance (capture-safe)
-            # precompute the bias-corrected lr_t on device (capture-safe)
+            # precompute the bias-corrected lr_t on device. Use log-space so
+            # the base is a device tensor (scalar**tensor would create a CPU
+            # tensor and a pageable H2D copy — a sync, and illegal in capture)
             tf32 = self._step_t.to(torch.float32)
-            st_t = (lr * torch.sqrt(1.0 - torch.pow(torch.tensor(
-                b2, device=self.device), tf32))
-                    / (1.0 - torch.pow(torch.tensor(b1, device=self.device),
-                                       tf32))).reshape(1)
+            st_t = (lr * torch.sqrt(1.0 - torch.exp(tf32 * math.log(b2)))
+                    / (1.0 - torch.exp(tf32 * math.log(b1)))).reshape(1)
         F.adam_sparse_rows_step(self.tok_table, tok_ids, tok_rows,
                                 self._adam_m['tok_table'], self._adam_v['tok_table'],
                                 t, lr, b1, b2, eps, lrt_t=st_t)
