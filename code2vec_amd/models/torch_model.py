@@ -39,6 +39,7 @@ class Code2VecModel(Code2VecModelBase):
         self.world_size = world_size
         self.rank = rank
         self._epochs_trained = 0
+        self._throughput_ewma = None
         super().__init__(config)
 
     # ---- inner model lifecycle ----
@@ -119,9 +120,20 @@ class Code2VecModel(Code2VecModelBase):
             if batch_num % cfg.NUM_BATCHES_TO_LOG_PROGRESS == 0:
                 elapsed = time.time() - multi_batch_start
                 throughput = window_examples / max(elapsed, 1e-9)
-                self.log('Average loss at batch %d: %f, throughput: %d samples/sec'
-                         % (batch_num, sum_loss / cfg.NUM_BATCHES_TO_LOG_PROGRESS,
-                            throughput))
+                # EWMA + epoch ETA (reference P13 progress logger semantics,
+                # keras_checkpoint_saver_callback.py:92-127)
+                self._throughput_ewma = (
+                    throughput if self._throughput_ewma is None
+                    else 0.5 * self._throughput_ewma + 0.5 * throughput)
+                msg = ('Average loss at batch %d: %f, throughput: %d samples/sec'
+                       % (batch_num, sum_loss / cfg.NUM_BATCHES_TO_LOG_PROGRESS,
+                          throughput))
+                if steps_per_epoch > 0 and self._throughput_ewma > 0:
+                    remaining = (steps_per_epoch - batch_num % steps_per_epoch)
+                    eta_sec = remaining * cfg.TRAIN_BATCH_SIZE * self.world_size \
+                        / self._throughput_ewma
+                    msg += ', epoch ETA: %dm%02ds' % (eta_sec // 60, eta_sec % 60)
+                self.log(msg)
                 sum_loss = 0.0
                 window_examples = 0
                 multi_batch_start = time.time()

@@ -274,6 +274,147 @@ __global__ void k_gemm_bt(const ushort* __restrict__ A,
   }
 }
 
+// ---------------------------------------------------------------------------
+// K4/K8 v2: 256x256-tile bf16 MFMA GEMM (CDNA4 guide §5 template, 2-phase
+// double-buffered variant). C(N,M) = A(N,K) @ Bt(M,K)^T.
+// 8 waves (512 threads) as 2x4 over (128x64) per-wave outputs; BK=64;
+// K-tile staged to LDS via width-16 global_load_lds, double buffered with
+// next-tile prefetch issued before the current tile's compute.
+// Grid ordering is COLUMN-major with the XCD-bijective swizzle so the blocks
+// sharing a Bt panel run on the same XCD back-to-back (Bt streams once per
+// XCD from HBM instead of once per row-tile) — that is what makes this
+// kernel win on the skinny-N logits shape (N=1024, M=261246).
+// ---------------------------------------------------------------------------
+
+#define G256_BM 256
+#define G256_BN 256
+#define G256_BK 64
+
+template <bool TANH>
+__launch_bounds__(512, 1)
+__global__ void k_gemm256_bt(const ushort* __restrict__ A,
+                             const ushort* __restrict__ Bt,
+                             ushort* __restrict__ C, int N, int M, int K) {
+  // LDS: 2 buffers x (A[256][64] + B[256][64]) bf16 = 128 KiB
+  extern __shared__ ushort lds256[];
+  ushort* lds_a[2] = {lds256, lds256 + 2 * 16384};
+  ushort* lds_b[2] = {lds256 + 16384, lds256 + 3 * 16384};
+
+  const int n_tiles = (N + G256_BM - 1) / G256_BM;
+  const int m_tiles = (M + G256_BN - 1) / G256_BN;
+  const int nwg = n_tiles * m_tiles;
+  int wg = blockIdx.x;
+  {
+    const int nxcd = 8;
+    const int q = nwg / nxcd, r = nwg % nxcd;
+    const int xcd = wg % nxcd, orig = wg / nxcd;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig;
+  }
+  const int tile_m = wg / n_tiles;   // column-major order: Bt panel reuse
+  const int tile_n = wg % n_tiles;
+  const int row0 = tile_n * G256_BM;
+  const int col0 = tile_m * G256_BN;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;          // 8 waves: 2 (row) x 4 (col)
+  const int wrow = (wid >> 2) * 128;
+  const int wcol = (wid & 3) * 64;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  // staging: each wave issues 4 width-16 loads for A and 4 for B per K-tile;
+  // wave-issue q covers 1 KiB (8 rows of 64 bf16) of the [256][64] tile.
+  auto stage = [&](int buf, int ks) {
+    const int k0 = ks * G256_BK;
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      const int off = (wid * 4 + q) * 1024 + lane * 16;   // byte in tile
+      const int lrow = off >> 7;             // 128 B per row
+      const int lcol = (off & 127) >> 1;     // bf16 column
+      {
+        const int grow = min(row0 + lrow, N - 1);
+        const ushort* gp = A + (long)grow * K + k0 + lcol;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) u32*)gp,
+            (__attribute__((address_space(3))) u32*)(lds_a[buf] + (wid * 4 + q) * 512),
+            16, 0, 0);
+      }
+      {
+        const int grow = min(col0 + lrow, M - 1);
+        const ushort* gp = Bt + (long)grow * K + k0 + lcol;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) u32*)gp,
+            (__attribute__((address_space(3))) u32*)(lds_b[buf] + (wid * 4 + q) * 512),
+            16, 0, 0);
+      }
+    }
+  };
+
+  const int k_steps = K / G256_BK;
+  stage(0, 0);
+  __syncthreads();
+
+  int cur = 0;
+  for (int ks = 0; ks < k_steps; ++ks) {
+    if (ks + 1 < k_steps) stage(cur ^ 1, ks + 1);
+
+    // B fragments for this wave's 64 columns: 4 col-frags x 2 k-frags
+    bf16x8 bfrag[4][2];
+#pragma unroll
+    for (int n = 0; n < 4; ++n)
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        const int r = wcol + n * 16 + (lane & 15);
+        bfrag[n][kk] = *reinterpret_cast<const bf16x8*>(
+            lds_b[cur] + r * G256_BK + kk * 32 + (lane >> 4) * 8);
+      }
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int m = 0; m < 8; ++m) {
+      bf16x8 afrag[2];
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        const int r = wrow + m * 16 + (lane & 15);
+        afrag[kk] = *reinterpret_cast<const bf16x8*>(
+            lds_a[cur] + r * G256_BK + kk * 32 + (lane >> 4) * 8);
+      }
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[0], bfrag[n][0], acc[m][n], 0, 0, 0);
+        acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[1], bfrag[n][1], acc[m][n], 0, 0, 0);
+      }
+    }
+    __builtin_amdgcn_s_setprio(0);
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // epilogue: C/D layout col=lane&15, row=(lane>>4)*4+reg
+#pragma unroll
+  for (int m = 0; m < 8; ++m) {
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      const int col = col0 + wcol + n * 16 + (lane & 15);
+      if (col >= M) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = row0 + wrow + m * 16 + (lane >> 4) * 4 + r;
+        if (row >= N) continue;
+        float v = acc[m][n][r];
+        if (TANH) v = tanhf(v);
+        C[(long)row * M + col] = f2bf(v);
+      }
+    }
+  }
+}
+
 // elementwise tanh backward: d_z = d_y * (1 - y^2), bf16
 __global__ void k_tanh_bwd_mul(const ushort* __restrict__ dy,
                                const ushort* __restrict__ y,
@@ -981,14 +1122,50 @@ torch::Tensor gather_concat_bwd(torch::Tensor d_ctx, double keep_prob,
   return out;
 }
 
-torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor Bt, bool tanh_epilogue) {
+torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor Bt, bool tanh_epilogue,
+                      int variant) {
   CHECK_DEV(A); CHECK_CONT(A); CHECK_DEV(Bt); CHECK_CONT(Bt);
   TORCH_CHECK(A.scalar_type() == torch::kBFloat16 &&
               Bt.scalar_type() == torch::kBFloat16);
   const int N = A.size(0), K = A.size(1), M = Bt.size(0);
   TORCH_CHECK(Bt.size(1) == K, "K mismatch");
-  TORCH_CHECK(K % GEMM_BK == 0, "K must be a multiple of 32");
+
+  // variant: 0 = auto, 1 = 128x128/BK32, 2 = 256x256/BK64 (needs K%64==0)
+  bool use256 = false;
+  if (variant == 2) use256 = true;
+  else if (variant == 0)
+    use256 = (K % G256_BK == 0) && ((long)N * M >= 4L * 1024 * 1024) &&
+             N >= G256_BM;
+  if (use256) {
+    TORCH_CHECK(K % G256_BK == 0, "256-tile variant needs K%64==0");
+  } else {
+    TORCH_CHECK(K % GEMM_BK == 0, "K must be a multiple of 32");
+  }
+
   auto C = torch::empty({N, M}, A.options());
+  if (use256) {
+    const int n_tiles = (N + G256_BM - 1) / G256_BM;
+    const int m_tiles = (M + G256_BN - 1) / G256_BN;
+    dim3 grid(n_tiles * m_tiles);
+    const size_t lds = 4 * 16384 * 2;  // 128 KiB
+    static bool lds_configured = false;
+    if (!lds_configured) {
+      (void)hipFuncSetAttribute((const void*)k_gemm256_bt<true>,
+                                hipFuncAttributeMaxDynamicSharedMemorySize,
+                                (int)lds);
+      (void)hipFuncSetAttribute((const void*)k_gemm256_bt<false>,
+                                hipFuncAttributeMaxDynamicSharedMemorySize,
+                                (int)lds);
+      lds_configured = true;
+    }
+    if (tanh_epilogue)
+      k_gemm256_bt<true><<<grid, 512, lds, cur_stream()>>>(
+          bf_ptr(A), bf_ptr(Bt), bf_ptr_mut(C), N, M, K);
+    else
+      k_gemm256_bt<false><<<grid, 512, lds, cur_stream()>>>(
+          bf_ptr(A), bf_ptr(Bt), bf_ptr_mut(C), N, M, K);
+    return C;
+  }
   const int n_tiles = (N + GEMM_BM - 1) / GEMM_BM;
   const int m_tiles = (M + GEMM_BN - 1) / GEMM_BN;
   dim3 grid(n_tiles * m_tiles);
@@ -1002,11 +1179,16 @@ torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor Bt, bool tanh_epilogue) {
 }
 
 torch::Tensor transform_tanh_fwd(torch::Tensor ctx, torch::Tensor w_oi) {
-  return gemm_bt(ctx, w_oi, true);
+  return gemm_bt(ctx, w_oi, true, 1);
 }
 
 torch::Tensor gemm_bt_bf16(torch::Tensor A, torch::Tensor Bt) {
-  return gemm_bt(A, Bt, false);
+  return gemm_bt(A, Bt, false, 1);
+}
+
+torch::Tensor gemm_bt_v(torch::Tensor A, torch::Tensor Bt, bool tanh_ep,
+                        int64_t variant) {
+  return gemm_bt(A, Bt, tanh_ep, (int)variant);
 }
 
 torch::Tensor tanh_bwd_mul(torch::Tensor dy, torch::Tensor y) {
@@ -1272,6 +1454,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gather_concat_bwd", &gather_concat_bwd);
   mod.def("transform_tanh_fwd", &transform_tanh_fwd);
   mod.def("gemm_bt_bf16", &gemm_bt_bf16);
+  mod.def("gemm_bt_v", &gemm_bt_v);
   mod.def("tanh_bwd_mul", &tanh_bwd_mul);
   mod.def("attention_fwd", &attention_fwd);
   mod.def("attention_bwd", &attention_bwd);

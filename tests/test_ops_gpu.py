@@ -283,3 +283,20 @@ def test_full_train_step_gpu_vs_cpu():
                                       mask.cuda(), labels.cuda()))
         lc = float(net_cpu.train_step(src, pth, tgt, mask, labels))
         assert abs(lg - lc) < 0.05 * max(1.0, abs(lc)), (step, lg, lc)
+
+
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("N,M,K", [(256, 256, 64), (512, 384, 384),
+                                   (1024, 2048, 384), (300, 260, 128)])
+def test_gemm256_vs_matmul(N, M, K):
+    """256x256-tile 2-phase GEMM: random asymmetric refcheck (guide G9)."""
+    A = randn(N, K, dtype=torch.bfloat16, scale=0.5, seed=N * 3 + M)
+    Bt = randn(M, K, dtype=torch.bfloat16, scale=0.5)
+    C = ext().gemm_bt_v(A, Bt, False, 2)
+    ref = A.float() @ Bt.float().t()
+    err = (C.float() - ref).abs().max().item()
+    denom = ref.abs().max().item()
+    assert err / denom < 0.02, 'max err %g vs scale %g' % (err, denom)
+    Ct = ext().gemm_bt_v(A, Bt, True, 2)
+    assert (Ct.float() - torch.tanh(ref)).abs().max().item() < 0.02
