@@ -297,8 +297,9 @@ __global__ void k_gemm256_bt(const ushort* __restrict__ A,
                              ushort* __restrict__ C, int N, int M, int K) {
   // LDS: 2 buffers x (A[256][64] + B[256][64]) bf16 = 128 KiB
   extern __shared__ ushort lds256[];
-  ushort* lds_a[2] = {lds256, lds256 + 2 * 16384};
-  ushort* lds_b[2] = {lds256 + 16384, lds256 + 3 * 16384};
+  // buffer b: A at b*32768, B at b*32768+16384 (ushort units)
+#define LDS_A(b) (lds256 + (b) * 32768)
+#define LDS_B(b) (lds256 + (b) * 32768 + 16384)
 
   const int n_tiles = (N + G256_BM - 1) / G256_BM;
   const int m_tiles = (M + G256_BN - 1) / G256_BN;
@@ -341,7 +342,7 @@ __global__ void k_gemm256_bt(const ushort* __restrict__ A,
         const ushort* gp = A + (long)grow * K + k0 + lcol;
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) u32*)gp,
-            (__attribute__((address_space(3))) u32*)(lds_a[buf] + (wid * 4 + q) * 512),
+            (__attribute__((address_space(3))) u32*)(LDS_A(buf) + (wid * 4 + q) * 512),
             16, 0, 0);
       }
       {
@@ -349,7 +350,7 @@ __global__ void k_gemm256_bt(const ushort* __restrict__ A,
         const ushort* gp = Bt + (long)grow * K + k0 + lcol;
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) u32*)gp,
-            (__attribute__((address_space(3))) u32*)(lds_b[buf] + (wid * 4 + q) * 512),
+            (__attribute__((address_space(3))) u32*)(LDS_B(buf) + (wid * 4 + q) * 512),
             16, 0, 0);
       }
     }
@@ -371,7 +372,7 @@ __global__ void k_gemm256_bt(const ushort* __restrict__ A,
       for (int kk = 0; kk < 2; ++kk) {
         const int r = wcol + n * 16 + (lane & 15);
         bfrag[n][kk] = *reinterpret_cast<const bf16x8*>(
-            lds_b[cur] + r * G256_BK + kk * 32 + (lane >> 4) * 8);
+            LDS_B(cur) + r * G256_BK + kk * 32 + (lane >> 4) * 8);
       }
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -381,7 +382,7 @@ __global__ void k_gemm256_bt(const ushort* __restrict__ A,
       for (int kk = 0; kk < 2; ++kk) {
         const int r = wrow + m * 16 + (lane & 15);
         afrag[kk] = *reinterpret_cast<const bf16x8*>(
-            lds_a[cur] + r * G256_BK + kk * 32 + (lane >> 4) * 8);
+            LDS_A(cur) + r * G256_BK + kk * 32 + (lane >> 4) * 8);
       }
 #pragma unroll
       for (int n = 0; n < 4; ++n) {
