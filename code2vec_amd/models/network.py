@@ -174,9 +174,9 @@ class Code2VecNetwork:
                             seed=seed, seed_t=seed_t)
 
     def logits(self, code: torch.Tensor) -> torch.Tensor:
-        """code (B,D) fp32 → (B, V_tgt) compute dtype via hipBLASLt
-        (plain library GEMM)."""
-        return code.to(self.compute_dtype) @ self.target_shadow.t()
+        """code (B,D) fp32 → (B, V_tgt) compute dtype:
+        256-tile MFMA kernel on big shapes, hipBLASLt otherwise."""
+        return F.logits_gemm(code.to(self.compute_dtype), self.target_shadow)
 
     # ---- full training step ----
 
@@ -217,7 +217,7 @@ class Code2VecNetwork:
             d_code = (d_cand @ w_cand).float()                   # (B,D)
             d_target = None
         else:
-            logits = code_c @ self.target_shadow.t()
+            logits = F.logits_gemm(code_c, self.target_shadow)
             loss_rows, lse = F.ce_fwd(logits, labels)
             loss = loss_rows.float().mean()
             d_logits = F.ce_bwd(logits, lse, labels, 1.0 / B)

@@ -909,6 +909,18 @@ __global__ void k_hash_lookup(const I* __restrict__ ids, long n,
   }
 }
 
+__global__ void k_zero_rows_dyn(float* __restrict__ acc,
+                                const int* __restrict__ n_uniq_ptr, int d) {
+  const long total = (long)(*n_uniq_ptr) * d;
+  const long n4 = total / 4;
+  for (long i4 = blockIdx.x * blockDim.x + threadIdx.x; i4 < n4;
+       i4 += (long)gridDim.x * blockDim.x)
+    *reinterpret_cast<float4*>(acc + i4 * 4) = float4{0.f, 0.f, 0.f, 0.f};
+  for (long i = n4 * 4 + blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x)
+    acc[i] = 0.f;
+}
+
 // dynamic-count variant of k_adam_rows: row count read from device memory
 // (no host sync on the dedup result)
 __global__ void k_adam_rows_dyn(float* __restrict__ p,
@@ -1363,7 +1375,9 @@ void adam_sparse_rows_hash(torch::Tensor p, torch::Tensor ids,
   auto uniq = torch::empty({n}, p.options().dtype(torch::kInt64));
   auto n_uniq = torch::zeros({1}, opts_i32);
   auto inverse = torch::empty({n}, opts_i32);
-  auto acc = torch::zeros({n, (long)d}, p.options().dtype(torch::kFloat32));
+  // acc rows beyond the (device-side) unique count are never read; zero only
+  // the live prefix after compact instead of a full torch::zeros fill
+  auto acc = torch::empty({n, (long)d}, p.options().dtype(torch::kFloat32));
 
   const u32 mask_ = cap - 1;
   if (ids_c.scalar_type() == torch::kInt32) {
@@ -1386,6 +1400,8 @@ void adam_sparse_rows_hash(torch::Tensor p, torch::Tensor ids,
         tbl_cidx.data_ptr<int>(), inverse.data_ptr<int>(), mask_);
   }
 
+  k_zero_rows_dyn<<<grid_1d(n * d / 4, 256), 256, 0, cur_stream()>>>(
+      acc.data_ptr<float>(), n_uniq.data_ptr<int>(), d);
   if (rows_c.scalar_type() == torch::kBFloat16)
     k_rows_accum<ushort><<<grid_1d(n * d, 256), 256, 0, cur_stream()>>>(
         bf_ptr(rows_c), inverse.data_ptr<int>(), acc.data_ptr<float>(), n, d);

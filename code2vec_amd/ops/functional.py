@@ -88,6 +88,17 @@ def ce_bwd(logits, lse, labels, scale: float) -> torch.Tensor:
     return ref.ce_bwd(logits, lse, labels, scale)
 
 
+def logits_gemm(code_c, shadow):
+    """logits = code @ shadow^T. On GPU with java14m-like shapes this runs the
+    256x256-tile MFMA kernel (column-major XCD-chunked grid streams the big
+    shadow panel once per XCD); falls back to hipBLASLt elsewhere."""
+    if (backend_for(code_c) == 'hip' and code_c.dtype == torch.bfloat16
+            and code_c.shape[1] % 64 == 0 and code_c.shape[0] >= 256
+            and shadow.shape[0] >= 4096):
+        return hip_ext(True).gemm_bt_v(code_c, shadow, False, 2)
+    return code_c @ shadow.t()
+
+
 def topk(logits, k: int):
     """Per-row top-k (values fp32 desc, tie → lower index; int64 indices)."""
     if backend_for(logits) == 'hip' and logits.dtype == torch.bfloat16 and k <= 32:
