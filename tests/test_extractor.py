@@ -199,3 +199,24 @@ def test_extractor_bridge_roundtrip(extractor, tmp_path, monkeypatch):
     assert len(ctx) == 3
     assert ctx[1] in unhash           # hashed -> original path string
     assert unhash[ctx[1]].startswith('(')
+
+
+def test_extractor_thread_pool_tsan(tmp_path):
+    """Race detection: the extractor's thread pool under
+    ThreadSanitizer must report no data races."""
+    r = subprocess.run(['make', '-C', EXTRACTOR_DIR, 'c2v-extract-tsan'],
+                       capture_output=True, text=True)
+    if r.returncode != 0:
+        pytest.skip('tsan build unavailable: ' + r.stderr[-300:])
+    d = tmp_path / 'proj'
+    d.mkdir()
+    for i in range(12):
+        (d / ('T%d.java' % i)).write_text(
+            'class T%d { int g(int v) { return v * %d; } }' % (i, i))
+    out = subprocess.run(
+        [os.path.join(EXTRACTOR_DIR, 'c2v-extract-tsan'), '--dir', str(d),
+         '--max_path_length', '8', '--max_path_width', '2', '--no_hash',
+         '--num_threads', '8'], capture_output=True, text=True)
+    assert out.returncode == 0
+    assert 'WARNING: ThreadSanitizer' not in out.stderr
+    assert len([l for l in out.stdout.splitlines() if l]) == 12
