@@ -118,12 +118,25 @@ class Code2VecNetwork:
         return sum(self.get_param(n).numel() for n in self.param_names())
 
     def _refresh_shadows(self, only_w: bool = False):
+        """Refresh the compute-dtype shadows IN PLACE. The buffers must keep
+        their identity: a hipGraph-captured step records reads of these exact
+        allocations, so rebinding fresh tensors here would leave every replay
+        computing with stale capture-time weights (found the hard way — the
+        graph-vs-eager test caught a 2e-3 parameter drift)."""
         cd = self.compute_dtype
-        self.w_oi = self.w.t().contiguous().to(cd)   # (out,in): fwd GEMM B^T operand
-        self.w_io = self.w.contiguous().to(cd)       # (in,out): bwd GEMM B^T operand
-        self.a_c = self.a.to(torch.float32)
+        if getattr(self, 'w_oi', None) is None:
+            self.w_oi = self.w.t().contiguous().to(cd)  # (out,in): fwd B^T
+            self.w_io = self.w.contiguous().to(cd)      # (in,out): bwd B^T
+            self.a_c = self.a.to(torch.float32)
+        else:
+            self.w_oi.copy_(self.w.t())
+            self.w_io.copy_(self.w)
+            self.a_c.copy_(self.a)
         if not only_w:
-            self.target_shadow = self.target_table.to(cd)
+            if getattr(self, 'target_shadow', None) is None:
+                self.target_shadow = self.target_table.to(cd)
+            else:
+                self.target_shadow.copy_(self.target_table)
 
     def state_dict(self) -> Dict[str, torch.Tensor]:
         sd = {n: self.get_param(n).cpu() for n in self.param_names()}
