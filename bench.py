@@ -106,7 +106,7 @@ def main():
         from code2vec_amd.parallel.ddp import Reducer
         backend = 'nccl' if device.startswith('cuda') else 'gloo'
         dist.init_process_group(backend=backend, rank=rank, world_size=world_size)
-        reducer = Reducer()
+        reducer = Reducer(assume_equal_shards=True)  # fixed per-rank batch
 
     cfg = make_config(device)
     n_gpus = world_size if distributed else args.gpus

@@ -47,8 +47,13 @@ class Reducer:
     reductions are launched async and waited for right before the matching
     Adam step."""
 
-    def __init__(self, process_group=None):
+    def __init__(self, process_group=None, assume_equal_shards: bool = False):
+        """assume_equal_shards: skip the per-step shard-size exchange (and its
+        host syncs) when every rank contributes identically-shaped sparse
+        grads — true whenever the per-rank batch size is fixed (the bench and
+        steady-state training; ragged tails need False)."""
         self.group = process_group
+        self.assume_equal_shards = assume_equal_shards
         self._handles: Dict[str, object] = {}
 
     @property
@@ -80,6 +85,13 @@ class Reducer:
         ws = self.world_size
         if ws <= 1:
             return ids, rows
+        if self.assume_equal_shards:
+            n = ids.numel()
+            ids_out = [torch.empty_like(ids) for _ in range(ws)]
+            rows_out = [torch.empty_like(rows) for _ in range(ws)]
+            dist.all_gather(ids_out, ids.contiguous(), group=self.group)
+            dist.all_gather(rows_out, rows.contiguous(), group=self.group)
+            return torch.cat(ids_out), torch.cat(rows_out) / ws
         n_local = torch.tensor([ids.numel()], dtype=torch.int64, device=ids.device)
         counts = [torch.zeros_like(n_local) for _ in range(ws)]
         dist.all_gather(counts, n_local, group=self.group)

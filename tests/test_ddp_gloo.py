@@ -76,3 +76,39 @@ def test_dp2_matches_dp1(tmp_path):
         assert torch.allclose(net1.get_param(name), dp2[name], atol=1e-6), name
         assert torch.allclose(net1._adam_m[name], dp2['adam_m.' + name],
                               atol=1e-6), name
+
+
+@pytest.mark.timeout(240)
+def test_dp2_equal_shards_fast_path(tmp_path):
+    """The size-exchange-free all-gather used by the bench must give the
+    same result as the general path."""
+    os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+    init_file = str(tmp_path / 'pg_init2')
+    mp.spawn(_worker_fast, args=(2, init_file, str(tmp_path)), nprocs=2,
+             join=True)
+    torch.manual_seed(7)
+    net1 = Code2VecNetwork(tiny_cfg(), V_TOK, V_PATH, V_TGT, device='cpu')
+    src, pth, tgt, mask, labels = make_batch()
+    for _ in range(2):
+        net1.train_step(src, pth, tgt, mask, labels)
+    dp2 = torch.load(str(tmp_path / 'dp2fast.pt'), weights_only=False)
+    for name in net1.param_names():
+        assert torch.allclose(net1.get_param(name), dp2[name], atol=1e-6), name
+
+
+def _worker_fast(rank, world_size, init_file, result_dir):
+    dist.init_process_group('gloo', init_method='file://' + init_file,
+                            rank=rank, world_size=world_size)
+    from code2vec_amd.parallel.ddp import Reducer
+    torch.manual_seed(7)
+    net = Code2VecNetwork(tiny_cfg(), V_TOK, V_PATH, V_TGT, device='cpu')
+    src, pth, tgt, mask, labels = make_batch()
+    sl = slice(rank * (B // world_size), (rank + 1) * (B // world_size))
+    reducer = Reducer(assume_equal_shards=True)
+    for _ in range(2):
+        net.train_step(src[sl], pth[sl], tgt[sl], mask[sl], labels[sl],
+                       reducer=reducer)
+    if rank == 0:
+        torch.save(net.state_dict(), os.path.join(result_dir, 'dp2fast.pt'))
+    dist.barrier()
+    dist.destroy_process_group()
