@@ -250,14 +250,19 @@ class Code2VecNetwork:
                                     seed_t=st.seed_t)
 
         # Sparse embedding grads: (ids, rows) pairs; under DP these are
-        # all-gathered (not dense-all-reduced) — SURVEY §2.4.
+        # all-gathered (not dense-all-reduced) — SURVEY §2.4. Single-process
+        # runs skip row materialization entirely (grads read straight out of
+        # the d_ctx slices by the scatter-Adam kernel).
         dt = cfg.TOKEN_EMBEDDINGS_SIZE
         tok_ids = torch.cat([src_ids.reshape(-1), tgt_ids.reshape(-1)])
-        tok_rows = torch.cat([d_ctx[:, :dt], d_ctx[:, 2 * dt:]], dim=0)
         path_ids_flat = path_ids.reshape(-1)
-        path_rows = d_ctx[:, dt:2 * dt]
-        tok_ids, tok_rows = reducer.allgather_sparse(tok_ids, tok_rows)
-        path_ids_flat, path_rows = reducer.allgather_sparse(path_ids_flat, path_rows)
+        ctx_direct = reducer.world_size == 1
+        if not ctx_direct:
+            tok_rows = torch.cat([d_ctx[:, :dt], d_ctx[:, 2 * dt:]], dim=0)
+            path_rows = d_ctx[:, dt:2 * dt]
+            tok_ids, tok_rows = reducer.allgather_sparse(tok_ids, tok_rows)
+            path_ids_flat, path_rows = reducer.allgather_sparse(path_ids_flat,
+                                                                path_rows)
 
         # ---- optimizer (TF AdamOptimizer formulation) ----
         self.adam_step += 1
@@ -273,12 +278,22 @@ class Code2VecNetwork:
             tf32 = self._step_t.to(torch.float32)
             st_t = (lr * torch.sqrt(1.0 - torch.exp(tf32 * math.log(b2)))
                     / (1.0 - torch.exp(tf32 * math.log(b1)))).reshape(1)
-        F.adam_sparse_rows_step(self.tok_table, tok_ids, tok_rows,
-                                self._adam_m['tok_table'], self._adam_v['tok_table'],
-                                t, lr, b1, b2, eps, lrt_t=st_t)
-        F.adam_sparse_rows_step(self.path_table, path_ids_flat, path_rows,
-                                self._adam_m['path_table'], self._adam_v['path_table'],
-                                t, lr, b1, b2, eps, lrt_t=st_t)
+        if ctx_direct:
+            F.adam_sparse_rows_from_ctx(
+                self.tok_table, tok_ids, d_ctx, 0, 2 * dt, 2, dt,
+                self._adam_m['tok_table'], self._adam_v['tok_table'],
+                t, lr, b1, b2, eps, lrt_t=st_t)
+            F.adam_sparse_rows_from_ctx(
+                self.path_table, path_ids_flat, d_ctx, dt, dt, 1, dt,
+                self._adam_m['path_table'], self._adam_v['path_table'],
+                t, lr, b1, b2, eps, lrt_t=st_t)
+        else:
+            F.adam_sparse_rows_step(self.tok_table, tok_ids, tok_rows,
+                                    self._adam_m['tok_table'], self._adam_v['tok_table'],
+                                    t, lr, b1, b2, eps, lrt_t=st_t)
+            F.adam_sparse_rows_step(self.path_table, path_ids_flat, path_rows,
+                                    self._adam_m['path_table'], self._adam_v['path_table'],
+                                    t, lr, b1, b2, eps, lrt_t=st_t)
         reducer.wait('w')
         F.adam_dense_step(self.w, d_w, self._adam_m['w'], self._adam_v['w'],
                           t, lr, b1, b2, eps, lrt_t=st_t)

@@ -876,22 +876,34 @@ __global__ void k_hash_compact(const int* __restrict__ tbl_id,
                                int* __restrict__ tbl_cidx,
                                long* __restrict__ uniq_out,
                                int* __restrict__ n_uniq, u32 cap) {
+  // block-aggregated counter: ballots per wave, one atomicAdd per BLOCK per
+  // sweep (a per-wave atomic still serialized ~32K same-address adds/step)
+  __shared__ int wave_cnt[4];
+  __shared__ int block_base;
   const int lane = threadIdx.x & 63;
-  for (u32 s = blockIdx.x * blockDim.x + threadIdx.x; s < cap;
-       s += gridDim.x * blockDim.x) {
+  const int wid = threadIdx.x >> 6;
+  for (u32 s0 = blockIdx.x * blockDim.x; s0 < cap;
+       s0 += gridDim.x * blockDim.x) {
+    const u32 s = s0 + threadIdx.x;
     const int id = (s < cap) ? tbl_id[s] : -1;
     const bool occ = id != -1;
-    // wave-aggregated counter: one atomicAdd per wave, offsets via ballot
     const unsigned long long ball = __ballot(occ);
     const int cnt = __popcll(ball);
-    int base = 0;
-    if (lane == 0 && cnt) base = atomicAdd(n_uniq, cnt);
-    base = __shfl(base, 0, 64);
-    if (occ) {
-      const int off = __popcll(ball & ((1ull << lane) - 1ull));
-      tbl_cidx[s] = base + off;
-      uniq_out[base + off] = id;
+    if (lane == 0) wave_cnt[wid] = cnt;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      const int total = wave_cnt[0] + wave_cnt[1] + wave_cnt[2] + wave_cnt[3];
+      block_base = total ? atomicAdd(n_uniq, total) : 0;
     }
+    __syncthreads();
+    if (occ) {
+      int wave_base = block_base;
+      for (int w = 0; w < wid; ++w) wave_base += wave_cnt[w];
+      const int off = wave_base + __popcll(ball & ((1ull << lane) - 1ull));
+      tbl_cidx[s] = off;
+      uniq_out[off] = id;
+    }
+    __syncthreads();
   }
 }
 
@@ -946,6 +958,26 @@ __global__ void k_adam_rows_dyn(float* __restrict__ p,
     const float pv = p[off] - lr_t * mi / (sqrtf(vi) + eps);
     p[off] = pv;
     if (shadow != nullptr) shadow[off] = f2bf(pv);
+  }
+}
+
+// accumulate token grads straight from the (N,3d) d_ctx layout: logical row
+// i < n_per_seg reads d_ctx[i][off0..off0+d); i >= n_per_seg reads
+// d_ctx[i-n][off1..) — avoids materializing the 105 MB torch.cat of the
+// source/target slices every step
+__global__ void k_rows_accum_ctx(const ushort* __restrict__ d_ctx, int ld,
+                                 int off0, int off1, long n_per_seg,
+                                 int n_seg, const int* __restrict__ inverse,
+                                 float* __restrict__ acc, int d) {
+  const long total = n_per_seg * n_seg * (long)d;
+  for (long s = blockIdx.x * blockDim.x + threadIdx.x; s < total;
+       s += (long)gridDim.x * blockDim.x) {
+    const long r = s / d;
+    const int col = (int)(s % d);
+    const long src_row = (r < n_per_seg) ? r : r - n_per_seg;
+    const int src_off = (r < n_per_seg) ? off0 : off1;
+    const float gv = bf2f(d_ctx[src_row * ld + src_off + col]);
+    if (gv != 0.f) atomicAdd(acc + (long)inverse[r] * d + col, gv);
   }
 }
 
@@ -1422,6 +1454,55 @@ void adam_sparse_rows_hash(torch::Tensor p, torch::Tensor ids,
       lrt_ptr);
 }
 
+void adam_sparse_rows_hash_ctx(torch::Tensor p, torch::Tensor ids,
+                               torch::Tensor d_ctx, int64_t off0, int64_t off1,
+                               int64_t n_seg, int64_t d, torch::Tensor m,
+                               torch::Tensor v, int64_t step, double lr,
+                               double beta1, double beta2, double eps,
+                               torch::Tensor lrt_t) {
+  CHECK_DEV(p); CHECK_CONT(p); CHECK_DEV(d_ctx); CHECK_CONT(d_ctx);
+  const float* lrt_ptr = (lrt_t.defined() && lrt_t.numel() == 1)
+                             ? lrt_t.data_ptr<float>() : nullptr;
+  auto ids_c = ids.contiguous();
+  const long n = ids_c.numel();
+  const long n_per_seg = d_ctx.size(0);
+  TORCH_CHECK(n == n_per_seg * n_seg, "ids length mismatch");
+  const int ld = (int)d_ctx.size(1);
+  const long V = p.size(0);
+  long want = 2 * std::min(n, V);
+  u32 cap = 1;
+  while (cap < (u32)want) cap <<= 1;
+  auto opts_i32 = p.options().dtype(torch::kInt32);
+  auto tbl_id = torch::full({(long)cap}, -1, opts_i32);
+  auto tbl_cidx = torch::full({(long)cap}, -1, opts_i32);
+  auto uniq = torch::empty({n}, p.options().dtype(torch::kInt64));
+  auto n_uniq = torch::zeros({1}, opts_i32);
+  auto inverse = torch::empty({n}, opts_i32);
+  auto acc = torch::empty({n, d}, p.options().dtype(torch::kFloat32));
+  const u32 mask_ = cap - 1;
+  TORCH_CHECK(ids_c.scalar_type() == torch::kInt32, "ctx path expects int32 ids");
+  k_hash_claim<int><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
+      ids_c.data_ptr<int>(), n, tbl_id.data_ptr<int>(), mask_);
+  k_hash_compact<<<grid_1d(cap, 256), 256, 0, cur_stream()>>>(
+      tbl_id.data_ptr<int>(), tbl_cidx.data_ptr<int>(), uniq.data_ptr<long>(),
+      n_uniq.data_ptr<int>(), cap);
+  k_hash_lookup<int><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
+      ids_c.data_ptr<int>(), n, tbl_id.data_ptr<int>(),
+      tbl_cidx.data_ptr<int>(), inverse.data_ptr<int>(), mask_);
+  k_zero_rows_dyn<<<grid_1d(n * d / 4, 256), 256, 0, cur_stream()>>>(
+      acc.data_ptr<float>(), n_uniq.data_ptr<int>(), (int)d);
+  k_rows_accum_ctx<<<grid_1d(n * d, 256), 256, 0, cur_stream()>>>(
+      bf_ptr(d_ctx), ld, (int)off0, (int)off1, n_per_seg, (int)n_seg,
+      inverse.data_ptr<int>(), acc.data_ptr<float>(), (int)d);
+  const float lr_t = (float)(lr * std::sqrt(1.0 - std::pow(beta2, (double)step)) /
+                             (1.0 - std::pow(beta1, (double)step)));
+  k_adam_rows_dyn<<<grid_1d(n * d, 256), 256, 0, cur_stream()>>>(
+      p.data_ptr<float>(), uniq.data_ptr<long>(), acc.data_ptr<float>(),
+      m.data_ptr<float>(), v.data_ptr<float>(), nullptr,
+      n_uniq.data_ptr<int>(), (int)d, lr_t, (float)beta1, (float)beta2,
+      (float)eps, lrt_ptr);
+}
+
 std::vector<torch::Tensor> sampled_ce_fwd(torch::Tensor logits_cand,
                                           torch::Tensor labels,
                                           torch::Tensor sampled,
@@ -1480,6 +1561,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("adam_dense_step", &adam_dense_step);
   mod.def("adam_sparse_rows_step", &adam_sparse_rows_step);
   mod.def("adam_sparse_rows_hash", &adam_sparse_rows_hash);
+  mod.def("adam_sparse_rows_hash_ctx", &adam_sparse_rows_hash_ctx);
   mod.def("sampled_ce_fwd", &sampled_ce_fwd);
   mod.def("sampled_ce_bwd", &sampled_ce_bwd);
   mod.def("topk", &topk);

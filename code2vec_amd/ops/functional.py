@@ -136,6 +136,26 @@ def adam_sparse_rows_step(p, ids, grad_rows, m, v, step: int, lr: float,
                               eps, shadow)
 
 
+def adam_sparse_rows_from_ctx(p, ids, d_ctx, off0, off1, n_seg, d, m, v,
+                              step: int, lr: float, beta1: float,
+                              beta2: float, eps: float, lrt_t=None):
+    """Sparse-row Adam reading grad rows directly from the (N,3d) d_ctx
+    layout (no cat/contiguous materialization). Single-process path; DP uses
+    the gathered-rows variant."""
+    if backend_for(p) == 'hip':
+        hip_ext(True).adam_sparse_rows_hash_ctx(
+            p, ids, d_ctx, int(off0), int(off1), int(n_seg), int(d), m, v,
+            int(step), float(lr), float(beta1), float(beta2), float(eps),
+            lrt_t if lrt_t is not None else torch.empty(0))
+        return
+    n = d_ctx.shape[0]
+    if n_seg == 2:
+        rows = torch.cat([d_ctx[:, off0:off0 + d], d_ctx[:, off1:off1 + d]], 0)
+    else:
+        rows = d_ctx[:, off0:off0 + d]
+    ref.adam_sparse_rows_step(p, ids, rows, m, v, step, lr, beta1, beta2, eps)
+
+
 def sampled_ce_fwd(logits_cand, labels, sampled, corr_true, corr_samp):
     if backend_for(logits_cand) == 'hip':
         return hip_ext(True).sampled_ce_fwd(logits_cand, labels, sampled,

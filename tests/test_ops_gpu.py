@@ -300,3 +300,34 @@ def test_gemm256_vs_matmul(N, M, K):
     assert err / denom < 0.02, 'max err %g vs scale %g' % (err, denom)
     Ct = ext().gemm_bt_v(A, Bt, True, 2)
     assert (Ct.float() - torch.tanh(ref)).abs().max().item() < 0.02
+
+
+def test_adam_sparse_from_ctx_matches_rows_path():
+    """ctx-direct scatter-Adam (reads grad rows straight from the (N,3d)
+    activation layout) ≡ the materialized-rows path."""
+    torch.manual_seed(21)
+    Vr, dt, N = 96, 16, 64
+    d_ctx = randn(N, 3 * dt, dtype=torch.bfloat16, seed=77)
+    src = torch.randint(0, Vr, (N,), dtype=torch.int32).cuda()
+    tgt = torch.randint(0, Vr, (N,), dtype=torch.int32).cuda()
+    ids = torch.cat([src, tgt])
+    p1 = randn(Vr, dt); p2 = p1.clone()
+    m1 = torch.zeros_like(p1); v1 = torch.zeros_like(p1)
+    m2 = m1.clone(); v2 = v1.clone()
+    from code2vec_amd.ops import functional as F
+    F.adam_sparse_rows_from_ctx(p1, ids, d_ctx, 0, 2 * dt, 2, dt, m1, v1,
+                                1, 1e-3, 0.9, 0.999, 1e-8)
+    rows = torch.cat([d_ctx[:, :dt], d_ctx[:, 2 * dt:]], 0).contiguous()
+    F.adam_sparse_rows_step(p2, ids, rows, m2, v2, 1, 1e-3, 0.9, 0.999, 1e-8)
+    assert (p1 - p2).abs().max().item() < 1e-6
+    assert (m1 - m2).abs().max().item() < 1e-6
+    # single-segment (path table) form
+    pp1 = randn(Vr, dt, seed=5); pp2 = pp1.clone()
+    mm1 = torch.zeros_like(pp1); vv1 = torch.zeros_like(pp1)
+    mm2 = mm1.clone(); vv2 = vv1.clone()
+    pids = torch.randint(0, Vr, (N,), dtype=torch.int32).cuda()
+    F.adam_sparse_rows_from_ctx(pp1, pids, d_ctx, dt, dt, 1, dt, mm1, vv1,
+                                1, 1e-3, 0.9, 0.999, 1e-8)
+    F.adam_sparse_rows_step(pp2, pids, d_ctx[:, dt:2 * dt].contiguous(),
+                            mm2, vv2, 1, 1e-3, 0.9, 0.999, 1e-8)
+    assert (pp1 - pp2).abs().max().item() < 1e-6
