@@ -87,6 +87,9 @@ def main():
     ap.add_argument('--batch', type=int, default=1024, help='per-GPU batch size')
     ap.add_argument('--no-graph', action='store_true',
                     help='disable hipGraph-captured stepping (single-GPU only)')
+    ap.add_argument('--sampled-softmax', type=int, default=0,
+                    help='train with sampled softmax over N negatives '
+                         '(BASELINE config 4); 0 = full softmax (default)')
     args = ap.parse_args()
 
     world_size = int(os.environ.get('WORLD_SIZE', '1'))
@@ -109,6 +112,7 @@ def main():
         reducer = Reducer(assume_equal_shards=True)  # fixed per-rank batch
 
     cfg = make_config(device)
+    cfg.SAMPLED_SOFTMAX_SIZE = args.sampled_softmax
     n_gpus = world_size if distributed else args.gpus
     net = Code2VecNetwork(cfg,
                           token_vocab_size=cfg.MAX_TOKEN_VOCAB_SIZE + 1,
@@ -176,7 +180,8 @@ def main():
                 'global_batch': global_batch,
                 'seq_len': cfg.MAX_CONTEXTS,
                 'parallelism': 'dp%d' % n_gpus,
-                'softmax': 'full',
+                'softmax': ('sampled-%d' % args.sampled_softmax)
+                           if args.sampled_softmax else 'full',
                 'stepping': 'hipgraph' if use_graph else 'eager',
                 'token_vocab': cfg.MAX_TOKEN_VOCAB_SIZE,
                 'path_vocab': cfg.MAX_PATH_VOCAB_SIZE,
