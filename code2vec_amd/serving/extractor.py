@@ -15,7 +15,7 @@ import shutil
 import subprocess
 from typing import Dict, List, Tuple
 
-EXTRACTION_API = 'extract'
+
 
 
 def java_string_hashcode(s: str) -> int:

@@ -4,7 +4,6 @@ Run in fp32 with dropout off so the comparison is tight."""
 
 import math
 
-import pytest
 import torch
 
 from code2vec_amd.config import Config

@@ -4,7 +4,6 @@ The op's gradient is checked against autograd on the same candidate-logit
 formulation; the engine path is checked for learning and for sparse-only
 target-table updates."""
 
-import math
 
 import torch
 
