@@ -101,9 +101,13 @@ class Code2VecNetwork:
         # sees fresh values (a host scalar would be baked into the capture).
         self._seed_t = None
         self._step_t = None
+        self._side_stream = None
         if self.device.type == 'cuda':
             self._seed_t = torch.zeros(1, dtype=torch.int64, device=self.device)
             self._step_t = torch.zeros(1, dtype=torch.int32, device=self.device)
+            # side stream for the independent target-table chain
+            # (d_target GEMM -> all-reduce -> Adam) overlapping main backward
+            self._side_stream = torch.cuda.Stream()
 
     # ---- parameters ----
 
@@ -234,11 +238,49 @@ class Code2VecNetwork:
             loss_rows, lse = F.ce_fwd(logits, labels)
             loss = loss_rows.float().mean()
             d_logits = F.ce_bwd(logits, lse, labels, 1.0 / B)
-            # Dense target-table grad first: it is the big all-reduce (≈200 MB
-            # bf16 on java14m), launched async to overlap the rest of backward.
-            d_target = (d_logits.t() @ code_c)                   # (V,D)
-            reducer.allreduce_dense('target_table', d_target)
             d_code = (d_logits @ self.target_shadow).float()     # (B,D)
+            d_target = None  # see target-chain dispatch below
+
+        side_done = None
+        if not use_sampled:
+            # lr_t must exist before the side chain runs its Adam step
+            self.adam_step += 1
+            t_pre, lr = self.adam_step, cfg.ADAM_LR
+            b1, b2, eps = cfg.ADAM_BETA1, cfg.ADAM_BETA2, cfg.ADAM_EPS
+            st_t_pre = None
+            if self._step_t is not None:
+                self._step_t.add_(1)
+                tf32 = self._step_t.to(torch.float32)
+                st_t_pre = (lr * torch.sqrt(1.0 - torch.exp(tf32 * math.log(b2)))
+                            / (1.0 - torch.exp(tf32 * math.log(b1)))).reshape(1)
+            if self._side_stream is not None:
+                # the whole target chain — the big GEMM, the (DP) all-reduce
+                # and the 100M-param Adam — is independent of the rest of
+                # backward; run it concurrently on the side stream
+                ev = torch.cuda.Event()
+                ev.record()
+                with torch.cuda.stream(self._side_stream):
+                    self._side_stream.wait_event(ev)
+                    d_target = d_logits.t() @ code_c             # (V,D)
+                    reducer.allreduce_dense('target_table', d_target)
+                    reducer.wait('target_table')
+                    F.adam_dense_step(self.target_table, d_target,
+                                      self._adam_m['target_table'],
+                                      self._adam_v['target_table'],
+                                      t_pre, lr, cfg.ADAM_BETA1,
+                                      cfg.ADAM_BETA2, cfg.ADAM_EPS,
+                                      shadow=self.target_shadow,
+                                      lrt_t=st_t_pre)
+                    side_done = torch.cuda.Event()
+                    side_done.record()
+                if not torch.cuda.is_current_stream_capturing():
+                    d_logits.record_stream(self._side_stream)
+                    code_c.record_stream(self._side_stream)
+                    if st_t_pre is not None:
+                        st_t_pre.record_stream(self._side_stream)
+            else:
+                d_target = (d_logits.t() @ code_c)
+                reducer.allreduce_dense('target_table', d_target)
 
         d_comb3, d_a = F.attention_bwd(st.comb.reshape(B, C, D), self.a_c,
                                        st.alpha, d_code)
@@ -265,19 +307,19 @@ class Code2VecNetwork:
                                                                 path_rows)
 
         # ---- optimizer (TF AdamOptimizer formulation) ----
-        self.adam_step += 1
+        if use_sampled:
+            self.adam_step += 1
+            st_t = None
+            if self._step_t is not None:
+                self._step_t.add_(1)
+                tf32 = self._step_t.to(torch.float32)
+                lr0, b10, b20 = cfg.ADAM_LR, cfg.ADAM_BETA1, cfg.ADAM_BETA2
+                st_t = (lr0 * torch.sqrt(1.0 - torch.exp(tf32 * math.log(b20)))
+                        / (1.0 - torch.exp(tf32 * math.log(b10)))).reshape(1)
+        else:
+            st_t = st_t_pre  # counters already advanced pre-backward
         t, lr = self.adam_step, cfg.ADAM_LR
         b1, b2, eps = cfg.ADAM_BETA1, cfg.ADAM_BETA2, cfg.ADAM_EPS
-
-        st_t = None
-        if self._step_t is not None:
-            self._step_t.add_(1)            # in-graph advance (capture-safe)
-            # precompute the bias-corrected lr_t on device. Use log-space so
-            # the base is a device tensor (scalar**tensor would create a CPU
-            # tensor and a pageable H2D copy — a sync, and illegal in capture)
-            tf32 = self._step_t.to(torch.float32)
-            st_t = (lr * torch.sqrt(1.0 - torch.exp(tf32 * math.log(b2)))
-                    / (1.0 - torch.exp(tf32 * math.log(b1)))).reshape(1)
         if ctx_direct:
             F.adam_sparse_rows_from_ctx(
                 self.tok_table, tok_ids, d_ctx, 0, 2 * dt, 2, dt,
@@ -307,6 +349,9 @@ class Code2VecNetwork:
                                     self._adam_v['target_table'],
                                     t, lr, b1, b2, eps,
                                     shadow=self.target_shadow, lrt_t=st_t)
+        elif side_done is not None:
+            # join the side-stream target chain
+            torch.cuda.current_stream().wait_event(side_done)
         else:
             reducer.wait('target_table')
             F.adam_dense_step(self.target_table, d_target,
