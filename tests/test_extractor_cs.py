@@ -1,0 +1,101 @@
+"""c2v-extract-cs contract tests: Roslyn-kind path grammar, variable
+grouping, METHOD_NAME masking, comment contexts, normalization
+(reference CSharpExtractor semantics; divergences in c2v_extract_cs.cpp
+header)."""
+
+import os
+import subprocess
+
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+EXTRACTOR_DIR = os.path.join(ROOT, 'extractor')
+BIN = os.path.join(EXTRACTOR_DIR, 'c2v-extract-cs')
+
+
+@pytest.fixture(scope='module')
+def cs_extractor():
+    if not os.path.isfile(BIN):
+        r = subprocess.run(['make', '-C', EXTRACTOR_DIR, 'c2v-extract-cs'],
+                           capture_output=True, text=True)
+        if r.returncode != 0:
+            pytest.skip('cannot build c2v-extract-cs: ' + r.stderr[-400:])
+    return BIN
+
+
+def run_cs(cs_extractor, code, tmp_path, extra=('--no_hash',)):
+    src = tmp_path / 'T.cs'
+    src.write_text(code)
+    out = subprocess.run([cs_extractor, '--path', str(src), *extra],
+                         capture_output=True, text=True)
+    assert out.returncode == 0, out.stderr
+    return [l for l in out.stdout.splitlines() if l]
+
+
+def test_basic_method(cs_extractor, tmp_path):
+    lines = run_cs(cs_extractor, '''
+class C {
+    public int AddValue(int x) { return x + 1; }
+}''', tmp_path)
+    assert len(lines) == 1
+    parts = lines[0].split(' ')
+    assert parts[0] == 'add|value'
+    joined = ' '.join(parts[1:])
+    assert 'METHOD_NAME' in joined
+    assert 'AddExpression' in joined
+    assert 'PredefinedType^MethodDeclaration' in joined
+    # numeric whitelist keeps 1
+    assert ',1' in joined or '1,' in joined
+
+
+def test_member_access_child_ids(cs_extractor, tmp_path):
+    lines = run_cs(cs_extractor, '''
+class C {
+    void Go(string s) { s.Trim(); }
+}''', tmp_path)
+    joined = lines[0]
+    # member-access / invocation parents add child ids
+    assert 'SimpleMemberAccessExpression0' in joined or \
+           'IdentifierName0' in joined
+
+
+def test_comments_become_contexts(cs_extractor, tmp_path):
+    lines = run_cs(cs_extractor, '''
+class C {
+    // compute the running total now
+    int F(int a) { return a; }
+}''', tmp_path)
+    joined = lines[0]
+    assert ',COMMENT,' in joined
+    assert 'compute|the|running|total|now' in joined
+
+
+def test_var_not_a_leaf(cs_extractor, tmp_path):
+    lines = run_cs(cs_extractor, '''
+class C {
+    int F() { var q = 3; return q; }
+}''', tmp_path)
+    joined = lines[0]
+    assert ',var' not in joined and 'var,' not in joined
+
+
+def test_hash_mode_is_deterministic_int(cs_extractor, tmp_path):
+    code = 'class C { int F(int a) { return a; } }'
+    hashed1 = run_cs(cs_extractor, code, tmp_path, extra=())
+    hashed2 = run_cs(cs_extractor, code, tmp_path, extra=())
+    assert hashed1 == hashed2
+    ctxs = [c for c in hashed1[0].split(' ')[1:] if c and 'COMMENT' not in c]
+    for c in ctxs:
+        int(c.split(',')[1])  # 32-bit int hash
+
+
+def test_dir_mode(cs_extractor, tmp_path):
+    d = tmp_path / 'proj'
+    d.mkdir()
+    for i in range(6):
+        (d / ('F%d.cs' % i)).write_text(
+            'class F%d { int M%d(int v) { return v * %d; } }' % (i, i, i))
+    out = subprocess.run([cs_extractor, '--path', str(d), '--no_hash',
+                          '--threads', '4'], capture_output=True, text=True)
+    lines = [l for l in out.stdout.splitlines() if l]
+    assert len(lines) == 6
