@@ -234,8 +234,9 @@ class Code2VecNetwork:
             d_code = (d_cand @ w_cand).float()                   # (B,D)
             d_target = None
         else:
-            logits = F.logits_gemm(code_c, self.target_shadow)
-            loss_rows, lse = F.ce_fwd(logits, labels)
+            logits, loss_rows, lse = F.logits_ce_fused(code_c,
+                                                       self.target_shadow,
+                                                       labels)
             loss = loss_rows.float().mean()
             d_logits = F.ce_bwd(logits, lse, labels, 1.0 / B)
             d_code = (d_logits @ self.target_shadow).float()     # (B,D)

@@ -290,11 +290,17 @@ __global__ void k_gemm_bt(const ushort* __restrict__ A,
 #define G256_BN 256
 #define G256_BK 64
 
-template <bool TANH>
+// CE_PART: fused CE-forward partials — each (row-tile, col-tile) block also
+// emits (rowmax, sumexp) over its 256-col slice into partials[(row)*T+tile_m]
+// so the separate 535 MB ce_fwd pass over the logits disappears;
+// k_ce_reduce_partials folds the T partials per row into (loss, lse).
+template <bool TANH, bool CE_PART = false>
 __launch_bounds__(512, 1)
 __global__ void k_gemm256_bt(const ushort* __restrict__ A,
                              const ushort* __restrict__ Bt,
-                             ushort* __restrict__ C, int N, int M, int K) {
+                             ushort* __restrict__ C, int N, int M, int K,
+                             float* __restrict__ partials = nullptr,
+                             int n_col_tiles = 0) {
   // LDS: 2 buffers x (A[256][64] + B[256][64]) bf16 = 128 KiB
   extern __shared__ ushort lds256[];
   // buffer b: A at b*32768, B at b*32768+16384 (ushort units)
@@ -397,22 +403,85 @@ __global__ void k_gemm256_bt(const ushort* __restrict__ A,
     cur ^= 1;
   }
 
-  // epilogue: C/D layout col=lane&15, row=(lane>>4)*4+reg
+  // epilogue: C/D layout col=lane&15, row=(lane>>4)*4+reg. Iterated (m,r,n)
+  // so the CE variant can reduce each row's 64-col slice across the 16-lane
+  // group that holds it (lanes 0-15 share rows; shfl width 16).
 #pragma unroll
   for (int m = 0; m < 8; ++m) {
 #pragma unroll
-    for (int n = 0; n < 4; ++n) {
-      const int col = col0 + wcol + n * 16 + (lane & 15);
-      if (col >= M) continue;
+    for (int r = 0; r < 4; ++r) {
+      const int row = row0 + wrow + m * 16 + (lane >> 4) * 4 + r;
+      float lmax = -3.0e38f;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int row = row0 + wrow + m * 16 + (lane >> 4) * 4 + r;
-        if (row >= N) continue;
+      for (int n = 0; n < 4; ++n) {
+        const int col = col0 + wcol + n * 16 + (lane & 15);
         float v = acc[m][n][r];
         if (TANH) v = tanhf(v);
-        C[(long)row * M + col] = f2bf(v);
+        if (col < M && row < N) C[(long)row * M + col] = f2bf(v);
+        if (CE_PART && col < M) lmax = fmaxf(lmax, v);
+      }
+      if (CE_PART) {
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1)
+          lmax = fmaxf(lmax, __shfl_xor(lmax, off, 16));
+        float lsum = 0.f;
+#pragma unroll
+        for (int n = 0; n < 4; ++n) {
+          const int col = col0 + wcol + n * 16 + (lane & 15);
+          if (col < M) lsum += __expf(acc[m][n][r] - lmax);
+        }
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1)
+          lsum += __shfl_xor(lsum, off, 16);
+        if ((lane & 15) == 0 && row < N) {
+          float* p = partials + 2 * ((long)row * n_col_tiles + tile_m);
+          p[0] = lmax;
+          p[1] = lsum;
+        }
       }
     }
+  }
+}
+
+// fold the per-(row, col-tile) (max, sumexp) partials into (loss, lse)
+__launch_bounds__(256)
+__global__ void k_ce_reduce_partials(const float* __restrict__ partials,
+                                     const ushort* __restrict__ logits,
+                                     const long* __restrict__ labels,
+                                     float* __restrict__ loss,
+                                     float* __restrict__ lse, int N, int T,
+                                     int V) {
+  __shared__ float sm[4], ss[4];
+  const int row = blockIdx.x;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  float m = -3.0e38f, s = 0.f;
+  for (int t = tid; t < T; t += blockDim.x) {
+    const float pm = partials[2 * ((long)row * T + t)];
+    const float psum = partials[2 * ((long)row * T + t) + 1];
+    if (pm > m) { s = s * __expf(m - pm) + psum; m = pm; }
+    else s += psum * __expf(pm - m);
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    const float mo = __shfl_down(m, off, 64);
+    const float so = __shfl_down(s, off, 64);
+    if (mo > m) { s = s * __expf(m - mo) + so; m = mo; }
+    else s += so * __expf(mo - m);
+  }
+  if (lane == 0) { sm[wid] = m; ss[wid] = s; }
+  __syncthreads();
+  if (tid == 0) {
+    float M2 = sm[0], S2 = ss[0];
+#pragma unroll
+    for (int w = 1; w < 4; ++w) {
+      if (sm[w] > M2) { S2 = S2 * __expf(M2 - sm[w]) + ss[w]; M2 = sm[w]; }
+      else S2 += ss[w] * __expf(sm[w] - M2);
+    }
+    const float l = M2 + __logf(S2);
+    lse[row] = l;
+    loss[row] = l - bf2f(logits[(long)row * V + labels[row]]);
   }
 }
 
@@ -1223,6 +1292,37 @@ torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor Bt, bool tanh_epilogue,
   return C;
 }
 
+std::vector<torch::Tensor> logits_ce_fused(torch::Tensor code,
+                                           torch::Tensor shadow,
+                                           torch::Tensor labels) {
+  CHECK_DEV(code); CHECK_CONT(code); CHECK_DEV(shadow); CHECK_CONT(shadow);
+  const int N = code.size(0), K = code.size(1), M = shadow.size(0);
+  TORCH_CHECK(shadow.size(1) == K && K % G256_BK == 0);
+  auto labels_c = labels.contiguous();
+  auto C = torch::empty({N, M}, code.options());
+  const int n_tiles = (N + G256_BM - 1) / G256_BM;
+  const int m_tiles = (M + G256_BN - 1) / G256_BN;
+  auto partials = torch::empty({(long)N, (long)m_tiles, 2},
+                               code.options().dtype(torch::kFloat32));
+  auto loss = torch::empty({N}, code.options().dtype(torch::kFloat32));
+  auto lse = torch::empty({N}, code.options().dtype(torch::kFloat32));
+  const size_t lds = 4 * 16384 * 2;
+  static bool configured = false;
+  if (!configured) {
+    (void)hipFuncSetAttribute((const void*)k_gemm256_bt<false, true>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              (int)lds);
+    configured = true;
+  }
+  k_gemm256_bt<false, true><<<n_tiles * m_tiles, 512, lds, cur_stream()>>>(
+      bf_ptr(code), bf_ptr(shadow), bf_ptr_mut(C), N, M, K,
+      partials.data_ptr<float>(), m_tiles);
+  k_ce_reduce_partials<<<N, 256, 0, cur_stream()>>>(
+      partials.data_ptr<float>(), bf_ptr(C), labels_c.data_ptr<long>(),
+      loss.data_ptr<float>(), lse.data_ptr<float>(), N, m_tiles, M);
+  return {C, loss, lse};
+}
+
 torch::Tensor transform_tanh_fwd(torch::Tensor ctx, torch::Tensor w_oi) {
   return gemm_bt(ctx, w_oi, true, 1);
 }
@@ -1553,6 +1653,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("transform_tanh_fwd", &transform_tanh_fwd);
   mod.def("gemm_bt_bf16", &gemm_bt_bf16);
   mod.def("gemm_bt_v", &gemm_bt_v);
+  mod.def("logits_ce_fused", &logits_ce_fused);
   mod.def("tanh_bwd_mul", &tanh_bwd_mul);
   mod.def("attention_fwd", &attention_fwd);
   mod.def("attention_bwd", &attention_bwd);

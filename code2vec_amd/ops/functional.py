@@ -99,6 +99,22 @@ def logits_gemm(code_c, shadow):
     return code_c @ shadow.t()
 
 
+def logits_ce_fused(code_c, shadow, labels):
+    """Fused K8+K9 forward: the 256-tile logits GEMM also emits per-tile
+    (rowmax, sumexp) partials, folded into (loss, lse) by a small reduce —
+    no separate full-vocab CE pass. Returns (logits, loss_rows, lse).
+    The lse comes from the fp32 accumulator (pre-bf16-rounding), which is
+    closer to the reference's fp32 TF math than the unfused bf16 path."""
+    if (backend_for(code_c) == 'hip' and code_c.dtype == torch.bfloat16
+            and code_c.shape[1] % 64 == 0 and code_c.shape[0] >= 256
+            and shadow.shape[0] >= 4096):
+        logits, loss, lse = hip_ext(True).logits_ce_fused(code_c, shadow, labels)
+        return logits, loss, lse
+    logits = logits_gemm(code_c, shadow)
+    loss, lse = ce_fwd(logits, labels)
+    return logits, loss, lse
+
+
 def topk(logits, k: int):
     """Per-row top-k (values fp32 desc, tie → lower index; int64 indices)."""
     if backend_for(logits) == 'hip' and logits.dtype == torch.bfloat16 and k <= 32:

@@ -331,3 +331,20 @@ def test_adam_sparse_from_ctx_matches_rows_path():
     F.adam_sparse_rows_step(pp2, pids, d_ctx[:, dt:2 * dt].contiguous(),
                             mm2, vv2, 1, 1e-3, 0.9, 0.999, 1e-8)
     assert (pp1 - pp2).abs().max().item() < 1e-6
+
+
+def test_logits_ce_fused_matches_unfused():
+    """Fused K8+K9 forward vs 256-tile GEMM + reference CE. The fused lse is
+    computed from the fp32 accumulator (pre-bf16 rounding), so tolerances
+    span the bf16 rounding of the unfused path."""
+    torch.manual_seed(31)
+    B, D, V = 1024, 384, 261246
+    code = randn(B, D, dtype=torch.bfloat16, scale=0.3, seed=31)
+    shadow = randn(V, D, dtype=torch.bfloat16, scale=0.3)
+    labels = torch.randint(0, V, (B,)).cuda()
+    logits_f, loss_f, lse_f = ext().logits_ce_fused(code, shadow, labels)
+    logits_u = ext().gemm_bt_v(code, shadow, False, 2)
+    assert torch.equal(logits_f, logits_u)
+    loss_u, lse_u = R.ce_fwd(logits_u, labels)
+    assert (lse_f - lse_u).abs().max().item() < 5e-3
+    assert (loss_f - loss_u).abs().max().item() < 1e-2
