@@ -333,10 +333,11 @@ def test_adam_sparse_from_ctx_matches_rows_path():
     assert (pp1 - pp2).abs().max().item() < 1e-6
 
 
-def test_logits_ce_fused_matches_unfused():
-    """Fused K8+K9 forward vs 256-tile GEMM + reference CE. The fused lse is
-    computed from the fp32 accumulator (pre-bf16 rounding), so tolerances
-    span the bf16 rounding of the unfused path."""
+def test_logits_ce_fused_matches_fp32_oracle():
+    """Fused K8+K9 forward vs an fp32 matmul+logsumexp oracle. The fused lse
+    comes from the fp32 MFMA accumulator (pre-bf16 rounding), so it must track
+    the fp32 oracle tightly — NOT the bf16-rounded unfused path, whose lse
+    differs by up to bf16-eps at logit scale (~0.06 at |logit|≈8)."""
     torch.manual_seed(31)
     B, D, V = 1024, 384, 261246
     code = randn(B, D, dtype=torch.bfloat16, scale=0.3, seed=31)
@@ -344,7 +345,10 @@ def test_logits_ce_fused_matches_unfused():
     labels = torch.randint(0, V, (B,)).cuda()
     logits_f, loss_f, lse_f = ext().logits_ce_fused(code, shadow, labels)
     logits_u = ext().gemm_bt_v(code, shadow, False, 2)
-    assert torch.equal(logits_f, logits_u)
-    loss_u, lse_u = R.ce_fwd(logits_u, labels)
-    assert (lse_f - lse_u).abs().max().item() < 5e-3
-    assert (loss_f - loss_u).abs().max().item() < 1e-2
+    assert torch.equal(logits_f, logits_u)   # same bf16 logits out
+    ref32 = code.float() @ shadow.float().t()
+    lse32 = torch.logsumexp(ref32, dim=1)
+    assert (lse_f - lse32).abs().max().item() < 2e-3
+    # loss = fp32 lse − bf16-rounded label logit (the stored logits feed bwd)
+    picked = logits_f.float().gather(1, labels.reshape(-1, 1)).squeeze(1)
+    assert (loss_f - (lse32 - picked)).abs().max().item() < 3e-3
