@@ -105,9 +105,11 @@ def logits_ce_fused(code_c, shadow, labels):
     no separate full-vocab CE pass. Returns (logits, loss_rows, lse).
     The lse comes from the fp32 accumulator (pre-bf16-rounding), which is
     closer to the reference's fp32 TF math than the unfused bf16 path."""
+    import os as _os
     if (backend_for(code_c) == 'hip' and code_c.dtype == torch.bfloat16
             and code_c.shape[1] % 64 == 0 and code_c.shape[0] >= 256
-            and shadow.shape[0] >= 4096):
+            and shadow.shape[0] >= 4096
+            and _os.environ.get('C2V_FUSED_CE', '1') == '1'):
         logits, loss, lse = hip_ext(True).logits_ce_fused(code_c, shadow, labels)
         return logits, loss, lse
     logits = logits_gemm(code_c, shadow)
