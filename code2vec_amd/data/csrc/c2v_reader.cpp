@@ -20,8 +20,63 @@
 
 namespace {
 
+// allocation-free open-addressing string->int map: lookups hash a
+// string_view and compare against interned keys (the std::unordered_map
+// path allocated a std::string per lookup — 600 allocations per data row)
+struct StrMap {
+  struct Slot {
+    uint64_t h = 0;
+    int32_t key = -1;   // index into keys
+    int32_t val = 0;
+  };
+  std::vector<std::string> keys;
+  std::vector<Slot> slots;
+  uint64_t mask = 0;
+
+  static uint64_t hash(std::string_view s) {
+    uint64_t h = 1469598103934665603ull;  // FNV-1a 64
+    for (unsigned char c : s) {
+      h ^= c;
+      h *= 1099511628211ull;
+    }
+    return h | 1;  // nonzero
+  }
+
+  void build(const std::unordered_map<std::string, int>& m) {
+    size_t cap = 16;
+    while (cap < m.size() * 2) cap <<= 1;
+    slots.assign(cap, Slot{});
+    keys.reserve(m.size());
+    mask = cap - 1;
+    for (const auto& [k, v] : m) {
+      const uint64_t h = hash(k);
+      uint64_t s = h & mask;
+      while (slots[s].key != -1) s = (s + 1) & mask;
+      slots[s].h = h;
+      slots[s].key = (int32_t)keys.size();
+      slots[s].val = v;
+      keys.push_back(k);
+    }
+  }
+
+  int look(std::string_view s, int dflt) const {
+    const uint64_t h = hash(s);
+    uint64_t i = h & mask;
+    for (;;) {
+      const Slot& sl = slots[i];
+      if (sl.key == -1) return dflt;
+      if (sl.h == h && keys[sl.key] == s) return sl.val;
+      i = (i + 1) & mask;
+    }
+  }
+};
+
+}  // namespace
+
+namespace {
+
 struct Parser {
-  std::unordered_map<std::string, int> tok, path, tgt;
+  StrMap tok, path, tgt;
   int tok_pad = 0, tok_oov = 0, path_pad = 0, path_oov = 0, tgt_oov = 0;
   int max_contexts = 200;
   int n_threads = 6;  // reference READER_NUM_PARALLEL_BATCHES default
@@ -31,15 +86,16 @@ struct Parser {
          const std::unordered_map<std::string, int>& tgt_map, int tok_pad_,
          int tok_oov_, int path_pad_, int path_oov_, int tgt_oov_,
          int max_contexts_, int n_threads_)
-      : tok(tok_map), path(path_map), tgt(tgt_map), tok_pad(tok_pad_),
-        tok_oov(tok_oov_), path_pad(path_pad_), path_oov(path_oov_),
-        tgt_oov(tgt_oov_), max_contexts(max_contexts_),
-        n_threads(std::max(1, n_threads_)) {}
+      : tok_pad(tok_pad_), tok_oov(tok_oov_), path_pad(path_pad_),
+        path_oov(path_oov_), tgt_oov(tgt_oov_), max_contexts(max_contexts_),
+        n_threads(std::max(1, n_threads_)) {
+    tok.build(tok_map);
+    path.build(path_map);
+    tgt.build(tgt_map);
+  }
 
-  inline int look(const std::unordered_map<std::string, int>& m,
-                  std::string_view s, int oov) const {
-    auto it = m.find(std::string(s));
-    return it == m.end() ? oov : it->second;
+  inline int look(const StrMap& m, std::string_view s, int oov) const {
+    return m.look(s, oov);
   }
 
   // Parse one line into the row buffers. Returns target index.
@@ -94,6 +150,51 @@ struct Parser {
     return target_idx;
   }
 
+  // Parse a whole text buffer (zero-copy from python bytes): lines are
+  // split here and parsed thread-parallel — no per-line python objects.
+  py::tuple parse_buffer(py::bytes data) const {
+    std::string_view buf = std::string_view(data);
+    // line offsets
+    std::vector<std::pair<size_t, size_t>> lines;
+    size_t start = 0;
+    while (start < buf.size()) {
+      size_t nl = buf.find('\n', start);
+      size_t end = (nl == std::string_view::npos) ? buf.size() : nl;
+      if (end > start) lines.push_back({start, end - start});
+      start = end + 1;
+    }
+    const int64_t B = (int64_t)lines.size();
+    const int64_t C = max_contexts;
+    auto opts = torch::TensorOptions().dtype(torch::kInt32);
+    auto src = torch::empty({B, C}, opts);
+    auto pth = torch::empty({B, C}, opts);
+    auto tgt = torch::empty({B, C}, opts);
+    auto mask = torch::empty({B, C}, torch::TensorOptions().dtype(torch::kFloat32));
+    auto tidx = torch::empty({B}, torch::TensorOptions().dtype(torch::kInt64));
+    int* src_p = src.data_ptr<int>();
+    int* pth_p = pth.data_ptr<int>();
+    int* tgt_p = tgt.data_ptr<int>();
+    float* mask_p = mask.data_ptr<float>();
+    int64_t* tidx_p = tidx.data_ptr<int64_t>();
+    {
+      py::gil_scoped_release release;
+      const int nt = (int)std::min<int64_t>(n_threads, std::max<int64_t>(1, B));
+      std::atomic<int64_t> next(0);
+      auto work = [&]() {
+        int64_t i;
+        while ((i = next.fetch_add(1)) < B) {
+          tidx_p[i] = parse_line(buf.substr(lines[i].first, lines[i].second),
+                                 src_p + i * C, pth_p + i * C, tgt_p + i * C,
+                                 mask_p + i * C);
+        }
+      };
+      std::vector<std::thread> pool;
+      for (int t = 0; t < nt; ++t) pool.emplace_back(work);
+      for (auto& th : pool) th.join();
+    }
+    return py::make_tuple(src, pth, tgt, mask, tidx);
+  }
+
   // Parse a batch of lines (thread-parallel across rows).
   py::tuple parse_batch(const std::vector<std::string>& lines) const {
     const int64_t B = (int64_t)lines.size();
@@ -143,5 +244,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
                     const std::unordered_map<std::string, int>&,
                     const std::unordered_map<std::string, int>&, int, int, int,
                     int, int, int, int>())
-      .def("parse_batch", &Parser::parse_batch);
+      .def("parse_batch", &Parser::parse_batch)
+      .def("parse_buffer", &Parser::parse_buffer);
 }

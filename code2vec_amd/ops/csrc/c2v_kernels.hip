@@ -434,7 +434,10 @@ __global__ void k_gemm256_bt(const ushort* __restrict__ A,
         for (int off = 1; off < 16; off <<= 1)
           lsum += __shfl_xor(lsum, off, 16);
         if ((lane & 15) == 0 && row < N) {
-          float* p = partials + 2 * ((long)row * n_col_tiles + tile_m);
+          // per-wave-column slot: 4 waves cover the 256-col tile, each owns
+          // a quarter (a shared slot would race and drop 192 columns)
+          float* p = partials +
+                     2 * (((long)row * n_col_tiles + tile_m) * 4 + (wid & 3));
           p[0] = lmax;
           p[1] = lsum;
         }
@@ -1302,7 +1305,7 @@ std::vector<torch::Tensor> logits_ce_fused(torch::Tensor code,
   auto C = torch::empty({N, M}, code.options());
   const int n_tiles = (N + G256_BM - 1) / G256_BM;
   const int m_tiles = (M + G256_BN - 1) / G256_BN;
-  auto partials = torch::empty({(long)N, (long)m_tiles, 2},
+  auto partials = torch::empty({(long)N, (long)m_tiles, 4, 2},
                                code.options().dtype(torch::kFloat32));
   auto loss = torch::empty({N}, code.options().dtype(torch::kFloat32));
   auto lse = torch::empty({N}, code.options().dtype(torch::kFloat32));
@@ -1319,7 +1322,7 @@ std::vector<torch::Tensor> logits_ce_fused(torch::Tensor code,
       partials.data_ptr<float>(), m_tiles);
   k_ce_reduce_partials<<<N, 256, 0, cur_stream()>>>(
       partials.data_ptr<float>(), bf_ptr(C), labels_c.data_ptr<long>(),
-      loss.data_ptr<float>(), lse.data_ptr<float>(), N, m_tiles, M);
+      loss.data_ptr<float>(), lse.data_ptr<float>(), N, m_tiles * 4, M);
   return {C, loss, lse};
 }
 
