@@ -106,6 +106,14 @@ class Code2VecModel(Code2VecModelBase):
         window_examples = 0
         start = time.time()
         multi_batch_start = start
+        scalar_log = None
+        if cfg.USE_TENSORBOARD:
+            # tensorboard isn't installed in this environment; the flag writes
+            # the same scalars to a CSV instead (batch, avg_loss, samples/sec)
+            scalar_path = (cfg.MODEL_SAVE_PATH or 'model') + '.scalars.csv'
+            os.makedirs(os.path.dirname(scalar_path) or '.', exist_ok=True)
+            scalar_log = open(scalar_path, 'a')
+            scalar_log.write('batch,avg_loss,samples_per_sec\n')
         from ..data.prefetcher import BatchPrefetcher
         prefetcher = BatchPrefetcher(reader.iter_batches(), device,
                                      depth=cfg.READER_QUEUE_DEPTH)
@@ -134,6 +142,11 @@ class Code2VecModel(Code2VecModelBase):
                         / self._throughput_ewma
                     msg += ', epoch ETA: %dm%02ds' % (eta_sec // 60, eta_sec % 60)
                 self.log(msg)
+                if scalar_log is not None:
+                    scalar_log.write('%d,%f,%f\n' % (
+                        batch_num, sum_loss / cfg.NUM_BATCHES_TO_LOG_PROGRESS,
+                        throughput))
+                    scalar_log.flush()
                 sum_loss = 0.0
                 window_examples = 0
                 multi_batch_start = time.time()
@@ -150,6 +163,8 @@ class Code2VecModel(Code2VecModelBase):
                     if results is not None:
                         self.log('After %d epochs -- %s' % (epoch_num, str(results)))
 
+        if scalar_log is not None:
+            scalar_log.close()
         elapsed = int(time.time() - start)
         self.log('Done training')
         self.log('Training time: %sH:%sM:%sS'
