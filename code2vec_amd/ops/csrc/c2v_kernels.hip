@@ -848,7 +848,7 @@ __global__ void k_sampled_ce_bwd(const ushort* __restrict__ L,
 // lazy update of p/m/v.
 // ---------------------------------------------------------------------------
 
-template <typename G>
+template <typename G, bool NT = false>
 __global__ void k_adam_dense(float* __restrict__ p, const G* __restrict__ g,
                              float* __restrict__ m, float* __restrict__ v,
                              ushort* __restrict__ shadow, long n, float lr_t,
@@ -884,12 +884,28 @@ __global__ void k_adam_dense(float* __restrict__ p, const G* __restrict__ g,
       pp[j] = pn;
       sh[j] = f2bf(pn);
     }
-    *reinterpret_cast<float4*>(p + i) = pv;
-    *reinterpret_cast<float4*>(m + i) = mv;
-    *reinterpret_cast<float4*>(v + i) = vv;
-    if (shadow != nullptr)
-      *reinterpret_cast<ulonglong1*>(shadow + i) =
-          *reinterpret_cast<ulonglong1*>(sh);
+    if (NT) {
+      // streaming stores: p/m/v/shadow are not re-read in this kernel and
+      // the tables far exceed L2 — keep them out of the caches.
+      // (ext_vector types: HIP_vector_type isn't accepted by the builtin)
+      using f32x4v = __attribute__((ext_vector_type(4))) float;
+      __builtin_nontemporal_store(*reinterpret_cast<f32x4v*>(&pv),
+                                  reinterpret_cast<f32x4v*>(p + i));
+      __builtin_nontemporal_store(*reinterpret_cast<f32x4v*>(&mv),
+                                  reinterpret_cast<f32x4v*>(m + i));
+      __builtin_nontemporal_store(*reinterpret_cast<f32x4v*>(&vv),
+                                  reinterpret_cast<f32x4v*>(v + i));
+      if (shadow != nullptr)
+        __builtin_nontemporal_store(*reinterpret_cast<unsigned long long*>(sh),
+                                    reinterpret_cast<unsigned long long*>(shadow + i));
+    } else {
+      *reinterpret_cast<float4*>(p + i) = pv;
+      *reinterpret_cast<float4*>(m + i) = mv;
+      *reinterpret_cast<float4*>(v + i) = vv;
+      if (shadow != nullptr)
+        *reinterpret_cast<ulonglong1*>(shadow + i) =
+            *reinterpret_cast<ulonglong1*>(sh);
+    }
   }
   // scalar tail
   for (long i = n4 * 4 + blockIdx.x * blockDim.x + threadIdx.x; i < n;
@@ -1442,7 +1458,18 @@ void adam_dense_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
   ushort* shadow_ptr = nullptr;
   if (shadow.defined() && shadow.numel() == n)
     shadow_ptr = reinterpret_cast<ushort*>(shadow.data_ptr<at::BFloat16>());
+  static const bool use_nt = [] {
+    const char* e = getenv("C2V_ADAM_NT");
+    return e && e[0] == '1';
+  }();
   const int grid = grid_1d(std::max<long>(n / 4, 1), 256);
+  if (use_nt && g_c.scalar_type() == torch::kBFloat16) {
+    k_adam_dense<ushort, true><<<grid, 256, 0, cur_stream()>>>(
+        p.data_ptr<float>(), bf_ptr(g_c), m.data_ptr<float>(),
+        v.data_ptr<float>(), shadow_ptr, n, lr_t, (float)beta1, (float)beta2,
+        (float)eps, lrt_ptr);
+    return;
+  }
   if (g_c.scalar_type() == torch::kBFloat16)
     k_adam_dense<ushort><<<grid, 256, 0, cur_stream()>>>(
         p.data_ptr<float>(), bf_ptr(g_c), m.data_ptr<float>(),
