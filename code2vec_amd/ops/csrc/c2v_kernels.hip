@@ -1202,6 +1202,7 @@ inline hipStream_t cur_stream() {
 
 inline int grid_1d(long total, int block, int cap = 2048) {
   long g = (total + block - 1) / block;
+  if (g < 1) g = 1;  // zero-sized work still needs a valid launch config
   return (int)std::min<long>(g, cap);
 }
 
