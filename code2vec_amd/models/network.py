@@ -283,14 +283,17 @@ class Code2VecNetwork:
                 d_target = (d_logits.t() @ code_c)
                 reducer.allreduce_dense('target_table', d_target)
 
-        d_comb3, d_a = F.attention_bwd(st.comb.reshape(B, C, D), self.a_c,
-                                       st.alpha, d_code)
-        d_comb = d_comb3.reshape(B * C, D)
-        d_ctx, d_w = F.transform_tanh_bwd(st.ctx, self.w_io, st.comb, d_comb)
+        # attention backward emits dL/dz directly (tanh' fused into its
+        # d_comb write); the dX GEMM applies the dropout mask in its epilogue
+        # — no separate tanh_bwd or dropout_bwd passes
+        d_z3, d_a = F.attention_bwd(st.comb.reshape(B, C, D), self.a_c,
+                                    st.alpha, d_code, fuse_tanh_bwd=True)
+        d_z = d_z3.reshape(B * C, D)
+        d_ctx, d_w = F.linear_bwd_dropout(d_z, self.w_io, st.ctx,
+                                          cfg.DROPOUT_KEEP_RATE, st.seed,
+                                          seed_t=st.seed_t, training=True)
         reducer.allreduce_dense('w', d_w)
         reducer.allreduce_dense('a', d_a)
-        d_ctx = F.gather_concat_bwd(d_ctx, cfg.DROPOUT_KEEP_RATE, st.seed, True,
-                                    seed_t=st.seed_t)
 
         # Sparse embedding grads: (ids, rows) pairs; under DP these are
         # all-gathered (not dense-all-reduced) — SURVEY §2.4. Single-process

@@ -162,11 +162,15 @@ __global__ void k_dropout_bwd(const ushort* __restrict__ g_in,
 #define GEMM_BN 128
 #define GEMM_BK 32
 
-template <bool TANH>
+template <bool TANH, bool DROPOUT = false>
 __launch_bounds__(256)
 __global__ void k_gemm_bt(const ushort* __restrict__ A,
                           const ushort* __restrict__ Bt,
-                          ushort* __restrict__ C, int N, int M, int K) {
+                          ushort* __restrict__ C, int N, int M, int K,
+                          float keep_prob = 1.f, u64 seed_scalar = 0,
+                          const long* __restrict__ seed_ptr = nullptr) {
+  const u64 seed = DROPOUT ? (seed_ptr ? (u64)*seed_ptr : seed_scalar) : 0;
+  const float inv_keep = DROPOUT ? (1.f / keep_prob) : 1.f;
   __shared__ ushort lds_a[GEMM_BM * GEMM_BK];
   __shared__ ushort lds_b[GEMM_BN * GEMM_BK];
 
@@ -268,6 +272,13 @@ __global__ void k_gemm_bt(const ushort* __restrict__ A,
         if (row >= N) continue;
         float v = acc[mi][ni][r];
         if (TANH) v = tanhf(v);
+        if (DROPOUT) {
+          // fused dropout backward: apply the keep-mask/scale for element
+          // (row, col) of the (N, 3d) d_ctx directly in the epilogue
+          const bool keep =
+              hash_uniform(seed, (u64)row * M + col) < keep_prob;
+          v = keep ? v * inv_keep : 0.f;
+        }
         C[(long)row * M + col] = f2bf(v);
       }
     }
@@ -604,7 +615,7 @@ __global__ void k_attn_bwd(const ushort* __restrict__ comb,
                            const float* __restrict__ d_code,
                            ushort* __restrict__ d_comb,
                            float* __restrict__ d_a_partial,  // (grid,D)
-                           int B, int C, int D) {
+                           int B, int C, int D, int fuse_tanh_bwd) {
   float* dal = reinterpret_cast<float*>(smem);  // C f32: d_alpha then d_e
   __shared__ float s_inner;
   const int b = blockIdx.x;
@@ -661,8 +672,14 @@ __global__ void k_attn_bwd(const ushort* __restrict__ comb,
     float da = 0.f;
     for (int c = 0; c < C; ++c) {
       const float de = dal[c];
-      da += de * bf2f(rp[(long)c * D]);
-      wp[(long)c * D] = f2bf(alpha_row[c] * dvv + de * av);
+      const float y = bf2f(rp[(long)c * D]);
+      da += de * y;
+      float g = alpha_row[c] * dvv + de * av;
+      // fused tanh backward: comb = tanh(z) is already in hand, so emit
+      // dL/dz = dL/dcomb * (1 - comb^2) directly (kills the separate
+      // 157 MB tanh_bwd_mul pass)
+      if (fuse_tanh_bwd) g *= (1.f - y * y);
+      wp[(long)c * D] = f2bf(g);
     }
     d_a_partial[(long)blockIdx.x * D + col] = da;
   }
@@ -1356,6 +1373,24 @@ torch::Tensor gemm_bt_v(torch::Tensor A, torch::Tensor Bt, bool tanh_ep,
   return gemm_bt(A, Bt, tanh_ep, (int)variant);
 }
 
+// dX GEMM with the dropout keep-mask fused into the epilogue (128-tile path)
+torch::Tensor gemm_bt_dropout(torch::Tensor A, torch::Tensor Bt,
+                              double keep_prob, int64_t seed,
+                              torch::Tensor seed_t) {
+  CHECK_DEV(A); CHECK_CONT(A); CHECK_DEV(Bt); CHECK_CONT(Bt);
+  const int N = A.size(0), K = A.size(1), M = Bt.size(0);
+  TORCH_CHECK(Bt.size(1) == K && K % GEMM_BK == 0);
+  const long* seed_ptr = (seed_t.defined() && seed_t.numel() == 1)
+                             ? seed_t.data_ptr<long>() : nullptr;
+  auto C = torch::empty({N, M}, A.options());
+  const int n_tiles = (N + GEMM_BM - 1) / GEMM_BM;
+  const int m_tiles = (M + GEMM_BN - 1) / GEMM_BN;
+  k_gemm_bt<false, true><<<n_tiles * m_tiles, 256, 0, cur_stream()>>>(
+      bf_ptr(A), bf_ptr(Bt), bf_ptr_mut(C), N, M, K, (float)keep_prob,
+      (u64)seed, seed_ptr);
+  return C;
+}
+
 torch::Tensor tanh_bwd_mul(torch::Tensor dy, torch::Tensor y) {
   CHECK_DEV(dy); CHECK_CONT(dy); CHECK_DEV(y); CHECK_CONT(y);
   auto dz = torch::empty_like(dy);
@@ -1386,7 +1421,8 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor comb, torch::Tensor a,
 
 std::vector<torch::Tensor> attention_bwd(torch::Tensor comb, torch::Tensor a,
                                          torch::Tensor alpha,
-                                         torch::Tensor d_code) {
+                                         torch::Tensor d_code,
+                                         bool fuse_tanh_bwd) {
   CHECK_DEV(comb); CHECK_CONT(comb);
   const int B = comb.size(0), C = comb.size(1), D = comb.size(2);
   TORCH_CHECK(D % 64 == 0 && D <= 512);
@@ -1399,7 +1435,7 @@ std::vector<torch::Tensor> attention_bwd(torch::Tensor comb, torch::Tensor a,
   k_attn_bwd<<<B, 256, lds, cur_stream()>>>(
       bf_ptr(comb), a32.data_ptr<float>(), alpha32.data_ptr<float>(),
       dcode32.data_ptr<float>(), bf_ptr_mut(d_comb),
-      d_a_partial.data_ptr<float>(), B, C, D);
+      d_a_partial.data_ptr<float>(), B, C, D, fuse_tanh_bwd ? 1 : 0);
   auto d_a = d_a_partial.sum(0);
   return {d_comb, d_a};
 }
@@ -1684,6 +1720,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("transform_tanh_fwd", &transform_tanh_fwd);
   mod.def("gemm_bt_bf16", &gemm_bt_bf16);
   mod.def("gemm_bt_v", &gemm_bt_v);
+  mod.def("gemm_bt_dropout", &gemm_bt_dropout);
   mod.def("logits_ce_fused", &logits_ce_fused);
   mod.def("tanh_bwd_mul", &tanh_bwd_mul);
   mod.def("attention_fwd", &attention_fwd);

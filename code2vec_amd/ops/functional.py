@@ -70,10 +70,45 @@ def attention_fwd(comb, a, valid_mask) -> Tuple[torch.Tensor, torch.Tensor]:
     return ref.attention_fwd(comb, a, valid_mask)
 
 
-def attention_bwd(comb, a, alpha, d_code) -> Tuple[torch.Tensor, torch.Tensor]:
+def attention_bwd(comb, a, alpha, d_code,
+                  fuse_tanh_bwd: bool = False) -> Tuple[torch.Tensor, torch.Tensor]:
+    """With fuse_tanh_bwd, returns dL/dz (z = pre-tanh transform output)
+    instead of dL/dcomb — the tanh' factor is applied inside the kernel's
+    d_comb write where comb is already loaded."""
     if backend_for(comb) == 'hip':
-        return hip_ext(True).attention_bwd(comb, a, alpha, d_code)
-    return ref.attention_bwd(comb, a, alpha, d_code)
+        return hip_ext(True).attention_bwd(comb, a, alpha, d_code,
+                                           bool(fuse_tanh_bwd))
+    d_comb, d_a = ref.attention_bwd(comb, a, alpha, d_code)
+    if fuse_tanh_bwd:
+        d_comb = (d_comb.float() * (1.0 - comb.float() ** 2)).to(d_comb.dtype)
+    return d_comb, d_a
+
+
+def linear_bwd_dropout(d_z, w_io, ctx, keep_prob: float, seed: int,
+                       seed_t=None,
+                       training: bool = True) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Backward of [dropout -> Linear]: d_ctx = dropout_mask ⊙ (d_z @ W^T)
+    with the mask fused into the GEMM epilogue on HIP; d_w = ctx^T @ d_z via
+    hipBLASLt. (The tanh' factor is expected to already be in d_z — see
+    attention_bwd(fuse_tanh_bwd=True).)"""
+    drop = training and keep_prob < 1.0
+    if backend_for(d_z) == 'hip':
+        e = hip_ext(True)
+        if drop:
+            d_ctx = e.gemm_bt_dropout(d_z, w_io, float(keep_prob), int(seed),
+                                      seed_t if seed_t is not None
+                                      else torch.empty(0))
+        else:
+            d_ctx = e.gemm_bt_bf16(d_z, w_io)
+        d_w = (ctx.t().to(d_z.dtype) @ d_z).float()
+        return d_ctx, d_w
+    d_ctx = d_z @ w_io.to(d_z.dtype).t()
+    if drop:
+        if seed_t is not None:
+            seed = int(seed_t.item())
+        d_ctx = ref.gather_concat_bwd(d_ctx, keep_prob, seed, True)
+    d_w = (ctx.t().float() @ d_z.float())
+    return d_ctx, d_w
 
 
 def ce_fwd(logits, labels) -> Tuple[torch.Tensor, torch.Tensor]:

@@ -134,7 +134,7 @@ def test_attention_bwd(B, C, D):
     mask[:, 0] = 1.0
     _, alpha = R.attention_fwd(comb, a, mask)
     d_code = randn(B, D, scale=0.5)
-    d_comb, d_a = ext().attention_bwd(comb, a, alpha, d_code)
+    d_comb, d_a = ext().attention_bwd(comb, a, alpha, d_code, False)
     d_comb_ref, d_a_ref = R.attention_bwd(comb, a, alpha, d_code)
     assert (d_comb.float() - d_comb_ref.float()).abs().max().item() < 1e-2
     assert (d_a - d_a_ref).abs().max().item() / max(d_a_ref.abs().max().item(), 1e-3) < 1e-2
@@ -352,3 +352,34 @@ def test_logits_ce_fused_matches_fp32_oracle():
     # loss = fp32 lse − bf16-rounded label logit (the stored logits feed bwd)
     picked = logits_f.float().gather(1, labels.reshape(-1, 1)).squeeze(1)
     assert (loss_f - (lse32 - picked)).abs().max().item() < 3e-3
+
+
+def test_attention_bwd_fused_tanh():
+    from code2vec_amd.ops import functional as F
+    B, C, D = 6, 33, 128
+    comb = randn(B, C, D, dtype=torch.bfloat16, scale=0.7, seed=41)
+    a = randn(D, scale=0.3)
+    mask = (torch.rand(B, C) > 0.2).float().cuda()
+    mask[:, 0] = 1.0
+    _, alpha = R.attention_fwd(comb, a, mask)
+    d_code = randn(B, D, scale=0.5)
+    d_z, d_a = F.attention_bwd(comb, a, alpha, d_code, fuse_tanh_bwd=True)
+    d_comb_ref, d_a_ref = R.attention_bwd(comb, a, alpha, d_code)
+    d_z_ref = (d_comb_ref.float() * (1 - comb.float() ** 2))
+    assert (d_z.float() - d_z_ref).abs().max().item() < 1e-2
+    assert (d_a - d_a_ref).abs().max().item() < 1e-2
+
+
+def test_gemm_bt_dropout_epilogue():
+    from code2vec_amd.ops import functional as F
+    N, K, M = 256, 64, 48
+    dz = randn(N, K, dtype=torch.bfloat16, scale=0.5, seed=43)
+    w = randn(M, K, dtype=torch.bfloat16, scale=0.5)
+    ctx = randn(N, M, dtype=torch.bfloat16, scale=0.5)
+    seed = 777
+    d_ctx, d_w = F.linear_bwd_dropout(dz, w, ctx, 0.75, seed, training=True)
+    plain = dz.float() @ w.float().t()
+    masked = R.gather_concat_bwd(plain.to(torch.bfloat16), 0.75, seed, True)
+    # same mask pattern: zeros agree exactly; kept values within bf16 noise
+    assert torch.equal(d_ctx == 0, masked == 0)
+    assert (d_ctx.float() - masked.float()).abs().max().item() < 0.05
