@@ -76,5 +76,5 @@ def test_gpu_train_evaluate_predict(tmp_path):
     line = 'zz tk0,ph0,tk1 tk2,ph1,tk3'
     preds = model.predict([line])
     assert preds[0].original_name == 'zz'
-    assert len(preds[0].topk_predicted_words) == 4
+    assert len(preds[0].topk_predicted_words) == 5  # 4 targets + PAD_OR_OOV
     assert torch.isfinite(torch.tensor(preds[0].code_vector)).all()

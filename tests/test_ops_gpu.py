@@ -379,7 +379,10 @@ def test_gemm_bt_dropout_epilogue():
     seed = 777
     d_ctx, d_w = F.linear_bwd_dropout(dz, w, ctx, 0.75, seed, training=True)
     plain = dz.float() @ w.float().t()
-    masked = R.gather_concat_bwd(plain.to(torch.bfloat16), 0.75, seed, True)
-    # same mask pattern: zeros agree exactly; kept values within bf16 noise
-    assert torch.equal(d_ctx == 0, masked == 0)
-    assert (d_ctx.float() - masked.float()).abs().max().item() < 0.05
+    mask = R.dropout_keep_mask(seed, N * M, 0.75, 'cuda').reshape(N, M)
+    # dropped positions are exactly zero
+    assert torch.all(d_ctx[~mask] == 0)
+    # kept positions match plain/keep within bf16 noise of the value scale
+    expected = plain / 0.75
+    err = (d_ctx.float() - expected)[mask].abs().max().item()
+    assert err < 0.02 * expected.abs().max().item(), err
