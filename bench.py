@@ -53,6 +53,13 @@ def make_config(device):
     cfg = Config(set_defaults=True)
     cfg.DEVICE = device
     cfg.COMPUTE_DTYPE = 'bf16' if device.startswith('cuda') else 'fp32'
+    if os.environ.get('C2V_BENCH_TINY') == '1':
+        # test-only shrink (tests/test_bench_cli.py): exercises the exact
+        # bench code path — distributed init, reducer, timing, JSON — on CPU
+        cfg.MAX_TOKEN_VOCAB_SIZE = 1000
+        cfg.MAX_PATH_VOCAB_SIZE = 800
+        cfg.MAX_TARGET_VOCAB_SIZE = 500
+        cfg.MAX_CONTEXTS = 16
     return cfg
 
 
@@ -69,7 +76,8 @@ def synth_batches(cfg, device, batch_size, n_batches=8, seed=0):
         pth = torch.randint(1, V_path, (batch_size, C), generator=g, dtype=torch.int32)
         tgt = torch.randint(1, V_tok, (batch_size, C), generator=g, dtype=torch.int32)
         # realistic context-count distribution: valid prefix of U[64, 200]
-        n_valid = torch.randint(64, C + 1, (batch_size,), generator=g)
+        lo = min(64, max(1, C // 2))
+        n_valid = torch.randint(lo, C + 1, (batch_size,), generator=g)
         mask = (torch.arange(C).unsqueeze(0) < n_valid.unsqueeze(1)).float()
         src = torch.where(mask.bool(), src, torch.zeros_like(src))
         pth = torch.where(mask.bool(), pth, torch.zeros_like(pth))
