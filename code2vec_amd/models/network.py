@@ -239,7 +239,7 @@ class Code2VecNetwork:
                                                        labels)
             loss = loss_rows.float().mean()
             d_logits = F.ce_bwd(logits, lse, labels, 1.0 / B)
-            d_code = (d_logits @ self.target_shadow).float()     # (B,D)
+            d_code = F.logits_bwd_code(d_logits, self.target_shadow)  # (B,D)
             d_target = None  # see target-chain dispatch below
 
         side_done = None

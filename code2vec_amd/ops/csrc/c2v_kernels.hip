@@ -499,6 +499,220 @@ __global__ void k_ce_reduce_partials(const float* __restrict__ partials,
   }
 }
 
+// ---------------------------------------------------------------------------
+// d_code GEMM (nn operands): C(N,M) = A(N,K) @ B(K,M), A and B ROW-major
+// bf16, C fp32. Shaped for dL/dcode = d_logits(B,V) @ target_shadow(V,D):
+// N small (batch), M <= 384 (code dim), K huge (vocab) — so the kernel is
+// SPLIT-K: the K axis is carved into S chunks, each block accumulates a
+// (256 x M) fp32 partial for its chunk, and k_splitk_reduce folds the S
+// slices. B fragments need column-major k-runs, so the B tile is staged
+// TRANSPOSED into LDS ([col][k], padded stride) from coalesced row-major
+// b128 global reads; frag reads are then contiguous b128 like the A side.
+// Grid mapping (m204-style): dispatch round-robins blockIdx over the 8 XCDs,
+// so we decode (k_chunk, row_tile) such that all row-tiles of one k-chunk
+// land on ONE XCD back-to-back — the chunk's B panel (~3 MB) streams from
+// HBM once per XCD instead of once per row-tile.
+// ---------------------------------------------------------------------------
+
+#define GNN_BM 256
+#define GNN_BN 384   // max M (full code dim); cols beyond M are zero-padded
+#define GNN_BK 32
+#define GNN_PKB 40   // padded k-stride of the transposed B tile
+
+__launch_bounds__(512, 1)
+__global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
+                                 const ushort* __restrict__ B,
+                                 float* __restrict__ P,  // (S, N, M) partials
+                                 int N, int M, int K, int S,
+                                 int ksteps_per_chunk, int row_tiles) {
+  // double-buffered: A[2][256][32] + Bt[2][384][40] bf16 = 92 KiB dynamic
+  extern __shared__ ushort ldsnn[];
+#define NLDS_A(b) (ldsnn + (b) * 8192)
+#define NLDS_B(b) (ldsnn + 16384 + (b) * (GNN_BN * GNN_PKB))
+
+  // XCD-aware decode: blocks dispatch round-robin over 8 XCDs, so
+  // xcd = bid % 8 and slot = bid / 8 enumerate one XCD's blocks; chunks are
+  // dealt to XCDs in groups of row_tiles consecutive slots.
+  const int bid = blockIdx.x;
+  const int xcd = bid & 7;
+  const int slot = bid >> 3;
+  const int chunk = (slot / row_tiles) * 8 + xcd;
+  const int tile_n = slot % row_tiles;
+  const int row0 = tile_n * GNN_BM;
+
+  const int total_ksteps = (K + GNN_BK - 1) / GNN_BK;
+  const int ks0 = chunk * ksteps_per_chunk;
+  const int ks1 = min(total_ksteps, ks0 + ksteps_per_chunk);
+  const int ks1_main = min(ks1, K / GNN_BK);  // full 32-k steps only
+  // ks0 >= ks1 (empty chunk) still falls through: acc stays zero and the
+  // epilogue writes a zero partial slice (the reducer reads every slice)
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;              // 8 waves: 2 (row) x 4 (col)
+  const int wrow = (wid >> 2) * 128;
+  const int wcol = (wid & 3) * 96;
+
+  f32x4 acc[8][6];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 6; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  // B tile 32x384 = 1536 b128 chunks -> 3/thread; krow = idx/48, c8 = idx%48*8
+  int b_kr[3], b_c8[3];
+#pragma unroll
+  for (int r = 0; r < 3; ++r) {
+    const int idx = tid + r * 512;
+    b_kr[r] = idx / 48;
+    b_c8[r] = (idx % 48) * 8;
+  }
+
+  // A staging: async width-16 global->LDS, 2 wave-issues cover the 16 KiB tile
+  auto stage_a = [&](int buf, int ks) {
+    const int k0 = ks * GNN_BK;
+#pragma unroll
+    for (int q = 0; q < 2; ++q) {
+      const int off = (wid * 2 + q) * 1024 + lane * 16;  // byte in tile
+      const int lrow = off >> 6;              // 64 B per row of 32 bf16
+      const int lcol = (off & 63) >> 1;
+      const int grow = min(row0 + lrow, N - 1);
+      const ushort* gp = A + (long)grow * K + k0 + lcol;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) u32*)gp,
+          (__attribute__((address_space(3))) u32*)(NLDS_A(buf) +
+                                                   (wid * 2 + q) * 512),
+          16, 0, 0);
+    }
+  };
+
+  // B staging: b128 row-major global reads into regs (prefetched 2 tiles
+  // ahead), written TRANSPOSED into LDS one barrier before their compute
+  bf16x8 rb[3];
+  auto load_b = [&](int ks) {
+    const int k0 = ks * GNN_BK;
+#pragma unroll
+    for (int r = 0; r < 3; ++r) {
+      if (b_c8[r] + 8 <= M) {
+        rb[r] = *reinterpret_cast<const bf16x8*>(
+            B + (long)(k0 + b_kr[r]) * M + b_c8[r]);
+      } else {
+        const long base = (long)(k0 + b_kr[r]) * M;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          rb[r][j] = (b_c8[r] + j < M) ? (short)B[base + b_c8[r] + j]
+                                       : (short)0;
+      }
+    }
+  };
+  auto store_b = [&](int buf) {
+#pragma unroll
+    for (int r = 0; r < 3; ++r)
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        NLDS_B(buf)[(b_c8[r] + j) * GNN_PKB + b_kr[r]] = rb[r][j];
+  };
+
+  auto compute = [&](int buf) {
+    bf16x8 bfrag[6];
+#pragma unroll
+    for (int n = 0; n < 6; ++n) {
+      const int col = wcol + n * 16 + (lane & 15);
+      bfrag[n] = *reinterpret_cast<const bf16x8*>(
+          NLDS_B(buf) + col * GNN_PKB + (lane >> 4) * 8);
+    }
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int m = 0; m < 8; ++m) {
+      const int row = wrow + m * 16 + (lane & 15);
+      const bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
+          NLDS_A(buf) + row * GNN_BK + (lane >> 4) * 8);
+#pragma unroll
+      for (int n = 0; n < 6; ++n)
+        acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag, bfrag[n], acc[m][n], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+  };
+
+  if (ks0 < ks1_main) {
+    load_b(ks0);
+    stage_a(0, ks0);
+    store_b(0);
+    if (ks0 + 1 < ks1_main) load_b(ks0 + 1);
+    __syncthreads();
+    for (int ks = ks0; ks < ks1_main; ++ks) {
+      const int cur = (ks - ks0) & 1;
+      if (ks + 1 < ks1_main) {
+        stage_a(cur ^ 1, ks + 1);   // async, lands by next barrier
+        store_b(cur ^ 1);           // regs loaded one iteration ago
+        if (ks + 2 < ks1_main) load_b(ks + 2);
+      }
+      compute(cur);
+      __syncthreads();
+    }
+  }
+
+  // partial last k-step (K % 32): cooperative zero-filled manual staging,
+  // one guarded iteration outside the hot loop
+  if ((K % GNN_BK) && ks1 == total_ksteps && ks0 < ks1) {
+    const int k0 = (total_ksteps - 1) * GNN_BK;
+#pragma unroll
+    for (int q = 0; q < 2; ++q) {
+      const int off = (wid * 2 + q) * 1024 + lane * 16;
+      const int lrow = off >> 6;
+      const int lcol = (off & 63) >> 1;
+      const int grow = min(row0 + lrow, N - 1);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        NLDS_A(0)[(wid * 2 + q) * 512 + lane * 8 + j] =
+            (k0 + lcol + j < K) ? A[(long)grow * K + k0 + lcol + j]
+                                : (ushort)0;
+    }
+#pragma unroll
+    for (int r = 0; r < 3; ++r) {
+      const int gk = k0 + b_kr[r];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        NLDS_B(0)[(b_c8[r] + j) * GNN_PKB + b_kr[r]] =
+            (gk < K && b_c8[r] + j < M) ? B[(long)gk * M + b_c8[r] + j]
+                                        : (ushort)0;
+    }
+    __syncthreads();
+    compute(0);
+  }
+
+  float* Pc = P + (long)chunk * N * M;
+#pragma unroll
+  for (int m = 0; m < 8; ++m) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = row0 + wrow + m * 16 + (lane >> 4) * 4 + r;
+      if (row >= N) continue;
+#pragma unroll
+      for (int n = 0; n < 6; ++n) {
+        const int col = wcol + n * 16 + (lane & 15);
+        if (col < M) Pc[(long)row * M + col] = acc[m][n][r];
+      }
+    }
+  }
+}
+
+// fold the S split-K slices: C[i] = sum_s P[s][i] (fp32 out)
+__launch_bounds__(256)
+__global__ void k_splitk_reduce(const float* __restrict__ P,
+                                float* __restrict__ C, int S, long total) {
+  for (long i4 = blockIdx.x * blockDim.x + threadIdx.x; i4 * 4 < total;
+       i4 += (long)gridDim.x * blockDim.x) {
+    f32x4 s = {0.f, 0.f, 0.f, 0.f};
+    for (int c = 0; c < S; ++c) {
+      const f32x4 v = *reinterpret_cast<const f32x4*>(P + c * total + i4 * 4);
+      s[0] += v[0]; s[1] += v[1]; s[2] += v[2]; s[3] += v[3];
+    }
+    *reinterpret_cast<f32x4*>(C + i4 * 4) = s;
+  }
+}
+
 // elementwise tanh backward: d_z = d_y * (1 - y^2), bf16
 __global__ void k_tanh_bwd_mul(const ushort* __restrict__ dy,
                                const ushort* __restrict__ y,
@@ -1391,6 +1605,43 @@ torch::Tensor gemm_bt_dropout(torch::Tensor A, torch::Tensor Bt,
   return C;
 }
 
+// split-K nn GEMM: C(N,M) fp32 = A(N,K) @ B(K,M), row-major bf16 operands.
+// Shaped for d_code = d_logits @ target_shadow (N=batch, M=code dim <= 384,
+// K=vocab). Partials workspace (S, N, M) fp32 comes from the caching
+// allocator; S is a multiple of 8 so the XCD-grouped chunk decode is
+// bijective, sized so the grid is ~256 blocks (1 per CU).
+torch::Tensor gemm_nn_splitk(torch::Tensor A, torch::Tensor B) {
+  CHECK_DEV(A); CHECK_CONT(A); CHECK_DEV(B); CHECK_CONT(B);
+  TORCH_CHECK(A.scalar_type() == torch::kBFloat16 &&
+              B.scalar_type() == torch::kBFloat16);
+  const int N = A.size(0), K = A.size(1), M = B.size(1);
+  TORCH_CHECK(B.size(0) == K, "K mismatch");
+  TORCH_CHECK(M <= GNN_BN && M % 8 == 0, "nn GEMM: M must be <=384, mult of 8");
+  const int row_tiles = (N + GNN_BM - 1) / GNN_BM;
+  int S = 256 / (row_tiles * 8) * 8;          // grid ~256, S multiple of 8
+  if (S < 8) S = 8;
+  const int total_ksteps = (K + GNN_BK - 1) / GNN_BK;
+  const int kpc = (total_ksteps + S - 1) / S;
+  auto P = torch::empty({(long)S, (long)N, (long)M},
+                        A.options().dtype(torch::kFloat32));
+  auto C = torch::empty({N, M}, A.options().dtype(torch::kFloat32));
+  const size_t lds = (16384 + 2 * (size_t)GNN_BN * GNN_PKB) * 2;  // 92 KiB
+  static bool nn_configured = false;
+  if (!nn_configured) {
+    (void)hipFuncSetAttribute((const void*)k_gemm_nn_splitk,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              (int)lds);
+    nn_configured = true;
+  }
+  k_gemm_nn_splitk<<<S * row_tiles, 512, lds, cur_stream()>>>(
+      bf_ptr(A), bf_ptr(B), P.data_ptr<float>(), N, M, K, S, kpc, row_tiles);
+  const long total = (long)N * M;
+  TORCH_CHECK(total % 4 == 0);
+  k_splitk_reduce<<<grid_1d(total / 4, 256), 256, 0, cur_stream()>>>(
+      P.data_ptr<float>(), C.data_ptr<float>(), S, total);
+  return C;
+}
+
 torch::Tensor tanh_bwd_mul(torch::Tensor dy, torch::Tensor y) {
   CHECK_DEV(dy); CHECK_CONT(dy); CHECK_DEV(y); CHECK_CONT(y);
   auto dz = torch::empty_like(dy);
@@ -1722,6 +1973,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gemm_bt_v", &gemm_bt_v);
   mod.def("gemm_bt_dropout", &gemm_bt_dropout);
   mod.def("logits_ce_fused", &logits_ce_fused);
+  mod.def("gemm_nn_splitk", &gemm_nn_splitk);
   mod.def("tanh_bwd_mul", &tanh_bwd_mul);
   mod.def("attention_fwd", &attention_fwd);
   mod.def("attention_bwd", &attention_bwd);

@@ -134,6 +134,21 @@ def logits_gemm(code_c, shadow):
     return code_c @ shadow.t()
 
 
+def logits_bwd_code(d_logits, shadow):
+    """d_code = d_logits @ shadow (nn operands, fp32 out). On GPU with the
+    vocab-sized K this runs the split-K MFMA kernel (k_gemm_nn_splitk:
+    transposed-LDS B staging, XCD-grouped k-chunks, fp32 partials + reduce);
+    elsewhere a plain matmul."""
+    import os as _os
+    if (backend_for(d_logits) == 'hip' and d_logits.dtype == torch.bfloat16
+            and shadow.dtype == torch.bfloat16
+            and shadow.shape[1] <= 384 and shadow.shape[1] % 8 == 0
+            and shadow.shape[0] >= 4096
+            and _os.environ.get('C2V_NN_GEMM', '1') == '1'):
+        return hip_ext(True).gemm_nn_splitk(d_logits, shadow)
+    return (d_logits @ shadow).float()
+
+
 def logits_ce_fused(code_c, shadow, labels):
     """Fused K8+K9 forward: the 256-tile logits GEMM also emits per-tile
     (rowmax, sumexp) partials, folded into (loss, lse) by a small reduce —

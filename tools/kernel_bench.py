@@ -54,6 +54,7 @@ def main():
 
     dl = torch.randn(B, V, device=dev).to(torch.bfloat16)
     timeit('d_code hipBLASLt', lambda: dl @ shadow)
+    timeit('d_code our split-K nn', lambda: ext.gemm_nn_splitk(dl, shadow))
     timeit('d_target hipBLASLt', lambda: dl.t() @ code)
 
     timeit('attn_fwd', lambda: ext.attention_fwd(comb, a, mask))
