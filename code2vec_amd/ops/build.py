@@ -41,7 +41,11 @@ def build(verbose: bool = True) -> str:
         sources=[src],
         build_directory=build_dir,
         extra_cflags=['-O3'],
-        extra_cuda_cflags=['-O3', '-std=c++17'],
+        # AGPR-form MFMA: accumulators allocate in the AGPR half of the
+        # unified gfx950 register file — without this the 192-reg split-K
+        # accumulator tile spills (25 in-loop scratch ops)
+        extra_cuda_cflags=['-O3', '-std=c++17',
+                           '-mllvm', '-amdgpu-mfma-vgpr-form=0'],
         verbose=verbose,
         is_python_module=False,
         with_cuda=True,

@@ -26,6 +26,8 @@
 #define DEVI __device__ __forceinline__
 
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using u16x8 = __attribute__((ext_vector_type(8))) unsigned short;
+using u16x4 = __attribute__((ext_vector_type(4))) unsigned short;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 using u32 = uint32_t;
 using u64 = uint64_t;
@@ -514,25 +516,50 @@ __global__ void k_ce_reduce_partials(const float* __restrict__ partials,
 // HBM once per XCD instead of once per row-tile.
 // ---------------------------------------------------------------------------
 
-#define GNN_BM 256
+#define GNN_BM 128
 #define GNN_BN 384   // max M (full code dim); cols beyond M are zero-padded
 #define GNN_BK 32
-#define GNN_PKB 40   // padded k-stride of the transposed B tile
+// padded LDS strides, chosen for bank spread (32 banks x 4B):
+//  A rows stride 36 ushorts (72 B): afrag b64 reads hit 16 distinct bank
+//    starts across the 16 lanes of a row group (18*r mod 32 is a bijection
+//    on even banks) -> conflict-free;
+//  B rows stride 440 ushorts with the row image of k ROTATED right by
+//    16*(k/8) cols: b128 stores stay 16 B-aligned and contiguous
+//    (conflict-free), and bfrag u16 reads get their four k-groups on
+//    disjoint bank quarters via the +16*(k/8) term (bank = 28j + col/2
+//    + 8q mod 32) with paired cols sharing a dword (LDS broadcast)
+//    -> conflict-free. Two earlier layouts measured worse: a transposed
+//    [col][k] tile made the scalar u16 stores 48-way same-bank
+//    (SQ_LDS_BANK_CONFLICT ~6x busy cycles, 897 us); an unrotated 386
+//    stride fixed the reads but its 4 B stores still hit only 8 banks
+//    (conflicts ~1.2x busy, 433 us).
+#define GNN_PKA 36
+#define GNN_PKB 440
+#define GNN_BROT(k) (16 * ((k) >> 3))
 
-__launch_bounds__(512, 1)
+__launch_bounds__(256, 1)
 __global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
                                  const ushort* __restrict__ B,
                                  float* __restrict__ P,  // (S, N, M) partials
                                  int N, int M, int K, int S,
                                  int ksteps_per_chunk, int row_tiles) {
-  // double-buffered: A[2][256][32] + Bt[2][384][40] bf16 = 92 KiB dynamic
+  // double-buffered: A[2][128][36] + B[2][32][440] bf16 = 73 KiB dynamic.
+  // 256 threads = 4 waves = 1 wave/SIMD: the per-SIMD register pool is 512
+  // regs/lane, the only occupancy at which the 192-reg accumulator tile
+  // plus staging fits unspilled (the gfx950 compiler does not move MFMA
+  // accumulators into the AGPR half of the unified file). Tile-size A/B:
+  // BM=32 with 2 blocks/CU amortized per-kstep costs 4x worse (607 us);
+  // BM=128 measured 409 us ~= the 2.1 GB HBM bound (B re-read once per
+  // row-tile; the XCD-grouped decode lets L2 absorb part of it).
+  // Non-temporal A loads / P stores measured worse (449 us) — reverted.
   extern __shared__ ushort ldsnn[];
-#define NLDS_A(b) (ldsnn + (b) * 8192)
-#define NLDS_B(b) (ldsnn + 16384 + (b) * (GNN_BN * GNN_PKB))
+#define NLDS_A(b) (ldsnn + (b) * (GNN_BM * GNN_PKA))
+#define NLDS_B(b) (ldsnn + 2 * GNN_BM * GNN_PKA + (b) * (GNN_BK * GNN_PKB))
 
   // XCD-aware decode: blocks dispatch round-robin over 8 XCDs, so
   // xcd = bid % 8 and slot = bid / 8 enumerate one XCD's blocks; chunks are
-  // dealt to XCDs in groups of row_tiles consecutive slots.
+  // dealt to XCDs in groups of row_tiles consecutive slots, so every
+  // row-tile of one k-chunk shares that XCD's L2 for the B panel.
   const int bid = blockIdx.x;
   const int xcd = bid & 7;
   const int slot = bid >> 3;
@@ -549,9 +576,8 @@ __global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
-  const int wid = tid >> 6;              // 8 waves: 2 (row) x 4 (col)
-  const int wrow = (wid >> 2) * 128;
-  const int wcol = (wid & 3) * 96;
+  const int wid = tid >> 6;              // 4 waves, one 96-col group each
+  const int wcol = wid * 96;
 
   f32x4 acc[8][6];
 #pragma unroll
@@ -559,74 +585,86 @@ __global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
 #pragma unroll
     for (int j = 0; j < 6; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  // B tile 32x384 = 1536 b128 chunks -> 3/thread; krow = idx/48, c8 = idx%48*8
-  int b_kr[3], b_c8[3];
+  // per-thread staging slices, raw u16 vectors (an arithmetic __bf16 type
+  // here would make LDS stores VALUE-convert instead of bit-copy — caught
+  // by tools/debug_nn.py: outputs were denormals whose bit patterns equaled
+  // the rounded reference values shifted left 16).
+  // A tile 128x32 = 512 b128 chunks -> 2/thread; row = idx/4, k8 = idx%4*8
+  const int a_row[2] = {(tid + 0) >> 2, (tid + 256) >> 2};
+  const int a_k8 = (tid & 3) * 8;
+  // B tile 32x384 = 1536 b128 chunks -> 6/thread; krow = idx/48, c8=idx%48*8
+  int b_kr[6], b_c8[6];
 #pragma unroll
-  for (int r = 0; r < 3; ++r) {
-    const int idx = tid + r * 512;
+  for (int r = 0; r < 6; ++r) {
+    const int idx = tid + r * 256;
     b_kr[r] = idx / 48;
     b_c8[r] = (idx % 48) * 8;
   }
 
-  // A staging: async width-16 global->LDS, 2 wave-issues cover the 16 KiB tile
-  auto stage_a = [&](int buf, int ks) {
+  u16x8 ra[2], rb[6];
+  auto load_ab = [&](int ks) {
     const int k0 = ks * GNN_BK;
 #pragma unroll
-    for (int q = 0; q < 2; ++q) {
-      const int off = (wid * 2 + q) * 1024 + lane * 16;  // byte in tile
-      const int lrow = off >> 6;              // 64 B per row of 32 bf16
-      const int lcol = (off & 63) >> 1;
-      const int grow = min(row0 + lrow, N - 1);
-      const ushort* gp = A + (long)grow * K + k0 + lcol;
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) u32*)gp,
-          (__attribute__((address_space(3))) u32*)(NLDS_A(buf) +
-                                                   (wid * 2 + q) * 512),
-          16, 0, 0);
+    for (int r = 0; r < 2; ++r) {
+      const int grow = min(row0 + a_row[r], N - 1);
+      ra[r] = *reinterpret_cast<const u16x8*>(A + (long)grow * K + k0 + a_k8);
     }
-  };
-
-  // B staging: b128 row-major global reads into regs (prefetched 2 tiles
-  // ahead), written TRANSPOSED into LDS one barrier before their compute
-  bf16x8 rb[3];
-  auto load_b = [&](int ks) {
-    const int k0 = ks * GNN_BK;
 #pragma unroll
-    for (int r = 0; r < 3; ++r) {
+    for (int r = 0; r < 6; ++r) {
       if (b_c8[r] + 8 <= M) {
-        rb[r] = *reinterpret_cast<const bf16x8*>(
+        rb[r] = *reinterpret_cast<const u16x8*>(
             B + (long)(k0 + b_kr[r]) * M + b_c8[r]);
       } else {
         const long base = (long)(k0 + b_kr[r]) * M;
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          rb[r][j] = (b_c8[r] + j < M) ? (short)B[base + b_c8[r] + j]
-                                       : (short)0;
+          rb[r][j] = (b_c8[r] + j < M) ? B[base + b_c8[r] + j] : (ushort)0;
       }
     }
   };
-  auto store_b = [&](int buf) {
+  auto store_ab = [&](int buf) {
 #pragma unroll
-    for (int r = 0; r < 3; ++r)
+    for (int r = 0; r < 2; ++r) {
+      ushort* p = NLDS_A(buf) + a_row[r] * GNN_PKA + a_k8;
+      // 2x b64 (72 B row stride keeps b64 alignment; b128 would misalign
+      // on odd rows)
+      *reinterpret_cast<u16x4*>(p) =
+          u16x4{ra[r][0], ra[r][1], ra[r][2], ra[r][3]};
+      *reinterpret_cast<u16x4*>(p + 4) =
+          u16x4{ra[r][4], ra[r][5], ra[r][6], ra[r][7]};
+    }
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        NLDS_B(buf)[(b_c8[r] + j) * GNN_PKB + b_kr[r]] = rb[r][j];
+    for (int r = 0; r < 6; ++r) {
+      ushort* p = NLDS_B(buf) + b_kr[r] * GNN_PKB + GNN_BROT(b_kr[r]) +
+                  b_c8[r];
+      *reinterpret_cast<u16x8*>(p) = rb[r];   // single aligned b128 store
+    }
   };
 
   auto compute = [&](int buf) {
+    // bfrag: 8 scalar u16 reads per fragment from the row-major B tile
+    // ([k][col], k-stride 386) — the k-groups land on disjoint bank
+    // quarters, and paired cols share a dword (LDS broadcast)
     bf16x8 bfrag[6];
 #pragma unroll
     for (int n = 0; n < 6; ++n) {
       const int col = wcol + n * 16 + (lane & 15);
-      bfrag[n] = *reinterpret_cast<const bf16x8*>(
-          NLDS_B(buf) + col * GNN_PKB + (lane >> 4) * 8);
+      const ushort* bp = NLDS_B(buf) + (lane >> 4) * (8 * GNN_PKB + 16) + col;
+      u16x8 t;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) t[j] = bp[j * GNN_PKB];
+      bfrag[n] = __builtin_bit_cast(bf16x8, t);
     }
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int m = 0; m < 8; ++m) {
-      const int row = wrow + m * 16 + (lane & 15);
-      const bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
-          NLDS_A(buf) + row * GNN_BK + (lane >> 4) * 8);
+      const int row = m * 16 + (lane & 15);
+      const ushort* ap = NLDS_A(buf) + row * GNN_PKA + (lane >> 4) * 8;
+      const u16x4 alo = *reinterpret_cast<const u16x4*>(ap);
+      const u16x4 ahi = *reinterpret_cast<const u16x4*>(ap + 4);
+      const bf16x8 afrag = __builtin_bit_cast(
+          bf16x8, u16x8{alo[0], alo[1], alo[2], alo[3],
+                        ahi[0], ahi[1], ahi[2], ahi[3]});
 #pragma unroll
       for (int n = 0; n < 6; ++n)
         acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -636,17 +674,15 @@ __global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
   };
 
   if (ks0 < ks1_main) {
-    load_b(ks0);
-    stage_a(0, ks0);
-    store_b(0);
-    if (ks0 + 1 < ks1_main) load_b(ks0 + 1);
+    load_ab(ks0);
+    store_ab(0);
+    if (ks0 + 1 < ks1_main) load_ab(ks0 + 1);
     __syncthreads();
     for (int ks = ks0; ks < ks1_main; ++ks) {
       const int cur = (ks - ks0) & 1;
       if (ks + 1 < ks1_main) {
-        stage_a(cur ^ 1, ks + 1);   // async, lands by next barrier
-        store_b(cur ^ 1);           // regs loaded one iteration ago
-        if (ks + 2 < ks1_main) load_b(ks + 2);
+        store_ab(cur ^ 1);          // regs loaded one iteration ago
+        if (ks + 2 < ks1_main) load_ab(ks + 2);
       }
       compute(cur);
       __syncthreads();
@@ -658,23 +694,20 @@ __global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
   if ((K % GNN_BK) && ks1 == total_ksteps && ks0 < ks1) {
     const int k0 = (total_ksteps - 1) * GNN_BK;
 #pragma unroll
-    for (int q = 0; q < 2; ++q) {
-      const int off = (wid * 2 + q) * 1024 + lane * 16;
-      const int lrow = off >> 6;
-      const int lcol = (off & 63) >> 1;
-      const int grow = min(row0 + lrow, N - 1);
+    for (int r = 0; r < 2; ++r) {
+      const int grow = min(row0 + a_row[r], N - 1);
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        NLDS_A(0)[(wid * 2 + q) * 512 + lane * 8 + j] =
-            (k0 + lcol + j < K) ? A[(long)grow * K + k0 + lcol + j]
+        NLDS_A(0)[a_row[r] * GNN_PKA + a_k8 + j] =
+            (k0 + a_k8 + j < K) ? A[(long)grow * K + k0 + a_k8 + j]
                                 : (ushort)0;
     }
 #pragma unroll
-    for (int r = 0; r < 3; ++r) {
+    for (int r = 0; r < 6; ++r) {
       const int gk = k0 + b_kr[r];
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        NLDS_B(0)[(b_c8[r] + j) * GNN_PKB + b_kr[r]] =
+        NLDS_B(0)[b_kr[r] * GNN_PKB + GNN_BROT(b_kr[r]) + b_c8[r] + j] =
             (gk < K && b_c8[r] + j < M) ? B[(long)gk * M + b_c8[r] + j]
                                         : (ushort)0;
     }
@@ -687,7 +720,7 @@ __global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
   for (int m = 0; m < 8; ++m) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int row = row0 + wrow + m * 16 + (lane >> 4) * 4 + r;
+      const int row = row0 + m * 16 + (lane >> 4) * 4 + r;
       if (row >= N) continue;
 #pragma unroll
       for (int n = 0; n < 6; ++n) {
@@ -1625,7 +1658,7 @@ torch::Tensor gemm_nn_splitk(torch::Tensor A, torch::Tensor B) {
   auto P = torch::empty({(long)S, (long)N, (long)M},
                         A.options().dtype(torch::kFloat32));
   auto C = torch::empty({N, M}, A.options().dtype(torch::kFloat32));
-  const size_t lds = (16384 + 2 * (size_t)GNN_BN * GNN_PKB) * 2;  // 92 KiB
+  const size_t lds = 2UL * (GNN_BM * GNN_PKA + GNN_BK * GNN_PKB) * 2;  // 66 KiB
   static bool nn_configured = false;
   if (!nn_configured) {
     (void)hipFuncSetAttribute((const void*)k_gemm_nn_splitk,
@@ -1633,7 +1666,7 @@ torch::Tensor gemm_nn_splitk(torch::Tensor A, torch::Tensor B) {
                               (int)lds);
     nn_configured = true;
   }
-  k_gemm_nn_splitk<<<S * row_tiles, 512, lds, cur_stream()>>>(
+  k_gemm_nn_splitk<<<S * row_tiles, 256, lds, cur_stream()>>>(
       bf_ptr(A), bf_ptr(B), P.data_ptr<float>(), N, M, K, S, kpc, row_tiles);
   const long total = (long)N * M;
   TORCH_CHECK(total % 4 == 0);
