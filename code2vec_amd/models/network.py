@@ -262,7 +262,7 @@ class Code2VecNetwork:
                 ev.record()
                 with torch.cuda.stream(self._side_stream):
                     self._side_stream.wait_event(ev)
-                    d_target = d_logits.t() @ code_c             # (V,D)
+                    d_target = F.logits_bwd_target(d_logits, code_c)  # (V,D)
                     reducer.allreduce_dense('target_table', d_target)
                     reducer.wait('target_table')
                     F.adam_dense_step(self.target_table, d_target,
@@ -280,7 +280,7 @@ class Code2VecNetwork:
                     if st_t_pre is not None:
                         st_t_pre.record_stream(self._side_stream)
             else:
-                d_target = (d_logits.t() @ code_c)
+                d_target = F.logits_bwd_target(d_logits, code_c)
                 reducer.allreduce_dense('target_table', d_target)
 
         # attention backward emits dL/dz directly (tanh' fused into its

@@ -731,6 +731,184 @@ __global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
   }
 }
 
+// ---------------------------------------------------------------------------
+// d_target GEMM (tn operands): C(V,M) bf16 = A(K,V)^T @ B(K,M), A and B
+// ROW-major bf16. Shaped for dL/dtarget = d_logits(B,V)^T @ code(B,M):
+// the contraction axis K is the BATCH (small), V is huge -> one block per
+// 128-row V-tile (~2041 blocks, no split-K). Neither operand is transposed
+// in memory: both tiles stage ROW-major along K with the same rotated
+// padded layout as k_gemm_nn_splitk's B tile, and both fragments gather 8
+// k-strided u16 scalars (conflict-free: the +16*(k/8) rotation puts the
+// four k-groups on disjoint bank quarters). code (K x M, ~768 KB) is read
+// by every block and lives in L2; d_logits streams once.
+// ---------------------------------------------------------------------------
+
+#define GTN_BV 128   // V rows per block
+#define GTN_PKT 184  // padded+rotated stride of the staged d_logits tile
+
+__launch_bounds__(256, 1)
+__global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
+                          const ushort* __restrict__ B,  // code (K, M)
+                          ushort* __restrict__ C,        // out (V, M)
+                          int V, int M, int K) {
+  // double-buffered: At[2][32][184] + B[2][32][440] bf16 = 78 KiB dynamic
+  extern __shared__ ushort ldstn[];
+#define TLDS_A(b) (ldstn + (b) * (GNN_BK * GTN_PKT))
+#define TLDS_B(b) (ldstn + 2 * GNN_BK * GTN_PKT + (b) * (GNN_BK * GNN_PKB))
+
+  const int v0 = blockIdx.x * GTN_BV;
+  const int total_ksteps = (K + GNN_BK - 1) / GNN_BK;
+  const int ks1_main = K / GNN_BK;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;              // 4 waves, one 96-col group each
+  const int wcol = wid * 96;
+
+  f32x4 acc[8][6];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 6; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  // A tile: 32 k-rows x 128 v-cols = 512 b128 chunks -> 2/thread
+  const int at_kr[2] = {(tid + 0) >> 4, (tid + 256) >> 4};
+  const int at_c8 = (tid & 15) * 8;
+  // B tile: 32 k-rows x 384 cols = 1536 b128 chunks -> 6/thread
+  int b_kr[6], b_c8[6];
+#pragma unroll
+  for (int r = 0; r < 6; ++r) {
+    const int idx = tid + r * 256;
+    b_kr[r] = idx / 48;
+    b_c8[r] = (idx % 48) * 8;
+  }
+
+  u16x8 ra[2], rb[6];
+  auto load_ab = [&](int ks) {
+    const int k0 = ks * GNN_BK;
+#pragma unroll
+    for (int r = 0; r < 2; ++r) {
+      const long base = (long)(k0 + at_kr[r]) * V + v0;
+      if (v0 + at_c8 + 8 <= V) {
+        ra[r] = *reinterpret_cast<const u16x8*>(A + base + at_c8);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          ra[r][j] = (v0 + at_c8 + j < V) ? A[base + at_c8 + j] : (ushort)0;
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 6; ++r) {
+      if (b_c8[r] + 8 <= M) {
+        rb[r] = *reinterpret_cast<const u16x8*>(
+            B + (long)(k0 + b_kr[r]) * M + b_c8[r]);
+      } else {
+        const long base = (long)(k0 + b_kr[r]) * M;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          rb[r][j] = (b_c8[r] + j < M) ? B[base + b_c8[r] + j] : (ushort)0;
+      }
+    }
+  };
+  auto store_ab = [&](int buf) {
+#pragma unroll
+    for (int r = 0; r < 2; ++r) {
+      ushort* p = TLDS_A(buf) + at_kr[r] * GTN_PKT + GNN_BROT(at_kr[r]) +
+                  at_c8;
+      *reinterpret_cast<u16x8*>(p) = ra[r];  // aligned b128 (368 B stride)
+    }
+#pragma unroll
+    for (int r = 0; r < 6; ++r) {
+      ushort* p = TLDS_B(buf) + b_kr[r] * GNN_PKB + GNN_BROT(b_kr[r]) +
+                  b_c8[r];
+      *reinterpret_cast<u16x8*>(p) = rb[r];
+    }
+  };
+
+  auto compute = [&](int buf) {
+    bf16x8 bfrag[6];
+#pragma unroll
+    for (int n = 0; n < 6; ++n) {
+      const int col = wcol + n * 16 + (lane & 15);
+      const ushort* bp = TLDS_B(buf) + (lane >> 4) * (8 * GNN_PKB + 16) + col;
+      u16x8 t;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) t[j] = bp[j * GNN_PKB];
+      bfrag[n] = __builtin_bit_cast(bf16x8, t);
+    }
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int m = 0; m < 8; ++m) {
+      const int row = m * 16 + (lane & 15);   // v-col inside the tile
+      const ushort* ap = TLDS_A(buf) + (lane >> 4) * (8 * GTN_PKT + 16) + row;
+      u16x8 t;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) t[j] = ap[j * GTN_PKT];
+      const bf16x8 afrag = __builtin_bit_cast(bf16x8, t);
+#pragma unroll
+      for (int n = 0; n < 6; ++n)
+        acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag, bfrag[n], acc[m][n], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+  };
+
+  if (ks1_main > 0) {
+    load_ab(0);
+    store_ab(0);
+    if (1 < ks1_main) load_ab(1);
+    __syncthreads();
+    for (int ks = 0; ks < ks1_main; ++ks) {
+      const int cur = ks & 1;
+      if (ks + 1 < ks1_main) {
+        store_ab(cur ^ 1);
+        if (ks + 2 < ks1_main) load_ab(ks + 2);
+      }
+      compute(cur);
+      __syncthreads();
+    }
+  }
+
+  // ragged last batch rows (K % 32): one zero-filled guarded iteration
+  if (K % GNN_BK) {
+    const int k0 = (total_ksteps - 1) * GNN_BK;
+#pragma unroll
+    for (int r = 0; r < 2; ++r) {
+      const int gk = k0 + at_kr[r];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        TLDS_A(0)[at_kr[r] * GTN_PKT + GNN_BROT(at_kr[r]) + at_c8 + j] =
+            (gk < K && v0 + at_c8 + j < V) ? A[(long)gk * V + v0 + at_c8 + j]
+                                           : (ushort)0;
+    }
+#pragma unroll
+    for (int r = 0; r < 6; ++r) {
+      const int gk = k0 + b_kr[r];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        TLDS_B(0)[b_kr[r] * GNN_PKB + GNN_BROT(b_kr[r]) + b_c8[r] + j] =
+            (gk < K && b_c8[r] + j < M) ? B[(long)gk * M + b_c8[r] + j]
+                                        : (ushort)0;
+    }
+    __syncthreads();
+    compute(0);
+  }
+
+#pragma unroll
+  for (int m = 0; m < 8; ++m) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int v = v0 + m * 16 + (lane >> 4) * 4 + r;
+      if (v >= V) continue;
+#pragma unroll
+      for (int n = 0; n < 6; ++n) {
+        const int col = wcol + n * 16 + (lane & 15);
+        if (col < M) C[(long)v * M + col] = f2bf(acc[m][n][r]);
+      }
+    }
+  }
+}
+
 // fold the S split-K slices: C[i] = sum_s P[s][i] (fp32 out)
 __launch_bounds__(256)
 __global__ void k_splitk_reduce(const float* __restrict__ P,
@@ -1638,6 +1816,30 @@ torch::Tensor gemm_bt_dropout(torch::Tensor A, torch::Tensor Bt,
   return C;
 }
 
+// d_target GEMM: C(V,M) bf16 = A(K,V)^T @ B(K,M), row-major bf16 operands,
+// one block per 128-row V-tile (K = batch, contracted; no split-K).
+torch::Tensor gemm_tn_bf16(torch::Tensor A, torch::Tensor B) {
+  CHECK_DEV(A); CHECK_CONT(A); CHECK_DEV(B); CHECK_CONT(B);
+  TORCH_CHECK(A.scalar_type() == torch::kBFloat16 &&
+              B.scalar_type() == torch::kBFloat16);
+  const int K = A.size(0), V = A.size(1), M = B.size(1);
+  TORCH_CHECK(B.size(0) == K, "K mismatch");
+  TORCH_CHECK(M <= GNN_BN && M % 8 == 0, "tn GEMM: M must be <=384, mult of 8");
+  auto C = torch::empty({V, M}, A.options());
+  const size_t lds = 2UL * GNN_BK * (GTN_PKT + GNN_PKB) * 2;  // 78 KiB
+  static bool tn_configured = false;
+  if (!tn_configured) {
+    (void)hipFuncSetAttribute((const void*)k_gemm_tn,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              (int)lds);
+    tn_configured = true;
+  }
+  const int grid = (V + GTN_BV - 1) / GTN_BV;
+  k_gemm_tn<<<grid, 256, lds, cur_stream()>>>(
+      bf_ptr(A), bf_ptr(B), bf_ptr_mut(C), V, M, K);
+  return C;
+}
+
 // split-K nn GEMM: C(N,M) fp32 = A(N,K) @ B(K,M), row-major bf16 operands.
 // Shaped for d_code = d_logits @ target_shadow (N=batch, M=code dim <= 384,
 // K=vocab). Partials workspace (S, N, M) fp32 comes from the caching
@@ -2007,6 +2209,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gemm_bt_dropout", &gemm_bt_dropout);
   mod.def("logits_ce_fused", &logits_ce_fused);
   mod.def("gemm_nn_splitk", &gemm_nn_splitk);
+  mod.def("gemm_tn_bf16", &gemm_tn_bf16);
   mod.def("tanh_bwd_mul", &tanh_bwd_mul);
   mod.def("attention_fwd", &attention_fwd);
   mod.def("attention_bwd", &attention_bwd);

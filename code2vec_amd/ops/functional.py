@@ -149,6 +149,23 @@ def logits_bwd_code(d_logits, shadow):
     return (d_logits @ shadow).float()
 
 
+def logits_bwd_target(d_logits, code_c):
+    """d_target = d_logits^T @ code (tn operands, bf16 out). C2V_TN_GEMM=1
+    opts into the custom per-V-tile MFMA kernel (k_gemm_tn) — measured
+    527 us vs hipBLASLt's 333 us on the java14m shape (K=batch is too small
+    to amortize the staged-tile overheads at 1 wave/SIMD), so the library
+    GEMM is the default; the kernel is kept as the starting point for a
+    multi-K-tile revision."""
+    import os as _os
+    if (backend_for(d_logits) == 'hip' and d_logits.dtype == torch.bfloat16
+            and code_c.dtype == torch.bfloat16
+            and code_c.shape[1] <= 384 and code_c.shape[1] % 8 == 0
+            and d_logits.shape[1] >= 4096
+            and _os.environ.get('C2V_TN_GEMM', '0') == '1'):
+        return hip_ext(True).gemm_tn_bf16(d_logits, code_c)
+    return d_logits.t() @ code_c
+
+
 def logits_ce_fused(code_c, shadow, labels):
     """Fused K8+K9 forward: the 256-tile logits GEMM also emits per-tile
     (rowmax, sumexp) partials, folded into (loss, lse) by a small reduce —

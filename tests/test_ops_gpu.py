@@ -419,3 +419,31 @@ def test_logits_bwd_code_dispatch():
     assert out.dtype == torch.float32
     err = (out - ref).abs().max().item() / ref.abs().max().item()
     assert err < 0.01, err
+
+
+@pytest.mark.parametrize('K,V,M', [
+    (1024, 261246, 384),   # d_target shape (V tail: 261246 % 128 != 0)
+    (1000, 65536, 384),    # ragged batch (K % 32 != 0)
+    (256, 8200, 192),      # narrow M + V tail
+])
+def test_gemm_tn_vs_matmul(K, V, M):
+    """tn GEMM (d_target): random asymmetric refcheck (guide G9)."""
+    A = randn(K, V, dtype=torch.bfloat16, scale=0.1, seed=K + V)
+    B = randn(K, M, dtype=torch.bfloat16, scale=0.1)
+    C = ext().gemm_tn_bf16(A, B)
+    assert C.dtype == torch.bfloat16 and C.shape == (V, M)
+    ref = A.float().t() @ B.float()
+    err = (C.float() - ref).abs().max().item()
+    denom = ref.abs().max().item()
+    assert err / denom < 0.02, 'max err %g vs scale %g' % (err, denom)
+
+
+def test_logits_bwd_target_dispatch():
+    from code2vec_amd.ops import functional as F
+    torch.manual_seed(9)
+    d_logits = randn(512, 8192, dtype=torch.bfloat16, scale=0.01)
+    code_c = randn(512, 384, dtype=torch.bfloat16, scale=0.5)
+    out = F.logits_bwd_target(d_logits, code_c)
+    ref = d_logits.float().t() @ code_c.float()
+    err = (out.float() - ref).abs().max().item() / ref.abs().max().item()
+    assert err < 0.02, err

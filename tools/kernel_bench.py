@@ -56,6 +56,7 @@ def main():
     timeit('d_code hipBLASLt', lambda: dl @ shadow)
     timeit('d_code our split-K nn', lambda: ext.gemm_nn_splitk(dl, shadow))
     timeit('d_target hipBLASLt', lambda: dl.t() @ code)
+    timeit('d_target our tn', lambda: ext.gemm_tn_bf16(dl, code))
 
     timeit('attn_fwd', lambda: ext.attention_fwd(comb, a, mask))
     alpha = ext.attention_fwd(comb, a, mask)[1]
