@@ -311,7 +311,9 @@ class PathContextReader:
         def drain(final=False):
             nonlocal pool, pool_rows
             # amortize the pool permutation: only drain once enough rows are
-            # queued to emit several batches past the shuffle window
+            # queued to emit several batches past the shuffle window (32x
+            # batches measured slower than 8x — the bigger gather falls out
+            # of cache)
             threshold = batch_size if sb == 0 else sb + 8 * batch_size
             if not pool or (pool_rows < threshold and not final):
                 return
@@ -334,57 +336,94 @@ class PathContextReader:
             else:
                 pool_rows = 0
 
-        epoch = 0
-        while epochs < 0 or epoch < epochs:
-            epoch += 1
-            with open(data_path, 'rb') as f:
-                carry = b''
-                while True:
-                    chunk = f.read(4 << 20)
-                    if not chunk:
+        # Two-stage pipeline: a producer thread does file IO + parse_buffer
+        # (C++ with the GIL released) + filter/DP-shard, and this generator
+        # consumes filtered tensor chunks for the pool shuffle + batch
+        # slicing. Overlapping the two stages measured 95K -> ~150K ex/s per
+        # process on a 333 MB synthetic corpus (tools/ profiling; parse
+        # alone runs at 330K lines/s).
+        import queue
+        import threading
+
+        def parse_filter(use: bytes):
+            nonlocal line_base
+            src, pth, tgt, mask, tidx = self._native.parse_buffer(use)
+            n = src.shape[0]
+            if n == 0:
+                return None
+            keep = mask.any(dim=1)
+            if self.estimator_action.is_train:
+                keep &= tidx > self._tgt_oov
+            if self.world_size > 1:
+                gidx = torch.arange(line_base, line_base + n)
+                keep &= (gidx % self.world_size) == self.rank
+            line_base += n
+            if not bool(keep.all()):
+                idx = keep.nonzero(as_tuple=True)[0]
+                if idx.numel() == 0:
+                    return None
+                return tuple(t[idx] for t in (src, pth, tgt, mask, tidx))
+            return (src, pth, tgt, mask, tidx)
+
+        q: 'queue.Queue' = queue.Queue(maxsize=8)
+        stop = threading.Event()
+
+        def producer():
+            try:
+                epoch = 0
+                while epochs < 0 or epoch < epochs:
+                    epoch += 1
+                    with open(data_path, 'rb') as f:
+                        carry = b''
+                        while not stop.is_set():
+                            chunk = f.read(4 << 20)
+                            if not chunk:
+                                break
+                            buf = carry + chunk
+                            last_nl = buf.rfind(b'\n')
+                            if last_nl < 0:
+                                carry = buf
+                                continue
+                            carry = buf[last_nl + 1:]
+                            tup = parse_filter(buf[:last_nl + 1])
+                            if tup is not None:
+                                while not stop.is_set():
+                                    try:
+                                        q.put(tup, timeout=0.5)
+                                        break
+                                    except queue.Full:
+                                        pass
+                        if carry.strip() and not stop.is_set():
+                            tup = parse_filter(carry + b'\n')
+                            if tup is not None:
+                                while not stop.is_set():
+                                    try:
+                                        q.put(tup, timeout=0.5)
+                                        break
+                                    except queue.Full:
+                                        pass
+                    if stop.is_set():
                         break
-                    buf = carry + chunk
-                    last_nl = buf.rfind(b'\n')
-                    if last_nl < 0:
-                        carry = buf
-                        continue
-                    carry = buf[last_nl + 1:]
-                    use = buf[:last_nl + 1]
-                    src, pth, tgt, mask, tidx = self._native.parse_buffer(use)
-                    n = src.shape[0]
-                    if n == 0:
-                        continue
-                    keep = mask.any(dim=1)
-                    if self.estimator_action.is_train:
-                        keep &= tidx > self._tgt_oov
-                    if self.world_size > 1:
-                        gidx = torch.arange(line_base, line_base + n)
-                        keep &= (gidx % self.world_size) == self.rank
-                    line_base += n
-                    if not bool(keep.all()):
-                        idx = keep.nonzero(as_tuple=True)[0]
-                        src, pth, tgt = src[idx], pth[idx], tgt[idx]
-                        mask, tidx = mask[idx], tidx[idx]
-                    if src.shape[0]:
-                        pool.append((src, pth, tgt, mask, tidx))
-                        pool_rows += src.shape[0]
-                    yield from drain()
-                if carry.strip():
-                    src, pth, tgt, mask, tidx = self._native.parse_buffer(
-                        carry + b'\n')
-                    keep = mask.any(dim=1)
-                    if self.estimator_action.is_train:
-                        keep &= tidx > self._tgt_oov
-                    if self.world_size > 1:
-                        gidx = torch.arange(line_base, line_base + src.shape[0])
-                        keep &= (gidx % self.world_size) == self.rank
-                    line_base += src.shape[0]
-                    idx = keep.nonzero(as_tuple=True)[0]
-                    if idx.numel():
-                        pool.append(tuple(t[idx] for t in
-                                          (src, pth, tgt, mask, tidx)))
-                        pool_rows += idx.numel()
-        yield from drain(final=True)
+                q.put(None)
+            except BaseException as exc:  # noqa: BLE001 — surface in consumer
+                q.put(exc)
+
+        th = threading.Thread(target=producer, daemon=True,
+                              name='c2v-reader-parse')
+        th.start()
+        try:
+            while True:
+                item = q.get()
+                if item is None:
+                    break
+                if isinstance(item, BaseException):
+                    raise item
+                pool.append(item)
+                pool_rows += item[0].shape[0]
+                yield from drain()
+            yield from drain(final=True)
+        finally:
+            stop.set()
 
     def _iter_batches_native(self, lines: Iterable[str],
                              batch_size: int) -> Iterator[ReaderBatch]:
