@@ -238,8 +238,21 @@ class Code2VecNetwork:
                                                        self.target_shadow,
                                                        labels)
             loss = loss_rows.float().mean()
-            d_logits = F.ce_bwd(logits, lse, labels, 1.0 / B)
-            d_code = F.logits_bwd_code(d_logits, self.target_shadow)  # (B,D)
+            # C2V_FUSED_CEBWD=1: CE backward computed inside the consumers'
+            # staging (gemm_nn_splitk_ce / gemm_tn_ce) — d_logits is never
+            # materialized (saves a 535 MB write + two 535 MB reads). Kept
+            # opt-in: measured net-neutral-to-slower at DP=1 because the tn
+            # d_target kernel it requires is slower than hipBLASLt there
+            # (profiles/r01_optimization_log.md).
+            ce_scale = 1.0 / B
+            ce_fused = F.ce_bwd_fused_available(logits)
+            if ce_fused:
+                d_logits = None
+                d_code = F.logits_bwd_code_ce(logits, self.target_shadow,
+                                              lse, labels, ce_scale)
+            else:
+                d_logits = F.ce_bwd(logits, lse, labels, ce_scale)
+                d_code = F.logits_bwd_code(d_logits, self.target_shadow)
             d_target = None  # see target-chain dispatch below
 
         side_done = None
@@ -262,7 +275,11 @@ class Code2VecNetwork:
                 ev.record()
                 with torch.cuda.stream(self._side_stream):
                     self._side_stream.wait_event(ev)
-                    d_target = F.logits_bwd_target(d_logits, code_c)  # (V,D)
+                    if ce_fused:
+                        d_target = F.logits_bwd_target_ce(
+                            logits, code_c, lse, labels, ce_scale)
+                    else:
+                        d_target = F.logits_bwd_target(d_logits, code_c)
                     reducer.allreduce_dense('target_table', d_target)
                     reducer.wait('target_table')
                     F.adam_dense_step(self.target_table, d_target,
@@ -275,12 +292,18 @@ class Code2VecNetwork:
                     side_done = torch.cuda.Event()
                     side_done.record()
                 if not torch.cuda.is_current_stream_capturing():
-                    d_logits.record_stream(self._side_stream)
-                    code_c.record_stream(self._side_stream)
+                    side_inputs = ((logits, lse, labels, code_c) if ce_fused
+                                   else (d_logits, code_c))
+                    for t in side_inputs:
+                        t.record_stream(self._side_stream)
                     if st_t_pre is not None:
                         st_t_pre.record_stream(self._side_stream)
             else:
-                d_target = F.logits_bwd_target(d_logits, code_c)
+                if ce_fused:
+                    d_target = F.logits_bwd_target_ce(logits, code_c, lse,
+                                                      labels, ce_scale)
+                else:
+                    d_target = F.logits_bwd_target(d_logits, code_c)
                 reducer.allreduce_dense('target_table', d_target)
 
         # attention backward emits dL/dz directly (tanh' fused into its

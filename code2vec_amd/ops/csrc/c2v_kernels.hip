@@ -537,12 +537,20 @@ __global__ void k_ce_reduce_partials(const float* __restrict__ partials,
 #define GNN_PKB 440
 #define GNN_BROT(k) (16 * ((k) >> 3))
 
+// CEB: CE-backward fused into the A staging — A holds LOGITS and each
+// staged element becomes d_logit = bf16((exp(l - lse[row]) - onehot)*scale),
+// bit-identical to k_ce_bwd's output, so the 535 MB d_logits tensor is
+// never materialized (k_ce_bwd read 535 + wrote 535 MB per step).
+template <bool CEB = false>
 __launch_bounds__(256, 1)
 __global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
                                  const ushort* __restrict__ B,
                                  float* __restrict__ P,  // (S, N, M) partials
                                  int N, int M, int K, int S,
-                                 int ksteps_per_chunk, int row_tiles) {
+                                 int ksteps_per_chunk, int row_tiles,
+                                 const float* __restrict__ lse = nullptr,
+                                 const long* __restrict__ labels = nullptr,
+                                 float ce_scale = 1.f) {
   // double-buffered: A[2][128][36] + B[2][32][440] bf16 = 73 KiB dynamic.
   // 256 threads = 4 waves = 1 wave/SIMD: the per-SIMD register pool is 512
   // regs/lane, the only occupancy at which the 192-reg accumulator tile
@@ -602,12 +610,26 @@ __global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
   }
 
   u16x8 ra[2], rb[6];
+  auto ce_map = [&](u16x8 v, int grow, int kbase) -> u16x8 {
+    if (!CEB) return v;
+    const float l = lse[grow];
+    const long lab = labels[grow];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float pr = __expf(bf2f(v[j]) - l);
+      if (kbase + j == lab) pr -= 1.f;
+      v[j] = f2bf(pr * ce_scale);
+    }
+    return v;
+  };
   auto load_ab = [&](int ks) {
     const int k0 = ks * GNN_BK;
 #pragma unroll
     for (int r = 0; r < 2; ++r) {
       const int grow = min(row0 + a_row[r], N - 1);
-      ra[r] = *reinterpret_cast<const u16x8*>(A + (long)grow * K + k0 + a_k8);
+      ra[r] = ce_map(*reinterpret_cast<const u16x8*>(
+                         A + (long)grow * K + k0 + a_k8),
+                     grow, k0 + a_k8);
     }
 #pragma unroll
     for (int r = 0; r < 6; ++r) {
@@ -696,11 +718,20 @@ __global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
 #pragma unroll
     for (int r = 0; r < 2; ++r) {
       const int grow = min(row0 + a_row[r], N - 1);
+      u16x8 v;
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        NLDS_A(0)[a_row[r] * GNN_PKA + a_k8 + j] =
-            (k0 + a_k8 + j < K) ? A[(long)grow * K + k0 + a_k8 + j]
-                                : (ushort)0;
+        v[j] = (k0 + a_k8 + j < K) ? A[(long)grow * K + k0 + a_k8 + j]
+                                   : (ushort)0;
+      if (CEB) {
+        const u16x8 m = ce_map(v, grow, k0 + a_k8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          v[j] = (k0 + a_k8 + j < K) ? m[j] : (ushort)0;
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        NLDS_A(0)[a_row[r] * GNN_PKA + a_k8 + j] = v[j];
     }
 #pragma unroll
     for (int r = 0; r < 6; ++r) {
@@ -746,11 +777,15 @@ __global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
 #define GTN_BV 128   // V rows per block
 #define GTN_PKT 184  // padded+rotated stride of the staged d_logits tile
 
+template <bool CEB = false>
 __launch_bounds__(256, 1)
 __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
                           const ushort* __restrict__ B,  // code (K, M)
                           ushort* __restrict__ C,        // out (V, M)
-                          int V, int M, int K) {
+                          int V, int M, int K,
+                          const float* __restrict__ lse = nullptr,
+                          const long* __restrict__ labels = nullptr,
+                          float ce_scale = 1.f) {
   // double-buffered: At[2][32][184] + B[2][32][440] bf16 = 78 KiB dynamic
   extern __shared__ ushort ldstn[];
 #define TLDS_A(b) (ldstn + (b) * (GNN_BK * GTN_PKT))
@@ -784,17 +819,35 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
   }
 
   u16x8 ra[2], rb[6];
+  auto ce_map = [&](u16x8 v, int gk, int vbase) -> u16x8 {
+    if (!CEB) return v;
+    const float l = lse[gk];
+    const long lab = labels[gk];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float pr = __expf(bf2f(v[j]) - l);
+      if (vbase + j == lab) pr -= 1.f;
+      v[j] = f2bf(pr * ce_scale);
+    }
+    return v;
+  };
   auto load_ab = [&](int ks) {
     const int k0 = ks * GNN_BK;
 #pragma unroll
     for (int r = 0; r < 2; ++r) {
       const long base = (long)(k0 + at_kr[r]) * V + v0;
       if (v0 + at_c8 + 8 <= V) {
-        ra[r] = *reinterpret_cast<const u16x8*>(A + base + at_c8);
+        ra[r] = ce_map(*reinterpret_cast<const u16x8*>(A + base + at_c8),
+                       k0 + at_kr[r], v0 + at_c8);
       } else {
+        u16x8 v;
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          ra[r][j] = (v0 + at_c8 + j < V) ? A[base + at_c8 + j] : (ushort)0;
+          v[j] = (v0 + at_c8 + j < V) ? A[base + at_c8 + j] : (ushort)0;
+        const u16x8 m = ce_map(v, k0 + at_kr[r], v0 + at_c8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          ra[r][j] = (v0 + at_c8 + j < V) ? m[j] : (ushort)0;
       }
     }
 #pragma unroll
@@ -875,11 +928,20 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
 #pragma unroll
     for (int r = 0; r < 2; ++r) {
       const int gk = k0 + at_kr[r];
+      u16x8 v;
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        TLDS_A(0)[at_kr[r] * GTN_PKT + GNN_BROT(at_kr[r]) + at_c8 + j] =
-            (gk < K && v0 + at_c8 + j < V) ? A[(long)gk * V + v0 + at_c8 + j]
-                                           : (ushort)0;
+        v[j] = (gk < K && v0 + at_c8 + j < V)
+                   ? A[(long)gk * V + v0 + at_c8 + j] : (ushort)0;
+      if (CEB && gk < K) {
+        const u16x8 m = ce_map(v, gk, v0 + at_c8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          v[j] = (v0 + at_c8 + j < V) ? m[j] : (ushort)0;
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        TLDS_A(0)[at_kr[r] * GTN_PKT + GNN_BROT(at_kr[r]) + at_c8 + j] = v[j];
     }
 #pragma unroll
     for (int r = 0; r < 6; ++r) {
@@ -1829,14 +1891,43 @@ torch::Tensor gemm_tn_bf16(torch::Tensor A, torch::Tensor B) {
   const size_t lds = 2UL * GNN_BK * (GTN_PKT + GNN_PKB) * 2;  // 78 KiB
   static bool tn_configured = false;
   if (!tn_configured) {
-    (void)hipFuncSetAttribute((const void*)k_gemm_tn,
+    (void)hipFuncSetAttribute((const void*)k_gemm_tn<false>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              (int)lds);
+    (void)hipFuncSetAttribute((const void*)k_gemm_tn<true>,
                               hipFuncAttributeMaxDynamicSharedMemorySize,
                               (int)lds);
     tn_configured = true;
   }
   const int grid = (V + GTN_BV - 1) / GTN_BV;
-  k_gemm_tn<<<grid, 256, lds, cur_stream()>>>(
+  k_gemm_tn<false><<<grid, 256, lds, cur_stream()>>>(
       bf_ptr(A), bf_ptr(B), bf_ptr_mut(C), V, M, K);
+  return C;
+}
+
+// d_target with CE-backward fused into the d_logits staging:
+// C(V,M) bf16 = ce_bwd(logits, lse, labels, scale)^T @ code — d_logits is
+// computed on the fly, never written to HBM.
+torch::Tensor gemm_tn_ce(torch::Tensor logits, torch::Tensor code,
+                         torch::Tensor lse, torch::Tensor labels,
+                         double scale) {
+  CHECK_DEV(logits); CHECK_CONT(logits); CHECK_DEV(code); CHECK_CONT(code);
+  const int K = logits.size(0), V = logits.size(1), M = code.size(1);
+  TORCH_CHECK(code.size(0) == K && M <= GNN_BN && M % 8 == 0);
+  auto labels_c = labels.contiguous();
+  auto C = torch::empty({V, M}, logits.options());
+  const size_t lds = 2UL * GNN_BK * (GTN_PKT + GNN_PKB) * 2;
+  static bool cfg2 = false;
+  if (!cfg2) {
+    (void)hipFuncSetAttribute((const void*)k_gemm_tn<true>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              (int)lds);
+    cfg2 = true;
+  }
+  const int grid = (V + GTN_BV - 1) / GTN_BV;
+  k_gemm_tn<true><<<grid, 256, lds, cur_stream()>>>(
+      bf_ptr(logits), bf_ptr(code), bf_ptr_mut(C), V, M, K,
+      lse.data_ptr<float>(), labels_c.data_ptr<long>(), (float)scale);
   return C;
 }
 
@@ -1863,13 +1954,53 @@ torch::Tensor gemm_nn_splitk(torch::Tensor A, torch::Tensor B) {
   const size_t lds = 2UL * (GNN_BM * GNN_PKA + GNN_BK * GNN_PKB) * 2;  // 66 KiB
   static bool nn_configured = false;
   if (!nn_configured) {
-    (void)hipFuncSetAttribute((const void*)k_gemm_nn_splitk,
+    (void)hipFuncSetAttribute((const void*)k_gemm_nn_splitk<false>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              (int)lds);
+    (void)hipFuncSetAttribute((const void*)k_gemm_nn_splitk<true>,
                               hipFuncAttributeMaxDynamicSharedMemorySize,
                               (int)lds);
     nn_configured = true;
   }
-  k_gemm_nn_splitk<<<S * row_tiles, 256, lds, cur_stream()>>>(
+  k_gemm_nn_splitk<false><<<S * row_tiles, 256, lds, cur_stream()>>>(
       bf_ptr(A), bf_ptr(B), P.data_ptr<float>(), N, M, K, S, kpc, row_tiles);
+  const long total = (long)N * M;
+  TORCH_CHECK(total % 4 == 0);
+  k_splitk_reduce<<<grid_1d(total / 4, 256), 256, 0, cur_stream()>>>(
+      P.data_ptr<float>(), C.data_ptr<float>(), S, total);
+  return C;
+}
+
+// d_code with CE-backward fused into the d_logits staging:
+// C(N,M) fp32 = ce_bwd(logits, lse, labels, scale) @ shadow.
+torch::Tensor gemm_nn_splitk_ce(torch::Tensor logits, torch::Tensor shadow,
+                                torch::Tensor lse, torch::Tensor labels,
+                                double scale) {
+  CHECK_DEV(logits); CHECK_CONT(logits);
+  CHECK_DEV(shadow); CHECK_CONT(shadow);
+  const int N = logits.size(0), K = logits.size(1), M = shadow.size(1);
+  TORCH_CHECK(shadow.size(0) == K && M <= GNN_BN && M % 8 == 0);
+  auto labels_c = labels.contiguous();
+  const int row_tiles = (N + GNN_BM - 1) / GNN_BM;
+  int S = 256 / (row_tiles * 8) * 8;
+  if (S < 8) S = 8;
+  const int total_ksteps = (K + GNN_BK - 1) / GNN_BK;
+  const int kpc = (total_ksteps + S - 1) / S;
+  auto P = torch::empty({(long)S, (long)N, (long)M},
+                        logits.options().dtype(torch::kFloat32));
+  auto C = torch::empty({N, M}, logits.options().dtype(torch::kFloat32));
+  const size_t lds = 2UL * (GNN_BM * GNN_PKA + GNN_BK * GNN_PKB) * 2;
+  static bool cfg2 = false;
+  if (!cfg2) {
+    (void)hipFuncSetAttribute((const void*)k_gemm_nn_splitk<true>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              (int)lds);
+    cfg2 = true;
+  }
+  k_gemm_nn_splitk<true><<<S * row_tiles, 256, lds, cur_stream()>>>(
+      bf_ptr(logits), bf_ptr(shadow), P.data_ptr<float>(), N, M, K, S, kpc,
+      row_tiles, lse.data_ptr<float>(), labels_c.data_ptr<long>(),
+      (float)scale);
   const long total = (long)N * M;
   TORCH_CHECK(total % 4 == 0);
   k_splitk_reduce<<<grid_1d(total / 4, 256), 256, 0, cur_stream()>>>(
@@ -2210,6 +2341,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("logits_ce_fused", &logits_ce_fused);
   mod.def("gemm_nn_splitk", &gemm_nn_splitk);
   mod.def("gemm_tn_bf16", &gemm_tn_bf16);
+  mod.def("gemm_tn_ce", &gemm_tn_ce);
+  mod.def("gemm_nn_splitk_ce", &gemm_nn_splitk_ce);
   mod.def("tanh_bwd_mul", &tanh_bwd_mul);
   mod.def("attention_fwd", &attention_fwd);
   mod.def("attention_bwd", &attention_bwd);

@@ -166,6 +166,42 @@ def logits_bwd_target(d_logits, code_c):
     return d_logits.t() @ code_c
 
 
+def ce_bwd_fused_available(logits):
+    """True when the fully-fused CE backward path (d_logits never
+    materialized: gemm_nn_splitk_ce for d_code + gemm_tn_ce for d_target)
+    should run. OPT-IN (C2V_FUSED_CEBWD=1): at DP=1 it measured slightly
+    slower than the materialized path because it forces d_target onto the
+    custom tn kernel (527 us vs hipBLASLt 333) and adds exp() latency to
+    the nn GEMM's staging chain — see profiles/r01_optimization_log.md."""
+    import os as _os
+    return (backend_for(logits) == 'hip' and logits.dtype == torch.bfloat16
+            and logits.shape[1] >= 4096
+            and _os.environ.get('C2V_FUSED_CEBWD', '0') == '1')
+
+
+def logits_bwd_code_ce(logits, shadow, lse, labels, scale):
+    """d_code = ce_bwd(logits, lse, labels, scale) @ shadow with the CE
+    backward computed inside the GEMM's A staging (bit-identical bf16
+    d_logit values to k_ce_bwd; the 535 MB d_logits tensor is skipped)."""
+    if ce_bwd_fused_available(logits) and shadow.shape[1] <= 384 \
+            and shadow.shape[1] % 8 == 0:
+        return hip_ext(True).gemm_nn_splitk_ce(logits, shadow, lse, labels,
+                                               float(scale))
+    d_logits = ce_bwd(logits, lse, labels, scale)
+    return logits_bwd_code(d_logits, shadow)
+
+
+def logits_bwd_target_ce(logits, code_c, lse, labels, scale):
+    """d_target = ce_bwd(logits, lse, labels, scale)^T @ code, CE backward
+    fused into the tn GEMM's staging (no d_logits materialization)."""
+    if ce_bwd_fused_available(logits) and code_c.shape[1] <= 384 \
+            and code_c.shape[1] % 8 == 0:
+        return hip_ext(True).gemm_tn_ce(logits, code_c, lse, labels,
+                                        float(scale))
+    d_logits = ce_bwd(logits, lse, labels, scale)
+    return logits_bwd_target(d_logits, code_c)
+
+
 def logits_ce_fused(code_c, shadow, labels):
     """Fused K8+K9 forward: the 256-tile logits GEMM also emits per-tile
     (rowmax, sumexp) partials, folded into (loss, lse) by a small reduce —

@@ -447,3 +447,53 @@ def test_logits_bwd_target_dispatch():
     ref = d_logits.float().t() @ code_c.float()
     err = (out.float() - ref).abs().max().item() / ref.abs().max().item()
     assert err < 0.02, err
+
+
+def test_ce_fused_consumers_match_materialized():
+    """gemm_nn_splitk_ce / gemm_tn_ce vs k_ce_bwd + plain GEMMs: the fused
+    staging computes bit-identical bf16 d_logit values, so the outputs must
+    match the materialized two-kernel compositions near-exactly."""
+    e = ext()
+    torch.manual_seed(11)
+    B, V, D = 512, 70030, 384
+    code = randn(B, D, dtype=torch.bfloat16, scale=0.3)
+    shadow = randn(V, D, dtype=torch.bfloat16, scale=0.3)
+    labels = torch.randint(0, V, (B,), device='cuda')
+    logits, _, lse = e.logits_ce_fused(code, shadow, labels)
+    scale = 1.0 / B
+
+    d_logits = e.ce_bwd(logits, lse, labels, scale)
+    ref_code = e.gemm_nn_splitk(d_logits, shadow)
+    fused_code = e.gemm_nn_splitk_ce(logits, shadow, lse, labels, scale)
+    err = (fused_code - ref_code).abs().max().item()
+    dn = ref_code.abs().max().item()
+    assert err / dn < 1e-3, (err, dn)
+
+    ref_tgt = e.gemm_tn_bf16(d_logits, code)
+    fused_tgt = e.gemm_tn_ce(logits, code, lse, labels, scale)
+    err = (fused_tgt.float() - ref_tgt.float()).abs().max().item()
+    dn = ref_tgt.float().abs().max().item()
+    assert err / dn < 1e-3, (err, dn)
+
+
+def test_ce_fused_consumers_vs_fp32_oracle():
+    """End math check: fused d_code/d_target vs an fp32 softmax-grad oracle."""
+    e = ext()
+    torch.manual_seed(12)
+    B, V, D = 256, 8192, 384
+    code = randn(B, D, dtype=torch.bfloat16, scale=0.3)
+    shadow = randn(V, D, dtype=torch.bfloat16, scale=0.3)
+    labels = torch.randint(0, V, (B,), device='cuda')
+    logits, _, lse = e.logits_ce_fused(code, shadow, labels)
+    scale = 1.0 / B
+    lf = logits.float()
+    dl = (torch.softmax(lf, dim=1)
+          - torch.nn.functional.one_hot(labels, V).float()) * scale
+    ref_code = dl @ shadow.float()
+    ref_tgt = dl.t() @ code.float()
+    fused_code = e.gemm_nn_splitk_ce(logits, shadow, lse, labels, scale)
+    fused_tgt = e.gemm_tn_ce(logits, code, lse, labels, scale).float()
+    assert (fused_code - ref_code).abs().max().item() \
+        / ref_code.abs().max().item() < 0.02
+    assert (fused_tgt - ref_tgt).abs().max().item() \
+        / ref_tgt.abs().max().item() < 0.02
