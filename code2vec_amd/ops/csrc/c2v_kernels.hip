@@ -777,7 +777,11 @@ __global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
 #define GTN_BV 128   // V rows per block
 #define GTN_PKT 184  // padded+rotated stride of the staged d_logits tile
 
-template <bool CEB = false>
+// SPLITK: for tn shapes whose contraction axis is HUGE and output tiny
+// (dW = ctx^T @ d_z: K = B*C ~ 205K, out 384x384) — blocks are
+// (row-tile, k-chunk) pairs like k_gemm_nn_splitk, accumulate fp32
+// partials P[(chunk, V, M)] folded by k_splitk_reduce. C is unused then.
+template <bool CEB = false, bool SPLITK = false>
 __launch_bounds__(256, 1)
 __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
                           const ushort* __restrict__ B,  // code (K, M)
@@ -785,15 +789,30 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
                           int V, int M, int K,
                           const float* __restrict__ lse = nullptr,
                           const long* __restrict__ labels = nullptr,
-                          float ce_scale = 1.f) {
+                          float ce_scale = 1.f,
+                          float* __restrict__ P = nullptr,
+                          int S = 1, int ksteps_per_chunk = 0,
+                          int row_tiles = 1) {
   // double-buffered: At[2][32][184] + B[2][32][440] bf16 = 78 KiB dynamic
   extern __shared__ ushort ldstn[];
 #define TLDS_A(b) (ldstn + (b) * (GNN_BK * GTN_PKT))
 #define TLDS_B(b) (ldstn + 2 * GNN_BK * GTN_PKT + (b) * (GNN_BK * GNN_PKB))
 
-  const int v0 = blockIdx.x * GTN_BV;
+  int v0, chunk = 0, ks0 = 0, ks1;
   const int total_ksteps = (K + GNN_BK - 1) / GNN_BK;
-  const int ks1_main = K / GNN_BK;
+  if (SPLITK) {
+    // XCD-aware decode (see k_gemm_nn_splitk)
+    const int xcd = blockIdx.x & 7;
+    const int slot = blockIdx.x >> 3;
+    chunk = (slot / row_tiles) * 8 + xcd;
+    v0 = (slot % row_tiles) * GTN_BV;
+    ks0 = chunk * ksteps_per_chunk;
+    ks1 = min(total_ksteps, ks0 + ksteps_per_chunk);
+  } else {
+    v0 = blockIdx.x * GTN_BV;
+    ks1 = total_ksteps;
+  }
+  const int ks1_main = min(ks1, K / GNN_BK);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -906,13 +925,13 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
     __builtin_amdgcn_s_setprio(0);
   };
 
-  if (ks1_main > 0) {
-    load_ab(0);
+  if (ks0 < ks1_main) {
+    load_ab(ks0);
     store_ab(0);
-    if (1 < ks1_main) load_ab(1);
+    if (ks0 + 1 < ks1_main) load_ab(ks0 + 1);
     __syncthreads();
-    for (int ks = 0; ks < ks1_main; ++ks) {
-      const int cur = ks & 1;
+    for (int ks = ks0; ks < ks1_main; ++ks) {
+      const int cur = (ks - ks0) & 1;
       if (ks + 1 < ks1_main) {
         store_ab(cur ^ 1);
         if (ks + 2 < ks1_main) load_ab(ks + 2);
@@ -922,8 +941,8 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
     }
   }
 
-  // ragged last batch rows (K % 32): one zero-filled guarded iteration
-  if (K % GNN_BK) {
+  // ragged last rows (K % 32): one zero-filled guarded iteration
+  if ((K % GNN_BK) && ks1 == total_ksteps && ks0 < ks1) {
     const int k0 = (total_ksteps - 1) * GNN_BK;
 #pragma unroll
     for (int r = 0; r < 2; ++r) {
@@ -956,6 +975,7 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
     compute(0);
   }
 
+  float* Pc = SPLITK ? P + (long)chunk * V * M : nullptr;
 #pragma unroll
   for (int m = 0; m < 8; ++m) {
 #pragma unroll
@@ -965,7 +985,9 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
 #pragma unroll
       for (int n = 0; n < 6; ++n) {
         const int col = wcol + n * 16 + (lane & 15);
-        if (col < M) C[(long)v * M + col] = f2bf(acc[m][n][r]);
+        if (col >= M) continue;
+        if (SPLITK) Pc[(long)v * M + col] = acc[m][n][r];
+        else C[(long)v * M + col] = f2bf(acc[m][n][r]);
       }
     }
   }
@@ -1931,6 +1953,42 @@ torch::Tensor gemm_tn_ce(torch::Tensor logits, torch::Tensor code,
   return C;
 }
 
+// dW GEMM: C(N2,M) fp32 = A(K,N2)^T @ B(K,M) with K huge (batch*contexts)
+// and a tiny output — split-K tn (k_gemm_tn<false,true>), partials folded
+// by k_splitk_reduce. Shaped for d_w = ctx^T @ d_z (K~205K, out 384x384).
+torch::Tensor gemm_tn_splitk(torch::Tensor A, torch::Tensor B) {
+  CHECK_DEV(A); CHECK_CONT(A); CHECK_DEV(B); CHECK_CONT(B);
+  TORCH_CHECK(A.scalar_type() == torch::kBFloat16 &&
+              B.scalar_type() == torch::kBFloat16);
+  const int K = A.size(0), N2 = A.size(1), M = B.size(1);
+  TORCH_CHECK(B.size(0) == K, "K mismatch");
+  TORCH_CHECK(M <= GNN_BN && M % 8 == 0 && N2 % 8 == 0);
+  const int row_tiles = (N2 + GTN_BV - 1) / GTN_BV;
+  int S = 256 / (row_tiles * 8) * 8;
+  if (S < 8) S = 8;
+  const int total_ksteps = (K + GNN_BK - 1) / GNN_BK;
+  const int kpc = (total_ksteps + S - 1) / S;
+  auto P = torch::empty({(long)S, (long)N2, (long)M},
+                        A.options().dtype(torch::kFloat32));
+  auto C = torch::empty({N2, M}, A.options().dtype(torch::kFloat32));
+  const size_t lds = 2UL * GNN_BK * (GTN_PKT + GNN_PKB) * 2;
+  static bool cfg3 = false;
+  if (!cfg3) {
+    (void)hipFuncSetAttribute((const void*)k_gemm_tn<false, true>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              (int)lds);
+    cfg3 = true;
+  }
+  k_gemm_tn<false, true><<<S * row_tiles, 256, lds, cur_stream()>>>(
+      bf_ptr(A), bf_ptr(B), nullptr, N2, M, K, nullptr, nullptr, 1.f,
+      P.data_ptr<float>(), S, kpc, row_tiles);
+  const long total = (long)N2 * M;
+  TORCH_CHECK(total % 4 == 0);
+  k_splitk_reduce<<<grid_1d(total / 4, 256), 256, 0, cur_stream()>>>(
+      P.data_ptr<float>(), C.data_ptr<float>(), S, total);
+  return C;
+}
+
 // split-K nn GEMM: C(N,M) fp32 = A(N,K) @ B(K,M), row-major bf16 operands.
 // Shaped for d_code = d_logits @ target_shadow (N=batch, M=code dim <= 384,
 // K=vocab). Partials workspace (S, N, M) fp32 comes from the caching
@@ -2342,6 +2400,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gemm_nn_splitk", &gemm_nn_splitk);
   mod.def("gemm_tn_bf16", &gemm_tn_bf16);
   mod.def("gemm_tn_ce", &gemm_tn_ce);
+  mod.def("gemm_tn_splitk", &gemm_tn_splitk);
   mod.def("gemm_nn_splitk_ce", &gemm_nn_splitk_ce);
   mod.def("tanh_bwd_mul", &tanh_bwd_mul);
   mod.def("attention_fwd", &attention_fwd);

@@ -56,12 +56,27 @@ def transform_tanh_bwd(ctx, w_io_bf16, y, d_y) -> Tuple[torch.Tensor, torch.Tens
         e = hip_ext(True)
         d_z = e.tanh_bwd_mul(d_y, y)          # d_z = d_y * (1 - y^2)
         d_ctx = e.gemm_bt_bf16(d_z, w_io_bf16)
-        d_w = (ctx.t().to(d_z.dtype) @ d_z).float()
-        return d_ctx, d_w
+        return d_ctx, grad_weight_gemm(ctx, d_z)
     d_zr = d_y * (1.0 - y.float() ** 2).to(d_y.dtype)
     d_ctx = d_zr @ w_io_bf16.to(d_zr.dtype).t()
     d_w = (ctx.t().float() @ d_zr.float())
     return d_ctx, d_w
+
+
+def grad_weight_gemm(ctx, d_z):
+    """d_w = ctx^T @ d_z (fp32 out). On GPU with the training-sized K
+    (batch*contexts ~ 205K rows) this runs the split-K tn MFMA kernel
+    (k_gemm_tn<false,true>: fp32 partials per XCD-grouped k-chunk +
+    reduce); elsewhere hipBLASLt / plain matmul."""
+    import os as _os
+    if (backend_for(ctx) == 'hip' and ctx.dtype == torch.bfloat16
+            and d_z.dtype == torch.bfloat16
+            and ctx.shape[1] <= 384 and ctx.shape[1] % 8 == 0
+            and d_z.shape[1] <= 384 and d_z.shape[1] % 8 == 0
+            and ctx.shape[0] >= 65536
+            and _os.environ.get('C2V_DW_GEMM', '1') == '1'):
+        return hip_ext(True).gemm_tn_splitk(ctx, d_z)
+    return (ctx.t().to(d_z.dtype) @ d_z).float()
 
 
 def attention_fwd(comb, a, valid_mask) -> Tuple[torch.Tensor, torch.Tensor]:
@@ -100,8 +115,7 @@ def linear_bwd_dropout(d_z, w_io, ctx, keep_prob: float, seed: int,
                                       else torch.empty(0))
         else:
             d_ctx = e.gemm_bt_bf16(d_z, w_io)
-        d_w = (ctx.t().to(d_z.dtype) @ d_z).float()
-        return d_ctx, d_w
+        return d_ctx, grad_weight_gemm(ctx, d_z)
     d_ctx = d_z @ w_io.to(d_z.dtype).t()
     if drop:
         if seed_t is not None:

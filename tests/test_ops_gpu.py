@@ -497,3 +497,20 @@ def test_ce_fused_consumers_vs_fp32_oracle():
         / ref_code.abs().max().item() < 0.02
     assert (fused_tgt - ref_tgt).abs().max().item() \
         / ref_tgt.abs().max().item() < 0.02
+
+
+@pytest.mark.parametrize('K,N2,M', [
+    (204800, 384, 384),    # dW shape (K = B*C)
+    (65560, 384, 384),     # K % 32 != 0 tail
+    (131072, 128, 192),    # narrow output
+])
+def test_gemm_tn_splitk_vs_matmul(K, N2, M):
+    """split-K tn GEMM (dW): random asymmetric refcheck (guide G9)."""
+    A = randn(K, N2, dtype=torch.bfloat16, scale=0.1, seed=K % 977)
+    B = randn(K, M, dtype=torch.bfloat16, scale=0.1)
+    C = ext().gemm_tn_splitk(A, B)
+    assert C.dtype == torch.float32 and C.shape == (N2, M)
+    ref = A.float().t() @ B.float()
+    err = (C - ref).abs().max().item()
+    dn = ref.abs().max().item()
+    assert err / dn < 0.01, 'max err %g vs scale %g' % (err, dn)
