@@ -245,11 +245,14 @@ class Code2VecNetwork:
             # d_target kernel it requires is slower than hipBLASLt there
             # (profiles/r01_optimization_log.md).
             ce_scale = 1.0 / B
-            ce_fused = F.ce_bwd_fused_available(logits)
-            if ce_fused:
+            ce_mode = F.ce_bwd_mode(logits)
+            if ce_mode == 1:
                 d_logits = None
                 d_code = F.logits_bwd_code_ce(logits, self.target_shadow,
                                               lse, labels, ce_scale)
+            elif ce_mode == 2:
+                d_code, d_logits = F.logits_bwd_code_ce_write(
+                    logits, self.target_shadow, lse, labels, ce_scale)
             else:
                 d_logits = F.ce_bwd(logits, lse, labels, ce_scale)
                 d_code = F.logits_bwd_code(d_logits, self.target_shadow)
@@ -275,7 +278,7 @@ class Code2VecNetwork:
                 ev.record()
                 with torch.cuda.stream(self._side_stream):
                     self._side_stream.wait_event(ev)
-                    if ce_fused:
+                    if ce_mode == 1:
                         d_target = F.logits_bwd_target_ce(
                             logits, code_c, lse, labels, ce_scale)
                     else:
@@ -292,14 +295,14 @@ class Code2VecNetwork:
                     side_done = torch.cuda.Event()
                     side_done.record()
                 if not torch.cuda.is_current_stream_capturing():
-                    side_inputs = ((logits, lse, labels, code_c) if ce_fused
-                                   else (d_logits, code_c))
+                    side_inputs = ((logits, lse, labels, code_c)
+                                   if ce_mode == 1 else (d_logits, code_c))
                     for t in side_inputs:
                         t.record_stream(self._side_stream)
                     if st_t_pre is not None:
                         st_t_pre.record_stream(self._side_stream)
             else:
-                if ce_fused:
+                if ce_mode == 1:
                     d_target = F.logits_bwd_target_ce(logits, code_c, lse,
                                                       labels, ce_scale)
                 else:

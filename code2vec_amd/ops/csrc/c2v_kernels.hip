@@ -539,9 +539,13 @@ __global__ void k_ce_reduce_partials(const float* __restrict__ partials,
 
 // CEB: CE-backward fused into the A staging — A holds LOGITS and each
 // staged element becomes d_logit = bf16((exp(l - lse[row]) - onehot)*scale),
-// bit-identical to k_ce_bwd's output, so the 535 MB d_logits tensor is
-// never materialized (k_ce_bwd read 535 + wrote 535 MB per step).
-template <bool CEB = false>
+// bit-identical to k_ce_bwd's output. CEB=1 keeps d_logits virtual;
+// CEB=2 ALSO streams the transformed tiles back to DL (each logits element
+// is staged by exactly one block, so coverage is bijective) — that retires
+// the separate k_ce_bwd pass (535 MB read + 535 MB write) for the cost of
+// one extra 535 MB write here, and d_target's hipBLASLt GEMM still gets a
+// materialized d_logits.
+template <int CEB = 0>
 __launch_bounds__(256, 1)
 __global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
                                  const ushort* __restrict__ B,
@@ -550,7 +554,8 @@ __global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
                                  int ksteps_per_chunk, int row_tiles,
                                  const float* __restrict__ lse = nullptr,
                                  const long* __restrict__ labels = nullptr,
-                                 float ce_scale = 1.f) {
+                                 float ce_scale = 1.f,
+                                 ushort* __restrict__ DL = nullptr) {
   // double-buffered: A[2][128][36] + B[2][32][440] bf16 = 73 KiB dynamic.
   // 256 threads = 4 waves = 1 wave/SIMD: the per-SIMD register pool is 512
   // regs/lane, the only occupancy at which the 192-reg accumulator tile
@@ -611,7 +616,7 @@ __global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
 
   u16x8 ra[2], rb[6];
   auto ce_map = [&](u16x8 v, int grow, int kbase) -> u16x8 {
-    if (!CEB) return v;
+    if (CEB == 0) return v;
     const float l = lse[grow];
     const long lab = labels[grow];
 #pragma unroll
@@ -630,6 +635,8 @@ __global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
       ra[r] = ce_map(*reinterpret_cast<const u16x8*>(
                          A + (long)grow * K + k0 + a_k8),
                      grow, k0 + a_k8);
+      if (CEB == 2 && row0 + a_row[r] < N)
+        *reinterpret_cast<u16x8*>(DL + (long)grow * K + k0 + a_k8) = ra[r];
     }
 #pragma unroll
     for (int r = 0; r < 6; ++r) {
@@ -723,11 +730,16 @@ __global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
       for (int j = 0; j < 8; ++j)
         v[j] = (k0 + a_k8 + j < K) ? A[(long)grow * K + k0 + a_k8 + j]
                                    : (ushort)0;
-      if (CEB) {
+      if (CEB != 0) {
         const u16x8 m = ce_map(v, grow, k0 + a_k8);
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           v[j] = (k0 + a_k8 + j < K) ? m[j] : (ushort)0;
+        if (CEB == 2 && row0 + a_row[r] < N) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            if (k0 + a_k8 + j < K) DL[(long)grow * K + k0 + a_k8 + j] = v[j];
+        }
       }
 #pragma unroll
       for (int j = 0; j < 8; ++j)
@@ -2012,15 +2024,18 @@ torch::Tensor gemm_nn_splitk(torch::Tensor A, torch::Tensor B) {
   const size_t lds = 2UL * (GNN_BM * GNN_PKA + GNN_BK * GNN_PKB) * 2;  // 66 KiB
   static bool nn_configured = false;
   if (!nn_configured) {
-    (void)hipFuncSetAttribute((const void*)k_gemm_nn_splitk<false>,
+    (void)hipFuncSetAttribute((const void*)k_gemm_nn_splitk<0>,
                               hipFuncAttributeMaxDynamicSharedMemorySize,
                               (int)lds);
-    (void)hipFuncSetAttribute((const void*)k_gemm_nn_splitk<true>,
+    (void)hipFuncSetAttribute((const void*)k_gemm_nn_splitk<1>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              (int)lds);
+    (void)hipFuncSetAttribute((const void*)k_gemm_nn_splitk<2>,
                               hipFuncAttributeMaxDynamicSharedMemorySize,
                               (int)lds);
     nn_configured = true;
   }
-  k_gemm_nn_splitk<false><<<S * row_tiles, 256, lds, cur_stream()>>>(
+  k_gemm_nn_splitk<0><<<S * row_tiles, 256, lds, cur_stream()>>>(
       bf_ptr(A), bf_ptr(B), P.data_ptr<float>(), N, M, K, S, kpc, row_tiles);
   const long total = (long)N * M;
   TORCH_CHECK(total % 4 == 0);
@@ -2050,12 +2065,12 @@ torch::Tensor gemm_nn_splitk_ce(torch::Tensor logits, torch::Tensor shadow,
   const size_t lds = 2UL * (GNN_BM * GNN_PKA + GNN_BK * GNN_PKB) * 2;
   static bool cfg2 = false;
   if (!cfg2) {
-    (void)hipFuncSetAttribute((const void*)k_gemm_nn_splitk<true>,
+    (void)hipFuncSetAttribute((const void*)k_gemm_nn_splitk<1>,
                               hipFuncAttributeMaxDynamicSharedMemorySize,
                               (int)lds);
     cfg2 = true;
   }
-  k_gemm_nn_splitk<true><<<S * row_tiles, 256, lds, cur_stream()>>>(
+  k_gemm_nn_splitk<1><<<S * row_tiles, 256, lds, cur_stream()>>>(
       bf_ptr(logits), bf_ptr(shadow), P.data_ptr<float>(), N, M, K, S, kpc,
       row_tiles, lse.data_ptr<float>(), labels_c.data_ptr<long>(),
       (float)scale);
@@ -2064,6 +2079,47 @@ torch::Tensor gemm_nn_splitk_ce(torch::Tensor logits, torch::Tensor shadow,
   k_splitk_reduce<<<grid_1d(total / 4, 256), 256, 0, cur_stream()>>>(
       P.data_ptr<float>(), C.data_ptr<float>(), S, total);
   return C;
+}
+
+// CE-backward fused d_code GEMM that ALSO materializes d_logits (CEB=2):
+// returns (d_code fp32 (N,M), d_logits bf16 (N,K)) — replaces the separate
+// k_ce_bwd pass; d_target consumes the returned d_logits.
+std::vector<torch::Tensor> gemm_nn_splitk_ce_write(torch::Tensor logits,
+                                                   torch::Tensor shadow,
+                                                   torch::Tensor lse,
+                                                   torch::Tensor labels,
+                                                   double scale) {
+  CHECK_DEV(logits); CHECK_CONT(logits);
+  CHECK_DEV(shadow); CHECK_CONT(shadow);
+  const int N = logits.size(0), K = logits.size(1), M = shadow.size(1);
+  TORCH_CHECK(shadow.size(0) == K && M <= GNN_BN && M % 8 == 0);
+  auto labels_c = labels.contiguous();
+  const int row_tiles = (N + GNN_BM - 1) / GNN_BM;
+  int S = 256 / (row_tiles * 8) * 8;
+  if (S < 8) S = 8;
+  const int total_ksteps = (K + GNN_BK - 1) / GNN_BK;
+  const int kpc = (total_ksteps + S - 1) / S;
+  auto P = torch::empty({(long)S, (long)N, (long)M},
+                        logits.options().dtype(torch::kFloat32));
+  auto C = torch::empty({N, M}, logits.options().dtype(torch::kFloat32));
+  auto DL = torch::empty_like(logits);
+  const size_t lds = 2UL * (GNN_BM * GNN_PKA + GNN_BK * GNN_PKB) * 2;
+  static bool cfg4 = false;
+  if (!cfg4) {
+    (void)hipFuncSetAttribute((const void*)k_gemm_nn_splitk<2>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              (int)lds);
+    cfg4 = true;
+  }
+  k_gemm_nn_splitk<2><<<S * row_tiles, 256, lds, cur_stream()>>>(
+      bf_ptr(logits), bf_ptr(shadow), P.data_ptr<float>(), N, M, K, S, kpc,
+      row_tiles, lse.data_ptr<float>(), labels_c.data_ptr<long>(),
+      (float)scale, bf_ptr_mut(DL));
+  const long total = (long)N * M;
+  TORCH_CHECK(total % 4 == 0);
+  k_splitk_reduce<<<grid_1d(total / 4, 256), 256, 0, cur_stream()>>>(
+      P.data_ptr<float>(), C.data_ptr<float>(), S, total);
+  return {C, DL};
 }
 
 torch::Tensor tanh_bwd_mul(torch::Tensor dy, torch::Tensor y) {
@@ -2402,6 +2458,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gemm_tn_ce", &gemm_tn_ce);
   mod.def("gemm_tn_splitk", &gemm_tn_splitk);
   mod.def("gemm_nn_splitk_ce", &gemm_nn_splitk_ce);
+  mod.def("gemm_nn_splitk_ce_write", &gemm_nn_splitk_ce_write);
   mod.def("tanh_bwd_mul", &tanh_bwd_mul);
   mod.def("attention_fwd", &attention_fwd);
   mod.def("attention_bwd", &attention_bwd);

@@ -180,17 +180,24 @@ def logits_bwd_target(d_logits, code_c):
     return d_logits.t() @ code_c
 
 
-def ce_bwd_fused_available(logits):
-    """True when the fully-fused CE backward path (d_logits never
-    materialized: gemm_nn_splitk_ce for d_code + gemm_tn_ce for d_target)
-    should run. OPT-IN (C2V_FUSED_CEBWD=1): at DP=1 it measured slightly
-    slower than the materialized path because it forces d_target onto the
-    custom tn kernel (527 us vs hipBLASLt 333) and adds exp() latency to
-    the nn GEMM's staging chain — see profiles/r01_optimization_log.md."""
+def ce_bwd_mode(logits):
+    """CE-backward fusion mode (C2V_FUSED_CEBWD):
+    '2' (default): gemm_nn_splitk_ce_write — CE backward fused into the
+        d_code GEMM's staging, which also streams d_logits out (bijective
+        tile coverage); retires the separate k_ce_bwd pass (~260 us) for
+        one extra 535 MB write (~85 us), and d_target keeps hipBLASLt.
+    '1': fully virtual d_logits (forces the slower tn d_target kernel —
+        measured net-negative at DP=1, see profiles/).
+    '0': materialized k_ce_bwd + plain GEMMs."""
     import os as _os
-    return (backend_for(logits) == 'hip' and logits.dtype == torch.bfloat16
-            and logits.shape[1] >= 4096
-            and _os.environ.get('C2V_FUSED_CEBWD', '0') == '1')
+    if not (backend_for(logits) == 'hip' and logits.dtype == torch.bfloat16
+            and logits.shape[1] >= 4096):
+        return 0
+    return int(_os.environ.get('C2V_FUSED_CEBWD', '2'))
+
+
+def ce_bwd_fused_available(logits):
+    return ce_bwd_mode(logits) == 1
 
 
 def logits_bwd_code_ce(logits, shadow, lse, labels, scale):
@@ -203,6 +210,18 @@ def logits_bwd_code_ce(logits, shadow, lse, labels, scale):
                                                float(scale))
     d_logits = ce_bwd(logits, lse, labels, scale)
     return logits_bwd_code(d_logits, shadow)
+
+
+def logits_bwd_code_ce_write(logits, shadow, lse, labels, scale):
+    """(d_code fp32, d_logits bf16) with CE backward fused into the GEMM
+    staging and d_logits streamed out as a byproduct (mode '2')."""
+    if ce_bwd_mode(logits) != 0 and shadow.shape[1] <= 384 \
+            and shadow.shape[1] % 8 == 0:
+        out = hip_ext(True).gemm_nn_splitk_ce_write(logits, shadow, lse,
+                                                    labels, float(scale))
+        return out[0], out[1]
+    d_logits = ce_bwd(logits, lse, labels, scale)
+    return logits_bwd_code(d_logits, shadow), d_logits
 
 
 def logits_bwd_target_ce(logits, code_c, lse, labels, scale):

@@ -514,3 +514,24 @@ def test_gemm_tn_splitk_vs_matmul(K, N2, M):
     err = (C - ref).abs().max().item()
     dn = ref.abs().max().item()
     assert err / dn < 0.01, 'max err %g vs scale %g' % (err, dn)
+
+
+def test_ce_fused_write_matches_ce_bwd():
+    """gemm_nn_splitk_ce_write: the streamed-out d_logits must be
+    bit-identical to k_ce_bwd's output, and d_code must match the
+    materialized composition."""
+    e = ext()
+    torch.manual_seed(13)
+    B, V, D = 512, 70030, 384
+    code = randn(B, D, dtype=torch.bfloat16, scale=0.3)
+    shadow = randn(V, D, dtype=torch.bfloat16, scale=0.3)
+    labels = torch.randint(0, V, (B,), device='cuda')
+    logits, _, lse = e.logits_ce_fused(code, shadow, labels)
+    scale = 1.0 / B
+    d_logits_ref = e.ce_bwd(logits, lse, labels, scale)
+    d_code, d_logits = e.gemm_nn_splitk_ce_write(logits, shadow, lse,
+                                                 labels, scale)
+    assert torch.equal(d_logits, d_logits_ref), 'd_logits not bit-identical'
+    ref_code = e.gemm_nn_splitk(d_logits_ref, shadow)
+    err = (d_code - ref_code).abs().max().item() / ref_code.abs().max().item()
+    assert err < 1e-3, err
