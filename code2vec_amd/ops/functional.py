@@ -181,19 +181,21 @@ def logits_bwd_target(d_logits, code_c):
 
 
 def ce_bwd_mode(logits):
-    """CE-backward fusion mode (C2V_FUSED_CEBWD):
-    '2' (default): gemm_nn_splitk_ce_write — CE backward fused into the
-        d_code GEMM's staging, which also streams d_logits out (bijective
-        tile coverage); retires the separate k_ce_bwd pass (~260 us) for
-        one extra 535 MB write (~85 us), and d_target keeps hipBLASLt.
-    '1': fully virtual d_logits (forces the slower tn d_target kernel —
-        measured net-negative at DP=1, see profiles/).
-    '0': materialized k_ce_bwd + plain GEMMs."""
+    """CE-backward fusion mode (C2V_FUSED_CEBWD). Default '0'
+    (materialized k_ce_bwd + plain GEMMs) — BOTH fusion modes measured
+    slower in-bench despite moving less HBM traffic, because the nn d_code
+    kernel is latency-bound at 1 wave/SIMD and the fused exp()/store chain
+    sits on its staging critical path (mode 2: 282K vs 302K ex/s; mode 1:
+    252K — see profiles/r01_optimization_log.md):
+    '2': gemm_nn_splitk_ce_write — CE backward fused into the d_code
+        GEMM's staging, which also streams d_logits out (bijective tile
+        coverage, bit-identical to k_ce_bwd); d_target keeps hipBLASLt.
+    '1': fully virtual d_logits (also forces the tn d_target kernel)."""
     import os as _os
     if not (backend_for(logits) == 'hip' and logits.dtype == torch.bfloat16
             and logits.shape[1] >= 4096):
         return 0
-    return int(_os.environ.get('C2V_FUSED_CEBWD', '2'))
+    return int(_os.environ.get('C2V_FUSED_CEBWD', '0'))
 
 
 def ce_bwd_fused_available(logits):
