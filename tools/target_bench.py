@@ -59,3 +59,9 @@ t('ce_bwd', lambda: ext.ce_bwd(
     ext.gemm_bt_v(code, shadow, False, 2), lse, labels, 1.0 / B), iters=5)
 
 t('dW our split-K tn', lambda: ext.gemm_tn_splitk(ctx, dz))
+
+w = (torch.randn(D, D, device='cuda') * 0.1).to(torch.bfloat16)
+t('transform GEMM 128-tile (current)', lambda: ext.gemm_bt_v(dz, w, False, 1))
+t('transform GEMM 256-tile', lambda: ext.gemm_bt_v(dz, w, False, 2))
+t('transform GEMM 128-tile tanh', lambda: ext.gemm_bt_v(dz, w, True, 1))
+t('transform GEMM 256-tile tanh', lambda: ext.gemm_bt_v(dz, w, True, 2))
