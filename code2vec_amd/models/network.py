@@ -52,6 +52,9 @@ class NullReducer:
     def allreduce_mean_scalar(self, value: float) -> float:
         return value
 
+    def all_continue(self, have_next: bool) -> bool:
+        return have_next
+
 
 class Code2VecNetwork:
     """Parameter store + step engine. Not an nn.Module: parameters are plain

@@ -117,7 +117,16 @@ class Code2VecModel(Code2VecModelBase):
         from ..data.prefetcher import BatchPrefetcher
         prefetcher = BatchPrefetcher(reader.iter_batches(), device,
                                      depth=cfg.READER_QUEUE_DEPTH)
-        for b in prefetcher:
+        batches = iter(prefetcher)
+        while True:
+            b = next(batches, None)
+            # DP termination consensus: the shard split is not batch-aligned,
+            # so ranks can finish with unequal batch counts — a rank stepping
+            # once more than its peers would hang in the gradient
+            # collectives. All ranks must agree to continue (ddp.Reducer
+            # .all_continue); longer ranks drop their surplus batch.
+            if not self.reducer.all_continue(b is not None):
+                break
             loss = self.network.train_step(
                 b.source_token_indices, b.path_indices, b.target_token_indices,
                 b.context_valid_mask, b.target_index, reducer=self.reducer)

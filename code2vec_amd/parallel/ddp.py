@@ -110,6 +110,21 @@ class Reducer:
         all_rows = torch.cat([t[:c] for t, c in zip(rows_out, counts)])
         return all_ids, all_rows / ws
 
+    def all_continue(self, have_next: bool) -> bool:
+        """Termination consensus for reader-driven DP training: ranks can
+        end an epoch with unequal batch counts (the DP shard split is not
+        batch-aligned), and a rank stepping once more than its peers would
+        deadlock in the gradient collectives. Every step each rank votes
+        whether it has a next batch; training continues only if ALL do
+        (surplus batches on longer ranks are dropped)."""
+        if self.world_size <= 1:
+            return have_next
+        t = torch.tensor([1 if have_next else 0], dtype=torch.int32)
+        if dist.get_backend(self.group) == 'nccl':
+            t = t.cuda()
+        dist.all_reduce(t, op=dist.ReduceOp.MIN, group=self.group)
+        return bool(int(t.item()))
+
     def allreduce_mean_scalar(self, value: float) -> float:
         if self.world_size <= 1:
             return value
