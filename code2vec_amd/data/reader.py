@@ -284,11 +284,12 @@ class PathContextReader:
 
     def _iter_batches_native_stream(self, data_path: str,
                                     batch_size: int) -> Iterator[ReaderBatch]:
-        """Training fast path: read the file in binary chunks, split/parse
-        lines inside the C++ extension (parse_buffer), filter + DP-shard with
-        tensor ops, and shuffle at tensor level — a windowed shuffle over at
-        least SHUFFLE_BUFFER_SIZE rows, equivalent in spirit to the
-        reference's 10000-line shuffle buffer."""
+        """Training fast path: read the file in binary chunks (DP rank takes
+        every world_size-th chunk), split/parse lines inside the C++
+        extension (parse_buffer), filter with tensor ops, and shuffle at
+        tensor level — a windowed shuffle over at least SHUFFLE_BUFFER_SIZE
+        rows, equivalent in spirit to the reference's 10000-line shuffle
+        buffer."""
         pin = torch.cuda.is_available()
         sb = self.config.SHUFFLE_BUFFER_SIZE if self.estimator_action.is_train else 0
         g = torch.Generator().manual_seed(1234 + self.rank)
@@ -297,7 +298,6 @@ class PathContextReader:
 
         pool = []           # list of 5-tuples of tensors
         pool_rows = 0
-        line_base = 0       # global line counter for DP sharding
 
         def make_batch(tensors):
             if pin:
@@ -345,6 +345,19 @@ class PathContextReader:
         import queue
         import threading
 
+        # DP sharding mode: big corpora shard by CHUNK (each rank parses
+        # only every world_size-th 4 MB chunk -> 1/N of the parse work per
+        # rank); small files (< 8 chunks per rank) fall back to line-modulo
+        # so every rank still sees data. Chunk shards are not batch-aligned;
+        # the train loop's per-step termination consensus
+        # (ddp.Reducer.all_continue) absorbs the ragged tail.
+        chunk_bytes = int(os.environ.get('C2V_READER_CHUNK_BYTES', 4 << 20))
+        chunk_shard = (self.world_size > 1 and
+                       os.path.getsize(data_path) >=
+                       self.world_size * 8 * chunk_bytes)
+        line_shard = self.world_size > 1 and not chunk_shard
+        line_base = 0
+
         def parse_filter(use: bytes):
             nonlocal line_base
             src, pth, tgt, mask, tidx = self._native.parse_buffer(use)
@@ -354,7 +367,7 @@ class PathContextReader:
             keep = mask.any(dim=1)
             if self.estimator_action.is_train:
                 keep &= tidx > self._tgt_oov
-            if self.world_size > 1:
+            if line_shard:
                 gidx = torch.arange(line_base, line_base + n)
                 keep &= (gidx % self.world_size) == self.rank
             line_base += n
@@ -373,10 +386,11 @@ class PathContextReader:
                 epoch = 0
                 while epochs < 0 or epoch < epochs:
                     epoch += 1
+                    chunk_i = 0
                     with open(data_path, 'rb') as f:
                         carry = b''
                         while not stop.is_set():
-                            chunk = f.read(4 << 20)
+                            chunk = f.read(chunk_bytes)
                             if not chunk:
                                 break
                             buf = carry + chunk
@@ -385,6 +399,11 @@ class PathContextReader:
                                 carry = buf
                                 continue
                             carry = buf[last_nl + 1:]
+                            mine = (not chunk_shard or
+                                    chunk_i % self.world_size == self.rank)
+                            chunk_i += 1
+                            if not mine:
+                                continue
                             tup = parse_filter(buf[:last_nl + 1])
                             if tup is not None:
                                 while not stop.is_set():
@@ -393,7 +412,9 @@ class PathContextReader:
                                         break
                                     except queue.Full:
                                         pass
-                        if carry.strip() and not stop.is_set():
+                        if carry.strip() and not stop.is_set() and (
+                                not chunk_shard
+                                or chunk_i % self.world_size == self.rank):
                             tup = parse_filter(carry + b'\n')
                             if tup is not None:
                                 while not stop.is_set():
