@@ -1480,7 +1480,8 @@ __global__ void k_adam_dense(float* __restrict__ p, const G* __restrict__ g,
 //   3. lookup: every input id re-probes and reads its compact index
 template <typename I>
 __global__ void k_hash_claim(const I* __restrict__ ids, long n,
-                             int* __restrict__ tbl_id, u32 mask_) {
+                             int* __restrict__ tbl_id, u32 mask_,
+                             int* __restrict__ cnt = nullptr) {
   const int lane = threadIdx.x & 63;
   for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (long)gridDim.x * blockDim.x) {
@@ -1491,7 +1492,17 @@ __global__ void k_hash_claim(const I* __restrict__ ids, long n,
     // (measured 2.3 ms/step). Lane L-1 is active whenever L is (i grows
     // with lane), so the shuffle is safe at the tail.
     const int prev = __shfl_up(id, 1, 64);
-    if (lane > 0 && prev == id) continue;  // leader of each run claims
+    const bool cont = (lane > 0 && prev == id);
+    // occurrence counts (for the accum single-store fast path): each run
+    // leader adds its run length at the id's slot
+    const unsigned long long cmask = __ballot(cont);
+    if (cont) continue;                    // leader of each run claims
+    int run_len = 1;
+    if (cnt) {
+      const unsigned long long rest =
+          (lane < 63) ? ~(cmask >> (lane + 1)) : ~0ull;
+      run_len = 1 + (rest ? __builtin_ctzll(rest) : 64);
+    }
     u32 slot = ((u32)id * 2654435761u) & mask_;
     for (;;) {
       const int cur = tbl_id[slot];
@@ -1502,6 +1513,7 @@ __global__ void k_hash_claim(const I* __restrict__ ids, long n,
       }
       slot = (slot + 1) & mask_;
     }
+    if (cnt) atomicAdd(cnt + slot, run_len);
   }
 }
 
@@ -1544,13 +1556,16 @@ template <typename I>
 __global__ void k_hash_lookup(const I* __restrict__ ids, long n,
                               const int* __restrict__ tbl_id,
                               const int* __restrict__ tbl_cidx,
-                              int* __restrict__ inverse_out, u32 mask_) {
+                              int* __restrict__ inverse_out, u32 mask_,
+                              const int* __restrict__ cnt = nullptr) {
   for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (long)gridDim.x * blockDim.x) {
     const int id = (int)ids[i];
     u32 slot = ((u32)id * 2654435761u) & mask_;
     while (tbl_id[slot] != id) slot = (slot + 1) & mask_;
-    inverse_out[i] = tbl_cidx[slot];
+    // bit 30 marks single-occurrence ids: their accum write needs no atomic
+    inverse_out[i] = tbl_cidx[slot] |
+                     ((cnt && cnt[slot] == 1) ? (1 << 30) : 0);
   }
 }
 
@@ -1610,7 +1625,14 @@ __global__ void k_rows_accum_ctx(const ushort* __restrict__ d_ctx, int ld,
     const long src_row = (r < n_per_seg) ? r : r - n_per_seg;
     const int src_off = (r < n_per_seg) ? off0 : off1;
     const float gv = bf2f(d_ctx[src_row * ld + src_off + col]);
-    if (gv != 0.f) atomicAdd(acc + (long)inverse[r] * d + col, gv);
+    const int inv = inverse[r];
+    if (inv & (1 << 30)) {
+      // single-occurrence id (~86% of unique rows on uniform ids): plain
+      // store — global atomicAdd throughput was the kernel's bound
+      acc[(long)(inv & 0x3FFFFFFF) * d + col] = gv;
+    } else if (gv != 0.f) {
+      atomicAdd(acc + (long)inv * d + col, gv);
+    }
   }
 }
 
@@ -1625,8 +1647,12 @@ __global__ void k_rows_accum(const G* __restrict__ rows,
     float gv;
     if constexpr (sizeof(G) == 2) gv = bf2f(((const ushort*)rows)[s]);
     else gv = ((const float*)rows)[s];
-    if (gv != 0.f)
-      atomicAdd(acc + (long)inverse[r] * d + col, gv);
+    const int inv = inverse[r];
+    if (inv & (1 << 30)) {
+      acc[(long)(inv & 0x3FFFFFFF) * d + col] = gv;
+    } else if (gv != 0.f) {
+      atomicAdd(acc + (long)inv * d + col, gv);
+    }
   }
 }
 
@@ -2309,6 +2335,7 @@ void adam_sparse_rows_hash(torch::Tensor p, torch::Tensor ids,
   auto opts_i32 = p.options().dtype(torch::kInt32);
   auto tbl_id = torch::full({(long)cap}, -1, opts_i32);
   auto tbl_cidx = torch::full({(long)cap}, -1, opts_i32);
+  auto tbl_cnt = torch::zeros({(long)cap}, opts_i32);
   auto uniq = torch::empty({n}, p.options().dtype(torch::kInt64));
   auto n_uniq = torch::zeros({1}, opts_i32);
   auto inverse = torch::empty({n}, opts_i32);
@@ -2319,22 +2346,26 @@ void adam_sparse_rows_hash(torch::Tensor p, torch::Tensor ids,
   const u32 mask_ = cap - 1;
   if (ids_c.scalar_type() == torch::kInt32) {
     k_hash_claim<int><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
-        ids_c.data_ptr<int>(), n, tbl_id.data_ptr<int>(), mask_);
+        ids_c.data_ptr<int>(), n, tbl_id.data_ptr<int>(), mask_,
+        tbl_cnt.data_ptr<int>());
     k_hash_compact<<<grid_1d(cap, 256), 256, 0, cur_stream()>>>(
         tbl_id.data_ptr<int>(), tbl_cidx.data_ptr<int>(),
         uniq.data_ptr<long>(), n_uniq.data_ptr<int>(), cap);
     k_hash_lookup<int><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
         ids_c.data_ptr<int>(), n, tbl_id.data_ptr<int>(),
-        tbl_cidx.data_ptr<int>(), inverse.data_ptr<int>(), mask_);
+        tbl_cidx.data_ptr<int>(), inverse.data_ptr<int>(), mask_,
+        tbl_cnt.data_ptr<int>());
   } else {
     k_hash_claim<long><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
-        ids_c.data_ptr<long>(), n, tbl_id.data_ptr<int>(), mask_);
+        ids_c.data_ptr<long>(), n, tbl_id.data_ptr<int>(), mask_,
+        tbl_cnt.data_ptr<int>());
     k_hash_compact<<<grid_1d(cap, 256), 256, 0, cur_stream()>>>(
         tbl_id.data_ptr<int>(), tbl_cidx.data_ptr<int>(),
         uniq.data_ptr<long>(), n_uniq.data_ptr<int>(), cap);
     k_hash_lookup<long><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
         ids_c.data_ptr<long>(), n, tbl_id.data_ptr<int>(),
-        tbl_cidx.data_ptr<int>(), inverse.data_ptr<int>(), mask_);
+        tbl_cidx.data_ptr<int>(), inverse.data_ptr<int>(), mask_,
+        tbl_cnt.data_ptr<int>());
   }
 
   k_zero_rows_dyn<<<grid_1d(n * d / 4, 256), 256, 0, cur_stream()>>>(
@@ -2380,6 +2411,7 @@ void adam_sparse_rows_hash_ctx(torch::Tensor p, torch::Tensor ids,
   auto opts_i32 = p.options().dtype(torch::kInt32);
   auto tbl_id = torch::full({(long)cap}, -1, opts_i32);
   auto tbl_cidx = torch::full({(long)cap}, -1, opts_i32);
+  auto tbl_cnt = torch::zeros({(long)cap}, opts_i32);
   auto uniq = torch::empty({n}, p.options().dtype(torch::kInt64));
   auto n_uniq = torch::zeros({1}, opts_i32);
   auto inverse = torch::empty({n}, opts_i32);
@@ -2387,13 +2419,15 @@ void adam_sparse_rows_hash_ctx(torch::Tensor p, torch::Tensor ids,
   const u32 mask_ = cap - 1;
   TORCH_CHECK(ids_c.scalar_type() == torch::kInt32, "ctx path expects int32 ids");
   k_hash_claim<int><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
-      ids_c.data_ptr<int>(), n, tbl_id.data_ptr<int>(), mask_);
+      ids_c.data_ptr<int>(), n, tbl_id.data_ptr<int>(), mask_,
+      tbl_cnt.data_ptr<int>());
   k_hash_compact<<<grid_1d(cap, 256), 256, 0, cur_stream()>>>(
       tbl_id.data_ptr<int>(), tbl_cidx.data_ptr<int>(), uniq.data_ptr<long>(),
       n_uniq.data_ptr<int>(), cap);
   k_hash_lookup<int><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
       ids_c.data_ptr<int>(), n, tbl_id.data_ptr<int>(),
-      tbl_cidx.data_ptr<int>(), inverse.data_ptr<int>(), mask_);
+      tbl_cidx.data_ptr<int>(), inverse.data_ptr<int>(), mask_,
+      tbl_cnt.data_ptr<int>());
   k_zero_rows_dyn<<<grid_1d(n * d / 4, 256), 256, 0, cur_stream()>>>(
       acc.data_ptr<float>(), n_uniq.data_ptr<int>(), (int)d);
   k_rows_accum_ctx<<<grid_1d(n * d, 256), 256, 0, cur_stream()>>>(

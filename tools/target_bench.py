@@ -65,3 +65,14 @@ t('transform GEMM 128-tile (current)', lambda: ext.gemm_bt_v(dz, w, False, 1))
 t('transform GEMM 256-tile', lambda: ext.gemm_bt_v(dz, w, False, 2))
 t('transform GEMM 128-tile tanh', lambda: ext.gemm_bt_v(dz, w, True, 1))
 t('transform GEMM 256-tile tanh', lambda: ext.gemm_bt_v(dz, w, True, 2))
+
+# sparse embedding-update chain, java14m-ish shapes (uniform ids)
+ids = torch.randint(1, 1301137, (2 * N,), device='cuda', dtype=torch.int32)
+dctx = (torch.randn(N, D, device='cuda') * 0.1).to(torch.bfloat16)
+tok_p = torch.randn(1301137, 128, device='cuda')
+tok_m = torch.zeros_like(tok_p)
+tok_v = torch.zeros_like(tok_p)
+t('sparse adam ctx-direct (tok table)',
+  lambda: ext.adam_sparse_rows_hash_ctx(
+      tok_p, ids, dctx, 0, 256, 2, 128, tok_m, tok_v, 1, 1e-3, 0.9, 0.999,
+      1e-8, torch.empty(0)))
