@@ -448,9 +448,12 @@ __global__ void k_gemm256_bt(const ushort* __restrict__ A,
           lsum += __shfl_xor(lsum, off, 16);
         if ((lane & 15) == 0 && row < N) {
           // per-wave-column slot: 4 waves cover the 256-col tile, each owns
-          // a quarter (a shared slot would race and drop 192 columns)
+          // a quarter (a shared slot would race and drop 192 columns).
+          // (tile, quarter)-MAJOR layout: this block's 256 rows land 8 B
+          // apart — the old row-major layout scattered every store 32 KB
+          // apart (one cache line each, ~150 us/step of write traffic)
           float* p = partials +
-                     2 * (((long)row * n_col_tiles + tile_m) * 4 + (wid & 3));
+                     2 * (((long)tile_m * 4 + (wid & 3)) * N + row);
           p[0] = lmax;
           p[1] = lsum;
         }
@@ -459,7 +462,31 @@ __global__ void k_gemm256_bt(const ushort* __restrict__ A,
   }
 }
 
-// fold the per-(row, col-tile) (max, sumexp) partials into (loss, lse)
+// stage 1: fold the (T, N, 2) partials down to (N, T2, 2) — each thread
+// owns one row inside a 256-row block and walks a T-chunk; reads are fully
+// coalesced (256 rows x 8 B contiguous per step)
+__launch_bounds__(256)
+__global__ void k_ce_part_reduce1(const float* __restrict__ partials,
+                                  float* __restrict__ out, int N, int T,
+                                  int t_chunks) {
+  const int row = blockIdx.x * 256 + threadIdx.x;
+  const int tc = blockIdx.y;
+  if (row >= N) return;
+  const int per = (T + t_chunks - 1) / t_chunks;
+  const int t0 = tc * per, t1 = min(T, t0 + per);
+  float m = -3.0e38f, ssum = 0.f;
+  for (int t = t0; t < t1; ++t) {
+    const float pm = partials[2 * ((long)t * N + row)];
+    const float ps = partials[2 * ((long)t * N + row) + 1];
+    if (pm > m) { ssum = ssum * __expf(m - pm) + ps; m = pm; }
+    else ssum += ps * __expf(pm - m);
+  }
+  out[2 * ((long)row * t_chunks + tc)] = m;
+  out[2 * ((long)row * t_chunks + tc) + 1] = ssum;
+}
+
+// stage 2 (and legacy single-stage path): fold per-(row, chunk) (max,
+// sumexp) partials in (N, T, 2) layout into (loss, lse)
 __launch_bounds__(256)
 __global__ void k_ce_reduce_partials(const float* __restrict__ partials,
                                      const ushort* __restrict__ logits,
@@ -1893,8 +1920,12 @@ std::vector<torch::Tensor> logits_ce_fused(torch::Tensor code,
   auto C = torch::empty({N, M}, code.options());
   const int n_tiles = (N + G256_BM - 1) / G256_BM;
   const int m_tiles = (M + G256_BN - 1) / G256_BN;
-  auto partials = torch::empty({(long)N, (long)m_tiles, 4, 2},
+  // (tile, quarter)-major partials (see the CE_PART epilogue comment)
+  auto partials = torch::empty({(long)m_tiles * 4, (long)N, 2},
                                code.options().dtype(torch::kFloat32));
+  const int t_chunks = 64;
+  auto partials2 = torch::empty({(long)N, (long)t_chunks, 2},
+                                code.options().dtype(torch::kFloat32));
   auto loss = torch::empty({N}, code.options().dtype(torch::kFloat32));
   auto lse = torch::empty({N}, code.options().dtype(torch::kFloat32));
   const size_t lds = 4 * 16384 * 2;
@@ -1908,9 +1939,13 @@ std::vector<torch::Tensor> logits_ce_fused(torch::Tensor code,
   k_gemm256_bt<false, true><<<n_tiles * m_tiles, 512, lds, cur_stream()>>>(
       bf_ptr(code), bf_ptr(shadow), bf_ptr_mut(C), N, M, K,
       partials.data_ptr<float>(), m_tiles);
+  dim3 rg1((N + 255) / 256, t_chunks);
+  k_ce_part_reduce1<<<rg1, 256, 0, cur_stream()>>>(
+      partials.data_ptr<float>(), partials2.data_ptr<float>(), N,
+      m_tiles * 4, t_chunks);
   k_ce_reduce_partials<<<N, 256, 0, cur_stream()>>>(
-      partials.data_ptr<float>(), bf_ptr(C), labels_c.data_ptr<long>(),
-      loss.data_ptr<float>(), lse.data_ptr<float>(), N, m_tiles * 4, M);
+      partials2.data_ptr<float>(), bf_ptr(C), labels_c.data_ptr<long>(),
+      loss.data_ptr<float>(), lse.data_ptr<float>(), N, t_chunks, M);
   return {C, loss, lse};
 }
 
