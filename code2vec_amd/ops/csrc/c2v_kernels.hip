@@ -288,6 +288,147 @@ __global__ void k_gemm_bt(const ushort* __restrict__ A,
 }
 
 // ---------------------------------------------------------------------------
+// K4 v2: 128x128-tile, BK=64, double-buffered (one barrier per k-step,
+// bt-kernel pipeline). The single-buffered BK=32 k_gemm_bt pays 2 barriers
+// and an exposed staging latency per step — tolerable at java14m's K=384
+// for throughput-bound shapes, but the transform GEMMs (K=384 -> only 6
+// BK=64 steps) are latency-bound, so prefetching the next tile during
+// compute matters. Same epilogue options (TANH / fused dropout mask).
+// ---------------------------------------------------------------------------
+
+template <bool TANH, bool DROPOUT = false>
+__launch_bounds__(256, 1)
+__global__ void k_gemm_bt2(const ushort* __restrict__ A,
+                           const ushort* __restrict__ Bt,
+                           ushort* __restrict__ C, int N, int M, int K,
+                           float keep_prob = 1.f, u64 seed_scalar = 0,
+                           const long* __restrict__ seed_ptr = nullptr) {
+  const u64 seed = DROPOUT ? (seed_ptr ? (u64)*seed_ptr : seed_scalar) : 0;
+  const float inv_keep = DROPOUT ? (1.f / keep_prob) : 1.f;
+  // 2 buffers x (A[128][64] + B[128][64]) bf16 = 64 KiB
+  extern __shared__ ushort lds2[];
+#define L2A(b) (lds2 + (b) * 16384)
+#define L2B(b) (lds2 + (b) * 16384 + 8192)
+
+  const int n_tiles = (N + GEMM_BM - 1) / GEMM_BM;
+  const int m_tiles = (M + GEMM_BN - 1) / GEMM_BN;
+  const int nwg = n_tiles * m_tiles;
+  int wg = blockIdx.x;
+  {
+    const int nxcd = 8;
+    const int q = nwg / nxcd, r = nwg % nxcd;
+    const int xcd = wg % nxcd, orig = wg / nxcd;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig;
+  }
+  const int tile_n = wg / m_tiles;
+  const int tile_m = wg % m_tiles;
+  const int row0 = tile_n * GEMM_BM;
+  const int col0 = tile_m * GEMM_BN;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;           // 4 waves: 2x2 over (64,64) subtiles
+  const int wrow = (wid >> 1) * 64;
+  const int wcol = (wid & 1) * 64;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  // stage one 128x64 tile pair per k-step: 16 KiB per operand, 4 waves x
+  // 4 width-16 issues each (one issue = 64 lanes x 16 B = 8 rows of 128 B)
+  auto stage = [&](int buf, int ks) {
+    const int k0 = ks * 64;
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      const int off = (wid * 4 + q) * 1024 + lane * 16;  // byte in tile
+      const int lrow = off >> 7;           // 128 B per row of 64 bf16
+      const int lcol = (off & 127) >> 1;
+      {
+        const int grow = min(row0 + lrow, N - 1);
+        const ushort* gp = A + (long)grow * K + k0 + lcol;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) u32*)gp,
+            (__attribute__((address_space(3))) u32*)(L2A(buf) +
+                                                     (wid * 4 + q) * 512),
+            16, 0, 0);
+      }
+      {
+        const int grow = min(col0 + lrow, M - 1);
+        const ushort* gp = Bt + (long)grow * K + k0 + lcol;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) u32*)gp,
+            (__attribute__((address_space(3))) u32*)(L2B(buf) +
+                                                     (wid * 4 + q) * 512),
+            16, 0, 0);
+      }
+    }
+  };
+
+  const int k_steps = K / 64;
+  stage(0, 0);
+  __syncthreads();
+  int cur = 0;
+  for (int ks = 0; ks < k_steps; ++ks) {
+    if (ks + 1 < k_steps) stage(cur ^ 1, ks + 1);
+    bf16x8 bfrag[4][2];
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni)
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        const int r = wcol + ni * 16 + (lane & 15);
+        bfrag[ni][kk] = *reinterpret_cast<const bf16x8*>(
+            L2B(cur) + r * 64 + kk * 32 + (lane >> 4) * 8);
+      }
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+      bf16x8 afrag[2];
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        const int r = wrow + mi * 16 + (lane & 15);
+        afrag[kk] = *reinterpret_cast<const bf16x8*>(
+            L2A(cur) + r * 64 + kk * 32 + (lane >> 4) * 8);
+      }
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[0], bfrag[ni][0], acc[mi][ni], 0, 0, 0);
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[1], bfrag[ni][1], acc[mi][ni], 0, 0, 0);
+      }
+    }
+    __builtin_amdgcn_s_setprio(0);
+    __syncthreads();
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      const int col = col0 + wcol + ni * 16 + (lane & 15);
+      if (col >= M) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = row0 + wrow + mi * 16 + (lane >> 4) * 4 + r;
+        if (row >= N) continue;
+        float v = acc[mi][ni][r];
+        if (TANH) v = tanhf(v);
+        if (DROPOUT) {
+          const bool keep =
+              hash_uniform(seed, (u64)row * M + col) < keep_prob;
+          v = keep ? v * inv_keep : 0.f;
+        }
+        C[(long)row * M + col] = f2bf(v);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // K4/K8 v2: 256x256-tile bf16 MFMA GEMM (CDNA4 guide §5 template, 2-phase
 // double-buffered variant). C(N,M) = A(N,K) @ Bt(M,K)^T.
 // 8 waves (512 threads) as 2x4 over (128x64) per-wave outputs; BK=64;
@@ -1862,7 +2003,8 @@ torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor Bt, bool tanh_epilogue,
   const int N = A.size(0), K = A.size(1), M = Bt.size(0);
   TORCH_CHECK(Bt.size(1) == K, "K mismatch");
 
-  // variant: 0 = auto, 1 = 128x128/BK32, 2 = 256x256/BK64 (needs K%64==0)
+  // variant: 0 = auto, 1 = 128x128/BK32 single-buffered,
+  // 2 = 256x256/BK64, 3 = 128x128/BK64 double-buffered (K%64==0)
   bool use256 = false;
   if (variant == 2) use256 = true;
   else if (variant == 0)
@@ -1870,6 +2012,8 @@ torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor Bt, bool tanh_epilogue,
              N >= G256_BM;
   if (use256) {
     TORCH_CHECK(K % G256_BK == 0, "256-tile variant needs K%64==0");
+  } else if (variant == 3) {
+    TORCH_CHECK(K % 64 == 0, "bt2 variant needs K%64==0");
   } else {
     TORCH_CHECK(K % GEMM_BK == 0, "K must be a multiple of 32");
   }
@@ -1901,6 +2045,29 @@ torch::Tensor gemm_bt(torch::Tensor A, torch::Tensor Bt, bool tanh_epilogue,
   const int n_tiles = (N + GEMM_BM - 1) / GEMM_BM;
   const int m_tiles = (M + GEMM_BN - 1) / GEMM_BN;
   dim3 grid(n_tiles * m_tiles);
+  if (variant == 3) {
+    const size_t lds = 64 * 1024;
+    static bool bt2_cfg = false;
+    if (!bt2_cfg) {
+      (void)hipFuncSetAttribute((const void*)k_gemm_bt2<true>,
+                                hipFuncAttributeMaxDynamicSharedMemorySize,
+                                (int)lds);
+      (void)hipFuncSetAttribute((const void*)k_gemm_bt2<false>,
+                                hipFuncAttributeMaxDynamicSharedMemorySize,
+                                (int)lds);
+      (void)hipFuncSetAttribute((const void*)k_gemm_bt2<false, true>,
+                                hipFuncAttributeMaxDynamicSharedMemorySize,
+                                (int)lds);
+      bt2_cfg = true;
+    }
+    if (tanh_epilogue)
+      k_gemm_bt2<true><<<grid, 256, lds, cur_stream()>>>(
+          bf_ptr(A), bf_ptr(Bt), bf_ptr_mut(C), N, M, K);
+    else
+      k_gemm_bt2<false><<<grid, 256, lds, cur_stream()>>>(
+          bf_ptr(A), bf_ptr(Bt), bf_ptr_mut(C), N, M, K);
+    return C;
+  }
   if (tanh_epilogue)
     k_gemm_bt<true><<<grid, 256, 0, cur_stream()>>>(
         bf_ptr(A), bf_ptr(Bt), bf_ptr_mut(C), N, M, K);

@@ -535,3 +535,16 @@ def test_ce_fused_write_matches_ce_bwd():
     ref_code = e.gemm_nn_splitk(d_logits_ref, shadow)
     err = (d_code - ref_code).abs().max().item() / ref_code.abs().max().item()
     assert err < 1e-3, err
+
+
+@pytest.mark.parametrize('N,M,K', [(1024, 384, 384), (500, 200, 128)])
+def test_gemm_bt2_vs_matmul(N, M, K):
+    """double-buffered BK=64 128-tile GEMM (variant 3) refcheck."""
+    A = randn(N, K, dtype=torch.bfloat16, scale=0.3, seed=N + M + K)
+    Bt = randn(M, K, dtype=torch.bfloat16, scale=0.3)
+    C = ext().gemm_bt_v(A, Bt, False, 3)
+    ref = A.float() @ Bt.float().t()
+    err = (C.float() - ref).abs().max().item() / ref.abs().max().item()
+    assert err < 0.02, err
+    Ct = ext().gemm_bt_v(A, Bt, True, 3)
+    assert (Ct.float() - torch.tanh(ref)).abs().max().item() < 0.02

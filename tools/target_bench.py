@@ -65,6 +65,8 @@ t('transform GEMM 128-tile (current)', lambda: ext.gemm_bt_v(dz, w, False, 1))
 t('transform GEMM 256-tile', lambda: ext.gemm_bt_v(dz, w, False, 2))
 t('transform GEMM 128-tile tanh', lambda: ext.gemm_bt_v(dz, w, True, 1))
 t('transform GEMM 256-tile tanh', lambda: ext.gemm_bt_v(dz, w, True, 2))
+t('transform GEMM bt2 dbuf', lambda: ext.gemm_bt_v(dz, w, False, 3))
+t('transform GEMM bt2 dbuf tanh', lambda: ext.gemm_bt_v(dz, w, True, 3))
 
 # sparse embedding-update chain, java14m-ish shapes (uniform ids)
 ids = torch.randint(1, 1301137, (2 * N,), device='cuda', dtype=torch.int32)
