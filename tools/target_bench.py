@@ -78,3 +78,11 @@ t('sparse adam ctx-direct (tok table)',
   lambda: ext.adam_sparse_rows_hash_ctx(
       tok_p, ids, dctx, 0, 256, 2, 128, tok_m, tok_v, 1, 1e-3, 0.9, 0.999,
       1e-8, torch.empty(0)))
+
+comb = (torch.randn(B, C, D, device='cuda') * 0.1).to(torch.bfloat16)
+av = torch.randn(D, device='cuda')
+maskf = torch.ones(B, C, device='cuda')
+codev, alphav = ext.attention_fwd(comb, av, maskf)
+dcv = torch.randn(B, D, device='cuda')
+t('attn_bwd', lambda: ext.attention_bwd(comb, av, alphav, dcv, True))
+t('attn_fwd', lambda: ext.attention_fwd(comb, av, maskf))
