@@ -1288,13 +1288,41 @@ __global__ void k_attn_fwd(const ushort* __restrict__ comb,
   }
   __syncthreads();
 
-  // ---- weighted sum: thread t owns output columns t, t+T, ...
-  for (int col = tid; col < D; col += blockDim.x) {
-    float acc = 0.f;
-    const ushort* p = base + col;
-    for (int c = 0; c < C; ++c) acc += sc[c] * bf2f(p[(long)c * D]);
-    code[(long)b * D + col] = acc;
+  // ---- weighted sum, vectorized: thread owns an 8-col octet x a C-slice
+  // (b128 row loads; per-element u16 was issue-bound); slices combine in
+  // LDS behind the sc scratch
+  float* acc_sh = sc + C;
+  for (int col = tid; col < D; col += blockDim.x) acc_sh[col] = 0.f;
+  __syncthreads();
+  {
+    const int octets = D >> 3;
+    const int slices = min(8, (int)blockDim.x / octets);
+    const int oct = tid % octets;
+    const int slice = tid / octets;
+    if (slice < slices) {
+      const int per = (C + slices - 1) / slices;
+      const int c0 = slice * per, c1 = min(C, c0 + per);
+      const int col0 = oct * 8;
+      float acc8[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc8[j] = 0.f;
+      for (int c = c0; c < c1; ++c) {
+        const float w = sc[c];
+        ulonglong2 packed = *reinterpret_cast<const ulonglong2*>(
+            base + (long)c * D + col0);
+        const ushort* u = reinterpret_cast<const ushort*>(&packed);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc8[j] += w * bf2f(u[j]);
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        if (slices > 1) atomicAdd(acc_sh + col0 + j, acc8[j]);
+        else acc_sh[col0 + j] = acc8[j];
+    }
   }
+  __syncthreads();
+  for (int col = tid; col < D; col += blockDim.x)
+    code[(long)b * D + col] = acc_sh[col];
 }
 
 __launch_bounds__(ATTN_THREADS)
@@ -2407,7 +2435,7 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor comb, torch::Tensor a,
   auto mask32 = mask.to(torch::kFloat32).contiguous();
   auto code = torch::empty({B, D}, comb.options().dtype(torch::kFloat32));
   auto alpha = torch::empty({B, C}, comb.options().dtype(torch::kFloat32));
-  const size_t lds = (size_t)C * 4;
+  const size_t lds = (size_t)(C + D) * 4;  // scores + code accumulator
   k_attn_fwd<<<B, 256, lds, cur_stream()>>>(
       bf_ptr(comb), a32.data_ptr<float>(), mask32.data_ptr<float>(),
       code.data_ptr<float>(), alpha.data_ptr<float>(), B, C, D);
