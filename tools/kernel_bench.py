@@ -61,7 +61,7 @@ def main():
     timeit('attn_fwd', lambda: ext.attention_fwd(comb, a, mask))
     alpha = ext.attention_fwd(comb, a, mask)[1]
     dcode = torch.randn(B, D, device=dev)
-    timeit('attn_bwd', lambda: ext.attention_bwd(comb, a, alpha, dcode))
+    timeit('attn_bwd', lambda: ext.attention_bwd(comb, a, alpha, dcode, False))
 
     timeit('ce_fwd', lambda: ext.ce_fwd(logits, labels))
     timeit('ce_bwd', lambda: ext.ce_bwd(logits, lse, labels, 1.0 / B))
