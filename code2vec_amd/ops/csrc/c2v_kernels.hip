@@ -1770,7 +1770,19 @@ __global__ void k_hash_claim(const I* __restrict__ ids, long n,
       }
       slot = (slot + 1) & mask_;
     }
-    if (cnt) atomicAdd(cnt + slot, run_len);
+    if (cnt) {
+      // saturating occurrence marker: only "exactly one" vs "more" matters
+      // downstream, so multi-element runs write 2 idempotently and
+      // single-element runs increment ONLY while cnt < 2 (read-before-add —
+      // an unconditional atomicAdd serialized 136K same-address adds when a
+      // hot id appeared as many short runs: 6.2 ms on the 33%-interleaved
+      // PAD micro-bench)
+      if (run_len > 1) {
+        if (cnt[slot] < 2) cnt[slot] = 2;
+      } else if (cnt[slot] < 2) {
+        atomicAdd(cnt + slot, 1);
+      }
+    }
   }
 }
 
