@@ -994,7 +994,7 @@ __global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
 // (row-tile, k-chunk) pairs like k_gemm_nn_splitk, accumulate fp32
 // partials P[(chunk, V, M)] folded by k_splitk_reduce. C is unused then.
 template <bool CEB = false, bool SPLITK = false>
-__launch_bounds__(256, 1)
+__launch_bounds__(512, 1)
 __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
                           const ushort* __restrict__ B,  // code (K, M)
                           ushort* __restrict__ C,        // out (V, M)
@@ -1028,28 +1028,32 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
-  const int wid = tid >> 6;              // 4 waves, one 96-col group each
-  const int wcol = wid * 96;
+  const int wid = tid >> 6;              // 8 waves: 2 (v-rows) x 4 (cols)
+  const int wrow = (wid >> 2) * 64;
+  const int wcol = (wid & 3) * 96;
 
-  f32x4 acc[8][6];
+  // 512 threads / 64x96 per-wave tile: 96 accumulator regs -> the whole
+  // block fits 2 waves/SIMD (the earlier 4-wave/128-row layout needed 192
+  // accs and ran 1 wave/SIMD, fully latency-exposed at only 32 k-steps)
+  f32x4 acc[4][6];
 #pragma unroll
-  for (int i = 0; i < 8; ++i)
+  for (int i = 0; i < 4; ++i)
 #pragma unroll
     for (int j = 0; j < 6; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  // A tile: 32 k-rows x 128 v-cols = 512 b128 chunks -> 2/thread
-  const int at_kr[2] = {(tid + 0) >> 4, (tid + 256) >> 4};
+  // A tile: 32 k-rows x 128 v-cols = 512 b128 chunks -> 1/thread
+  const int at_kr[1] = {tid >> 4};
   const int at_c8 = (tid & 15) * 8;
-  // B tile: 32 k-rows x 384 cols = 1536 b128 chunks -> 6/thread
-  int b_kr[6], b_c8[6];
+  // B tile: 32 k-rows x 384 cols = 1536 b128 chunks -> 3/thread
+  int b_kr[3], b_c8[3];
 #pragma unroll
-  for (int r = 0; r < 6; ++r) {
-    const int idx = tid + r * 256;
+  for (int r = 0; r < 3; ++r) {
+    const int idx = tid + r * 512;
     b_kr[r] = idx / 48;
     b_c8[r] = (idx % 48) * 8;
   }
 
-  u16x8 ra[2], rb[6];
+  u16x8 ra[1], rb[3];
   auto ce_map = [&](u16x8 v, int gk, int vbase) -> u16x8 {
     if (!CEB) return v;
     const float l = lse[gk];
@@ -1065,7 +1069,7 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
   auto load_ab = [&](int ks) {
     const int k0 = ks * GNN_BK;
 #pragma unroll
-    for (int r = 0; r < 2; ++r) {
+    for (int r = 0; r < 1; ++r) {
       const long base = (long)(k0 + at_kr[r]) * V + v0;
       if (v0 + at_c8 + 8 <= V) {
         ra[r] = ce_map(*reinterpret_cast<const u16x8*>(A + base + at_c8),
@@ -1082,7 +1086,7 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
       }
     }
 #pragma unroll
-    for (int r = 0; r < 6; ++r) {
+    for (int r = 0; r < 3; ++r) {
       if (b_c8[r] + 8 <= M) {
         rb[r] = *reinterpret_cast<const u16x8*>(
             B + (long)(k0 + b_kr[r]) * M + b_c8[r]);
@@ -1096,13 +1100,13 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
   };
   auto store_ab = [&](int buf) {
 #pragma unroll
-    for (int r = 0; r < 2; ++r) {
+    for (int r = 0; r < 1; ++r) {
       ushort* p = TLDS_A(buf) + at_kr[r] * GTN_PKT + GNN_BROT(at_kr[r]) +
                   at_c8;
       *reinterpret_cast<u16x8*>(p) = ra[r];  // aligned b128 (368 B stride)
     }
 #pragma unroll
-    for (int r = 0; r < 6; ++r) {
+    for (int r = 0; r < 3; ++r) {
       ushort* p = TLDS_B(buf) + b_kr[r] * GNN_PKB + GNN_BROT(b_kr[r]) +
                   b_c8[r];
       *reinterpret_cast<u16x8*>(p) = rb[r];
@@ -1122,8 +1126,8 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
     }
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int m = 0; m < 8; ++m) {
-      const int row = m * 16 + (lane & 15);   // v-col inside the tile
+    for (int m = 0; m < 4; ++m) {
+      const int row = wrow + m * 16 + (lane & 15);  // v-col inside the tile
       const ushort* ap = TLDS_A(buf) + (lane >> 4) * (8 * GTN_PKT + 16) + row;
       u16x8 t;
 #pragma unroll
@@ -1157,7 +1161,7 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
   if ((K % GNN_BK) && ks1 == total_ksteps && ks0 < ks1) {
     const int k0 = (total_ksteps - 1) * GNN_BK;
 #pragma unroll
-    for (int r = 0; r < 2; ++r) {
+    for (int r = 0; r < 1; ++r) {
       const int gk = k0 + at_kr[r];
       u16x8 v;
 #pragma unroll
@@ -1175,7 +1179,7 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
         TLDS_A(0)[at_kr[r] * GTN_PKT + GNN_BROT(at_kr[r]) + at_c8 + j] = v[j];
     }
 #pragma unroll
-    for (int r = 0; r < 6; ++r) {
+    for (int r = 0; r < 3; ++r) {
       const int gk = k0 + b_kr[r];
 #pragma unroll
       for (int j = 0; j < 8; ++j)
@@ -1189,10 +1193,10 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
 
   float* Pc = SPLITK ? P + (long)chunk * V * M : nullptr;
 #pragma unroll
-  for (int m = 0; m < 8; ++m) {
+  for (int m = 0; m < 4; ++m) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int v = v0 + m * 16 + (lane >> 4) * 4 + r;
+      const int v = v0 + wrow + m * 16 + (lane >> 4) * 4 + r;
       if (v >= V) continue;
 #pragma unroll
       for (int n = 0; n < 6; ++n) {
@@ -2270,7 +2274,7 @@ torch::Tensor gemm_tn_bf16(torch::Tensor A, torch::Tensor B) {
     tn_configured = true;
   }
   const int grid = (V + GTN_BV - 1) / GTN_BV;
-  k_gemm_tn<false><<<grid, 256, lds, cur_stream()>>>(
+  k_gemm_tn<false><<<grid, 512, lds, cur_stream()>>>(
       bf_ptr(A), bf_ptr(B), bf_ptr_mut(C), V, M, K);
   return C;
 }
@@ -2295,7 +2299,7 @@ torch::Tensor gemm_tn_ce(torch::Tensor logits, torch::Tensor code,
     cfg2 = true;
   }
   const int grid = (V + GTN_BV - 1) / GTN_BV;
-  k_gemm_tn<true><<<grid, 256, lds, cur_stream()>>>(
+  k_gemm_tn<true><<<grid, 512, lds, cur_stream()>>>(
       bf_ptr(logits), bf_ptr(code), bf_ptr_mut(C), V, M, K,
       lse.data_ptr<float>(), labels_c.data_ptr<long>(), (float)scale);
   return C;
@@ -2327,7 +2331,7 @@ torch::Tensor gemm_tn_splitk(torch::Tensor A, torch::Tensor B) {
                               (int)lds);
     cfg3 = true;
   }
-  k_gemm_tn<false, true><<<S * row_tiles, 256, lds, cur_stream()>>>(
+  k_gemm_tn<false, true><<<S * row_tiles, 512, lds, cur_stream()>>>(
       bf_ptr(A), bf_ptr(B), nullptr, N2, M, K, nullptr, nullptr, 1.f,
       P.data_ptr<float>(), S, kpc, row_tiles);
   const long total = (long)N2 * M;
