@@ -148,6 +148,9 @@ class Code2VecNetwork:
     def state_dict(self) -> Dict[str, torch.Tensor]:
         sd = {n: self.get_param(n).cpu() for n in self.param_names()}
         sd['adam_step'] = torch.tensor(self.adam_step)
+        # dropout stream position: resuming mid-training replays the exact
+        # per-step seeds the uninterrupted run would have used
+        sd['dropout_step'] = torch.tensor(self._step_ctr)
         for n in self.param_names():
             sd['adam_m.' + n] = self._adam_m[n].cpu()
             sd['adam_v.' + n] = self._adam_v[n].cpu()
@@ -161,6 +164,8 @@ class Code2VecNetwork:
     def load_state_dict(self, sd: Dict[str, torch.Tensor]):
         for n in self.param_names():
             self.get_param(n).copy_(sd[n].to(self.device))
+        if 'dropout_step' in sd:
+            self._step_ctr = int(sd['dropout_step'])
         if 'adam_step' in sd:
             self.adam_step = int(sd['adam_step'])
             for n in self.param_names():

@@ -182,3 +182,58 @@ def test_export_code_vectors(trained, tmp_path):
     with open(vec_file) as f:
         row = f.readline().split()
         assert len(row) == cfg.CODE_VECTOR_SIZE
+
+
+def test_resume_training_is_exact(tmp_path):
+    """Optimizer-state continuity: train 6 steps straight vs train 3 +
+    checkpoint + fresh-network load + 3 more — parameters must match
+    exactly (full Adam state, dropout-stream position and step counter all
+    round-trip, so the resumed run replays identical stochasticity)."""
+    import torch
+
+    from code2vec_amd.config import Config
+    from code2vec_amd.models.network import Code2VecNetwork
+
+    def make_cfg():
+        cfg = Config(set_defaults=True)
+        cfg.MAX_CONTEXTS = 6
+        cfg.TOKEN_EMBEDDINGS_SIZE = 8
+        cfg.PATH_EMBEDDINGS_SIZE = 8
+        cfg.CODE_VECTOR_SIZE = 24
+        cfg.TARGET_EMBEDDINGS_SIZE = 24
+        cfg.DROPOUT_KEEP_RATE = 0.75   # exercised: the seed stream resumes
+        cfg.COMPUTE_DTYPE = 'fp32'
+        cfg.DEVICE = 'cpu'
+        return cfg
+
+    def batches():
+        g = torch.Generator().manual_seed(11)
+        out = []
+        for _ in range(6):
+            src = torch.randint(0, 30, (4, 6), generator=g, dtype=torch.int32)
+            pth = torch.randint(0, 20, (4, 6), generator=g, dtype=torch.int32)
+            tgt = torch.randint(0, 30, (4, 6), generator=g, dtype=torch.int32)
+            mask = torch.ones(4, 6)
+            lab = torch.randint(1, 15, (4,), generator=g)
+            out.append((src, pth, tgt, mask, lab))
+        return out
+
+    bs = batches()
+    net_a = Code2VecNetwork(make_cfg(), 30, 20, 15, device='cpu')
+    for b in bs:
+        net_a.train_step(*b)
+
+    net_b = Code2VecNetwork(make_cfg(), 30, 20, 15, device='cpu')
+    for b in bs[:3]:
+        net_b.train_step(*b)
+    ckpt = tmp_path / 'mid.pt'
+    torch.save(net_b.state_dict(), ckpt)
+    net_c = Code2VecNetwork(make_cfg(), 30, 20, 15, device='cpu')
+    net_c.load_state_dict(torch.load(ckpt, weights_only=False))
+    assert net_c.adam_step == 3
+    for b in bs[3:]:
+        net_c.train_step(*b)
+
+    for n in net_a.param_names():
+        assert torch.equal(net_a.get_param(n), net_c.get_param(n)), n
+        assert torch.equal(net_a._adam_m[n], net_c._adam_m[n]), n
