@@ -100,6 +100,18 @@ def main():
                          '(BASELINE config 4); 0 = full softmax (default)')
     args = ap.parse_args()
 
+    # --gpus N without a torchrun rendezvous: self-launch one rank per GPU
+    # (a single process would otherwise run 1 GPU while reporting N-GPU
+    # aggregate throughput)
+    if args.gpus > 1 and 'WORLD_SIZE' not in os.environ:
+        import subprocess
+        import sys
+        cmd = [sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
+               '--nproc-per-node', str(args.gpus),
+               '--master-addr', '127.0.0.1', '--master-port', '29547',
+               os.path.abspath(__file__)] + sys.argv[1:]
+        raise SystemExit(subprocess.call(cmd))
+
     world_size = int(os.environ.get('WORLD_SIZE', '1'))
     rank = int(os.environ.get('RANK', '0'))
     local_rank = int(os.environ.get('LOCAL_RANK', str(rank)))

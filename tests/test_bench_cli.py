@@ -54,3 +54,14 @@ def test_bench_sampled_flag():
     out = run_bench([sys.executable, 'bench.py', '--steps', '3', '--warmup',
                      '1', '--batch', '8', '--sampled-softmax', '64'])
     assert out['config']['softmax'] == 'sampled-64'
+
+
+@pytest.mark.timeout(600)
+def test_bench_gpus_flag_self_launches_torchrun():
+    """`--gpus 2` without a torchrun rendezvous must run 2 ranks (a single
+    process would report inflated aggregate throughput)."""
+    out = run_bench([sys.executable, 'bench.py', '--gpus', '2', '--steps',
+                     '3', '--warmup', '1', '--batch', '8'])
+    assert out['n_gpus'] == 2
+    assert out['config']['parallelism'] == 'dp2'
+    assert out['config']['global_batch'] == 16
