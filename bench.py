@@ -104,11 +104,16 @@ def main():
     # (a single process would otherwise run 1 GPU while reporting N-GPU
     # aggregate throughput)
     if args.gpus > 1 and 'WORLD_SIZE' not in os.environ:
+        import socket
         import subprocess
         import sys
+        # pick a free rendezvous port so concurrent bench runs don't collide
+        with socket.socket() as s:
+            s.bind(('127.0.0.1', 0))
+            port = s.getsockname()[1]
         cmd = [sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
                '--nproc-per-node', str(args.gpus),
-               '--master-addr', '127.0.0.1', '--master-port', '29547',
+               '--master-addr', '127.0.0.1', '--master-port', str(port),
                os.path.abspath(__file__)] + sys.argv[1:]
         raise SystemExit(subprocess.call(cmd))
 

@@ -175,6 +175,15 @@ class Code2VecNetwork:
         self._refresh_shadows()
         if self._step_t is not None:
             self._step_t.fill_(self.adam_step)
+        if self._seed_t is not None:
+            # Restore the GPU dropout seed stream to where the uninterrupted
+            # run would be: forward() advances _seed_t by 2654435761/step from
+            # 0, so after _step_ctr steps it holds _step_ctr*2654435761 with
+            # int64 wraparound semantics.
+            v = (self._step_ctr * 2654435761) & 0xFFFFFFFFFFFFFFFF
+            if v >= 1 << 63:
+                v -= 1 << 64
+            self._seed_t.fill_(v)
 
     # ---- forward ----
 
@@ -412,6 +421,19 @@ class Code2VecNetwork:
     # ---- evaluation / prediction forward ----
 
     @torch.no_grad()
+    def eval_batch(self, src_ids, path_ids, tgt_ids, valid_mask, labels,
+                   top_k: int):
+        """predict_batch + per-row CE against `labels` (the reference Keras
+        backend reports evaluation loss — keras_model.py:166-228). Returns
+        (indices, scores, code, alpha, loss_sum) with loss_sum an unsynced
+        device scalar (sum over the batch, fp32)."""
+        st = self.forward(src_ids, path_ids, tgt_ids, valid_mask, training=False)
+        logits = self.logits(st.code)
+        loss_rows, _ = F.ce_fwd(logits, labels)
+        scores, indices = F.topk(logits, k=min(top_k, logits.shape[1]))
+        return indices, scores, st.code, st.alpha, loss_rows.float().sum()
+
+    @torch.no_grad()
     def predict_batch(self, src_ids, path_ids, tgt_ids, valid_mask, top_k: int,
                       normalize_scores: bool = False):
         """Returns (topk_indices (B,k) int64, topk_scores (B,k) fp32,
@@ -445,6 +467,17 @@ class GraphTrainStep:
         self.mask = torch.ones(batch_size, C, dtype=torch.float32, device=dev)
         self.labels = torch.ones(batch_size, dtype=torch.int64, device=dev)
 
+        # The warmup steps are REAL optimizer steps on dummy batches (the
+        # allocator must see the exact allocation pattern before capture);
+        # snapshot all mutable state so they leave no trace on the model.
+        snap_params = {n: net.get_param(n).detach().clone()
+                       for n in net.param_names()}
+        snap_m = {n: net._adam_m[n].detach().clone() for n in net.param_names()}
+        snap_v = {n: net._adam_v[n].detach().clone() for n in net.param_names()}
+        snap_adam_step, snap_step_ctr = net.adam_step, net._step_ctr
+        snap_seed_t = net._seed_t.clone()
+        snap_step_t = net._step_t.clone()
+
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side):
@@ -457,10 +490,19 @@ class GraphTrainStep:
         with torch.cuda.graph(self.graph):
             self.loss = net.train_step(self.src, self.pth, self.tgt,
                                        self.mask, self.labels)
-        # capture RECORDS the device ops without executing them, but the
-        # python body still bumped the host counter — undo that: the capture
-        # itself is not a training step.
-        net.adam_step_host_sync(-1)
+        # Capture RECORDS the device ops without executing them (only the
+        # python-side counters moved). Roll every piece of state the warmup
+        # touched back to the snapshot — in place, so the capture's recorded
+        # reads still point at the live allocations.
+        torch.cuda.synchronize()
+        for n in net.param_names():
+            net.get_param(n).copy_(snap_params[n])
+            net._adam_m[n].copy_(snap_m[n])
+            net._adam_v[n].copy_(snap_v[n])
+        net.adam_step, net._step_ctr = snap_adam_step, snap_step_ctr
+        net._seed_t.copy_(snap_seed_t)
+        net._step_t.copy_(snap_step_t)
+        net._refresh_shadows()
 
     def step(self, src, pth, tgt, mask, labels) -> torch.Tensor:
         self.src.copy_(src)

@@ -102,7 +102,10 @@ class Code2VecModel(Code2VecModelBase):
         save_every = max(1, steps_per_epoch * cfg.SAVE_EVERY_EPOCHS)
 
         batch_num = 0
-        sum_loss = 0.0
+        # device-resident loss accumulator: train_step returns an unsynced
+        # device scalar; forcing float() per batch would stall the pipeline,
+        # so the H2D sync is paid only at the logging boundary
+        sum_loss = None
         window_examples = 0
         start = time.time()
         multi_batch_start = start
@@ -131,10 +134,12 @@ class Code2VecModel(Code2VecModelBase):
                 b.source_token_indices, b.path_indices, b.target_token_indices,
                 b.context_valid_mask, b.target_index, reducer=self.reducer)
             batch_num += 1
-            sum_loss += float(loss)
+            loss = loss.detach()
+            sum_loss = loss if sum_loss is None else sum_loss + loss
             window_examples += b.source_token_indices.shape[0] * self.world_size
 
             if batch_num % cfg.NUM_BATCHES_TO_LOG_PROGRESS == 0:
+                avg_loss = float(sum_loss) / cfg.NUM_BATCHES_TO_LOG_PROGRESS
                 elapsed = time.time() - multi_batch_start
                 throughput = window_examples / max(elapsed, 1e-9)
                 # EWMA + epoch ETA (reference P13 progress logger semantics,
@@ -143,8 +148,7 @@ class Code2VecModel(Code2VecModelBase):
                     throughput if self._throughput_ewma is None
                     else 0.5 * self._throughput_ewma + 0.5 * throughput)
                 msg = ('Average loss at batch %d: %f, throughput: %d samples/sec'
-                       % (batch_num, sum_loss / cfg.NUM_BATCHES_TO_LOG_PROGRESS,
-                          throughput))
+                       % (batch_num, avg_loss, throughput))
                 if steps_per_epoch > 0 and self._throughput_ewma > 0:
                     remaining = (steps_per_epoch - batch_num % steps_per_epoch)
                     eta_sec = remaining * cfg.TRAIN_BATCH_SIZE * self.world_size \
@@ -152,11 +156,10 @@ class Code2VecModel(Code2VecModelBase):
                     msg += ', epoch ETA: %dm%02ds' % (eta_sec // 60, eta_sec % 60)
                 self.log(msg)
                 if scalar_log is not None:
-                    scalar_log.write('%d,%f,%f\n' % (
-                        batch_num, sum_loss / cfg.NUM_BATCHES_TO_LOG_PROGRESS,
-                        throughput))
+                    scalar_log.write('%d,%f,%f\n' % (batch_num, avg_loss,
+                                                     throughput))
                     scalar_log.flush()
-                sum_loss = 0.0
+                sum_loss = None
                 window_examples = 0
                 multi_batch_start = time.time()
 
@@ -212,9 +215,18 @@ class Code2VecModel(Code2VecModelBase):
             nr_examples = 0
             for batch in reader.iter_batches():
                 b = batch.to(device) if device.type != 'cpu' else batch
-                indices, scores, code, _alpha = self.network.predict_batch(
-                    b.source_token_indices, b.path_indices,
-                    b.target_token_indices, b.context_valid_mask, top_k=k)
+                if b.target_index is not None:
+                    indices, scores, code, _alpha, loss_sum = \
+                        self.network.eval_batch(
+                            b.source_token_indices, b.path_indices,
+                            b.target_token_indices, b.context_valid_mask,
+                            b.target_index, top_k=k)
+                    total_loss += float(loss_sum)
+                    total_rows += int(b.target_index.shape[0])
+                else:
+                    indices, scores, code, _alpha = self.network.predict_batch(
+                        b.source_token_indices, b.path_indices,
+                        b.target_token_indices, b.context_valid_mask, top_k=k)
                 idx_np = indices.cpu().numpy()
                 top_words = [[index_to_word.get(int(i), special_words.OOV)
                               for i in row] for row in idx_np]
