@@ -1,9 +1,11 @@
 """Extractor bridge for the interactive predict shell.
 
-Runs an AST path extractor over a single source file and re-hashes the path
-strings with a Java-`String#hashCode`-compatible hash so the displayed paths
-stay human-readable while the model sees the hashed vocabulary it was trained
-on (reference: extractor.py:12-49).
+Behavior contract (reference: extractor.py:11-49): run an AST path extractor
+over one source file with hashing disabled, then re-hash every path string
+with a java.lang.String#hashCode-compatible hash so the model sees the hashed
+vocabulary it was trained on while a reverse map keeps the raw path strings
+available for display. Lines are truncated to MAX_CONTEXTS contexts and
+space-padded like the offline preprocessor output.
 
 The extractor executable is resolved in order:
 1. our native C++ extractor `c2v-extract` (extractor/ build output), or
@@ -15,17 +17,18 @@ import shutil
 import subprocess
 from typing import Dict, List, Tuple
 
-
+_REPO_ROOT = os.path.dirname(os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
 
 
 def java_string_hashcode(s: str) -> int:
-    """Reimplementation of java.lang.String#hashCode (32-bit wrapping), the
-    hash the reference training data was produced with (extractor.py:40-49)."""
+    """java.lang.String#hashCode: h = 31*h + c over UTF-16 units, wrapping in
+    signed 32-bit. The training data's path hashes were produced with this
+    (reference extractor.py:40-49), so serving must match it exactly."""
     h = 0
     for ch in s:
-        h = (31 * h + ord(ch)) & 0xFFFFFFFF
-    h = h & 0xFFFFFFFF
-    return h - 0x100000000 if h > 0x7FFFFFFF else h
+        h = (h * 31 + ord(ch)) & 0xFFFFFFFF
+    return h - (1 << 32) if h >= (1 << 31) else h
 
 
 class Extractor:
@@ -37,47 +40,48 @@ class Extractor:
         self.max_contexts = config.MAX_CONTEXTS
         self.jar_path = jar_path or os.environ.get('EXTRACTOR_JAR')
         self.native_bin = shutil.which('c2v-extract') or os.path.join(
-            os.path.dirname(os.path.dirname(os.path.dirname(os.path.abspath(__file__)))),
-            'extractor', 'c2v-extract')
+            _REPO_ROOT, 'extractor', 'c2v-extract')
 
     def _command(self, path: str) -> List[str]:
+        limits = ['--max_path_length', str(self.max_path_length),
+                  '--max_path_width', str(self.max_path_width)]
         if os.path.isfile(self.native_bin) and os.access(self.native_bin, os.X_OK):
-            return [self.native_bin, '--file', path, '--no_hash',
-                    '--max_path_length', str(self.max_path_length),
-                    '--max_path_width', str(self.max_path_width)]
+            return [self.native_bin, '--file', path, '--no_hash'] + limits
         if self.jar_path:
-            return ['java', '-cp', self.jar_path, 'JavaExtractor.App',
-                    '--max_path_length', str(self.max_path_length),
-                    '--max_path_width', str(self.max_path_width),
-                    '--file', path, '--no_hash']
-        raise RuntimeError(
-            'No extractor available: build extractor/c2v-extract or set EXTRACTOR_JAR.')
+            return (['java', '-cp', self.jar_path, 'JavaExtractor.App']
+                    + limits + ['--file', path, '--no_hash'])
+        raise RuntimeError('No extractor available: build '
+                           'extractor/c2v-extract or set EXTRACTOR_JAR.')
 
-    def extract_paths(self, path: str) -> Tuple[List[str], Dict[int, str]]:
-        """Returns (model-input lines with hashed paths truncated to
-        MAX_CONTEXTS, hash→path-string dict for display)."""
-        out = subprocess.run(self._command(path), capture_output=True, text=True)
-        if out.returncode != 0:
-            raise ValueError(out.stderr.strip() or 'extractor failed')
-        output = out.stdout.splitlines()
-        if not output:
-            raise ValueError(out.stderr.strip() or 'extractor produced no output')
-        hash_to_string_dict: Dict[int, str] = {}
-        result: List[str] = []
-        for line in output:
-            parts = line.rstrip().split(' ')
-            method_name = parts[0]
-            current_result_line_parts = [method_name]
-            contexts = parts[1:]
-            for context in contexts[:self.max_contexts]:
-                context_parts = context.split(',')
-                if len(context_parts) != 3:
-                    continue
-                context_word1, context_path, context_word2 = context_parts
-                hashed_path = str(java_string_hashcode(context_path))
-                hash_to_string_dict[hashed_path] = context_path
-                current_result_line_parts.append(
-                    '%s,%s,%s' % (context_word1, hashed_path, context_word2))
-            space_padding = ' ' * (self.max_contexts - len(contexts))
-            result.append(' '.join(current_result_line_parts) + space_padding)
-        return result, hash_to_string_dict
+    def _rehash_line(self, line: str,
+                     unhash: Dict[str, str]) -> str:
+        """One raw extractor line -> model-input line: contexts truncated to
+        MAX_CONTEXTS, path strings replaced by their decimal hash (recorded in
+        `unhash`), and the line space-padded by the number of dropped-or-
+        missing context slots (matching preprocess.py:64-65 output)."""
+        fields = line.rstrip().split(' ')
+        method_name, contexts = fields[0], fields[1:]
+        kept = [method_name]
+        for ctx in contexts[:self.max_contexts]:
+            triple = ctx.split(',')
+            if len(triple) != 3:
+                continue
+            left, path_str, right = triple
+            hashed = str(java_string_hashcode(path_str))
+            unhash[hashed] = path_str
+            kept.append(f'{left},{hashed},{right}')
+        return ' '.join(kept) + ' ' * (self.max_contexts - len(contexts))
+
+    def extract_paths(self, path: str) -> Tuple[List[str], Dict[str, str]]:
+        """Returns (model-input lines, hash->path-string dict for display)."""
+        proc = subprocess.run(self._command(path), capture_output=True,
+                              text=True)
+        if proc.returncode != 0:
+            raise ValueError(proc.stderr.strip() or 'extractor failed')
+        raw_lines = proc.stdout.splitlines()
+        if not raw_lines:
+            raise ValueError(proc.stderr.strip()
+                             or 'extractor produced no output')
+        unhash: Dict[str, str] = {}
+        lines = [self._rehash_line(line, unhash) for line in raw_lines]
+        return lines, unhash
