@@ -121,15 +121,21 @@ class Code2VecModel(Code2VecModelBase):
         prefetcher = BatchPrefetcher(reader.iter_batches(), device,
                                      depth=cfg.READER_QUEUE_DEPTH)
         batches = iter(prefetcher)
+        # DP termination consensus with one-batch lookahead: the shard split
+        # is not batch-aligned, so ranks can finish with unequal batch counts
+        # — a rank stepping once more than its peers would hang in the
+        # gradient collectives. Each rank votes on its NEXT batch's existence
+        # asynchronously while the current step runs (Reducer.start_vote on
+        # the host-side group), so the consensus costs no per-step latency;
+        # longer ranks drop their surplus batches.
+        nxt = next(batches, None)
+        vote = self.reducer.start_vote(nxt is not None)
         while True:
-            b = next(batches, None)
-            # DP termination consensus: the shard split is not batch-aligned,
-            # so ranks can finish with unequal batch counts — a rank stepping
-            # once more than its peers would hang in the gradient
-            # collectives. All ranks must agree to continue (ddp.Reducer
-            # .all_continue); longer ranks drop their surplus batch.
-            if not self.reducer.all_continue(b is not None):
+            if not self.reducer.finish_vote(vote):
                 break
+            b = nxt
+            nxt = next(batches, None)
+            vote = self.reducer.start_vote(nxt is not None)
             loss = self.network.train_step(
                 b.source_token_indices, b.path_indices, b.target_token_indices,
                 b.context_valid_mask, b.target_index, reducer=self.reducer)

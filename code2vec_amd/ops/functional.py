@@ -312,6 +312,31 @@ def adam_sparse_rows_from_ctx(p, ids, d_ctx, off0, off1, n_seg, d, m, v,
     ref.adam_sparse_rows_step(p, ids, rows, m, v, step, lr, beta1, beta2, eps)
 
 
+def sparse_dedup_sum_ctx(ids, d_ctx, off0: int, off1: int, n_seg: int, d: int):
+    """Rank-local dedup+sum of embedding grad rows read straight from the
+    (N,3d) d_ctx layout (DP wire-volume reduction — SURVEY §2.4). Returns
+    (uniq_ids int64[capacity], acc fp32[capacity,d], count) where count is a
+    1-elem int32 device tensor on GPU and a plain int on CPU; only the first
+    `count` rows of the outputs are live."""
+    if backend_for(d_ctx) == 'hip':
+        uniq, acc, n_uniq = hip_ext(True).sparse_dedup_sum_ctx(
+            ids, d_ctx, int(off0), int(off1), int(n_seg), int(d))
+        return uniq, acc, n_uniq
+    if n_seg == 2:
+        rows = torch.cat([d_ctx[:, off0:off0 + d], d_ctx[:, off1:off1 + d]], 0)
+    else:
+        rows = d_ctx[:, off0:off0 + d]
+    return ref.sparse_dedup_sum(ids, rows)
+
+
+def sparse_dedup_sum(ids, rows):
+    """As sparse_dedup_sum_ctx but for materialized (n,d) grad rows."""
+    if backend_for(rows) == 'hip':
+        uniq, acc, n_uniq = hip_ext(True).sparse_dedup_sum_rows(ids, rows)
+        return uniq, acc, n_uniq
+    return ref.sparse_dedup_sum(ids, rows)
+
+
 def sampled_ce_fwd(logits_cand, labels, sampled, corr_true, corr_samp):
     if backend_for(logits_cand) == 'hip':
         return hip_ext(True).sampled_ce_fwd(logits_cand, labels, sampled,

@@ -223,6 +223,19 @@ def sampled_ce_bwd(logits_cand: torch.Tensor, labels: torch.Tensor,
 # Adam (TF AdamOptimizer formulation) — dense and sparse-row
 # ---------------------------------------------------------------------------
 
+def sparse_dedup_sum(ids: torch.Tensor,
+                     rows: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor, int]:
+    """Rank-local dedup+sum of sparse (id, grad-row) contributions before a
+    DP all-gather: returns (unique ids int64, fp32 summed rows, count).
+    Semantics-preserving vs shipping the raw rows — duplicate contributions
+    are summed either way; only the fp addition order changes."""
+    uniq, inverse = torch.unique(ids.long(), return_inverse=True)
+    acc = torch.zeros(uniq.numel(), rows.shape[1], dtype=torch.float32,
+                      device=rows.device)
+    acc.index_add_(0, inverse, rows.float())
+    return uniq, acc, int(uniq.numel())
+
+
 def adam_dense_step(p: torch.Tensor, g: torch.Tensor, m: torch.Tensor,
                     v: torch.Tensor, step: int, lr: float, beta1: float,
                     beta2: float, eps: float, shadow: torch.Tensor = None):

@@ -51,10 +51,24 @@ class Reducer:
         """assume_equal_shards: skip the per-step shard-size exchange (and its
         host syncs) when every rank contributes identically-shaped sparse
         grads — true whenever the per-rank batch size is fixed (the bench and
-        steady-state training; ragged tails need False)."""
+        steady-state training; ragged tails need False). Only meaningful for
+        the raw (non-dedup) gather path: deduped counts always differ."""
         self.group = process_group
         self.assume_equal_shards = assume_equal_shards
         self._handles: Dict[str, object] = {}
+        self._host_group = None
+        self._count_pin = None
+
+    def _host_pg(self):
+        """Process group for tiny host-side exchanges (sparse shard counts,
+        termination votes). Under RCCL these run on a separate gloo group so
+        they neither enqueue device work nor force a stream sync; under gloo
+        the main group already is host-side."""
+        if dist.get_backend(self.group) != 'nccl':
+            return self.group
+        if self._host_group is None:
+            self._host_group = dist.new_group(backend='gloo')
+        return self._host_group
 
     @property
     def world_size(self) -> int:
@@ -110,20 +124,94 @@ class Reducer:
         all_rows = torch.cat([t[:c] for t, c in zip(rows_out, counts)])
         return all_ids, all_rows / ws
 
-    def all_continue(self, have_next: bool) -> bool:
-        """Termination consensus for reader-driven DP training: ranks can
-        end an epoch with unequal batch counts (the DP shard split is not
-        batch-aligned), and a rank stepping once more than its peers would
-        deadlock in the gradient collectives. Every step each rank votes
-        whether it has a next batch; training continues only if ALL do
-        (surplus batches on longer ranks are dropped)."""
+    def allgather_sparse_dedup(self, entries):
+        """All-gather rank-locally deduped (unique ids, summed rows) sparse
+        gradients. `entries` is a list of (uniq_ids, acc_rows, count) triples
+        as returned by ops.functional.sparse_dedup_sum*; all entries share
+        ONE shard-count exchange. Returns a list of (ids, rows) with the DP
+        mean (1/ws) already folded into the gathered rows.
+
+        Wire format: rows travel bf16 on GPU (halves the xGMI bytes; the
+        fp32 local sums are rounded once) and fp32 on CPU (exactness tests);
+        ranks pad to the max count with a REAL id (their first unique id)
+        carrying all-zero rows — a no-op contribution under the hash-dedup
+        sparse Adam, so no slicing is needed after the gather."""
+        ws = self.world_size
+        assert ws > 1
+        dev = entries[0][0].device
+        cuda = dev.type == 'cuda'
+        # local counts to host: one batched async D2H + event (GPU), no-op (CPU)
+        if cuda:
+            cnt_dev = torch.cat([c.reshape(-1).to(torch.int64)
+                                 for _, _, c in entries])
+            if self._count_pin is None or self._count_pin.numel() < cnt_dev.numel():
+                self._count_pin = torch.empty(cnt_dev.numel(),
+                                              dtype=torch.int64,
+                                              pin_memory=True)
+            pin = self._count_pin[:cnt_dev.numel()]
+            pin.copy_(cnt_dev, non_blocking=True)
+            ev = torch.cuda.Event()
+            ev.record()
+            ev.synchronize()      # waits for the dedup kernels only: later
+            local = pin.tolist()  # device work keeps flowing behind this
+        else:
+            local = [int(c) if not torch.is_tensor(c) else int(c.item())
+                     for _, _, c in entries]
+        # count exchange on the host group (no device work, ~µs)
+        t = torch.tensor(local, dtype=torch.int64)
+        counts = [torch.empty_like(t) for _ in range(ws)]
+        dist.all_gather(counts, t, group=self._host_pg())
+        results = []
+        for j, (ids, rows, _) in enumerate(entries):
+            c = local[j]
+            n_max = max(int(cr[j]) for cr in counts)
+            wire_dt = torch.bfloat16 if cuda else rows.dtype
+            wire_ids = torch.empty(n_max, dtype=torch.int64, device=dev)
+            wire_rows = torch.empty(n_max, rows.shape[1], dtype=wire_dt,
+                                    device=dev)
+            if c > 0:
+                wire_ids[:c] = ids[:c]
+                wire_rows[:c] = rows[:c] * (1.0 / ws)
+            pad_id = ids[0] if c > 0 else torch.zeros(
+                (), dtype=torch.int64, device=dev)
+            if n_max > c:
+                wire_ids[c:] = pad_id
+                wire_rows[c:] = 0
+            ids_out = [torch.empty_like(wire_ids) for _ in range(ws)]
+            rows_out = [torch.empty_like(wire_rows) for _ in range(ws)]
+            dist.all_gather(ids_out, wire_ids, group=self.group)
+            dist.all_gather(rows_out, wire_rows, group=self.group)
+            results.append((torch.cat(ids_out), torch.cat(rows_out)))
+        return results
+
+    # -- termination consensus --------------------------------------------
+    # Ranks can end an epoch with unequal batch counts (the DP shard split is
+    # not batch-aligned); a rank stepping once more than its peers would
+    # deadlock in the gradient collectives. Each step every rank votes
+    # whether it has a NEXT batch; training continues only while ALL do
+    # (surplus batches on longer ranks are dropped). The vote is posted
+    # asynchronously on the host group one step ahead, so it rides along the
+    # step's compute instead of adding a blocking latency tax per step.
+
+    def start_vote(self, have_next: bool):
         if self.world_size <= 1:
             return have_next
         t = torch.tensor([1 if have_next else 0], dtype=torch.int32)
-        if dist.get_backend(self.group) == 'nccl':
-            t = t.cuda()
-        dist.all_reduce(t, op=dist.ReduceOp.MIN, group=self.group)
+        handle = dist.all_reduce(t, op=dist.ReduceOp.MIN,
+                                 group=self._host_pg(), async_op=True)
+        return handle, t
+
+    def finish_vote(self, vote) -> bool:
+        if self.world_size <= 1:
+            return bool(vote)
+        handle, t = vote
+        handle.wait()
         return bool(int(t.item()))
+
+    def all_continue(self, have_next: bool) -> bool:
+        """Blocking vote (start+finish back to back) — kept for callers
+        without a lookahead batch."""
+        return self.finish_vote(self.start_vote(have_next))
 
     def allreduce_mean_scalar(self, value: float) -> float:
         if self.world_size <= 1:

@@ -41,7 +41,8 @@ def make_batch():
     return src, pth, tgt, mask, labels
 
 
-def _worker(rank, world_size, init_file, result_dir):
+def _worker(rank, world_size, init_file, result_dir, dedup='1'):
+    os.environ['C2V_DP_DEDUP'] = dedup
     dist.init_process_group('gloo', init_method='file://' + init_file,
                             rank=rank, world_size=world_size)
     from code2vec_amd.parallel.ddp import Reducer
@@ -54,16 +55,22 @@ def _worker(rank, world_size, init_file, result_dir):
         net.train_step(src[sl], pth[sl], tgt[sl], mask[sl], labels[sl],
                        reducer=reducer)
     if rank == 0:
-        torch.save(net.state_dict(), os.path.join(result_dir, 'dp2.pt'))
+        torch.save(net.state_dict(), os.path.join(result_dir, 'dpN.pt'))
     dist.barrier()
     dist.destroy_process_group()
 
 
 @pytest.mark.timeout(240)
-def test_dp2_matches_dp1(tmp_path):
+@pytest.mark.parametrize('world_size,dedup', [(2, '1'), (2, '0'), (4, '1')])
+def test_dpn_matches_dp1(tmp_path, world_size, dedup):
+    """DP=N (batch split across ranks) must match DP=1 on the full batch —
+    with the default rank-local dedup+sum gather and with the raw-rows
+    gather (C2V_DP_DEDUP=0). Dedup changes only the fp summation
+    association, so 1e-6 covers it."""
     os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
     init_file = str(tmp_path / 'pg_init')
-    mp.spawn(_worker, args=(2, init_file, str(tmp_path)), nprocs=2, join=True)
+    mp.spawn(_worker, args=(world_size, init_file, str(tmp_path), dedup),
+             nprocs=world_size, join=True)
 
     torch.manual_seed(7)
     net1 = Code2VecNetwork(tiny_cfg(), V_TOK, V_PATH, V_TGT, device='cpu')
@@ -71,10 +78,10 @@ def test_dp2_matches_dp1(tmp_path):
     for _ in range(3):
         net1.train_step(src, pth, tgt, mask, labels)
 
-    dp2 = torch.load(str(tmp_path / 'dp2.pt'), weights_only=False)
+    dpn = torch.load(str(tmp_path / 'dpN.pt'), weights_only=False)
     for name in net1.param_names():
-        assert torch.allclose(net1.get_param(name), dp2[name], atol=1e-6), name
-        assert torch.allclose(net1._adam_m[name], dp2['adam_m.' + name],
+        assert torch.allclose(net1.get_param(name), dpn[name], atol=1e-6), name
+        assert torch.allclose(net1._adam_m[name], dpn['adam_m.' + name],
                               atol=1e-6), name
 
 

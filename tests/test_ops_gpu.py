@@ -576,3 +576,35 @@ def test_adam_sparse_heavy_duplicates_vs_reference():
     # float-atomic accumulation order differs from the reference sum
     assert (p.cpu() - p0).abs().max().item() < 2e-5
     assert (m.cpu() - m0).abs().max().item() < 2e-5
+
+
+def test_sparse_dedup_sum_matches_reference():
+    """Rank-local dedup+sum (DP wire-volume reduction): unique set and
+    per-id sums must match the torch.unique+index_add reference, for both
+    the materialized-rows and the d_ctx-slicing entry points."""
+    e = ext()
+    torch.manual_seed(33)
+    Vr, d, n = 500, 32, 2048
+    ids = torch.randint(0, Vr, (n,), dtype=torch.int32, device='cuda')
+    ids[::4] = 3          # hot id, non-contiguous
+    rows = (torch.randn(n, d, device='cuda') * 0.1).to(torch.bfloat16)
+    uniq, acc, cnt = e.sparse_dedup_sum_rows(ids, rows)
+    c = int(cnt.item())
+    ref_u, ref_acc, ref_c = R.sparse_dedup_sum(ids.cpu(), rows.cpu().float())
+    assert c == ref_c
+    order = torch.argsort(uniq[:c].cpu())
+    assert torch.equal(uniq[:c].cpu()[order], ref_u)
+    assert (acc[:c].cpu()[order] - ref_acc).abs().max().item() < 2e-2
+
+    # d_ctx two-segment layout (token grads: src cols [0,d), tgt cols [2d,3d))
+    N = 1024
+    d_ctx = (torch.randn(N, 3 * d, device='cuda') * 0.1).to(torch.bfloat16)
+    tok_ids = torch.randint(0, Vr, (2 * N,), dtype=torch.int32, device='cuda')
+    uniq2, acc2, cnt2 = e.sparse_dedup_sum_ctx(tok_ids, d_ctx, 0, 2 * d, 2, d)
+    c2 = int(cnt2.item())
+    rows2 = torch.cat([d_ctx[:, :d], d_ctx[:, 2 * d:]], 0).cpu().float()
+    ref_u2, ref_acc2, ref_c2 = R.sparse_dedup_sum(tok_ids.cpu(), rows2)
+    assert c2 == ref_c2
+    order2 = torch.argsort(uniq2[:c2].cpu())
+    assert torch.equal(uniq2[:c2].cpu()[order2], ref_u2)
+    assert (acc2[:c2].cpu()[order2] - ref_acc2).abs().max().item() < 2e-2

@@ -88,6 +88,35 @@ def main():
         ext.adam_sparse_rows_hash(tok, ids_pad, rows, mm, vv, 1, 1e-3, 0.9,
                                   0.999, 1e-8, torch.empty(0), torch.empty(0))
     timeit('sparse adam (hash, 33% PAD ids)', sparse_step_pad)
+    # realistic: PAD-slot grads are exactly zero (masked contexts produce no
+    # gradient), and the accumulate kernel skips atomics for zero values
+    rows_pad0 = rows.clone(); rows_pad0[::3] = 0
+    def sparse_step_pad0():
+        ext.adam_sparse_rows_hash(tok, ids_pad, rows_pad0, mm, vv, 1, 1e-3,
+                                  0.9, 0.999, 1e-8, torch.empty(0),
+                                  torch.empty(0))
+    timeit('sparse adam (33% PAD, zero grads)', sparse_step_pad0)
+
+    def zipf_ids(vocab, n, s=1.1, seed=7):
+        p = 1.0 / torch.arange(1, vocab + 1, dtype=torch.float64) ** s
+        cdf = torch.cumsum(p / p.sum(), 0).to(dev)
+        g = torch.Generator(device=dev).manual_seed(seed)
+        u = torch.rand(n, device=dev, generator=g, dtype=torch.float64)
+        return torch.searchsorted(cdf, u).clamp_(0, vocab - 1)
+
+    for s in (1.0, 1.1):
+        ids_z = zipf_ids(1301137, 2 * N, s)
+        uz = torch.unique(ids_z).numel()
+        def sparse_step_zipf():
+            ext.adam_sparse_rows_hash(tok, ids_z, rows, mm, vv, 1, 1e-3, 0.9,
+                                      0.999, 1e-8, torch.empty(0),
+                                      torch.empty(0))
+        timeit('sparse adam (zipf s=%.1f, %dK uniq)' % (s, uz // 1000),
+               sparse_step_zipf)
+        timeit('dedup_sum rows (zipf s=%.1f)' % s,
+               lambda: ext.sparse_dedup_sum_rows(ids_z, rows))
+    timeit('dedup_sum rows (uniform ids)',
+           lambda: ext.sparse_dedup_sum_rows(ids, rows))
     timeit('torch.unique only', lambda: torch.unique(ids, return_inverse=True))
 
 
