@@ -63,6 +63,28 @@ def make_config(device):
     return cfg
 
 
+ZIPF_S = 1.1   # id-frequency skew of the synthetic data (see synth_batches)
+
+_zipf_cdf_cache = {}
+
+
+def zipf_ids(vocab_size, shape, g, s=ZIPF_S):
+    """Zipf(s)-distributed ids in [1, vocab_size): real java14m token/path
+    frequencies are Zipf-like (a handful of identifiers and path shapes
+    dominate), and the id distribution is performance-relevant on both sides:
+    hot rows stress the sparse-Adam accumulate (replica path) and determine
+    the post-dedup DP gather volume. Uniform ids would be an unrealistically
+    EASY case for the accumulate and an unrealistically HARD one for DP."""
+    key = (vocab_size, s)
+    if key not in _zipf_cdf_cache:
+        p = 1.0 / torch.arange(1, vocab_size, dtype=torch.float64) ** s
+        _zipf_cdf_cache[key] = torch.cumsum(p / p.sum(), 0)
+    cdf = _zipf_cdf_cache[key]
+    u = torch.rand(shape, generator=g, dtype=torch.float64)
+    return (torch.searchsorted(cdf, u) + 1).clamp_(1, vocab_size - 1) \
+        .to(torch.int32)
+
+
 def synth_batches(cfg, device, batch_size, n_batches=8, seed=0):
     """Pool of synthetic java14m-shaped batches, resident on device."""
     g = torch.Generator(device='cpu').manual_seed(seed)
@@ -72,9 +94,9 @@ def synth_batches(cfg, device, batch_size, n_batches=8, seed=0):
     V_tgt = cfg.MAX_TARGET_VOCAB_SIZE + 1
     C = cfg.MAX_CONTEXTS
     for _ in range(n_batches):
-        src = torch.randint(1, V_tok, (batch_size, C), generator=g, dtype=torch.int32)
-        pth = torch.randint(1, V_path, (batch_size, C), generator=g, dtype=torch.int32)
-        tgt = torch.randint(1, V_tok, (batch_size, C), generator=g, dtype=torch.int32)
+        src = zipf_ids(V_tok, (batch_size, C), g)
+        pth = zipf_ids(V_path, (batch_size, C), g)
+        tgt = zipf_ids(V_tok, (batch_size, C), g)
         # realistic context-count distribution: valid prefix of U[64, 200]
         lo = min(64, max(1, C // 2))
         n_valid = torch.randint(lo, C + 1, (batch_size,), generator=g)
@@ -211,6 +233,7 @@ def main():
                 'token_vocab': cfg.MAX_TOKEN_VOCAB_SIZE,
                 'path_vocab': cfg.MAX_PATH_VOCAB_SIZE,
                 'target_vocab': cfg.MAX_TARGET_VOCAB_SIZE,
+                'id_dist': 'zipf-%s' % ZIPF_S,
             },
         }
         print(json.dumps(out))

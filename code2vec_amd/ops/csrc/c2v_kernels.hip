@@ -1739,6 +1739,19 @@ __global__ void k_adam_dense(float* __restrict__ p, const G* __restrict__ g,
 //   1. claim: CAS the id into its probe slot
 //   2. compact: every occupied slot takes a compact index
 //   3. lookup: every input id re-probes and reads its compact index
+// Hot-id replica accumulation: ids whose occurrence count crosses HOT_T get
+// HOT_R replica accumulator rows; contributions spread across replicas by
+// wave id and a fold pass sums them back. Without this, a Zipf-frequent id
+// (top java14m token ≈ 9% of a batch) serializes ~39K same-address fp32
+// atomics per column — measured 2.2 ms/step at Zipf s=1.1 vs 0.46 ms on
+// uniform ids. With replicas the worst per-address chain drops HOT_R-fold.
+#define HOT_T 48       // occurrences before an id is treated as hot
+#define HOT_R 64       // replica rows per hot id
+#define HOT_CAP 2048   // max tracked hot ids (beyond: plain atomics)
+#define INV_SINGLE (1 << 30)
+#define INV_HOT (1 << 29)
+#define INV_IDX 0x1FFFFFFF
+
 template <typename I>
 __global__ void k_hash_claim(const I* __restrict__ ids, long n,
                              int* __restrict__ tbl_id, u32 mask_,
@@ -1754,8 +1767,8 @@ __global__ void k_hash_claim(const I* __restrict__ ids, long n,
     // with lane), so the shuffle is safe at the tail.
     const int prev = __shfl_up(id, 1, 64);
     const bool cont = (lane > 0 && prev == id);
-    // occurrence counts (for the accum single-store fast path): each run
-    // leader adds its run length at the id's slot
+    // occurrence counts: drive the accum single-store fast path (cnt==1)
+    // and hot-id detection (cnt>=HOT_T); each run leader adds its run length
     const unsigned long long cmask = __ballot(cont);
     if (cont) continue;                    // leader of each run claims
     int run_len = 1;
@@ -1775,17 +1788,12 @@ __global__ void k_hash_claim(const I* __restrict__ ids, long n,
       slot = (slot + 1) & mask_;
     }
     if (cnt) {
-      // saturating occurrence marker: only "exactly one" vs "more" matters
-      // downstream, so multi-element runs write 2 idempotently and
-      // single-element runs increment ONLY while cnt < 2 (read-before-add —
-      // an unconditional atomicAdd serialized 136K same-address adds when a
-      // hot id appeared as many short runs: 6.2 ms on the 33%-interleaved
-      // PAD micro-bench)
-      if (run_len > 1) {
-        if (cnt[slot] < 2) cnt[slot] = 2;
-      } else if (cnt[slot] < 2) {
-        atomicAdd(cnt + slot, 1);
-      }
+      // saturating count: adds stop once the hot threshold is reached
+      // (read-before-add — an UNCONDITIONAL atomicAdd serialized 136K
+      // same-address adds when a hot id appeared as many short runs:
+      // 6.2 ms on the 33%-interleaved PAD micro-bench); races may
+      // overshoot HOT_T, which only matters as ">= HOT_T".
+      if (cnt[slot] < HOT_T) atomicAdd(cnt + slot, run_len);
     }
   }
 }
@@ -1793,7 +1801,11 @@ __global__ void k_hash_claim(const I* __restrict__ ids, long n,
 __global__ void k_hash_compact(const int* __restrict__ tbl_id,
                                int* __restrict__ tbl_cidx,
                                long* __restrict__ uniq_out,
-                               int* __restrict__ n_uniq, u32 cap) {
+                               int* __restrict__ n_uniq, u32 cap,
+                               const int* __restrict__ occ_cnt = nullptr,
+                               int* __restrict__ tbl_hot = nullptr,
+                               int* __restrict__ hot2cidx = nullptr,
+                               int* __restrict__ n_hot = nullptr) {
   // block-aggregated counter: ballots per wave, one atomicAdd per BLOCK per
   // sweep (a per-wave atomic still serialized ~32K same-address adds/step)
   __shared__ int wave_cnt[4];
@@ -1820,6 +1832,15 @@ __global__ void k_hash_compact(const int* __restrict__ tbl_id,
       const int off = wave_base + __popcll(ball & ((1ull << lane) - 1ull));
       tbl_cidx[s] = off;
       uniq_out[off] = id;
+      if (tbl_hot) {
+        int h = -1;
+        if (occ_cnt[s] >= HOT_T) {
+          h = atomicAdd(n_hot, 1);
+          if (h < HOT_CAP) hot2cidx[h] = off;
+          else h = -1;  // replica table full: plain atomics fallback
+        }
+        tbl_hot[s] = h;
+      }
     }
     __syncthreads();
   }
@@ -1830,15 +1851,38 @@ __global__ void k_hash_lookup(const I* __restrict__ ids, long n,
                               const int* __restrict__ tbl_id,
                               const int* __restrict__ tbl_cidx,
                               int* __restrict__ inverse_out, u32 mask_,
-                              const int* __restrict__ cnt = nullptr) {
+                              const int* __restrict__ cnt = nullptr,
+                              const int* __restrict__ tbl_hot = nullptr) {
   for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (long)gridDim.x * blockDim.x) {
     const int id = (int)ids[i];
     u32 slot = ((u32)id * 2654435761u) & mask_;
     while (tbl_id[slot] != id) slot = (slot + 1) & mask_;
-    // bit 30 marks single-occurrence ids: their accum write needs no atomic
-    inverse_out[i] = tbl_cidx[slot] |
-                     ((cnt && cnt[slot] == 1) ? (1 << 30) : 0);
+    // bit 30 marks single-occurrence ids (accum stores without atomics);
+    // bit 29 marks hot ids (accum spreads over HOT_R replica rows, payload
+    // is the hot index)
+    int inv = tbl_cidx[slot];
+    if (cnt && cnt[slot] == 1) inv |= INV_SINGLE;
+    else if (tbl_hot && tbl_hot[slot] >= 0) inv = tbl_hot[slot] | INV_HOT;
+    inverse_out[i] = inv;
+  }
+}
+
+// fold the HOT_R replica rows of each hot id back into its compact acc row
+__global__ void k_hot_fold(const float* __restrict__ hot_acc,
+                           const int* __restrict__ hot2cidx,
+                           const int* __restrict__ n_hot_ptr,
+                           float* __restrict__ acc, int d) {
+  const int n_hot = min(*n_hot_ptr, HOT_CAP);
+  for (long s = blockIdx.x * blockDim.x + threadIdx.x; s < (long)n_hot * d;
+       s += (long)gridDim.x * blockDim.x) {
+    const int h = (int)(s / d);
+    const int col = (int)(s % d);
+    float sum = 0.f;
+    const float* base = hot_acc + ((long)h * HOT_R) * d + col;
+#pragma unroll 4
+    for (int r = 0; r < HOT_R; ++r) sum += base[(long)r * d];
+    acc[(long)hot2cidx[h] * d + col] += sum;
   }
 }
 
@@ -1889,8 +1933,10 @@ __global__ void k_adam_rows_dyn(float* __restrict__ p,
 __global__ void k_rows_accum_ctx(const ushort* __restrict__ d_ctx, int ld,
                                  int off0, int off1, long n_per_seg,
                                  int n_seg, const int* __restrict__ inverse,
-                                 float* __restrict__ acc, int d) {
+                                 float* __restrict__ acc, int d,
+                                 float* __restrict__ hot_acc = nullptr) {
   const long total = n_per_seg * n_seg * (long)d;
+  const int wave = (int)((blockIdx.x * blockDim.x + threadIdx.x) >> 6);
   for (long s = blockIdx.x * blockDim.x + threadIdx.x; s < total;
        s += (long)gridDim.x * blockDim.x) {
     const long r = s / d;
@@ -1899,12 +1945,17 @@ __global__ void k_rows_accum_ctx(const ushort* __restrict__ d_ctx, int ld,
     const int src_off = (r < n_per_seg) ? off0 : off1;
     const float gv = bf2f(d_ctx[src_row * ld + src_off + col]);
     const int inv = inverse[r];
-    if (inv & (1 << 30)) {
+    if (inv & INV_SINGLE) {
       // single-occurrence id (~86% of unique rows on uniform ids): plain
       // store — global atomicAdd throughput was the kernel's bound
-      acc[(long)(inv & 0x3FFFFFFF) * d + col] = gv;
-    } else if (gv != 0.f) {
-      atomicAdd(acc + (long)inv * d + col, gv);
+      acc[(long)(inv & INV_IDX) * d + col] = gv;
+    } else if (gv == 0.f) {
+      // masked-context grads are exactly zero: skip the atomic
+    } else if (hot_acc && (inv & INV_HOT)) {
+      const long h = inv & INV_IDX;
+      atomicAdd(hot_acc + (h * HOT_R + (wave & (HOT_R - 1))) * d + col, gv);
+    } else {
+      atomicAdd(acc + (long)(inv & INV_IDX) * d + col, gv);
     }
   }
 }
@@ -1912,7 +1963,9 @@ __global__ void k_rows_accum_ctx(const ushort* __restrict__ d_ctx, int ld,
 template <typename G>
 __global__ void k_rows_accum(const G* __restrict__ rows,
                              const int* __restrict__ inverse,
-                             float* __restrict__ acc, long n_rows, int d) {
+                             float* __restrict__ acc, long n_rows, int d,
+                             float* __restrict__ hot_acc = nullptr) {
+  const int wave = (int)((blockIdx.x * blockDim.x + threadIdx.x) >> 6);
   for (long s = blockIdx.x * blockDim.x + threadIdx.x; s < n_rows * d;
        s += (long)gridDim.x * blockDim.x) {
     const long r = s / d;
@@ -1921,10 +1974,14 @@ __global__ void k_rows_accum(const G* __restrict__ rows,
     if constexpr (sizeof(G) == 2) gv = bf2f(((const ushort*)rows)[s]);
     else gv = ((const float*)rows)[s];
     const int inv = inverse[r];
-    if (inv & (1 << 30)) {
-      acc[(long)(inv & 0x3FFFFFFF) * d + col] = gv;
-    } else if (gv != 0.f) {
-      atomicAdd(acc + (long)inv * d + col, gv);
+    if (inv & INV_SINGLE) {
+      acc[(long)(inv & INV_IDX) * d + col] = gv;
+    } else if (gv == 0.f) {
+    } else if (hot_acc && (inv & INV_HOT)) {
+      const long h = inv & INV_IDX;
+      atomicAdd(hot_acc + (h * HOT_R + (wave & (HOT_R - 1))) * d + col, gv);
+    } else {
+      atomicAdd(acc + (long)(inv & INV_IDX) * d + col, gv);
     }
   }
 }
@@ -2590,6 +2647,120 @@ void adam_dense_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
         (float)eps, lrt_ptr);
 }
 
+// ---------------------------------------------------------------------------
+// Rank-local sparse-gradient dedup+sum for data-parallel training: the same
+// hash claim/compact/lookup/accumulate pipeline the sparse Adam uses, stopped
+// before the Adam update so the (unique ids, summed rows) pairs can go on the
+// wire instead of every raw contribution. On Zipf-distributed real-data ids
+// this cuts the per-rank all-gather volume by 3-4x (SURVEY §2.4 sparse
+// strategy; ROADMAP-R2 "Scaling" item b). Returns {uniq_ids int64[n],
+// acc fp32[n,d], n_uniq int32[1] device} — callers slice by the count.
+// ---------------------------------------------------------------------------
+
+struct DedupState {
+  torch::Tensor uniq, inverse, n_uniq;
+  torch::Tensor hot2cidx, n_hot;
+};
+
+static DedupState hash_dedup_ids(const torch::Tensor& ids_c) {
+  const long n = ids_c.numel();
+  long want = 2 * n;
+  u32 cap = 1;
+  while (cap < (u32)want) cap <<= 1;
+  auto opts_i32 = torch::TensorOptions().device(ids_c.device())
+                      .dtype(torch::kInt32);
+  auto tbl_id = torch::full({(long)cap}, -1, opts_i32);
+  auto tbl_cidx = torch::full({(long)cap}, -1, opts_i32);
+  auto tbl_cnt = torch::zeros({(long)cap}, opts_i32);
+  auto tbl_hot = torch::empty({(long)cap}, opts_i32);
+  auto hot2cidx = torch::empty({HOT_CAP}, opts_i32);
+  auto n_hot = torch::zeros({1}, opts_i32);
+  auto uniq = torch::empty({n}, opts_i32.dtype(torch::kInt64));
+  auto n_uniq = torch::zeros({1}, opts_i32);
+  auto inverse = torch::empty({n}, opts_i32);
+  const u32 mask_ = cap - 1;
+  if (ids_c.scalar_type() == torch::kInt32) {
+    k_hash_claim<int><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
+        ids_c.data_ptr<int>(), n, tbl_id.data_ptr<int>(), mask_,
+        tbl_cnt.data_ptr<int>());
+    k_hash_compact<<<grid_1d(cap, 256), 256, 0, cur_stream()>>>(
+        tbl_id.data_ptr<int>(), tbl_cidx.data_ptr<int>(),
+        uniq.data_ptr<long>(), n_uniq.data_ptr<int>(), cap,
+        tbl_cnt.data_ptr<int>(), tbl_hot.data_ptr<int>(),
+        hot2cidx.data_ptr<int>(), n_hot.data_ptr<int>());
+    k_hash_lookup<int><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
+        ids_c.data_ptr<int>(), n, tbl_id.data_ptr<int>(),
+        tbl_cidx.data_ptr<int>(), inverse.data_ptr<int>(), mask_,
+        tbl_cnt.data_ptr<int>(), tbl_hot.data_ptr<int>());
+  } else {
+    k_hash_claim<long><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
+        ids_c.data_ptr<long>(), n, tbl_id.data_ptr<int>(), mask_,
+        tbl_cnt.data_ptr<int>());
+    k_hash_compact<<<grid_1d(cap, 256), 256, 0, cur_stream()>>>(
+        tbl_id.data_ptr<int>(), tbl_cidx.data_ptr<int>(),
+        uniq.data_ptr<long>(), n_uniq.data_ptr<int>(), cap,
+        tbl_cnt.data_ptr<int>(), tbl_hot.data_ptr<int>(),
+        hot2cidx.data_ptr<int>(), n_hot.data_ptr<int>());
+    k_hash_lookup<long><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
+        ids_c.data_ptr<long>(), n, tbl_id.data_ptr<int>(),
+        tbl_cidx.data_ptr<int>(), inverse.data_ptr<int>(), mask_,
+        tbl_cnt.data_ptr<int>(), tbl_hot.data_ptr<int>());
+  }
+  return {uniq, inverse, n_uniq, hot2cidx, n_hot};
+}
+
+// zero acc prefix, accumulate rows (with hot-id replica spreading), fold
+static torch::Tensor accum_rows_common(const DedupState& st,
+                                       const torch::Tensor& rows_c, long n,
+                                       int d) {
+  auto acc = torch::empty({n, (long)d},
+                          rows_c.options().dtype(torch::kFloat32));
+  auto hot_acc = torch::empty({(long)HOT_CAP * HOT_R, (long)d},
+                              rows_c.options().dtype(torch::kFloat32));
+  k_zero_rows_dyn<<<grid_1d((long)HOT_CAP * HOT_R * d / 4, 256), 256, 0,
+                    cur_stream()>>>(
+      hot_acc.data_ptr<float>(), st.n_hot.data_ptr<int>(), HOT_R * d);
+  k_zero_rows_dyn<<<grid_1d(n * d / 4, 256), 256, 0, cur_stream()>>>(
+      acc.data_ptr<float>(), st.n_uniq.data_ptr<int>(), d);
+  if (rows_c.scalar_type() == torch::kBFloat16)
+    k_rows_accum<ushort><<<grid_1d(n * d, 256), 256, 0, cur_stream()>>>(
+        bf_ptr(rows_c), st.inverse.data_ptr<int>(), acc.data_ptr<float>(),
+        n, d, hot_acc.data_ptr<float>());
+  else
+    k_rows_accum<float><<<grid_1d(n * d, 256), 256, 0, cur_stream()>>>(
+        rows_c.data_ptr<float>(), st.inverse.data_ptr<int>(),
+        acc.data_ptr<float>(), n, d, hot_acc.data_ptr<float>());
+  k_hot_fold<<<grid_1d((long)HOT_CAP * d, 256), 256, 0, cur_stream()>>>(
+      hot_acc.data_ptr<float>(), st.hot2cidx.data_ptr<int>(),
+      st.n_hot.data_ptr<int>(), acc.data_ptr<float>(), d);
+  return acc;
+}
+
+static torch::Tensor accum_ctx_common(const DedupState& st,
+                                      const torch::Tensor& d_ctx, int off0,
+                                      int off1, long n_per_seg, int n_seg,
+                                      int d) {
+  const long n = n_per_seg * n_seg;
+  const int ld = (int)d_ctx.size(1);
+  auto acc = torch::empty({n, (long)d},
+                          d_ctx.options().dtype(torch::kFloat32));
+  auto hot_acc = torch::empty({(long)HOT_CAP * HOT_R, (long)d},
+                              d_ctx.options().dtype(torch::kFloat32));
+  k_zero_rows_dyn<<<grid_1d((long)HOT_CAP * HOT_R * d / 4, 256), 256, 0,
+                    cur_stream()>>>(
+      hot_acc.data_ptr<float>(), st.n_hot.data_ptr<int>(), HOT_R * d);
+  k_zero_rows_dyn<<<grid_1d(n * d / 4, 256), 256, 0, cur_stream()>>>(
+      acc.data_ptr<float>(), st.n_uniq.data_ptr<int>(), d);
+  k_rows_accum_ctx<<<grid_1d(n * d, 256), 256, 0, cur_stream()>>>(
+      bf_ptr(d_ctx), ld, off0, off1, n_per_seg, n_seg,
+      st.inverse.data_ptr<int>(), acc.data_ptr<float>(), d,
+      hot_acc.data_ptr<float>());
+  k_hot_fold<<<grid_1d((long)HOT_CAP * d, 256), 256, 0, cur_stream()>>>(
+      hot_acc.data_ptr<float>(), st.hot2cidx.data_ptr<int>(),
+      st.n_hot.data_ptr<int>(), acc.data_ptr<float>(), d);
+  return acc;
+}
+
 void adam_sparse_rows_step(torch::Tensor p, torch::Tensor uniq_ids,
                            torch::Tensor inverse, torch::Tensor grad_rows,
                            torch::Tensor m, torch::Tensor v, int64_t step,
@@ -2634,67 +2805,18 @@ void adam_sparse_rows_hash(torch::Tensor p, torch::Tensor ids,
   auto rows_c = grad_rows.contiguous();
   const long n = ids_c.numel();
   const int d = rows_c.size(1);
-  const long V = p.size(0);
-  // table capacity: power of two >= 2 * min(n, V)
-  long want = 2 * std::min(n, V);
-  u32 cap = 1;
-  while (cap < (u32)want) cap <<= 1;
-  auto opts_i32 = p.options().dtype(torch::kInt32);
-  auto tbl_id = torch::full({(long)cap}, -1, opts_i32);
-  auto tbl_cidx = torch::full({(long)cap}, -1, opts_i32);
-  auto tbl_cnt = torch::zeros({(long)cap}, opts_i32);
-  auto uniq = torch::empty({n}, p.options().dtype(torch::kInt64));
-  auto n_uniq = torch::zeros({1}, opts_i32);
-  auto inverse = torch::empty({n}, opts_i32);
-  // acc rows beyond the (device-side) unique count are never read; zero only
-  // the live prefix after compact instead of a full torch::zeros fill
-  auto acc = torch::empty({n, (long)d}, p.options().dtype(torch::kFloat32));
-
-  const u32 mask_ = cap - 1;
-  if (ids_c.scalar_type() == torch::kInt32) {
-    k_hash_claim<int><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
-        ids_c.data_ptr<int>(), n, tbl_id.data_ptr<int>(), mask_,
-        tbl_cnt.data_ptr<int>());
-    k_hash_compact<<<grid_1d(cap, 256), 256, 0, cur_stream()>>>(
-        tbl_id.data_ptr<int>(), tbl_cidx.data_ptr<int>(),
-        uniq.data_ptr<long>(), n_uniq.data_ptr<int>(), cap);
-    k_hash_lookup<int><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
-        ids_c.data_ptr<int>(), n, tbl_id.data_ptr<int>(),
-        tbl_cidx.data_ptr<int>(), inverse.data_ptr<int>(), mask_,
-        tbl_cnt.data_ptr<int>());
-  } else {
-    k_hash_claim<long><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
-        ids_c.data_ptr<long>(), n, tbl_id.data_ptr<int>(), mask_,
-        tbl_cnt.data_ptr<int>());
-    k_hash_compact<<<grid_1d(cap, 256), 256, 0, cur_stream()>>>(
-        tbl_id.data_ptr<int>(), tbl_cidx.data_ptr<int>(),
-        uniq.data_ptr<long>(), n_uniq.data_ptr<int>(), cap);
-    k_hash_lookup<long><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
-        ids_c.data_ptr<long>(), n, tbl_id.data_ptr<int>(),
-        tbl_cidx.data_ptr<int>(), inverse.data_ptr<int>(), mask_,
-        tbl_cnt.data_ptr<int>());
-  }
-
-  k_zero_rows_dyn<<<grid_1d(n * d / 4, 256), 256, 0, cur_stream()>>>(
-      acc.data_ptr<float>(), n_uniq.data_ptr<int>(), d);
-  if (rows_c.scalar_type() == torch::kBFloat16)
-    k_rows_accum<ushort><<<grid_1d(n * d, 256), 256, 0, cur_stream()>>>(
-        bf_ptr(rows_c), inverse.data_ptr<int>(), acc.data_ptr<float>(), n, d);
-  else
-    k_rows_accum<float><<<grid_1d(n * d, 256), 256, 0, cur_stream()>>>(
-        rows_c.data_ptr<float>(), inverse.data_ptr<int>(), acc.data_ptr<float>(),
-        n, d);
-
+  auto st = hash_dedup_ids(ids_c);
+  auto acc = accum_rows_common(st, rows_c, n, d);
   const float lr_t = (float)(lr * std::sqrt(1.0 - std::pow(beta2, (double)step)) /
                              (1.0 - std::pow(beta1, (double)step)));
   ushort* shadow_ptr = nullptr;
   if (shadow.defined() && shadow.numel() == p.numel())
     shadow_ptr = reinterpret_cast<ushort*>(shadow.data_ptr<at::BFloat16>());
   k_adam_rows_dyn<<<grid_1d(n * d, 256), 256, 0, cur_stream()>>>(
-      p.data_ptr<float>(), uniq.data_ptr<long>(), acc.data_ptr<float>(),
+      p.data_ptr<float>(), st.uniq.data_ptr<long>(), acc.data_ptr<float>(),
       m.data_ptr<float>(), v.data_ptr<float>(), shadow_ptr,
-      n_uniq.data_ptr<int>(), d, lr_t, (float)beta1, (float)beta2, (float)eps,
-      lrt_ptr);
+      st.n_uniq.data_ptr<int>(), d, lr_t, (float)beta1, (float)beta2,
+      (float)eps, lrt_ptr);
 }
 
 void adam_sparse_rows_hash_ctx(torch::Tensor p, torch::Tensor ids,
@@ -2710,97 +2832,17 @@ void adam_sparse_rows_hash_ctx(torch::Tensor p, torch::Tensor ids,
   const long n = ids_c.numel();
   const long n_per_seg = d_ctx.size(0);
   TORCH_CHECK(n == n_per_seg * n_seg, "ids length mismatch");
-  const int ld = (int)d_ctx.size(1);
-  const long V = p.size(0);
-  long want = 2 * std::min(n, V);
-  u32 cap = 1;
-  while (cap < (u32)want) cap <<= 1;
-  auto opts_i32 = p.options().dtype(torch::kInt32);
-  auto tbl_id = torch::full({(long)cap}, -1, opts_i32);
-  auto tbl_cidx = torch::full({(long)cap}, -1, opts_i32);
-  auto tbl_cnt = torch::zeros({(long)cap}, opts_i32);
-  auto uniq = torch::empty({n}, p.options().dtype(torch::kInt64));
-  auto n_uniq = torch::zeros({1}, opts_i32);
-  auto inverse = torch::empty({n}, opts_i32);
-  auto acc = torch::empty({n, d}, p.options().dtype(torch::kFloat32));
-  const u32 mask_ = cap - 1;
   TORCH_CHECK(ids_c.scalar_type() == torch::kInt32, "ctx path expects int32 ids");
-  k_hash_claim<int><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
-      ids_c.data_ptr<int>(), n, tbl_id.data_ptr<int>(), mask_,
-      tbl_cnt.data_ptr<int>());
-  k_hash_compact<<<grid_1d(cap, 256), 256, 0, cur_stream()>>>(
-      tbl_id.data_ptr<int>(), tbl_cidx.data_ptr<int>(), uniq.data_ptr<long>(),
-      n_uniq.data_ptr<int>(), cap);
-  k_hash_lookup<int><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
-      ids_c.data_ptr<int>(), n, tbl_id.data_ptr<int>(),
-      tbl_cidx.data_ptr<int>(), inverse.data_ptr<int>(), mask_,
-      tbl_cnt.data_ptr<int>());
-  k_zero_rows_dyn<<<grid_1d(n * d / 4, 256), 256, 0, cur_stream()>>>(
-      acc.data_ptr<float>(), n_uniq.data_ptr<int>(), (int)d);
-  k_rows_accum_ctx<<<grid_1d(n * d, 256), 256, 0, cur_stream()>>>(
-      bf_ptr(d_ctx), ld, (int)off0, (int)off1, n_per_seg, (int)n_seg,
-      inverse.data_ptr<int>(), acc.data_ptr<float>(), (int)d);
+  auto st = hash_dedup_ids(ids_c);
+  auto acc = accum_ctx_common(st, d_ctx, (int)off0, (int)off1, n_per_seg,
+                              (int)n_seg, (int)d);
   const float lr_t = (float)(lr * std::sqrt(1.0 - std::pow(beta2, (double)step)) /
                              (1.0 - std::pow(beta1, (double)step)));
   k_adam_rows_dyn<<<grid_1d(n * d, 256), 256, 0, cur_stream()>>>(
-      p.data_ptr<float>(), uniq.data_ptr<long>(), acc.data_ptr<float>(),
+      p.data_ptr<float>(), st.uniq.data_ptr<long>(), acc.data_ptr<float>(),
       m.data_ptr<float>(), v.data_ptr<float>(), nullptr,
-      n_uniq.data_ptr<int>(), (int)d, lr_t, (float)beta1, (float)beta2,
+      st.n_uniq.data_ptr<int>(), (int)d, lr_t, (float)beta1, (float)beta2,
       (float)eps, lrt_ptr);
-}
-
-// ---------------------------------------------------------------------------
-// Rank-local sparse-gradient dedup+sum for data-parallel training: the same
-// hash claim/compact/lookup/accumulate pipeline the sparse Adam uses, stopped
-// before the Adam update so the (unique ids, summed rows) pairs can go on the
-// wire instead of every raw contribution. On Zipf-distributed real-data ids
-// this cuts the per-rank all-gather volume by 3-4x (SURVEY §2.4 sparse
-// strategy; ROADMAP-R2 "Scaling" item b). Returns {uniq_ids int64[n],
-// acc fp32[n,d], n_uniq int32[1] device} — callers slice by the count.
-// ---------------------------------------------------------------------------
-
-struct DedupState {
-  torch::Tensor uniq, inverse, n_uniq;
-};
-
-static DedupState hash_dedup_ids(const torch::Tensor& ids_c) {
-  const long n = ids_c.numel();
-  long want = 2 * n;
-  u32 cap = 1;
-  while (cap < (u32)want) cap <<= 1;
-  auto opts_i32 = torch::TensorOptions().device(ids_c.device())
-                      .dtype(torch::kInt32);
-  auto tbl_id = torch::full({(long)cap}, -1, opts_i32);
-  auto tbl_cidx = torch::full({(long)cap}, -1, opts_i32);
-  auto tbl_cnt = torch::zeros({(long)cap}, opts_i32);
-  auto uniq = torch::empty({n}, opts_i32.dtype(torch::kInt64));
-  auto n_uniq = torch::zeros({1}, opts_i32);
-  auto inverse = torch::empty({n}, opts_i32);
-  const u32 mask_ = cap - 1;
-  if (ids_c.scalar_type() == torch::kInt32) {
-    k_hash_claim<int><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
-        ids_c.data_ptr<int>(), n, tbl_id.data_ptr<int>(), mask_,
-        tbl_cnt.data_ptr<int>());
-    k_hash_compact<<<grid_1d(cap, 256), 256, 0, cur_stream()>>>(
-        tbl_id.data_ptr<int>(), tbl_cidx.data_ptr<int>(),
-        uniq.data_ptr<long>(), n_uniq.data_ptr<int>(), cap);
-    k_hash_lookup<int><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
-        ids_c.data_ptr<int>(), n, tbl_id.data_ptr<int>(),
-        tbl_cidx.data_ptr<int>(), inverse.data_ptr<int>(), mask_,
-        tbl_cnt.data_ptr<int>());
-  } else {
-    k_hash_claim<long><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
-        ids_c.data_ptr<long>(), n, tbl_id.data_ptr<int>(), mask_,
-        tbl_cnt.data_ptr<int>());
-    k_hash_compact<<<grid_1d(cap, 256), 256, 0, cur_stream()>>>(
-        tbl_id.data_ptr<int>(), tbl_cidx.data_ptr<int>(),
-        uniq.data_ptr<long>(), n_uniq.data_ptr<int>(), cap);
-    k_hash_lookup<long><<<grid_1d(n, 256), 256, 0, cur_stream()>>>(
-        ids_c.data_ptr<long>(), n, tbl_id.data_ptr<int>(),
-        tbl_cidx.data_ptr<int>(), inverse.data_ptr<int>(), mask_,
-        tbl_cnt.data_ptr<int>());
-  }
-  return {uniq, inverse, n_uniq};
 }
 
 std::vector<torch::Tensor> sparse_dedup_sum_ctx(torch::Tensor ids,
@@ -2812,14 +2854,9 @@ std::vector<torch::Tensor> sparse_dedup_sum_ctx(torch::Tensor ids,
   const long n = ids_c.numel();
   const long n_per_seg = d_ctx.size(0);
   TORCH_CHECK(n == n_per_seg * n_seg, "ids length mismatch");
-  const int ld = (int)d_ctx.size(1);
   auto st = hash_dedup_ids(ids_c);
-  auto acc = torch::empty({n, d}, d_ctx.options().dtype(torch::kFloat32));
-  k_zero_rows_dyn<<<grid_1d(n * d / 4, 256), 256, 0, cur_stream()>>>(
-      acc.data_ptr<float>(), st.n_uniq.data_ptr<int>(), (int)d);
-  k_rows_accum_ctx<<<grid_1d(n * d, 256), 256, 0, cur_stream()>>>(
-      bf_ptr(d_ctx), ld, (int)off0, (int)off1, n_per_seg, (int)n_seg,
-      st.inverse.data_ptr<int>(), acc.data_ptr<float>(), (int)d);
+  auto acc = accum_ctx_common(st, d_ctx, (int)off0, (int)off1, n_per_seg,
+                              (int)n_seg, (int)d);
   return {st.uniq, acc, st.n_uniq};
 }
 
@@ -2832,16 +2869,7 @@ std::vector<torch::Tensor> sparse_dedup_sum_rows(torch::Tensor ids,
   const int d = rows_c.size(1);
   TORCH_CHECK(rows_c.size(0) == n, "row/id count mismatch");
   auto st = hash_dedup_ids(ids_c);
-  auto acc = torch::empty({n, (long)d}, rows_c.options().dtype(torch::kFloat32));
-  k_zero_rows_dyn<<<grid_1d(n * d / 4, 256), 256, 0, cur_stream()>>>(
-      acc.data_ptr<float>(), st.n_uniq.data_ptr<int>(), d);
-  if (rows_c.scalar_type() == torch::kBFloat16)
-    k_rows_accum<ushort><<<grid_1d(n * d, 256), 256, 0, cur_stream()>>>(
-        bf_ptr(rows_c), st.inverse.data_ptr<int>(), acc.data_ptr<float>(), n, d);
-  else
-    k_rows_accum<float><<<grid_1d(n * d, 256), 256, 0, cur_stream()>>>(
-        rows_c.data_ptr<float>(), st.inverse.data_ptr<int>(),
-        acc.data_ptr<float>(), n, d);
+  auto acc = accum_rows_common(st, rows_c, n, d);
   return {st.uniq, acc, st.n_uniq};
 }
 
