@@ -993,8 +993,8 @@ __global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
 // (dW = ctx^T @ d_z: K = B*C ~ 205K, out 384x384) — blocks are
 // (row-tile, k-chunk) pairs like k_gemm_nn_splitk, accumulate fp32
 // partials P[(chunk, V, M)] folded by k_splitk_reduce. C is unused then.
-template <bool CEB = false, bool SPLITK = false>
-__launch_bounds__(512, 1)
+template <bool CEB = false, bool SPLITK = false, int BV = GTN_BV>
+__launch_bounds__(4 * BV, 1)
 __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
                           const ushort* __restrict__ B,  // code (K, M)
                           ushort* __restrict__ C,        // out (V, M)
@@ -1005,10 +1005,15 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
                           float* __restrict__ P = nullptr,
                           int S = 1, int ksteps_per_chunk = 0,
                           int row_tiles = 1) {
-  // double-buffered: At[2][32][184] + B[2][32][440] bf16 = 78 KiB dynamic
+  // double-buffered: At[2][32][BV+56] + B[2][32][440] bf16 (78 KiB at
+  // BV=128 with one block/CU; 66 KiB at BV=64 with TWO blocks/CU, whose
+  // independent barriers overlap each other's load and compute phases)
   extern __shared__ ushort ldstn[];
-#define TLDS_A(b) (ldstn + (b) * (GNN_BK * GTN_PKT))
-#define TLDS_B(b) (ldstn + 2 * GNN_BK * GTN_PKT + (b) * (GNN_BK * GNN_PKB))
+  constexpr int PKT = BV + 56;          // padded+rotated A-tile stride
+  constexpr int TNT = 4 * BV;           // block threads (8 or 4 waves)
+  constexpr int NB = 1536 / TNT;        // B-tile b128 chunks per thread
+#define TLDS_A(b) (ldstn + (b) * (GNN_BK * PKT))
+#define TLDS_B(b) (ldstn + 2 * GNN_BK * PKT + (b) * (GNN_BK * GNN_PKB))
 
   int v0, chunk = 0, ks0 = 0, ks1;
   const int total_ksteps = (K + GNN_BK - 1) / GNN_BK;
@@ -1017,11 +1022,11 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
     const int xcd = blockIdx.x & 7;
     const int slot = blockIdx.x >> 3;
     chunk = (slot / row_tiles) * 8 + xcd;
-    v0 = (slot % row_tiles) * GTN_BV;
+    v0 = (slot % row_tiles) * BV;
     ks0 = chunk * ksteps_per_chunk;
     ks1 = min(total_ksteps, ks0 + ksteps_per_chunk);
   } else {
-    v0 = blockIdx.x * GTN_BV;
+    v0 = blockIdx.x * BV;
     ks1 = total_ksteps;
   }
   const int ks1_main = min(ks1, K / GNN_BK);
@@ -1041,19 +1046,19 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
 #pragma unroll
     for (int j = 0; j < 6; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  // A tile: 32 k-rows x 128 v-cols = 512 b128 chunks -> 1/thread
-  const int at_kr[1] = {tid >> 4};
-  const int at_c8 = (tid & 15) * 8;
-  // B tile: 32 k-rows x 384 cols = 1536 b128 chunks -> 3/thread
-  int b_kr[3], b_c8[3];
+  // A tile: 32 k-rows x BV v-cols -> 1 b128 chunk/thread
+  const int at_kr[1] = {tid / (BV / 8)};
+  const int at_c8 = (tid % (BV / 8)) * 8;
+  // B tile: 32 k-rows x 384 cols = 1536 b128 chunks -> NB/thread
+  int b_kr[NB], b_c8[NB];
 #pragma unroll
-  for (int r = 0; r < 3; ++r) {
-    const int idx = tid + r * 512;
+  for (int r = 0; r < NB; ++r) {
+    const int idx = tid + r * TNT;
     b_kr[r] = idx / 48;
     b_c8[r] = (idx % 48) * 8;
   }
 
-  u16x8 ra[1], rb[3];
+  u16x8 ra[1], rb[NB];
   auto ce_map = [&](u16x8 v, int gk, int vbase) -> u16x8 {
     if (!CEB) return v;
     const float l = lse[gk];
@@ -1086,7 +1091,7 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
       }
     }
 #pragma unroll
-    for (int r = 0; r < 3; ++r) {
+    for (int r = 0; r < NB; ++r) {
       if (b_c8[r] + 8 <= M) {
         rb[r] = *reinterpret_cast<const u16x8*>(
             B + (long)(k0 + b_kr[r]) * M + b_c8[r]);
@@ -1101,12 +1106,12 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
   auto store_ab = [&](int buf) {
 #pragma unroll
     for (int r = 0; r < 1; ++r) {
-      ushort* p = TLDS_A(buf) + at_kr[r] * GTN_PKT + GNN_BROT(at_kr[r]) +
+      ushort* p = TLDS_A(buf) + at_kr[r] * PKT + GNN_BROT(at_kr[r]) +
                   at_c8;
       *reinterpret_cast<u16x8*>(p) = ra[r];  // aligned b128 (368 B stride)
     }
 #pragma unroll
-    for (int r = 0; r < 3; ++r) {
+    for (int r = 0; r < NB; ++r) {
       ushort* p = TLDS_B(buf) + b_kr[r] * GNN_PKB + GNN_BROT(b_kr[r]) +
                   b_c8[r];
       *reinterpret_cast<u16x8*>(p) = rb[r];
@@ -1128,7 +1133,7 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
 #pragma unroll
     for (int m = 0; m < 4; ++m) {
       const int row = wrow + m * 16 + (lane & 15);  // v-col inside the tile
-      const ushort* ap = TLDS_A(buf) + (lane >> 4) * (8 * GTN_PKT + 16) + row;
+      const ushort* ap = TLDS_A(buf) + (lane >> 4) * (8 * PKT + 16) + row;
       u16x8 t;
 #pragma unroll
       for (int j = 0; j < 8; ++j) t[j] = ap[j * GTN_PKT];
@@ -1176,10 +1181,10 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
       }
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        TLDS_A(0)[at_kr[r] * GTN_PKT + GNN_BROT(at_kr[r]) + at_c8 + j] = v[j];
+        TLDS_A(0)[at_kr[r] * PKT + GNN_BROT(at_kr[r]) + at_c8 + j] = v[j];
     }
 #pragma unroll
-    for (int r = 0; r < 3; ++r) {
+    for (int r = 0; r < NB; ++r) {
       const int gk = k0 + b_kr[r];
 #pragma unroll
       for (int j = 0; j < 8; ++j)
@@ -1726,6 +1731,98 @@ __global__ void k_adam_dense(float* __restrict__ p, const G* __restrict__ g,
     const float pv = p[i] - lr_t * mi / (sqrtf(vi) + eps);
     p[i] = pv;
     if (shadow != nullptr) shadow[i] = f2bf(pv);
+  }
+}
+
+// Width-2 dense Adam: each thread owns two lane-contiguous float4 chunks per
+// iteration with all 8 loads issued before any compute — doubles the
+// outstanding-miss count per wave to cover the p/m/v read->write turnaround
+// that keeps the width-1 kernel under the streaming floor.
+template <typename G, bool NT = false>
+__global__ void k_adam_dense_w2(float* __restrict__ p, const G* __restrict__ g,
+                                float* __restrict__ m, float* __restrict__ v,
+                                ushort* __restrict__ shadow, long n, float lr_t,
+                                float b1, float b2, float eps,
+                                const float* __restrict__ lrt_ptr) {
+  if (lrt_ptr) lr_t = *lrt_ptr;
+  const long n4 = n / 4;
+  using f32x4v = __attribute__((ext_vector_type(4))) float;
+  for (long base = (long)blockIdx.x * blockDim.x * 2; base < n4;
+       base += (long)gridDim.x * blockDim.x * 2) {
+    long i4s[2];
+    float4 pv[2], mv[2], vv[2];
+    float gv[2][4];
+    bool live[2];
+#pragma unroll
+    for (int w = 0; w < 2; ++w) {
+      const long i4 = base + w * blockDim.x + threadIdx.x;
+      i4s[w] = i4;
+      live[w] = i4 < n4;
+      if (!live[w]) continue;
+      const long i = i4 * 4;
+      pv[w] = *reinterpret_cast<float4*>(p + i);
+      mv[w] = *reinterpret_cast<float4*>(m + i);
+      vv[w] = *reinterpret_cast<float4*>(v + i);
+      if constexpr (sizeof(G) == 2) {
+        ulonglong1 packed =
+            *reinterpret_cast<const ulonglong1*>((const ushort*)g + i);
+        const ushort* u = reinterpret_cast<const ushort*>(&packed);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) gv[w][j] = bf2f(u[j]);
+      } else {
+        float4 gf = *reinterpret_cast<const float4*>((const float*)g + i);
+        gv[w][0] = gf.x; gv[w][1] = gf.y; gv[w][2] = gf.z; gv[w][3] = gf.w;
+      }
+    }
+#pragma unroll
+    for (int w = 0; w < 2; ++w) {
+      if (!live[w]) continue;
+      const long i = i4s[w] * 4;
+      float* pp = &pv[w].x; float* mp = &mv[w].x; float* vp = &vv[w].x;
+      ushort sh[4];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const float mi = b1 * mp[j] + (1.f - b1) * gv[w][j];
+        const float vi = b2 * vp[j] + (1.f - b2) * gv[w][j] * gv[w][j];
+        mp[j] = mi; vp[j] = vi;
+        const float pn = pp[j] - lr_t * mi / (sqrtf(vi) + eps);
+        pp[j] = pn;
+        sh[j] = f2bf(pn);
+      }
+      if (NT) {
+        __builtin_nontemporal_store(*reinterpret_cast<f32x4v*>(&pv[w]),
+                                    reinterpret_cast<f32x4v*>(p + i));
+        __builtin_nontemporal_store(*reinterpret_cast<f32x4v*>(&mv[w]),
+                                    reinterpret_cast<f32x4v*>(m + i));
+        __builtin_nontemporal_store(*reinterpret_cast<f32x4v*>(&vv[w]),
+                                    reinterpret_cast<f32x4v*>(v + i));
+        if (shadow != nullptr)
+          __builtin_nontemporal_store(
+              *reinterpret_cast<unsigned long long*>(sh),
+              reinterpret_cast<unsigned long long*>(shadow + i));
+      } else {
+        *reinterpret_cast<float4*>(p + i) = pv[w];
+        *reinterpret_cast<float4*>(m + i) = mv[w];
+        *reinterpret_cast<float4*>(v + i) = vv[w];
+        if (shadow != nullptr)
+          *reinterpret_cast<ulonglong1*>(shadow + i) =
+              *reinterpret_cast<ulonglong1*>(sh);
+      }
+    }
+  }
+  // scalar tail
+  for (long i = n4 * 4 + blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    float gvs;
+    if constexpr (sizeof(G) == 2) gvs = bf2f(((const ushort*)g)[i]);
+    else gvs = ((const float*)g)[i];
+    const float mi = b1 * m[i] + (1.f - b1) * gvs;
+    const float vi = b2 * v[i] + (1.f - b2) * gvs * gvs;
+    m[i] = mi;
+    v[i] = vi;
+    const float pvs = p[i] - lr_t * mi / (sqrtf(vi) + eps);
+    p[i] = pvs;
+    if (shadow != nullptr) shadow[i] = f2bf(pvs);
   }
 }
 
@@ -2319,20 +2416,31 @@ torch::Tensor gemm_tn_bf16(torch::Tensor A, torch::Tensor B) {
   TORCH_CHECK(B.size(0) == K, "K mismatch");
   TORCH_CHECK(M <= GNN_BN && M % 8 == 0, "tn GEMM: M must be <=384, mult of 8");
   auto C = torch::empty({V, M}, A.options());
-  const size_t lds = 2UL * GNN_BK * (GTN_PKT + GNN_PKB) * 2;  // 78 KiB
+  // BV=64 runs 4-wave blocks at 66 KiB LDS -> 2 independent blocks/CU
+  // overlapping each other's barrier phases; BV=128 is the 1-block/CU shape
+  static const int bv_env = []() {
+    const char* e = getenv("C2V_TN_BV");
+    return e ? atoi(e) : 64;
+  }();
   static bool tn_configured = false;
   if (!tn_configured) {
-    (void)hipFuncSetAttribute((const void*)k_gemm_tn<false>,
+    (void)hipFuncSetAttribute((const void*)k_gemm_tn<false, false, 128>,
                               hipFuncAttributeMaxDynamicSharedMemorySize,
-                              (int)lds);
-    (void)hipFuncSetAttribute((const void*)k_gemm_tn<true>,
+                              (int)(2UL * GNN_BK * (128 + 56 + GNN_PKB) * 2));
+    (void)hipFuncSetAttribute((const void*)k_gemm_tn<false, false, 64>,
                               hipFuncAttributeMaxDynamicSharedMemorySize,
-                              (int)lds);
+                              (int)(2UL * GNN_BK * (64 + 56 + GNN_PKB) * 2));
     tn_configured = true;
   }
-  const int grid = (V + GTN_BV - 1) / GTN_BV;
-  k_gemm_tn<false><<<grid, 512, lds, cur_stream()>>>(
-      bf_ptr(A), bf_ptr(B), bf_ptr_mut(C), V, M, K);
+  if (bv_env == 64) {
+    const size_t lds = 2UL * GNN_BK * (64 + 56 + GNN_PKB) * 2;  // 66 KiB
+    k_gemm_tn<false, false, 64><<<(V + 63) / 64, 256, lds, cur_stream()>>>(
+        bf_ptr(A), bf_ptr(B), bf_ptr_mut(C), V, M, K);
+  } else {
+    const size_t lds = 2UL * GNN_BK * (GTN_PKT + GNN_PKB) * 2;  // 78 KiB
+    k_gemm_tn<false><<<(V + GTN_BV - 1) / GTN_BV, 512, lds, cur_stream()>>>(
+        bf_ptr(A), bf_ptr(B), bf_ptr_mut(C), V, M, K);
+  }
   return C;
 }
 
@@ -2347,18 +2455,31 @@ torch::Tensor gemm_tn_ce(torch::Tensor logits, torch::Tensor code,
   TORCH_CHECK(code.size(0) == K && M <= GNN_BN && M % 8 == 0);
   auto labels_c = labels.contiguous();
   auto C = torch::empty({V, M}, logits.options());
-  const size_t lds = 2UL * GNN_BK * (GTN_PKT + GNN_PKB) * 2;
+  static const int bv_env = []() {
+    const char* e = getenv("C2V_TN_BV");
+    return e ? atoi(e) : 64;
+  }();
   static bool cfg2 = false;
   if (!cfg2) {
-    (void)hipFuncSetAttribute((const void*)k_gemm_tn<true>,
+    (void)hipFuncSetAttribute((const void*)k_gemm_tn<true, false, 128>,
                               hipFuncAttributeMaxDynamicSharedMemorySize,
-                              (int)lds);
+                              (int)(2UL * GNN_BK * (128 + 56 + GNN_PKB) * 2));
+    (void)hipFuncSetAttribute((const void*)k_gemm_tn<true, false, 64>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              (int)(2UL * GNN_BK * (64 + 56 + GNN_PKB) * 2));
     cfg2 = true;
   }
-  const int grid = (V + GTN_BV - 1) / GTN_BV;
-  k_gemm_tn<true><<<grid, 512, lds, cur_stream()>>>(
-      bf_ptr(logits), bf_ptr(code), bf_ptr_mut(C), V, M, K,
-      lse.data_ptr<float>(), labels_c.data_ptr<long>(), (float)scale);
+  if (bv_env == 64) {
+    const size_t lds = 2UL * GNN_BK * (64 + 56 + GNN_PKB) * 2;
+    k_gemm_tn<true, false, 64><<<(V + 63) / 64, 256, lds, cur_stream()>>>(
+        bf_ptr(logits), bf_ptr(code), bf_ptr_mut(C), V, M, K,
+        lse.data_ptr<float>(), labels_c.data_ptr<long>(), (float)scale);
+  } else {
+    const size_t lds = 2UL * GNN_BK * (GTN_PKT + GNN_PKB) * 2;
+    k_gemm_tn<true><<<(V + GTN_BV - 1) / GTN_BV, 512, lds, cur_stream()>>>(
+        bf_ptr(logits), bf_ptr(code), bf_ptr_mut(C), V, M, K,
+        lse.data_ptr<float>(), labels_c.data_ptr<long>(), (float)scale);
+  }
   return C;
 }
 
@@ -2627,7 +2748,24 @@ void adam_dense_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
     const char* e = getenv("C2V_ADAM_NT");
     return e && e[0] == '1';
   }();
+  static const bool use_w2 = [] {
+    const char* e = getenv("C2V_ADAM_W2");
+    return !e || e[0] == '1';   // width-2 default; set 0 for the old kernel
+  }();
   const int grid = grid_1d(std::max<long>(n / 4, 1), 256);
+  if (use_w2 && g_c.scalar_type() == torch::kBFloat16) {
+    if (use_nt)
+      k_adam_dense_w2<ushort, true><<<grid, 256, 0, cur_stream()>>>(
+          p.data_ptr<float>(), bf_ptr(g_c), m.data_ptr<float>(),
+          v.data_ptr<float>(), shadow_ptr, n, lr_t, (float)beta1,
+          (float)beta2, (float)eps, lrt_ptr);
+    else
+      k_adam_dense_w2<ushort, false><<<grid, 256, 0, cur_stream()>>>(
+          p.data_ptr<float>(), bf_ptr(g_c), m.data_ptr<float>(),
+          v.data_ptr<float>(), shadow_ptr, n, lr_t, (float)beta1,
+          (float)beta2, (float)eps, lrt_ptr);
+    return;
+  }
   if (use_nt && g_c.scalar_type() == torch::kBFloat16) {
     k_adam_dense<ushort, true><<<grid, 256, 0, cur_stream()>>>(
         p.data_ptr<float>(), bf_ptr(g_c), m.data_ptr<float>(),
