@@ -66,6 +66,7 @@ class Code2VecModel(Code2VecModelBase):
                 base_path]
 
     def _load_inner_model(self):
+        from ..utils.tf_bundle import TFCheckpointReader
         for path in self._checkpoint_candidates(self.config.MODEL_LOAD_PATH):
             if os.path.isfile(path):
                 payload = torch.load(path, map_location='cpu', weights_only=False)
@@ -75,8 +76,60 @@ class Code2VecModel(Code2VecModelBase):
                 self.log('Loaded model weights from: %s (epochs trained: %d)'
                          % (path, self._epochs_trained))
                 return
+            if TFCheckpointReader.is_tf_checkpoint(path):
+                # released reference models (TF Saver V2 bundle,
+                # tensorflow_model.py:370-377) load unchanged
+                self._load_tf_checkpoint(path)
+                self.log('Loaded TF-format model weights from: %s' % path)
+                return
         raise ValueError('No checkpoint found for load path: %s'
                          % self.config.MODEL_LOAD_PATH)
+
+    # TF graph variable names (reference tensorflow_model.py:205-220,249-250)
+    TF_NAME_MAP = {
+        'model/WORDS_VOCAB': 'tok_table',
+        'model/PATHS_VOCAB': 'path_table',
+        'model/TARGET_WORDS_VOCAB': 'target_table',
+        'model/TRANSFORM': 'w',
+        'model/ATTENTION': 'a',
+    }
+
+    def _load_tf_checkpoint(self, prefix: str):
+        """Load a reference-format TF V2 checkpoint (entire-model with Adam
+        slots, or a weights-only `.release`) into the engine."""
+        import math
+        from ..utils.tf_bundle import TFCheckpointReader
+        reader = TFCheckpointReader(prefix)
+        net = self.network
+        for tf_name, param_name in self.TF_NAME_MAP.items():
+            param = net.get_param(param_name)
+            arr = reader.get_tensor(tf_name)
+            if param_name == 'a':
+                arr = arr.reshape(-1)       # stored (D, 1)
+            if tuple(arr.shape) != tuple(param.shape):
+                raise ValueError(
+                    'TF tensor %s has shape %s; this model expects %s '
+                    '(is dictionaries.bin next to the checkpoint the one it '
+                    'was trained with?)'
+                    % (tf_name, tuple(arr.shape), tuple(param.shape)))
+            param.copy_(torch.from_numpy(arr).to(param.device))
+            for slot, store in (('/Adam', net._adam_m), ('/Adam_1', net._adam_v)):
+                if reader.has_tensor(tf_name + slot):
+                    s = reader.get_tensor(tf_name + slot)
+                    if param_name == 'a':
+                        s = s.reshape(-1)
+                    store[param_name].copy_(
+                        torch.from_numpy(s).to(param.device))
+        # Adam step from beta1_power = beta1^t (TF non-slot variable)
+        for name in reader.tensor_names():
+            if name.endswith('beta1_power'):
+                b1p = float(reader.get_tensor(name).reshape(-1)[0])
+                if 0.0 < b1p < 1.0:
+                    net.adam_step = max(1, round(math.log(b1p) / math.log(0.9)))
+                break
+        net._refresh_shadows()
+        if net._step_t is not None:
+            net._step_t.fill_(net.adam_step)
 
     def _save_inner_model(self, path: str):
         if self.config.RELEASE:

@@ -1984,8 +1984,11 @@ __global__ void k_hot_fold(const float* __restrict__ hot_acc,
 }
 
 __global__ void k_zero_rows_dyn(float* __restrict__ acc,
-                                const int* __restrict__ n_uniq_ptr, int d) {
-  const long total = (long)(*n_uniq_ptr) * d;
+                                const int* __restrict__ n_uniq_ptr, int d,
+                                int max_rows = 0x7FFFFFFF) {
+  // max_rows clamps a counter that may overshoot its buffer (the hot-id
+  // counter keeps atomicAdd-ing past HOT_CAP when the table is full)
+  const long total = (long)min(*n_uniq_ptr, max_rows) * d;
   const long n4 = total / 4;
   for (long i4 = blockIdx.x * blockDim.x + threadIdx.x; i4 < n4;
        i4 += (long)gridDim.x * blockDim.x)
@@ -2857,7 +2860,8 @@ static torch::Tensor accum_rows_common(const DedupState& st,
                               rows_c.options().dtype(torch::kFloat32));
   k_zero_rows_dyn<<<grid_1d((long)HOT_CAP * HOT_R * d / 4, 256), 256, 0,
                     cur_stream()>>>(
-      hot_acc.data_ptr<float>(), st.n_hot.data_ptr<int>(), HOT_R * d);
+      hot_acc.data_ptr<float>(), st.n_hot.data_ptr<int>(), HOT_R * d,
+      HOT_CAP);
   k_zero_rows_dyn<<<grid_1d(n * d / 4, 256), 256, 0, cur_stream()>>>(
       acc.data_ptr<float>(), st.n_uniq.data_ptr<int>(), d);
   if (rows_c.scalar_type() == torch::kBFloat16)
@@ -2886,7 +2890,8 @@ static torch::Tensor accum_ctx_common(const DedupState& st,
                               d_ctx.options().dtype(torch::kFloat32));
   k_zero_rows_dyn<<<grid_1d((long)HOT_CAP * HOT_R * d / 4, 256), 256, 0,
                     cur_stream()>>>(
-      hot_acc.data_ptr<float>(), st.n_hot.data_ptr<int>(), HOT_R * d);
+      hot_acc.data_ptr<float>(), st.n_hot.data_ptr<int>(), HOT_R * d,
+      HOT_CAP);
   k_zero_rows_dyn<<<grid_1d(n * d / 4, 256), 256, 0, cur_stream()>>>(
       acc.data_ptr<float>(), st.n_uniq.data_ptr<int>(), d);
   k_rows_accum_ctx<<<grid_1d(n * d, 256), 256, 0, cur_stream()>>>(

@@ -1,0 +1,153 @@
+"""TF TensorBundle (checkpoint V2) compatibility: the released reference
+models are tf.compat.v1.train.Saver output (tensorflow_model.py:370-377);
+they must load unchanged (BASELINE.json north star). TensorFlow is not
+installed here, so the format itself is exercised through write/read
+round-trips of utils/tf_bundle plus hand-built snappy/crc fixtures."""
+
+import os
+import pickle
+
+import numpy as np
+import pytest
+import torch
+
+from code2vec_amd.utils import tf_bundle as tfb
+
+
+def test_crc32c_known_vectors():
+    # RFC 3720 test vectors
+    assert tfb.crc32c(b'') == 0
+    assert tfb.crc32c(b'\x00' * 32) == 0x8A9136AA
+    assert tfb.crc32c(bytes(range(32))) == 0x46DD794E
+    assert tfb.crc32c(b'123456789') == 0xE3069283
+    c = tfb.crc32c(b'hello world')
+    assert tfb.crc_unmask(tfb.crc_mask(c)) == c
+
+
+def test_snappy_decompress_literals_and_copies():
+    # "abcabcabcabcX": literal 'abc' + copy(offset=3, len=9) + literal 'X'
+    comp = bytearray()
+    comp += tfb.write_varint(13)
+    comp += bytes([(3 - 1) << 2]) + b'abc'          # literal, len 3
+    comp += bytes([((9 - 4) << 2) | 1, 3])          # copy1: len 9, offset 3
+    comp += bytes([(1 - 1) << 2]) + b'X'            # literal, len 1
+    assert tfb.snappy_decompress(bytes(comp)) == b'abcabcabcabcX'
+
+
+def test_bundle_round_trip(tmp_path):
+    rng = np.random.default_rng(3)
+    tensors = {
+        'model/WORDS_VOCAB': rng.standard_normal((21, 16)).astype(np.float32),
+        'model/PATHS_VOCAB': rng.standard_normal((16, 16)).astype(np.float32),
+        'model/TARGET_WORDS_VOCAB':
+            rng.standard_normal((7, 48)).astype(np.float32),
+        'model/TRANSFORM': rng.standard_normal((48, 48)).astype(np.float32),
+        'model/ATTENTION': rng.standard_normal((48, 1)).astype(np.float32),
+        'model/beta1_power': np.array([0.9 ** 7], dtype=np.float32),
+        'ints': np.arange(12, dtype=np.int64).reshape(3, 4),
+    }
+    prefix = str(tmp_path / 'ckpt' / 'saved_model_iter8')
+    os.makedirs(os.path.dirname(prefix))
+    tfb.write_checkpoint(prefix, tensors)
+    assert os.path.isfile(prefix + '.index')
+    assert os.path.isfile(prefix + '.data-00000-of-00001')
+
+    r = tfb.TFCheckpointReader(prefix)
+    assert r.tensor_names() == sorted(tensors)
+    for name, arr in tensors.items():
+        got = r.get_tensor(name, verify=True)   # crc checked
+        assert got.dtype == arr.dtype
+        assert np.array_equal(got, arr), name
+
+
+def test_model_loads_tf_release_checkpoint(tmp_path):
+    """End to end: a TF-format `.release` checkpoint + dictionaries.bin next
+    to it -> Code2VecModel loads it and predicts with those exact weights."""
+    from code2vec_amd.config import Config
+    from code2vec_amd.models.torch_model import Code2VecModel
+
+    tokens = ['tok%d' % i for i in range(20)]
+    paths = ['p%d' % i for i in range(15)]
+    targets = ['alpha', 'beta', 'gamma|ray']
+    model_dir = tmp_path / 'models'
+    os.makedirs(model_dir)
+    load_path = str(model_dir / 'saved_model_iter8')
+    # dictionaries.bin framing: token, target, path order, each stored
+    # WITHOUT the single joined <PAD_OR_OOV> special (re-added on load at
+    # index 0), so stored indices start at 1
+    with open(str(model_dir / 'dictionaries.bin'), 'wb') as f:
+        for words in (tokens, targets, paths):
+            word_to_index = {w: i + 1 for i, w in enumerate(sorted(words))}
+            index_to_word = {i: w for w, i in word_to_index.items()}
+            pickle.dump(word_to_index, f)
+            pickle.dump(index_to_word, f)
+            pickle.dump(len(word_to_index), f)
+
+    d, D = 16, 48
+    rng = np.random.default_rng(5)
+    tensors = {
+        'model/WORDS_VOCAB':
+            rng.standard_normal((len(tokens) + 1, d)).astype(np.float32),
+        'model/PATHS_VOCAB':
+            rng.standard_normal((len(paths) + 1, d)).astype(np.float32),
+        'model/TARGET_WORDS_VOCAB':
+            rng.standard_normal((len(targets) + 1, D)).astype(np.float32),
+        'model/TRANSFORM': rng.standard_normal((3 * d, D)).astype(np.float32),
+        'model/ATTENTION': rng.standard_normal((D, 1)).astype(np.float32),
+        'model/WORDS_VOCAB/Adam':
+            rng.standard_normal((len(tokens) + 1, d)).astype(np.float32),
+        'model/WORDS_VOCAB/Adam_1':
+            np.abs(rng.standard_normal((len(tokens) + 1, d))).astype(np.float32),
+        'model/beta1_power': np.array([0.9 ** 11], dtype=np.float32),
+    }
+    tfb.write_checkpoint(load_path + '.release', tensors)
+
+    cfg = Config(set_defaults=True)
+    cfg.MODEL_LOAD_PATH = load_path
+    cfg.TOKEN_EMBEDDINGS_SIZE = d
+    cfg.PATH_EMBEDDINGS_SIZE = d
+    cfg.CODE_VECTOR_SIZE = D
+    cfg.TARGET_EMBEDDINGS_SIZE = D
+    cfg.MAX_CONTEXTS = 4
+    cfg.COMPUTE_DTYPE = 'fp32'
+    cfg.DEVICE = 'cpu'
+    cfg.VERBOSE_MODE = 0
+    model = Code2VecModel(cfg)
+
+    net = model.network
+    assert torch.equal(net.tok_table,
+                       torch.from_numpy(tensors['model/WORDS_VOCAB']))
+    assert torch.equal(net.w, torch.from_numpy(tensors['model/TRANSFORM']))
+    assert torch.equal(net.a,
+                       torch.from_numpy(tensors['model/ATTENTION'][:, 0]))
+    assert torch.equal(net._adam_m['tok_table'],
+                       torch.from_numpy(tensors['model/WORDS_VOCAB/Adam']))
+    assert net.adam_step == 11
+
+    # predict exercises the full forward with the loaded weights
+    line = 'whatever tok1,p2,tok3 tok4,p5,tok6   '
+    results = model.predict([line])
+    assert len(results) == 1
+    assert len(results[0].topk_predicted_words) > 0
+
+    # shape mismatch (wrong dictionaries for checkpoint) fails loudly
+    cfg2 = Config(set_defaults=True)
+    cfg2.MODEL_LOAD_PATH = load_path
+    cfg2.TOKEN_EMBEDDINGS_SIZE = d
+    cfg2.PATH_EMBEDDINGS_SIZE = d
+    cfg2.CODE_VECTOR_SIZE = D
+    cfg2.TARGET_EMBEDDINGS_SIZE = D
+    cfg2.MAX_CONTEXTS = 4
+    cfg2.COMPUTE_DTYPE = 'fp32'
+    cfg2.DEVICE = 'cpu'
+    cfg2.VERBOSE_MODE = 0
+    import shutil
+    model_dir2 = tmp_path / 'models2'
+    shutil.copytree(model_dir, model_dir2)
+    tensors_bad = dict(tensors)
+    tensors_bad['model/WORDS_VOCAB'] = tensors['model/WORDS_VOCAB'][:5]
+    tfb.write_checkpoint(str(model_dir2 / 'saved_model_iter8') + '.release',
+                         tensors_bad)
+    cfg2.MODEL_LOAD_PATH = str(model_dir2 / 'saved_model_iter8')
+    with pytest.raises(ValueError, match='shape'):
+        Code2VecModel(cfg2)
