@@ -1214,6 +1214,146 @@ __global__ void k_gemm_tn(const ushort* __restrict__ A,  // d_logits (K, V)
   }
 }
 
+// Direct-register tn GEMM: C(V,M) bf16 = A(K,V)^T @ B(K,M), M == 384.
+// No LDS, no barriers: both operands are k-major in memory (the transpose
+// problem), so instead of staging+transposing tiles through LDS, each lane
+// assembles its MFMA fragments with 8 k-strided u16 loads straight from
+// global — a wave's 16 consecutive v/col lanes make every load instruction a
+// set of 32 B row segments that L1/L2 serve from lines the neighboring
+// m/n-tiles complete. Waves run free (register double-buffering only), which
+// removes the 32 block-wide barrier pairs that left the staged kernel
+// latency-bound at 2 waves/SIMD.
+template <bool CEB = false>
+__launch_bounds__(512, 1)
+__global__ void k_gemm_tn_direct(const ushort* __restrict__ A,
+                                 const ushort* __restrict__ B,
+                                 ushort* __restrict__ C, int V, int M, int K,
+                                 const float* __restrict__ lse = nullptr,
+                                 const long* __restrict__ labels = nullptr,
+                                 float ce_scale = 1.f) {
+  const int v0 = blockIdx.x * 128;
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;      // 8 waves: 2 (v) x 4 (col)
+  const int wrow = (wid >> 2) * 64;
+  const int wcol = (wid & 3) * 96;
+  const int kg = (lane >> 4) * 8;        // k-offset of this lane's fragment
+  const int lr = lane & 15;
+  const bool vtail = v0 + 128 > V;
+
+  f32x4 acc[4][6];
+#pragma unroll
+  for (int m = 0; m < 4; ++m)
+#pragma unroll
+    for (int n = 0; n < 6; ++n) acc[m][n] = {0.f, 0.f, 0.f, 0.f};
+
+  u16x8 af[2][4], bf[2][6];
+  auto ce_map = [&](u16x8 t, int gk0, int v) -> u16x8 {
+    if (!CEB) return t;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float pr = __expf(bf2f(t[j]) - lse[gk0 + j]);
+      if (v == (int)labels[gk0 + j]) pr -= 1.f;
+      t[j] = f2bf(pr * ce_scale);
+    }
+    return t;
+  };
+  auto load_step = [&](int ks, int buf) {
+    const int k0 = ks * GNN_BK;
+#pragma unroll
+    for (int m = 0; m < 4; ++m) {
+      const int v = v0 + wrow + m * 16 + lr;
+      const long base = (long)(k0 + kg) * V + v;
+      u16x8 t;
+      if (!vtail || v < V) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) t[j] = A[base + (long)j * V];
+        t = ce_map(t, k0 + kg, v);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) t[j] = (ushort)0;
+      }
+      af[buf][m] = t;
+    }
+#pragma unroll
+    for (int n = 0; n < 6; ++n) {
+      const long base = (long)(k0 + kg) * M + wcol + n * 16 + lr;
+      u16x8 t;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) t[j] = B[base + (long)j * M];
+      bf[buf][n] = t;
+    }
+  };
+  auto compute = [&](int buf) {
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int m = 0; m < 4; ++m) {
+      const bf16x8 afr = __builtin_bit_cast(bf16x8, af[buf][m]);
+#pragma unroll
+      for (int n = 0; n < 6; ++n)
+        acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afr, __builtin_bit_cast(bf16x8, bf[buf][n]), acc[m][n], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+  };
+
+  const int ksteps = K / GNN_BK;
+  if (ksteps > 0) {
+    load_step(0, 0);
+    for (int ks = 0; ks < ksteps; ++ks) {
+      const int cur = ks & 1;
+      if (ks + 1 < ksteps) load_step(ks + 1, cur ^ 1);
+      compute(cur);
+    }
+  }
+  if (K % GNN_BK) {                       // ragged K tail, fully guarded
+    const int k0 = ksteps * GNN_BK;
+#pragma unroll
+    for (int m = 0; m < 4; ++m) {
+      const int v = v0 + wrow + m * 16 + lr;
+      u16x8 t;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        t[j] = (k0 + kg + j < K && v < V) ? A[(long)(k0 + kg + j) * V + v]
+                                          : (ushort)0;
+      if (CEB && v < V) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          if (k0 + kg + j >= K) continue;
+          float pr = __expf(bf2f(t[j]) - lse[k0 + kg + j]);
+          if (v == (int)labels[k0 + kg + j]) pr -= 1.f;
+          t[j] = f2bf(pr * ce_scale);
+        }
+      }
+      af[0][m] = t;
+    }
+#pragma unroll
+    for (int n = 0; n < 6; ++n) {
+      const int col = wcol + n * 16 + lr;
+      u16x8 t;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        t[j] = (k0 + kg + j < K) ? B[(long)(k0 + kg + j) * M + col]
+                                 : (ushort)0;
+      bf[0][n] = t;
+    }
+    compute(0);
+  }
+
+#pragma unroll
+  for (int m = 0; m < 4; ++m) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int v = v0 + wrow + m * 16 + (lane >> 4) * 4 + r;
+      if (v >= V) continue;
+#pragma unroll
+      for (int n = 0; n < 6; ++n) {
+        const int col = wcol + n * 16 + lr;
+        C[(long)v * M + col] = f2bf(acc[m][n][r]);
+      }
+    }
+  }
+}
+
 // fold the S split-K slices: C[i] = sum_s P[s][i] (fp32 out)
 __launch_bounds__(256)
 __global__ void k_splitk_reduce(const float* __restrict__ P,
@@ -2419,31 +2559,28 @@ torch::Tensor gemm_tn_bf16(torch::Tensor A, torch::Tensor B) {
   TORCH_CHECK(B.size(0) == K, "K mismatch");
   TORCH_CHECK(M <= GNN_BN && M % 8 == 0, "tn GEMM: M must be <=384, mult of 8");
   auto C = torch::empty({V, M}, A.options());
-  // BV=64 runs 4-wave blocks at 66 KiB LDS -> 2 independent blocks/CU
-  // overlapping each other's barrier phases; BV=128 is the 1-block/CU shape
-  static const int bv_env = []() {
-    const char* e = getenv("C2V_TN_BV");
-    return e ? atoi(e) : 64;
-  }();
+  const size_t lds = 2UL * GNN_BK * (GTN_PKT + GNN_PKB) * 2;  // 78 KiB
   static bool tn_configured = false;
   if (!tn_configured) {
-    (void)hipFuncSetAttribute((const void*)k_gemm_tn<false, false, 128>,
+    (void)hipFuncSetAttribute((const void*)k_gemm_tn<false>,
                               hipFuncAttributeMaxDynamicSharedMemorySize,
-                              (int)(2UL * GNN_BK * (128 + 56 + GNN_PKB) * 2));
-    (void)hipFuncSetAttribute((const void*)k_gemm_tn<false, false, 64>,
+                              (int)lds);
+    (void)hipFuncSetAttribute((const void*)k_gemm_tn<true>,
                               hipFuncAttributeMaxDynamicSharedMemorySize,
-                              (int)(2UL * GNN_BK * (64 + 56 + GNN_PKB) * 2));
+                              (int)lds);
     tn_configured = true;
   }
-  if (bv_env == 64) {
-    const size_t lds = 2UL * GNN_BK * (64 + 56 + GNN_PKB) * 2;  // 66 KiB
-    k_gemm_tn<false, false, 64><<<(V + 63) / 64, 256, lds, cur_stream()>>>(
+  static const bool direct = [] {
+    const char* e = getenv("C2V_TN_DIRECT");
+    return !e || e[0] == '1';
+  }();
+  if (direct && M == 384) {
+    k_gemm_tn_direct<false><<<(V + 127) / 128, 512, 0, cur_stream()>>>(
         bf_ptr(A), bf_ptr(B), bf_ptr_mut(C), V, M, K);
-  } else {
-    const size_t lds = 2UL * GNN_BK * (GTN_PKT + GNN_PKB) * 2;  // 78 KiB
-    k_gemm_tn<false><<<(V + GTN_BV - 1) / GTN_BV, 512, lds, cur_stream()>>>(
-        bf_ptr(A), bf_ptr(B), bf_ptr_mut(C), V, M, K);
+    return C;
   }
+  k_gemm_tn<false><<<(V + GTN_BV - 1) / GTN_BV, 512, lds, cur_stream()>>>(
+      bf_ptr(A), bf_ptr(B), bf_ptr_mut(C), V, M, K);
   return C;
 }
 
@@ -2458,31 +2595,27 @@ torch::Tensor gemm_tn_ce(torch::Tensor logits, torch::Tensor code,
   TORCH_CHECK(code.size(0) == K && M <= GNN_BN && M % 8 == 0);
   auto labels_c = labels.contiguous();
   auto C = torch::empty({V, M}, logits.options());
-  static const int bv_env = []() {
-    const char* e = getenv("C2V_TN_BV");
-    return e ? atoi(e) : 64;
-  }();
+  const size_t lds = 2UL * GNN_BK * (GTN_PKT + GNN_PKB) * 2;
   static bool cfg2 = false;
   if (!cfg2) {
-    (void)hipFuncSetAttribute((const void*)k_gemm_tn<true, false, 128>,
+    (void)hipFuncSetAttribute((const void*)k_gemm_tn<true>,
                               hipFuncAttributeMaxDynamicSharedMemorySize,
-                              (int)(2UL * GNN_BK * (128 + 56 + GNN_PKB) * 2));
-    (void)hipFuncSetAttribute((const void*)k_gemm_tn<true, false, 64>,
-                              hipFuncAttributeMaxDynamicSharedMemorySize,
-                              (int)(2UL * GNN_BK * (64 + 56 + GNN_PKB) * 2));
+                              (int)lds);
     cfg2 = true;
   }
-  if (bv_env == 64) {
-    const size_t lds = 2UL * GNN_BK * (64 + 56 + GNN_PKB) * 2;
-    k_gemm_tn<true, false, 64><<<(V + 63) / 64, 256, lds, cur_stream()>>>(
+  static const bool direct = [] {
+    const char* e = getenv("C2V_TN_DIRECT");
+    return !e || e[0] == '1';
+  }();
+  if (direct && M == 384) {
+    k_gemm_tn_direct<true><<<(V + 127) / 128, 512, 0, cur_stream()>>>(
         bf_ptr(logits), bf_ptr(code), bf_ptr_mut(C), V, M, K,
         lse.data_ptr<float>(), labels_c.data_ptr<long>(), (float)scale);
-  } else {
-    const size_t lds = 2UL * GNN_BK * (GTN_PKT + GNN_PKB) * 2;
-    k_gemm_tn<true><<<(V + GTN_BV - 1) / GTN_BV, 512, lds, cur_stream()>>>(
-        bf_ptr(logits), bf_ptr(code), bf_ptr_mut(C), V, M, K,
-        lse.data_ptr<float>(), labels_c.data_ptr<long>(), (float)scale);
+    return C;
   }
+  k_gemm_tn<true><<<(V + GTN_BV - 1) / GTN_BV, 512, lds, cur_stream()>>>(
+      bf_ptr(logits), bf_ptr(code), bf_ptr_mut(C), V, M, K,
+      lse.data_ptr<float>(), labels_c.data_ptr<long>(), (float)scale);
   return C;
 }
 
@@ -2747,13 +2880,15 @@ void adam_dense_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
   ushort* shadow_ptr = nullptr;
   if (shadow.defined() && shadow.numel() == n)
     shadow_ptr = reinterpret_cast<ushort*>(shadow.data_ptr<at::BFloat16>());
+  // A/B on MI355X (java14m target-table shape, r02_call4): width-1 NT
+  // stores 500 us, width-1 cached 530, width-2 532 — NT default, w2 off
   static const bool use_nt = [] {
     const char* e = getenv("C2V_ADAM_NT");
-    return e && e[0] == '1';
+    return !e || e[0] == '1';
   }();
   static const bool use_w2 = [] {
     const char* e = getenv("C2V_ADAM_W2");
-    return !e || e[0] == '1';   // width-2 default; set 0 for the old kernel
+    return e && e[0] == '1';
   }();
   const int grid = grid_1d(std::max<long>(n / 4, 1), 256);
   if (use_w2 && g_c.scalar_type() == torch::kBFloat16) {
