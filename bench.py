@@ -109,6 +109,184 @@ def synth_batches(cfg, device, batch_size, n_batches=8, seed=0):
     return batches
 
 
+def _tok_word(i):
+    return 'tok%07d' % i
+
+
+def _path_word(i):
+    # java14m paths are Java String#hashCode ints on the wire
+    h = (i * 2654435761) & 0xFFFFFFFF
+    return str(h - (1 << 32) if h >= (1 << 31) else h)
+
+
+def _target_word(i):
+    return 'name|w%06d' % i
+
+
+def generate_e2e_dataset(path, cfg, n_rows, seed=7):
+    """Write a java14m-shaped synthetic `.c2v` to disk: Zipf-frequency
+    token/path/target words (ids map rank->word), U[64,C] contexts per row,
+    trailing-space padding exactly like preprocess.py output."""
+    g = torch.Generator().manual_seed(seed)
+    C = cfg.MAX_CONTEXTS
+    with open(path, 'w') as f:
+        for base in range(0, n_rows, 4096):
+            nb = min(4096, n_rows - base)
+            n_ctx = torch.randint(max(1, C // 3), C + 1, (nb,), generator=g)
+            # ids are Zipf ranks; subtract 1 -> word index (0-based)
+            src = zipf_ids(cfg.MAX_TOKEN_VOCAB_SIZE + 1, (nb, C), g) - 1
+            pth = zipf_ids(cfg.MAX_PATH_VOCAB_SIZE + 1, (nb, C), g) - 1
+            tgt = zipf_ids(cfg.MAX_TOKEN_VOCAB_SIZE + 1, (nb, C), g) - 1
+            lab = zipf_ids(cfg.MAX_TARGET_VOCAB_SIZE + 1, (nb,), g) - 1
+            src_l, pth_l, tgt_l = src.tolist(), pth.tolist(), tgt.tolist()
+            lab_l, nc_l = lab.tolist(), n_ctx.tolist()
+            out = []
+            for r in range(nb):
+                n = nc_l[r]
+                ctxs = ['%s,%s,%s' % (_tok_word(src_l[r][c]),
+                                      _path_word(pth_l[r][c]),
+                                      _tok_word(tgt_l[r][c]))
+                        for c in range(n)]
+                out.append(_target_word(lab_l[r]) + ' ' + ' '.join(ctxs)
+                           + ' ' * (C - n) + '\n')
+            f.writelines(out)
+
+
+def run_end_to_end(args, device, rank, world_size, reducer, distributed):
+    """BASELINE end-to-end mode: reader -> H2D -> train step on on-disk
+    java14m-shaped data with the real vocab-scale string->index maps
+    (VERDICT r01 missing #3: nothing measured reader+H2D+step together)."""
+    import json as _json
+    from types import SimpleNamespace
+
+    from code2vec_amd.data.prefetcher import BatchPrefetcher
+    from code2vec_amd.data.reader import EstimatorAction, PathContextReader
+    from code2vec_amd.vocabularies import (
+        Vocab, VocabType, _SpecialVocabWords_JoinedOovPad)
+
+    cfg = make_config(device)
+    cfg.SAMPLED_SOFTMAX_SIZE = args.sampled_softmax
+    n_gpus = world_size if distributed else 1
+
+    data_path = os.path.join(os.environ.get('TMPDIR', '/tmp'),
+                             'c2v_e2e.train.c2v')
+    if rank == 0 and (not os.path.isfile(data_path)
+                      or os.path.getsize(data_path) < 1000):
+        t0 = time.perf_counter()
+        generate_e2e_dataset(data_path, cfg, args.e2e_rows, seed=7)
+        print('# generated %s (%.1f MB) in %.1fs'
+              % (data_path, os.path.getsize(data_path) / 1e6,
+                 time.perf_counter() - t0))
+    if distributed:
+        import torch.distributed as dist
+        dist.barrier()
+
+    # full-vocab-scale string->index maps (rank->word is the id mapping the
+    # generator used, so lookups resolve to the same Zipf-shaped ids)
+    def make_vocab(vt, n, word_fn):
+        # default regime: joined <PAD_OR_OOV> for every vocab
+        return Vocab(vt, [word_fn(i) for i in range(n)],
+                     _SpecialVocabWords_JoinedOovPad)
+
+    t0 = time.perf_counter()
+    vocabs = SimpleNamespace(
+        token_vocab=make_vocab(VocabType.Token, cfg.MAX_TOKEN_VOCAB_SIZE,
+                               _tok_word),
+        path_vocab=make_vocab(VocabType.Path, cfg.MAX_PATH_VOCAB_SIZE,
+                              _path_word),
+        target_vocab=make_vocab(VocabType.Target, cfg.MAX_TARGET_VOCAB_SIZE,
+                                _target_word))
+    if rank == 0:
+        print('# vocabs built in %.1fs' % (time.perf_counter() - t0))
+
+    net = Code2VecNetwork(cfg,
+                          token_vocab_size=vocabs.token_vocab.size,
+                          path_vocab_size=vocabs.path_vocab.size,
+                          target_vocab_size=vocabs.target_vocab.size,
+                          device=device)
+    reader = PathContextReader(vocabs=vocabs, config=cfg,
+                               estimator_action=EstimatorAction.Train,
+                               repeat_endlessly=True,
+                               world_size=world_size, rank=rank)
+    prefetcher = BatchPrefetcher(reader.iter_batches(data_path=data_path),
+                                 device, depth=cfg.READER_QUEUE_DEPTH)
+    batches = iter(prefetcher)
+
+    use_graph = device.startswith('cuda') and not distributed \
+        and not args.no_graph
+    if use_graph:
+        graph_step = net.make_graph_step(args.batch)
+
+    def do_step(b):
+        if b.source_token_indices.shape[0] != args.batch:
+            return  # ragged shuffle-tail batch: skip for fixed-shape timing
+        if use_graph:
+            graph_step.step(b.source_token_indices, b.path_indices,
+                            b.target_token_indices, b.context_valid_mask,
+                            b.target_index)
+        else:
+            net.train_step(b.source_token_indices, b.path_indices,
+                           b.target_token_indices, b.context_valid_mask,
+                           b.target_index, reducer=reducer)
+
+    def barrier_sync():
+        if distributed:
+            import torch.distributed as dist
+            dist.barrier()
+        if device.startswith('cuda'):
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        do_step(next(batches))
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        do_step(next(batches))
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+    if distributed:
+        import torch.distributed as dist
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if device.startswith('cuda') else 'cpu')
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank == 0:
+        global_batch = args.batch * n_gpus
+        ex_per_sec = global_batch * args.steps / elapsed
+        print(_json.dumps({
+            'metric': 'train_examples_per_sec',
+            'value': round(ex_per_sec, 1),
+            'unit': 'examples/s',
+            'n_gpus': n_gpus,
+            'steps': args.steps,
+            'warmup': args.warmup,
+            'ms_per_step': round(elapsed / args.steps * 1000.0, 3),
+            'higher_is_better': True,
+            'scaling': 'weak',
+            'vs_baseline': round(ex_per_sec / BASELINE_V100_EX_PER_SEC, 2),
+            'dtype': cfg.COMPUTE_DTYPE,
+            'data': 'synthetic-on-disk (end-to-end reader->H2D->step)',
+            'config': {
+                'model': 'code2vec-java14m',
+                'global_batch': global_batch,
+                'seq_len': cfg.MAX_CONTEXTS,
+                'parallelism': 'dp%d' % n_gpus,
+                'softmax': ('sampled-%d' % args.sampled_softmax)
+                           if args.sampled_softmax else 'full',
+                'stepping': 'hipgraph' if use_graph else 'eager',
+                'token_vocab': cfg.MAX_TOKEN_VOCAB_SIZE,
+                'path_vocab': cfg.MAX_PATH_VOCAB_SIZE,
+                'target_vocab': cfg.MAX_TARGET_VOCAB_SIZE,
+                'id_dist': 'zipf-%s' % ZIPF_S,
+                'end_to_end': True,
+            },
+        }))
+    if distributed:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument('--gpus', type=int, default=1)
@@ -120,6 +298,11 @@ def main():
     ap.add_argument('--sampled-softmax', type=int, default=0,
                     help='train with sampled softmax over N negatives '
                          '(BASELINE config 4); 0 = full softmax (default)')
+    ap.add_argument('--end-to-end', action='store_true',
+                    help='time reader->H2D->step on generated on-disk data '
+                         'instead of resident synthetic batches')
+    ap.add_argument('--e2e-rows', type=int, default=250000,
+                    help='rows of on-disk data to generate for --end-to-end')
     args = ap.parse_args()
 
     # --gpus N without a torchrun rendezvous: self-launch one rank per GPU
@@ -157,6 +340,10 @@ def main():
         backend = 'nccl' if device.startswith('cuda') else 'gloo'
         dist.init_process_group(backend=backend, rank=rank, world_size=world_size)
         reducer = Reducer(assume_equal_shards=True)  # fixed per-rank batch
+
+    if args.end_to_end:
+        run_end_to_end(args, device, rank, world_size, reducer, distributed)
+        return
 
     cfg = make_config(device)
     cfg.SAMPLED_SOFTMAX_SIZE = args.sampled_softmax

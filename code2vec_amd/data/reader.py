@@ -336,12 +336,16 @@ class PathContextReader:
             else:
                 pool_rows = 0
 
-        # Two-stage pipeline: a producer thread does file IO + parse_buffer
-        # (C++ with the GIL released) + filter/DP-shard, and this generator
-        # consumes filtered tensor chunks for the pool shuffle + batch
-        # slicing. Overlapping the two stages measured 95K -> ~150K ex/s per
-        # process on a 333 MB synthetic corpus (tools/ profiling; parse
-        # alone runs at 330K lines/s).
+        # Three-stage pipeline (r01 was two-stage and topped out at ~150K
+        # ex/s with IO and parse serialized on one producer thread while the
+        # C++ pool sat underfed):
+        #   1. one IO thread reads line-aligned chunks (page cache, GB/s)
+        #      and applies DP chunk-sharding,
+        #   2. READER_WORKERS parser threads each call parse_buffer (C++
+        #      thread pool with the GIL released) + filter,
+        #   3. this generator pools the unordered tensor chunks for the
+        #      windowed shuffle + batch slicing (order is irrelevant under
+        #      the shuffle; epoch totals stay exact).
         import queue
         import threading
 
@@ -350,13 +354,18 @@ class PathContextReader:
         # rank); small files (< 8 chunks per rank) fall back to line-modulo
         # so every rank still sees data. Chunk shards are not batch-aligned;
         # the train loop's per-step termination consensus
-        # (ddp.Reducer.all_continue) absorbs the ragged tail.
+        # (ddp.Reducer.start_vote/finish_vote) absorbs the ragged tail.
+        # Line-modulo needs a global line counter, so it keeps one parser
+        # worker (small files only — throughput is irrelevant there).
         chunk_bytes = int(os.environ.get('C2V_READER_CHUNK_BYTES', 4 << 20))
         chunk_shard = (self.world_size > 1 and
                        os.path.getsize(data_path) >=
                        self.world_size * 8 * chunk_bytes)
         line_shard = self.world_size > 1 and not chunk_shard
         line_base = 0
+        n_workers = 1 if line_shard else max(
+            1, int(os.environ.get('C2V_READER_WORKERS',
+                                  getattr(self.config, 'READER_WORKERS', 4))))
 
         def parse_filter(use: bytes):
             nonlocal line_base
@@ -378,13 +387,23 @@ class PathContextReader:
                 return tuple(t[idx] for t in (src, pth, tgt, mask, tidx))
             return (src, pth, tgt, mask, tidx)
 
-        q: 'queue.Queue' = queue.Queue(maxsize=8)
+        raw_q: 'queue.Queue' = queue.Queue(maxsize=2 * n_workers + 2)
+        out_q: 'queue.Queue' = queue.Queue(maxsize=2 * n_workers + 2)
         stop = threading.Event()
 
-        def producer():
+        def q_put(q, item) -> bool:
+            while not stop.is_set():
+                try:
+                    q.put(item, timeout=0.5)
+                    return True
+                except queue.Full:
+                    pass
+            return False
+
+        def io_thread():
             try:
                 epoch = 0
-                while epochs < 0 or epoch < epochs:
+                while (epochs < 0 or epoch < epochs) and not stop.is_set():
                     epoch += 1
                     chunk_i = 0
                     with open(data_path, 'rb') as f:
@@ -402,41 +421,45 @@ class PathContextReader:
                             mine = (not chunk_shard or
                                     chunk_i % self.world_size == self.rank)
                             chunk_i += 1
-                            if not mine:
-                                continue
-                            tup = parse_filter(buf[:last_nl + 1])
-                            if tup is not None:
-                                while not stop.is_set():
-                                    try:
-                                        q.put(tup, timeout=0.5)
-                                        break
-                                    except queue.Full:
-                                        pass
+                            if mine and not q_put(raw_q, buf[:last_nl + 1]):
+                                return
                         if carry.strip() and not stop.is_set() and (
                                 not chunk_shard
                                 or chunk_i % self.world_size == self.rank):
-                            tup = parse_filter(carry + b'\n')
-                            if tup is not None:
-                                while not stop.is_set():
-                                    try:
-                                        q.put(tup, timeout=0.5)
-                                        break
-                                    except queue.Full:
-                                        pass
-                    if stop.is_set():
-                        break
-                q.put(None)
+                            if not q_put(raw_q, carry + b'\n'):
+                                return
+                for _ in range(n_workers):
+                    q_put(raw_q, None)
             except BaseException as exc:  # noqa: BLE001 — surface in consumer
-                q.put(exc)
+                q_put(out_q, exc)
 
-        th = threading.Thread(target=producer, daemon=True,
-                              name='c2v-reader-parse')
-        th.start()
+        def parser_thread():
+            try:
+                while True:
+                    buf = raw_q.get()
+                    if buf is None:
+                        q_put(out_q, None)
+                        return
+                    tup = parse_filter(buf)
+                    if tup is not None and not q_put(out_q, tup):
+                        return
+            except BaseException as exc:  # noqa: BLE001
+                q_put(out_q, exc)
+
+        threads = [threading.Thread(target=io_thread, daemon=True,
+                                    name='c2v-reader-io')]
+        threads += [threading.Thread(target=parser_thread, daemon=True,
+                                     name='c2v-reader-parse%d' % i)
+                    for i in range(n_workers)]
+        for th in threads:
+            th.start()
         try:
-            while True:
-                item = q.get()
+            done = 0
+            while done < n_workers:
+                item = out_q.get()
                 if item is None:
-                    break
+                    done += 1
+                    continue
                 if isinstance(item, BaseException):
                     raise item
                 pool.append(item)
