@@ -204,6 +204,13 @@ def run_end_to_end(args, device, rank, world_size, reducer, distributed):
                           path_vocab_size=vocabs.path_vocab.size,
                           target_vocab_size=vocabs.target_vocab.size,
                           device=device)
+    # capture the graph BEFORE the reader/prefetcher threads start: graph
+    # capture and background pin_memory/H2D activity must not overlap
+    use_graph = device.startswith('cuda') and not distributed \
+        and not args.no_graph
+    if use_graph:
+        graph_step = net.make_graph_step(args.batch)
+
     reader = PathContextReader(vocabs=vocabs, config=cfg,
                                estimator_action=EstimatorAction.Train,
                                repeat_endlessly=True,
@@ -211,11 +218,6 @@ def run_end_to_end(args, device, rank, world_size, reducer, distributed):
     prefetcher = BatchPrefetcher(reader.iter_batches(data_path=data_path),
                                  device, depth=cfg.READER_QUEUE_DEPTH)
     batches = iter(prefetcher)
-
-    use_graph = device.startswith('cuda') and not distributed \
-        and not args.no_graph
-    if use_graph:
-        graph_step = net.make_graph_step(args.batch)
 
     def do_step(b):
         if b.source_token_indices.shape[0] != args.batch:

@@ -511,7 +511,9 @@ class GraphTrainStep:
         torch.cuda.current_stream().wait_stream(side)
 
         self.graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.graph):
+        # thread_local: background threads (reader workers pinning batches)
+        # may touch the CUDA API while this thread captures
+        with torch.cuda.graph(self.graph, capture_error_mode='thread_local'):
             self.loss = net.train_step(self.src, self.pth, self.tgt,
                                        self.mask, self.labels)
         # Capture RECORDS the device ops without executing them (only the
