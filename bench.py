@@ -164,6 +164,9 @@ def run_end_to_end(args, device, rank, world_size, reducer, distributed):
     from code2vec_amd.vocabularies import (
         Vocab, VocabType, _SpecialVocabWords_JoinedOovPad)
 
+    # small CPU ops (reader filters, pool shuffle slicing) thrash when
+    # torch's intra-op pool spans a 256-core box; the GPU does the math here
+    torch.set_num_threads(min(8, os.cpu_count() or 8))
     cfg = make_config(device)
     cfg.SAMPLED_SOFTMAX_SIZE = args.sampled_softmax
     n_gpus = world_size if distributed else 1
@@ -176,7 +179,7 @@ def run_end_to_end(args, device, rank, world_size, reducer, distributed):
         generate_e2e_dataset(data_path, cfg, args.e2e_rows, seed=7)
         print('# generated %s (%.1f MB) in %.1fs'
               % (data_path, os.path.getsize(data_path) / 1e6,
-                 time.perf_counter() - t0))
+                 time.perf_counter() - t0), flush=True)
     if distributed:
         import torch.distributed as dist
         dist.barrier()
@@ -197,7 +200,8 @@ def run_end_to_end(args, device, rank, world_size, reducer, distributed):
         target_vocab=make_vocab(VocabType.Target, cfg.MAX_TARGET_VOCAB_SIZE,
                                 _target_word))
     if rank == 0:
-        print('# vocabs built in %.1fs' % (time.perf_counter() - t0))
+        print('# vocabs built in %.1fs' % (time.perf_counter() - t0),
+              flush=True)
 
     net = Code2VecNetwork(cfg,
                           token_vocab_size=vocabs.token_vocab.size,
@@ -284,6 +288,12 @@ def run_end_to_end(args, device, rank, world_size, reducer, distributed):
                 'end_to_end': True,
             },
         }))
+    # orderly shutdown: signal the reader threads, then join the prefetch
+    # worker (a daemon thread still inside the C++ parser at interpreter
+    # finalization aborts the process)
+    reader.stop_streaming(join=False)
+    prefetcher.stop()
+    reader.stop_streaming(join=True)
     if distributed:
         import torch.distributed as dist
         dist.destroy_process_group()

@@ -26,6 +26,7 @@ class BatchPrefetcher:
         self.copy_stream = torch.cuda.Stream() if self.use_gpu else None
         self.q: "queue.Queue" = queue.Queue(maxsize=max(2, depth))
         self._err: Optional[BaseException] = None
+        self._stopping = False
         self._thread = threading.Thread(target=self._worker,
                                         args=(batch_iter,), daemon=True)
         self._thread.start()
@@ -33,11 +34,35 @@ class BatchPrefetcher:
     def _worker(self, batch_iter):
         try:
             for batch in batch_iter:
-                self.q.put(batch)
+                placed = False
+                while not self._stopping and not placed:
+                    try:
+                        self.q.put(batch, timeout=0.5)
+                        placed = True
+                    except queue.Full:
+                        pass
+                if self._stopping:
+                    break
         except BaseException as e:  # noqa: BLE001
             self._err = e
         finally:
-            self.q.put(self._SENTINEL)
+            try:
+                self.q.put_nowait(self._SENTINEL)
+            except queue.Full:
+                pass
+
+    def stop(self, timeout: float = 5.0):
+        """Terminate the worker before interpreter shutdown (a daemon thread
+        still executing extension code during finalization aborts the
+        process). The underlying reader should be signalled to stop first so
+        the worker's batch_iter returns."""
+        self._stopping = True
+        try:
+            while True:
+                self.q.get_nowait()
+        except queue.Empty:
+            pass
+        self._thread.join(timeout=timeout)
 
     def __iter__(self):
         while True:
@@ -53,5 +78,14 @@ class BatchPrefetcher:
                 moved = item.to(self.device, non_blocking=True)
                 event = torch.cuda.Event()
                 event.record(self.copy_stream)
-            torch.cuda.current_stream().wait_event(event)
+            cur = torch.cuda.current_stream()
+            cur.wait_event(event)
+            # The device tensors were allocated under copy_stream; mark them
+            # in use by the consumer stream so the caching allocator cannot
+            # hand their blocks to a LATER H2D copy while this stream still
+            # reads them (without this, a fast reader recycles a batch
+            # mid-step: garbage ids -> embedding-gather memory fault).
+            for t in moved:
+                if torch.is_tensor(t) and t.is_cuda:
+                    t.record_stream(cur)
             yield moved

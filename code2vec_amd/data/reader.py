@@ -435,8 +435,11 @@ class PathContextReader:
 
         def parser_thread():
             try:
-                while True:
-                    buf = raw_q.get()
+                while not stop.is_set():
+                    try:
+                        buf = raw_q.get(timeout=0.5)
+                    except queue.Empty:
+                        continue
                     if buf is None:
                         q_put(out_q, None)
                         return
@@ -451,12 +454,17 @@ class PathContextReader:
         threads += [threading.Thread(target=parser_thread, daemon=True,
                                      name='c2v-reader-parse%d' % i)
                     for i in range(n_workers)]
+        self._stream_stop = stop
+        self._stream_threads = threads
         for th in threads:
             th.start()
         try:
             done = 0
-            while done < n_workers:
-                item = out_q.get()
+            while done < n_workers and not stop.is_set():
+                try:
+                    item = out_q.get(timeout=0.5)
+                except queue.Empty:
+                    continue
                 if item is None:
                     done += 1
                     continue
@@ -465,9 +473,21 @@ class PathContextReader:
                 pool.append(item)
                 pool_rows += item[0].shape[0]
                 yield from drain()
-            yield from drain(final=True)
+            if not stop.is_set():
+                yield from drain(final=True)
         finally:
             stop.set()
+
+    def stop_streaming(self, join: bool = True, timeout: float = 5.0):
+        """Stop the stream-path threads. Call before interpreter shutdown:
+        a daemon thread re-entering the pybind11 extension during
+        finalization aborts the process (GIL reacquire -> std::terminate)."""
+        stopev = getattr(self, '_stream_stop', None)
+        if stopev is not None:
+            stopev.set()
+        if join:
+            for th in getattr(self, '_stream_threads', []) or []:
+                th.join(timeout=timeout)
 
     def _iter_batches_native(self, lines: Iterable[str],
                              batch_size: int) -> Iterator[ReaderBatch]:
