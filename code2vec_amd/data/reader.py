@@ -308,6 +308,9 @@ class PathContextReader:
                                context_valid_mask=tensors[3],
                                target_index=tensors[4])
 
+        nat_mod = _load_native_reader_module()
+        nat_gather = getattr(nat_mod, 'shuffle_gather', None)
+
         def drain(final=False):
             nonlocal pool, pool_rows
             # amortize the pool permutation: only drain once enough rows are
@@ -317,27 +320,35 @@ class PathContextReader:
             threshold = batch_size if sb == 0 else sb + 8 * batch_size
             if not pool or (pool_rows < threshold and not final):
                 return
-            cat = [torch.cat([p[i] for p in pool]) for i in range(5)]
-            pool = []
-            n = cat[0].shape[0]
-            if sb > 0:
+            n = pool_rows
+            if sb > 0 and nat_gather is not None:
+                # fused concat + shuffle gather in C++ with the GIL released,
+                # STRAIGHT INTO pinned memory: batches become views of one
+                # pinned block — no per-batch pin copy, and the ~1 ms/batch
+                # of GIL-holding torch CPU copies stops serializing against
+                # the training loop's launch thread
                 perm = torch.randperm(n, generator=g)
-                if pin:
-                    # gather the shuffle permutation STRAIGHT INTO pinned
-                    # memory: batches are then views of one pinned block and
-                    # the per-batch pin_memory copy (a serial ~13 MB memcpy
-                    # that capped the e2e pipeline at ~200 batches/s)
-                    # disappears
-                    shuf = []
-                    for t in cat:
-                        out = torch.empty_like(t, pin_memory=True)
-                        torch.index_select(t, 0, perm, out=out)
-                        shuf.append(out)
-                    cat = shuf
-                else:
-                    cat = [t[perm] for t in cat]
-            elif pin:
-                cat = [t.pin_memory() for t in cat]  # one bulk pin per drain
+                cat = [torch.empty((n,) + tuple(pool[0][i].shape[1:]),
+                                   dtype=pool[0][i].dtype, pin_memory=pin)
+                       for i in range(5)]
+                nat_gather([[p[i] for p in pool] for i in range(5)],
+                           perm, cat, 6)
+            else:
+                cat = [torch.cat([p[i] for p in pool]) for i in range(5)]
+                if sb > 0:
+                    perm = torch.randperm(n, generator=g)
+                    if pin:
+                        shuf = []
+                        for t in cat:
+                            out = torch.empty_like(t, pin_memory=True)
+                            torch.index_select(t, 0, perm, out=out)
+                            shuf.append(out)
+                        cat = shuf
+                    else:
+                        cat = [t[perm] for t in cat]
+                elif pin:
+                    cat = [t.pin_memory() for t in cat]  # one bulk pin
+            pool = []
             emit_until = n if final else max(0, n - sb)
             start = 0
             while emit_until - start >= batch_size or \

@@ -1057,7 +1057,9 @@ class Parser {
 
   // ---- expressions ----
 
-  Node* parse_expression() { return parse_assignment(); }
+  // Any expression position may hold a lambda (JavaParser accepts
+  // `return x -> e;`, initializers, ternary arms — not just call args)
+  Node* parse_expression() { return parse_lambda_or_expr(); }
 
   Node* parse_assignment() {
     Node* lhs = parse_ternary();
@@ -1072,7 +1074,7 @@ class Parser {
         Node* a = ast_.mk("AssignExpr");
         a->op = name;
         ast_.add(a, lhs);
-        ast_.add(a, parse_assignment());
+        ast_.add(a, parse_expression());  // RHS may be a lambda
         return a;
       }
     }
@@ -1320,7 +1322,7 @@ class Parser {
   void parse_args(Node* call) {
     expect("(");
     while (!is_punct(")") && !at_end()) {
-      ast_.add(call, parse_lambda_or_expr());
+      ast_.add(call, parse_expression());
       if (is_punct(",")) lx_.advance();
     }
     expect(")");
@@ -1376,7 +1378,7 @@ class Parser {
         return le;
       }
     }
-    return parse_expression();
+    return parse_assignment();
   }
 
   Node* parse_lambda_body() {

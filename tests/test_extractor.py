@@ -220,3 +220,142 @@ def test_extractor_thread_pool_tsan(tmp_path):
     assert out.returncode == 0
     assert 'WARNING: ThreadSanitizer' not in out.stderr
     assert len([l for l in out.stdout.splitlines() if l]) == 12
+
+
+# ---- grammar-edge coverage (round-2: lambdas, anonymous classes, bounded
+# generics, nested arrays, switch fallthrough, method refs, varargs, twr) ----
+
+def _paths(lines):
+    out = []
+    for line in lines:
+        _, ctxs = contexts_of(line)
+        out.extend(c.split(',')[1] for c in ctxs)
+    return out
+
+
+def test_lambda_positions(extractor, tmp_path):
+    """Lambdas in return / initializer / call-arg positions all parse and
+    produce LambdaExpr nodes on paths (JavaParser accepts all three)."""
+    code = '''
+import java.util.function.Function;
+import java.util.List;
+class In {
+    Function<Integer, Integer> fromReturn(int x) { return y -> y * x; }
+    void fromInit(int x) {
+        Function<Integer, Integer> f = y -> y + x;
+        f.apply(3);
+    }
+    void fromArg(List<Integer> xs) { xs.forEach(v -> System.out.println(v)); }
+}
+'''
+    lines = run_extract(extractor, code, tmp_path)
+    names = [contexts_of(l)[0] for l in lines]
+    assert names == ['from|return', 'from|init', 'from|arg']
+    assert any('LambdaExpr' in p for p in _paths(lines))
+
+
+def test_anonymous_class_inner_method(extractor, tmp_path):
+    """Methods inside anonymous classes are extracted as their own examples
+    (JavaParser's visitor descends into ObjectCreationExpr bodies)."""
+    code = '''
+class In {
+    Runnable makeRunner(int x) {
+        Runnable r = new Runnable() {
+            public void run() { System.out.println(x); }
+        };
+        return r;
+    }
+}
+'''
+    lines = run_extract(extractor, code, tmp_path)
+    names = [contexts_of(l)[0] for l in lines]
+    assert names == ['make|runner', 'run']
+    assert any('ObjectCreationExpr' in p for p in _paths([lines[0]]))
+
+
+def test_bounded_generics(extractor, tmp_path):
+    code = '''
+class In {
+    <T extends Comparable<T>> T maxOf(T a, T b) {
+        return a.compareTo(b) > 0 ? a : b;
+    }
+}
+'''
+    lines = run_extract(extractor, code, tmp_path)
+    name, ctxs = contexts_of(lines[0])
+    assert name == 'max|of'
+    assert any('ConditionalExpr' in c for c in ctxs)
+    terminals = {p for c in ctxs for p in (c.split(',')[0], c.split(',')[2])}
+    assert 't' in terminals  # the type-variable leaf, lowercased
+
+
+def test_nested_arrays(extractor, tmp_path):
+    code = '''
+class In {
+    int[][] grid = new int[3][4];
+    int get(int i, int j) { return grid[i][j]; }
+}
+'''
+    lines = run_extract(extractor, code, tmp_path)
+    name, ctxs = contexts_of(lines[0])
+    assert name == 'get'
+    # nested indexing produces ArrayAccessExpr parents with child ids
+    assert any('ArrayAccessExpr' in c for c in ctxs)
+
+
+def test_switch_fallthrough(extractor, tmp_path):
+    code = '''
+class In {
+    int classify(int v) {
+        switch (v) {
+            case 0:
+            case 1:
+                return 10;
+            case 2:
+                v += 1;
+            default:
+                return v;
+        }
+    }
+}
+'''
+    lines = run_extract(extractor, code, tmp_path)
+    name, ctxs = contexts_of(lines[0])
+    assert name == 'classify'
+    assert any('SwitchStmt' in c for c in ctxs)
+    assert any('SwitchEntryStmt' in c for c in ctxs)
+
+
+def test_method_reference_and_varargs(extractor, tmp_path):
+    code = '''
+import java.util.List;
+class In {
+    void refs(List<String> xs) { xs.forEach(System.out::println); }
+    int sum(int... vals) {
+        int s = 0;
+        for (int v : vals) s += v;
+        return s;
+    }
+}
+'''
+    lines = run_extract(extractor, code, tmp_path)
+    names = [contexts_of(l)[0] for l in lines]
+    assert names == ['refs', 'sum']
+    assert any('ForeachStmt' in c for _, l in enumerate(lines)
+               for c in contexts_of(l)[1])
+
+
+def test_try_with_resources_and_string_switch(extractor, tmp_path):
+    code = '''
+class In {
+    String pick(String k) {
+        try (java.io.StringReader r = new java.io.StringReader(k)) {
+            switch (k) { case "a": return "x"; default: return "y"; }
+        }
+    }
+}
+'''
+    lines = run_extract(extractor, code, tmp_path)
+    name, ctxs = contexts_of(lines[0])
+    assert name == 'pick'
+    assert any('TryStmt' in c for c in ctxs)

@@ -157,3 +157,30 @@ def test_chunk_sharding_covers_all_rows(tmp_path):
     def key(t):
         return sorted((int(a), int(b)) for a, b in t.tolist())
     assert key(got) == key(reft)
+
+
+def test_shuffle_gather_matches_torch():
+    """Fused C++ concat+perm gather == torch.cat + index_select."""
+    from code2vec_amd.data.reader import _load_native_reader_module
+    mod = _load_native_reader_module()
+    if mod is None or not hasattr(mod, 'shuffle_gather'):
+        import pytest
+        pytest.skip('native reader module not built')
+    import torch
+    g = torch.Generator().manual_seed(3)
+    chunks = [5, 1, 7, 3]
+    fields = []
+    fields.append([torch.randint(0, 1000, (c, 11), dtype=torch.int32,
+                                 generator=g) for c in chunks])
+    fields.append([torch.rand(c, 11, generator=g) for c in chunks])
+    fields.append([torch.randint(0, 9, (c,), dtype=torch.int64, generator=g)
+                   for c in chunks])
+    n = sum(chunks)
+    perm = torch.randperm(n, generator=g)
+    outs = [torch.empty((n, 11), dtype=torch.int32),
+            torch.empty((n, 11)),
+            torch.empty((n,), dtype=torch.int64)]
+    mod.shuffle_gather(fields, perm, outs, 3)
+    for f, out in zip(fields, outs):
+        ref = torch.cat(f)[perm]
+        assert torch.equal(out, ref)
