@@ -1,20 +1,21 @@
 #!/usr/bin/env bash
 ###########################################################
-# Dataset preprocessing driver — the equivalent of the reference's
-# preprocess.sh (extract -> histograms -> truncate/pad/pickle), using the
-# native C++ extractor (extractor/c2v-extract) instead of the Java jar and
-# code2vec_amd.data.preprocess (which builds histograms itself, replacing the
-# awk pipeline).
+# C# dataset preprocessing driver — the equivalent of the reference's
+# preprocess_csharp.sh (extract -> histograms -> truncate/pad/pickle) using
+# the native C++ C# extractor (extractor/c2v-extract-cs, Roslyn-kind trees)
+# instead of `dotnet run`, and code2vec_amd.data.preprocess instead of the
+# awk histogram pipeline.
 #
-# TRAIN_DIR / VAL_DIR / TEST_DIR: directories of .java files (recursive).
+# TRAIN_DIR / VAL_DIR / TEST_DIR: directories of .cs files (recursive).
 ###########################################################
 set -euo pipefail
 
 TRAIN_DIR=${TRAIN_DIR:-my_train_dir}
 VAL_DIR=${VAL_DIR:-my_val_dir}
 TEST_DIR=${TEST_DIR:-my_test_dir}
-DATASET_NAME=${DATASET_NAME:-my_dataset}
+DATASET_NAME=${DATASET_NAME:-my_cs_dataset}
 MAX_CONTEXTS=${MAX_CONTEXTS:-200}
+MAX_EXTRACT_CONTEXTS=${MAX_EXTRACT_CONTEXTS:-30000}  # reservoir cap (Extractor.cs)
 WORD_VOCAB_SIZE=${WORD_VOCAB_SIZE:-1301136}
 PATH_VOCAB_SIZE=${PATH_VOCAB_SIZE:-911417}
 TARGET_VOCAB_SIZE=${TARGET_VOCAB_SIZE:-261245}
@@ -22,8 +23,8 @@ NUM_THREADS=${NUM_THREADS:-64}
 PYTHON=${PYTHON:-python3}
 
 HERE="$(cd "$(dirname "$0")/.." && pwd)"
-EXTRACTOR="${HERE}/extractor/c2v-extract"
-[ -x "${EXTRACTOR}" ] || make -C "${HERE}/extractor"
+EXTRACTOR="${HERE}/extractor/c2v-extract-cs"
+[ -x "${EXTRACTOR}" ] || make -C "${HERE}/extractor" c2v-extract-cs
 
 TRAIN_DATA_FILE=${DATASET_NAME}.train.raw.txt
 VAL_DATA_FILE=${DATASET_NAME}.val.raw.txt
@@ -31,25 +32,20 @@ TEST_DATA_FILE=${DATASET_NAME}.test.raw.txt
 
 mkdir -p data/${DATASET_NAME}
 
-# FAULT_TOLERANT=1 routes extraction through the batch driver
-# (code2vec_amd.data.extract_batch): per-directory retry with partial-output
-# cleanup, per-process kill timers, stuck-batch skip — the reference
-# JavaExtractor/extract.py semantics for dirty real-world corpora.
-if [ "${FAULT_TOLERANT:-0}" = "1" ]; then
-  extract() { PYTHONPATH="${HERE}" ${PYTHON} -m code2vec_amd.data.extract_batch \
-    --bin "${EXTRACTOR}" --dir "$1" --max_path_length 8 --max_path_width 2 \
-    --num_threads "${NUM_THREADS}"; }
-else
-  extract() { "${EXTRACTOR}" --dir "$1" --max_path_length 8 \
-    --max_path_width 2 --num_threads "${NUM_THREADS}"; }
-fi
-
 echo "Extracting paths from validation set..."
-extract "${VAL_DIR}" > "${VAL_DATA_FILE}"
+"${EXTRACTOR}" --path "${VAL_DIR}" --max_length 8 --max_width 2 \
+  --max_contexts "${MAX_EXTRACT_CONTEXTS}" --threads "${NUM_THREADS}" \
+  --ofile_name "${VAL_DATA_FILE}"
 echo "Extracting paths from test set..."
-extract "${TEST_DIR}" > "${TEST_DATA_FILE}"
+"${EXTRACTOR}" --path "${TEST_DIR}" --max_length 8 --max_width 2 \
+  --max_contexts "${MAX_EXTRACT_CONTEXTS}" --threads "${NUM_THREADS}" \
+  --ofile_name "${TEST_DATA_FILE}"
 echo "Extracting paths from training set..."
-extract "${TRAIN_DIR}" | shuf > "${TRAIN_DATA_FILE}"
+"${EXTRACTOR}" --path "${TRAIN_DIR}" --max_length 8 --max_width 2 \
+  --max_contexts "${MAX_EXTRACT_CONTEXTS}" --threads "${NUM_THREADS}" \
+  --ofile_name "${TRAIN_DATA_FILE}.unshuf"
+shuf "${TRAIN_DATA_FILE}.unshuf" > "${TRAIN_DATA_FILE}"
+rm -f "${TRAIN_DATA_FILE}.unshuf"
 
 echo "Preprocessing (histograms + truncate/pad + dictionaries)..."
 PYTHONPATH="${HERE}" ${PYTHON} -m code2vec_amd.data.preprocess \
