@@ -243,8 +243,11 @@ class Code2VecNetwork:
             from ..ops.reference import log_uniform_probs, sample_log_uniform
             sampled = sample_log_uniform(S, V, code_c.device)
             cand = torch.cat([labels, sampled])                  # (B+S,)
-            w_cand = self.target_shadow.index_select(0, cand)    # (B+S, D)
-            logits_cand = code_c @ w_cand.t()                    # (B, B+S)
+            # candidate logits/backward GEMMs run with the candidate-row
+            # gather fused into the kernels' B staging — no w_cand
+            # materialization, no torch GEMM in the sampled path
+            logits_cand = F.sampled_logits_gemm(code_c, self.target_shadow,
+                                                cand)            # (B, B+S)
             corr_true = torch.log(log_uniform_probs(labels, V) * S)
             corr_samp = torch.log(log_uniform_probs(sampled, V) * S)
             loss_rows, lse = F.sampled_ce_fwd(logits_cand, labels, sampled,
@@ -252,7 +255,7 @@ class Code2VecNetwork:
             loss = loss_rows.float().mean()
             d_cand = F.sampled_ce_bwd(logits_cand, labels, sampled, corr_true,
                                       corr_samp, lse, 1.0 / B)
-            d_target_rows = d_cand.t() @ code_c                  # (B+S, D)
+            d_target_rows = F.sampled_bwd_target_rows(d_cand, code_c)
             import os as _os
             if (reducer.world_size > 1
                     and _os.environ.get('C2V_DP_DEDUP', '1') == '1'):
@@ -262,7 +265,8 @@ class Code2VecNetwork:
             else:
                 cand_g, target_rows_g = reducer.allgather_sparse(
                     cand, d_target_rows)
-            d_code = (d_cand @ w_cand).float()                   # (B,D)
+            d_code = F.sampled_bwd_code(d_cand, self.target_shadow,
+                                        cand)                    # (B,D)
             d_target = None
         else:
             logits, loss_rows, lse = F.logits_ce_fused(code_c,

@@ -170,7 +170,11 @@ __global__ void k_gemm_bt(const ushort* __restrict__ A,
                           const ushort* __restrict__ Bt,
                           ushort* __restrict__ C, int N, int M, int K,
                           float keep_prob = 1.f, u64 seed_scalar = 0,
-                          const long* __restrict__ seed_ptr = nullptr) {
+                          const long* __restrict__ seed_ptr = nullptr,
+                          const long* __restrict__ b_rows = nullptr) {
+  // b_rows: optional row-gather on the Bt operand — Bt row r sources table
+  // row b_rows[r] (the sampled-softmax candidate GEMM: logits_cand =
+  // code @ gather(targets, cand)^T without materializing w_cand)
   const u64 seed = DROPOUT ? (seed_ptr ? (u64)*seed_ptr : seed_scalar) : 0;
   const float inv_keep = DROPOUT ? (1.f / keep_prob) : 1.f;
   __shared__ ushort lds_a[GEMM_BM * GEMM_BK];
@@ -226,7 +230,8 @@ __global__ void k_gemm_bt(const ushort* __restrict__ A,
               16, 0, 0);
         }
         {
-          const int grow = min(col0 + lrow, M - 1);
+          int grow = min(col0 + lrow, M - 1);
+          if (b_rows) grow = (int)b_rows[grow];
           const ushort* gp = Bt + (long)grow * K + k0 + kb;
           __builtin_amdgcn_global_load_lds(
               (const __attribute__((address_space(1))) u32*)gp,
@@ -755,7 +760,8 @@ __global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
                                  const float* __restrict__ lse = nullptr,
                                  const long* __restrict__ labels = nullptr,
                                  float ce_scale = 1.f,
-                                 ushort* __restrict__ DL = nullptr) {
+                                 ushort* __restrict__ DL = nullptr,
+                                 const long* __restrict__ b_rows = nullptr) {
   // double-buffered: A[2][128][36] + B[2][32][440] bf16 = 73 KiB dynamic.
   // 256 threads = 4 waves = 1 wave/SIMD: the per-SIMD register pool is 512
   // regs/lane, the only occupancy at which the 192-reg accumulator tile
@@ -840,11 +846,13 @@ __global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
     }
 #pragma unroll
     for (int r = 0; r < 6; ++r) {
+      long krow = k0 + b_kr[r];
+      if (b_rows) krow = b_rows[krow];
       if (b_c8[r] + 8 <= M) {
         rb[r] = *reinterpret_cast<const u16x8*>(
-            B + (long)(k0 + b_kr[r]) * M + b_c8[r]);
+            B + krow * M + b_c8[r]);
       } else {
-        const long base = (long)(k0 + b_kr[r]) * M;
+        const long base = krow * M;
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           rb[r][j] = (b_c8[r] + j < M) ? B[base + b_c8[r] + j] : (ushort)0;
@@ -948,10 +956,11 @@ __global__ void k_gemm_nn_splitk(const ushort* __restrict__ A,
 #pragma unroll
     for (int r = 0; r < 6; ++r) {
       const int gk = k0 + b_kr[r];
+      const long bk = (gk < K && b_rows) ? b_rows[gk] : gk;
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         NLDS_B(0)[b_kr[r] * GNN_PKB + GNN_BROT(b_kr[r]) + b_c8[r] + j] =
-            (gk < K && b_c8[r] + j < M) ? B[(long)gk * M + b_c8[r] + j]
+            (gk < K && b_c8[r] + j < M) ? B[bk * M + b_c8[r] + j]
                                         : (ushort)0;
     }
     __syncthreads();
@@ -2526,6 +2535,26 @@ torch::Tensor gemm_bt_bf16(torch::Tensor A, torch::Tensor Bt) {
   return gemm_bt(A, Bt, false, 1);
 }
 
+// Sampled-softmax candidate logits: C (N, n_idx) = A @ gather(table, idx)^T
+// — the gather rides the GEMM's B staging (no w_cand materialization).
+torch::Tensor gemm_bt_gather(torch::Tensor A, torch::Tensor table,
+                             torch::Tensor idx) {
+  CHECK_DEV(A); CHECK_CONT(A); CHECK_DEV(table); CHECK_CONT(table);
+  TORCH_CHECK(A.scalar_type() == torch::kBFloat16 &&
+              table.scalar_type() == torch::kBFloat16);
+  auto idx_c = idx.contiguous();
+  TORCH_CHECK(idx_c.scalar_type() == torch::kInt64, "idx must be int64");
+  const int N = A.size(0), K = A.size(1), M = (int)idx_c.numel();
+  TORCH_CHECK(table.size(1) == K && K % GEMM_BK == 0);
+  auto C = torch::empty({N, M}, A.options());
+  const int n_tiles = (N + GEMM_BM - 1) / GEMM_BM;
+  const int m_tiles = (M + GEMM_BN - 1) / GEMM_BN;
+  k_gemm_bt<false, false><<<n_tiles * m_tiles, 256, 0, cur_stream()>>>(
+      bf_ptr(A), bf_ptr(table), bf_ptr_mut(C), N, M, K, 1.f, 0, nullptr,
+      idx_c.data_ptr<long>());
+  return C;
+}
+
 torch::Tensor gemm_bt_v(torch::Tensor A, torch::Tensor Bt, bool tanh_ep,
                         int64_t variant) {
   return gemm_bt(A, Bt, tanh_ep, (int)variant);
@@ -2668,6 +2697,44 @@ torch::Tensor gemm_tn_splitk(torch::Tensor A, torch::Tensor B) {
 // K=vocab). Partials workspace (S, N, M) fp32 comes from the caching
 // allocator; S is a multiple of 8 so the XCD-grouped chunk decode is
 // bijective, sized so the grid is ~256 blocks (1 per CU).
+// Sampled-softmax d_code: C (N, M) fp32 = A @ gather(table, idx) — the
+// candidate-row gather rides the split-K nn GEMM's B staging.
+torch::Tensor gemm_nn_splitk_gather(torch::Tensor A, torch::Tensor table,
+                                    torch::Tensor idx) {
+  CHECK_DEV(A); CHECK_CONT(A); CHECK_DEV(table); CHECK_CONT(table);
+  TORCH_CHECK(A.scalar_type() == torch::kBFloat16 &&
+              table.scalar_type() == torch::kBFloat16);
+  auto idx_c = idx.contiguous();
+  TORCH_CHECK(idx_c.scalar_type() == torch::kInt64, "idx must be int64");
+  const int N = A.size(0), K = A.size(1), M = table.size(1);
+  TORCH_CHECK((int)idx_c.numel() == K, "idx length must equal A cols");
+  TORCH_CHECK(M <= GNN_BN && M % 8 == 0);
+  const int row_tiles = (N + GNN_BM - 1) / GNN_BM;
+  int S = 256 / (row_tiles * 8) * 8;
+  if (S < 8) S = 8;
+  const int total_ksteps = (K + GNN_BK - 1) / GNN_BK;
+  const int kpc = (total_ksteps + S - 1) / S;
+  auto P = torch::empty({(long)S, (long)N, (long)M},
+                        A.options().dtype(torch::kFloat32));
+  auto C = torch::empty({N, M}, A.options().dtype(torch::kFloat32));
+  const size_t lds = 2UL * (GNN_BM * GNN_PKA + GNN_BK * GNN_PKB) * 2;
+  static bool cfg_g = false;
+  if (!cfg_g) {
+    (void)hipFuncSetAttribute((const void*)k_gemm_nn_splitk<0>,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              (int)lds);
+    cfg_g = true;
+  }
+  k_gemm_nn_splitk<0><<<S * row_tiles, 256, lds, cur_stream()>>>(
+      bf_ptr(A), bf_ptr(table), P.data_ptr<float>(), N, M, K, S, kpc,
+      row_tiles, nullptr, nullptr, 1.f, nullptr, idx_c.data_ptr<long>());
+  const long total = (long)N * M;
+  TORCH_CHECK(total % 4 == 0);
+  k_splitk_reduce<<<grid_1d(total / 4, 256), 256, 0, cur_stream()>>>(
+      P.data_ptr<float>(), C.data_ptr<float>(), S, total);
+  return C;
+}
+
 torch::Tensor gemm_nn_splitk(torch::Tensor A, torch::Tensor B) {
   CHECK_DEV(A); CHECK_CONT(A); CHECK_DEV(B); CHECK_CONT(B);
   TORCH_CHECK(A.scalar_type() == torch::kBFloat16 &&
@@ -3212,6 +3279,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gemm_bt_dropout", &gemm_bt_dropout);
   mod.def("logits_ce_fused", &logits_ce_fused);
   mod.def("gemm_nn_splitk", &gemm_nn_splitk);
+  mod.def("gemm_bt_gather", &gemm_bt_gather);
+  mod.def("gemm_nn_splitk_gather", &gemm_nn_splitk_gather);
   mod.def("gemm_tn_bf16", &gemm_tn_bf16);
   mod.def("gemm_tn_ce", &gemm_tn_ce);
   mod.def("gemm_tn_splitk", &gemm_tn_splitk);

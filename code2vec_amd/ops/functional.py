@@ -337,6 +337,37 @@ def sparse_dedup_sum(ids, rows):
     return ref.sparse_dedup_sum(ids, rows)
 
 
+def sampled_logits_gemm(code_c, shadow, cand):
+    """Candidate logits for the sampled-softmax path: code @ gather(targets,
+    cand)^T with the row gather fused into the GEMM's B staging (BASELINE
+    config 4 deliverable: no torch GEMM and no w_cand materialization)."""
+    if (backend_for(code_c) == 'hip' and code_c.dtype == torch.bfloat16
+            and shadow.dtype == torch.bfloat16
+            and code_c.shape[1] % 32 == 0):
+        return hip_ext(True).gemm_bt_gather(code_c, shadow, cand)
+    return code_c @ shadow.index_select(0, cand).t()
+
+
+def sampled_bwd_code(d_cand, shadow, cand):
+    """d_code = d_cand @ gather(targets, cand) (fp32 out), gather fused into
+    the split-K nn GEMM's B staging."""
+    if (backend_for(d_cand) == 'hip' and d_cand.dtype == torch.bfloat16
+            and shadow.dtype == torch.bfloat16
+            and shadow.shape[1] <= 384 and shadow.shape[1] % 8 == 0):
+        return hip_ext(True).gemm_nn_splitk_gather(d_cand, shadow, cand)
+    return (d_cand @ shadow.index_select(0, cand)).float()
+
+
+def sampled_bwd_target_rows(d_cand, code_c):
+    """Per-candidate target-table grad rows: d_cand^T @ code (tn MFMA)."""
+    if (backend_for(d_cand) == 'hip' and d_cand.dtype == torch.bfloat16
+            and code_c.dtype == torch.bfloat16
+            and code_c.shape[1] <= 384 and code_c.shape[1] % 8 == 0
+            and d_cand.shape[1] % 8 == 0):
+        return hip_ext(True).gemm_tn_bf16(d_cand, code_c)
+    return d_cand.t() @ code_c
+
+
 def sampled_ce_fwd(logits_cand, labels, sampled, corr_true, corr_samp):
     if backend_for(logits_cand) == 'hip':
         return hip_ext(True).sampled_ce_fwd(logits_cand, labels, sampled,
