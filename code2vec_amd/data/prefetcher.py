@@ -46,10 +46,15 @@ class BatchPrefetcher:
         except BaseException as e:  # noqa: BLE001
             self._err = e
         finally:
-            try:
-                self.q.put_nowait(self._SENTINEL)
-            except queue.Full:
-                pass
+            # blocking-with-stop-check: the sentinel MUST reach the consumer
+            # on normal completion (a dropped sentinel deadlocks the train
+            # loop); under stop() the queue is being drained anyway
+            while not self._stopping:
+                try:
+                    self.q.put(self._SENTINEL, timeout=0.5)
+                    break
+                except queue.Full:
+                    pass
 
     def stop(self, timeout: float = 5.0):
         """Terminate the worker before interpreter shutdown (a daemon thread

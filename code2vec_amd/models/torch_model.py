@@ -222,6 +222,16 @@ class Code2VecModel(Code2VecModelBase):
                 window_examples = 0
                 multi_batch_start = time.time()
 
+            if (cfg.is_testing and self.rank == 0
+                    and cfg.NUM_TRAIN_BATCHES_TO_EVALUATE > 0
+                    and batch_num % cfg.NUM_TRAIN_BATCHES_TO_EVALUATE == 0
+                    and batch_num % save_every != 0):
+                # mid-epoch evaluation cadence (reference: every 1800 batches
+                # via ModelEvaluationCallback, keras_model.py:326-369)
+                results = self.evaluate()
+                if results is not None:
+                    self.log('After %d batches -- %s' % (batch_num, results))
+
             if batch_num % save_every == 0:
                 self._epochs_trained += cfg.SAVE_EVERY_EPOCHS
                 epoch_num = self._epochs_trained
