@@ -151,3 +151,60 @@ def test_model_loads_tf_release_checkpoint(tmp_path):
     cfg2.MODEL_LOAD_PATH = str(model_dir2 / 'saved_model_iter8')
     with pytest.raises(ValueError, match='shape'):
         Code2VecModel(cfg2)
+
+
+def test_reader_handles_snappy_compressed_blocks(tmp_path):
+    """TF's table writer may snappy-compress blocks; the reader must accept
+    compression type 1. Re-pack an uncompressed index with all-literal
+    snappy framing (valid per the format) and re-read it."""
+    import struct
+
+    rng = np.random.default_rng(9)
+    tensors = {'model/TRANSFORM': rng.standard_normal((8, 8)).astype(np.float32)}
+    prefix = str(tmp_path / 'ck')
+    tfb.write_checkpoint(prefix, tensors)
+
+    def snappy_compress_literals(data: bytes) -> bytes:
+        out = bytearray(tfb.write_varint(len(data)))
+        pos = 0
+        while pos < len(data):
+            chunk = data[pos:pos + 60]
+            out += bytes([(len(chunk) - 1) << 2]) + chunk
+            pos += len(chunk)
+        return bytes(out)
+
+    raw = open(prefix + '.index', 'rb').read()
+    # parse the original footer to find block handles
+    footer = raw[-48:]
+    p = 0
+    meta_off, p = tfb.read_varint(footer, p)
+    _meta_size, p = tfb.read_varint(footer, p)
+
+    # rebuild the file with every block snappy-compressed (handles encode
+    # the COMPRESSED sizes, so the index entry is re-encoded too)
+    new2 = bytearray()
+    data_content = raw[0:meta_off - 5]
+    comp = snappy_compress_literals(data_content)
+    new2 += comp + b'\x01'
+    new2 += struct.pack('<I', tfb.crc_mask(tfb.crc32c(comp + b'\x01')))
+    data_handle = (0, len(comp))
+    idx_content = tfb._build_block(
+        [(b'\xff', tfb.write_varint(data_handle[0])
+          + tfb.write_varint(data_handle[1]))])
+    icomp = snappy_compress_literals(idx_content)
+    idx_handle = (len(new2), len(icomp))
+    new2 += icomp + b'\x01'
+    new2 += struct.pack('<I', tfb.crc_mask(tfb.crc32c(icomp + b'\x01')))
+    meta_content = tfb._build_block([])
+    mh = (len(new2), len(meta_content))
+    new2 += meta_content + b'\x00'
+    new2 += struct.pack('<I', tfb.crc_mask(tfb.crc32c(meta_content + b'\x00')))
+    foot = tfb.write_varint(mh[0]) + tfb.write_varint(mh[1]) \
+        + tfb.write_varint(idx_handle[0]) + tfb.write_varint(idx_handle[1])
+    new2 += foot.ljust(40, b'\x00') + struct.pack('<Q', tfb.TABLE_MAGIC)
+    with open(prefix + '.index', 'wb') as f:
+        f.write(bytes(new2))
+
+    r = tfb.TFCheckpointReader(prefix)
+    got = r.get_tensor('model/TRANSFORM', verify=True)
+    assert np.array_equal(got, tensors['model/TRANSFORM'])
