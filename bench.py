@@ -169,9 +169,9 @@ def run_end_to_end(args, device, rank, world_size, reducer, distributed):
     torch.set_num_threads(min(8, os.cpu_count() or 8))
     cfg = make_config(device)
     cfg.SAMPLED_SOFTMAX_SIZE = args.sampled_softmax
-    # the windowed shuffle emits batches in drain bursts of ~sb/batch+8; a
-    # prefetch queue shallower than one burst starves the GPU between bursts
-    cfg.READER_QUEUE_DEPTH = int(os.environ.get('C2V_E2E_QDEPTH', 24))
+    # prefetch depth A/B on one box (r02_call10): 8 -> 244K ex/s,
+    # 24 -> 228K; the deeper queue only added pinned-memory pressure
+    cfg.READER_QUEUE_DEPTH = int(os.environ.get('C2V_E2E_QDEPTH', 8))
     n_gpus = world_size if distributed else 1
 
     data_path = os.path.join(os.environ.get('TMPDIR', '/tmp'),
