@@ -261,7 +261,7 @@ class Code2VecNetwork:
                     and _os.environ.get('C2V_DP_DEDUP', '1') == '1'):
                 # labels repeat across the candidate set: dedup before gather
                 (cand_g, target_rows_g), = reducer.allgather_sparse_dedup(
-                    [F.sparse_dedup_sum(cand, d_target_rows)])
+                    [F.sparse_dedup_sum(cand, d_target_rows)]).wait()
             else:
                 cand_g, target_rows_g = reducer.allgather_sparse(
                     cand, d_target_rows)
@@ -364,16 +364,19 @@ class Code2VecNetwork:
         tok_ids = torch.cat([src_ids.reshape(-1), tgt_ids.reshape(-1)])
         path_ids_flat = path_ids.reshape(-1)
         ctx_direct = reducer.world_size == 1
+        pending_gather = None
         if not ctx_direct:
             import os as _os
             if _os.environ.get('C2V_DP_DEDUP', '1') == '1':
                 # rank-local dedup+sum before the gather: ships each unique
-                # row once (3-4x fewer xGMI bytes on Zipf-shaped real ids)
+                # row once (3-5x fewer xGMI bytes on Zipf-shaped real ids);
+                # the gather is launched async and waited just before the
+                # sparse Adam so the dense w/a chain overlaps the comm
                 tok_e = F.sparse_dedup_sum_ctx(tok_ids, d_ctx, 0, 2 * dt, 2, dt)
                 path_e = F.sparse_dedup_sum_ctx(path_ids_flat, d_ctx,
                                                 dt, dt, 1, dt)
-                (tok_ids, tok_rows), (path_ids_flat, path_rows) = \
-                    reducer.allgather_sparse_dedup([tok_e, path_e])
+                pending_gather = reducer.allgather_sparse_dedup(
+                    [tok_e, path_e])
             else:
                 tok_rows = torch.cat([d_ctx[:, :dt], d_ctx[:, 2 * dt:]], dim=0)
                 path_rows = d_ctx[:, dt:2 * dt]
@@ -404,13 +407,9 @@ class Code2VecNetwork:
                 self.path_table, path_ids_flat, d_ctx, dt, dt, 1, dt,
                 self._adam_m['path_table'], self._adam_v['path_table'],
                 t, lr, b1, b2, eps, lrt_t=st_t)
-        else:
-            F.adam_sparse_rows_step(self.tok_table, tok_ids, tok_rows,
-                                    self._adam_m['tok_table'], self._adam_v['tok_table'],
-                                    t, lr, b1, b2, eps, lrt_t=st_t)
-            F.adam_sparse_rows_step(self.path_table, path_ids_flat, path_rows,
-                                    self._adam_m['path_table'], self._adam_v['path_table'],
-                                    t, lr, b1, b2, eps, lrt_t=st_t)
+        # dense w/a Adam before the sparse-table updates: under DP this work
+        # (and its small all-reduce waits) overlaps the in-flight sparse
+        # gather instead of queueing behind it
         reducer.wait('w')
         F.adam_dense_step(self.w, d_w, self._adam_m['w'], self._adam_v['w'],
                           t, lr, b1, b2, eps, lrt_t=st_t)
@@ -418,6 +417,16 @@ class Code2VecNetwork:
         F.adam_dense_step(self.a, d_a, self._adam_m['a'], self._adam_v['a'],
                           t, lr, b1, b2, eps, lrt_t=st_t)
         self._refresh_shadows(only_w=True)
+        if not ctx_direct:
+            if pending_gather is not None:
+                (tok_ids, tok_rows), (path_ids_flat, path_rows) = \
+                    pending_gather.wait()
+            F.adam_sparse_rows_step(self.tok_table, tok_ids, tok_rows,
+                                    self._adam_m['tok_table'], self._adam_v['tok_table'],
+                                    t, lr, b1, b2, eps, lrt_t=st_t)
+            F.adam_sparse_rows_step(self.path_table, path_ids_flat, path_rows,
+                                    self._adam_m['path_table'], self._adam_v['path_table'],
+                                    t, lr, b1, b2, eps, lrt_t=st_t)
         if use_sampled:
             F.adam_sparse_rows_step(self.target_table, cand_g, target_rows_g,
                                     self._adam_m['target_table'],

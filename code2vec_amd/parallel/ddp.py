@@ -25,6 +25,23 @@ import torch
 import torch.distributed as dist
 
 
+class _PendingSparseGather:
+    """Handle for in-flight sparse-grad all-gathers: callers enqueue
+    independent device work (the dense w/a Adam chain) between launch and
+    wait(), so the comm overlaps compute instead of fencing the stream."""
+
+    def __init__(self, pending):
+        self._pending = pending
+
+    def wait(self):
+        results = []
+        for h1, h2, ids_out, rows_out in self._pending:
+            h1.wait()
+            h2.wait()
+            results.append((torch.cat(ids_out), torch.cat(rows_out)))
+        return results
+
+
 def init_distributed_from_env(backend: Optional[str] = None) -> Tuple[int, int]:
     """Initialize torch.distributed from torchrun env vars if present.
     Returns (rank, world_size); (0, 1) when not distributed."""
@@ -161,7 +178,7 @@ class Reducer:
         t = torch.tensor(local, dtype=torch.int64)
         counts = [torch.empty_like(t) for _ in range(ws)]
         dist.all_gather(counts, t, group=self._host_pg())
-        results = []
+        pending = []
         for j, (ids, rows, _) in enumerate(entries):
             c = local[j]
             n_max = max(int(cr[j]) for cr in counts)
@@ -179,10 +196,12 @@ class Reducer:
                 wire_rows[c:] = 0
             ids_out = [torch.empty_like(wire_ids) for _ in range(ws)]
             rows_out = [torch.empty_like(wire_rows) for _ in range(ws)]
-            dist.all_gather(ids_out, wire_ids, group=self.group)
-            dist.all_gather(rows_out, wire_rows, group=self.group)
-            results.append((torch.cat(ids_out), torch.cat(rows_out)))
-        return results
+            h1 = dist.all_gather(ids_out, wire_ids, group=self.group,
+                                 async_op=True)
+            h2 = dist.all_gather(rows_out, wire_rows, group=self.group,
+                                 async_op=True)
+            pending.append((h1, h2, ids_out, rows_out))
+        return _PendingSparseGather(pending)
 
     # -- termination consensus --------------------------------------------
     # Ranks can end an epoch with unequal batch counts (the DP shard split is

@@ -176,3 +176,60 @@ def test_dp2_ragged_shard_terminates(tmp_path):
     init_file = str(tmp_path / 'init_ragged')
     mp.spawn(_worker_ragged, args=(2, init_file, str(tmp_path)), nprocs=2,
              join=True)
+
+
+def _worker_sampled(rank, world_size, init_file, result_dir):
+    os.environ['C2V_DP_DEDUP'] = '1'
+    dist.init_process_group('gloo', init_method='file://' + init_file,
+                            rank=rank, world_size=world_size)
+    from code2vec_amd.parallel.ddp import Reducer
+    import code2vec_amd.ops.reference as RR
+    # pin the negative draw: all ranks (and the DP=1 oracle) share it
+    fixed = torch.arange(2, 10, dtype=torch.int64)
+    RR.sample_log_uniform = lambda n, V, device, generator=None: fixed.clone()
+    torch.manual_seed(7)
+    cfg = tiny_cfg()
+    cfg.SAMPLED_SOFTMAX_SIZE = 8
+    net = Code2VecNetwork(cfg, V_TOK, V_PATH, V_TGT, device='cpu')
+    src, pth, tgt, mask, labels = make_batch()
+    sl = slice(rank * (B // world_size), (rank + 1) * (B // world_size))
+    reducer = Reducer()
+    for _ in range(3):
+        net.train_step(src[sl], pth[sl], tgt[sl], mask[sl], labels[sl],
+                       reducer=reducer)
+    if rank == 0:
+        torch.save(net.state_dict(), os.path.join(result_dir, 'dps.pt'))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+def test_dp2_sampled_softmax_matches_dp1(tmp_path):
+    """Sampled-softmax DP: candidate-row grads are deduped+gathered like the
+    embedding tables; with a pinned negative draw DP=2 must match DP=1.
+    NOTE the loss normalization is per-rank batch (1/B_local), so the DP
+    semantic is the mean of per-rank mean losses — the DP=1 oracle uses
+    batch B with scale 1/B which matches when shards are equal."""
+    os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+    init_file = str(tmp_path / 'pg_init_s')
+    mp.spawn(_worker_sampled, args=(2, init_file, str(tmp_path)), nprocs=2,
+             join=True)
+
+    import code2vec_amd.ops.reference as RR
+    fixed = torch.arange(2, 10, dtype=torch.int64)
+    orig = RR.sample_log_uniform
+    RR.sample_log_uniform = lambda n, V, device, generator=None: fixed.clone()
+    try:
+        torch.manual_seed(7)
+        cfg = tiny_cfg()
+        cfg.SAMPLED_SOFTMAX_SIZE = 8
+        net1 = Code2VecNetwork(cfg, V_TOK, V_PATH, V_TGT, device='cpu')
+        src, pth, tgt, mask, labels = make_batch()
+        for _ in range(3):
+            net1.train_step(src, pth, tgt, mask, labels)
+    finally:
+        RR.sample_log_uniform = orig
+
+    dps = torch.load(str(tmp_path / 'dps.pt'), weights_only=False)
+    for name in net1.param_names():
+        assert torch.allclose(net1.get_param(name), dps[name], atol=1e-5), name
