@@ -99,3 +99,34 @@ def test_dir_mode(cs_extractor, tmp_path):
                           '--threads', '4'], capture_output=True, text=True)
     lines = [l for l in out.stdout.splitlines() if l]
     assert len(lines) == 6
+
+
+def test_properties_and_interpolated_strings_dont_break_file(cs_extractor,
+                                                             tmp_path):
+    """Files with property accessors and interpolated strings (documented
+    dialect gaps) must still yield their ordinary methods — a dialect gap
+    may cost its construct, never the whole file. (The reference's Roslyn
+    extractor enumerates MethodDeclarationSyntax, so property accessors are
+    not method examples there either.)"""
+    code = '''
+using System;
+class P {
+    private int _x;
+    public int X {
+        get { return _x + 1; }
+        set { _x = value; }
+    }
+    public string Name { get; set; }
+    public int Add(int a, int b) { return a + b; }
+    public string Greet(string who) { return $"hello {who} ({_x})"; }
+}
+'''
+    src = tmp_path / 'P.cs'
+    src.write_text(code)
+    import subprocess
+    out = subprocess.run([cs_extractor, '--path', str(src), '--max_length',
+                          '8', '--max_width', '2', '--no_hash'],
+                         capture_output=True, text=True)
+    assert out.returncode == 0, out.stderr
+    names = [l.split(' ')[0] for l in out.stdout.strip().split('\n') if l]
+    assert names == ['add', 'greet']
