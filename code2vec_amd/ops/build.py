@@ -56,6 +56,42 @@ def build(verbose: bool = True) -> str:
     return dest
 
 
+def build_asan(verbose: bool = True) -> str:
+    """Device-ASAN instrumented build of the kernel extension
+    (gfx950:xnack+): the GPU-side sanitizer pass. Loaded only by
+    tests/test_gpu_sanitizer.py in a subprocess with HSA_XNACK=1 and the
+    host ASAN runtime LD_PRELOADed."""
+    os.environ['PYTORCH_ROCM_ARCH'] = 'gfx950:xnack+'
+    here = os.path.dirname(os.path.abspath(__file__))
+    build_dir = os.path.join(here, '_build_asan')
+    os.makedirs(build_dir, exist_ok=True)
+    src = os.path.join(here, 'csrc', 'c2v_kernels.hip')
+    dest = os.path.join(here, '_c2v_hip_asan.so')
+    try:
+        if _is_fresh(dest, src):
+            return dest
+        from torch.utils import cpp_extension
+        cpp_extension.load(
+            name='_c2v_hip_asan',
+            sources=[src],
+            build_directory=build_dir,
+            extra_cflags=['-O1', '-g', '-fsanitize=address',
+                          '-shared-libsan'],
+            extra_cuda_cflags=['-O1', '-g', '-std=c++17',
+                               '-fsanitize=address', '-shared-libsan'],
+            extra_ldflags=['-fsanitize=address', '-shared-libsan',
+                           '-L/opt/rocm/lib/llvm/lib/clang/22/lib/linux',
+                           '-Wl,-rpath,/opt/rocm/lib/llvm/lib/clang/22/lib/linux'],
+            verbose=verbose,
+            is_python_module=False,
+            with_cuda=True,
+        )
+        shutil.copy2(os.path.join(build_dir, '_c2v_hip_asan.so'), dest)
+    finally:
+        os.environ['PYTORCH_ROCM_ARCH'] = 'gfx950'
+    return dest
+
+
 def build_reader(verbose: bool = True) -> str:
     """CPU-only native reader extension (no HIP): built into
     code2vec_amd/data/_c2v_reader.so."""
