@@ -70,23 +70,32 @@ def build_asan(verbose: bool = True) -> str:
     try:
         if _is_fresh(dest, src):
             return dest
-        from torch.utils import cpp_extension
-        cpp_extension.load(
-            name='_c2v_hip_asan',
-            sources=[src],
-            build_directory=build_dir,
-            extra_cflags=['-O1', '-g', '-fsanitize=address',
-                          '-shared-libsan'],
-            extra_cuda_cflags=['-O1', '-g', '-std=c++17',
-                               '-fsanitize=address', '-shared-libsan'],
-            extra_ldflags=['-fsanitize=address', '-shared-libsan',
-                           '-L/opt/rocm/lib/llvm/lib/clang/22/lib/linux',
-                           '-Wl,-rpath,/opt/rocm/lib/llvm/lib/clang/22/lib/linux'],
-            verbose=verbose,
-            is_python_module=False,
-            with_cuda=True,
-        )
-        shutil.copy2(os.path.join(build_dir, '_c2v_hip_asan.so'), dest)
+        # The build runs in a SACRIFICIAL subprocess: cpp_extension.load
+        # dlopens the result after building, and loading an ASAN .so into a
+        # non-ASAN python _exit()s the process ("ASan runtime does not come
+        # first"). The ninja build completes before that, so the artifact
+        # survives; consumers LD_PRELOAD the runtime.
+        import subprocess
+        import sys
+        code = (
+            "from torch.utils import cpp_extension\n"
+            "cpp_extension.load(name='_c2v_hip_asan', sources=[%r],\n"
+            "    build_directory=%r,\n"
+            "    extra_cflags=['-O1', '-g'],\n"
+            "    extra_cuda_cflags=['-O1', '-g', '-std=c++17',\n"
+            "                       '-fsanitize=address', '-shared-libsan'],\n"
+            "    extra_ldflags=['-L/opt/rocm/lib/llvm/lib/clang/22/lib/linux',\n"
+            "                   '-lclang_rt.asan-x86_64',\n"
+            "                   '-Wl,-rpath,/opt/rocm/lib/llvm/lib/clang/22/lib/linux'],\n"
+            "    verbose=%r, is_python_module=False, with_cuda=True)\n"
+        ) % (src, build_dir, verbose)
+        env = dict(os.environ, PYTORCH_ROCM_ARCH='gfx950:xnack+')
+        subprocess.run([sys.executable, '-c', code], env=env,
+                       capture_output=not verbose)
+        built = os.path.join(build_dir, '_c2v_hip_asan.so')
+        if not os.path.isfile(built):
+            raise RuntimeError('ASAN extension build failed')
+        shutil.copy2(built, dest)
     finally:
         os.environ['PYTORCH_ROCM_ARCH'] = 'gfx950'
     return dest
