@@ -21,7 +21,7 @@ DRIVER = r'''
 import importlib.util
 import torch
 
-spec = importlib.util.spec_from_file_location('c2v_asan', {so!r})
+spec = importlib.util.spec_from_file_location('_c2v_hip_asan', {so!r})
 ext = importlib.util.module_from_spec(spec)
 spec.loader.exec_module(ext)
 torch.manual_seed(0)
