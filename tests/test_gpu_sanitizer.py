@@ -64,7 +64,8 @@ def test_hot_kernels_under_device_asan():
     env = dict(os.environ,
                LD_PRELOAD=ASAN_RT,
                HSA_XNACK='1',
-               ASAN_OPTIONS='detect_leaks=0:abort_on_error=0')
+               ASAN_OPTIONS='detect_leaks=0:abort_on_error=0:'
+                            'allocator_may_return_null=1')
     proc = subprocess.run(
         [sys.executable, '-c', DRIVER.format(so=ASAN_SO)],
         env=env, capture_output=True, text=True, timeout=600)
