@@ -111,12 +111,16 @@ class Code2VecNetwork:
         self._seed_t = None
         self._step_t = None
         self._side_stream = None
+        self._hash_stream = None
         if self.device.type == 'cuda':
             self._seed_t = torch.zeros(1, dtype=torch.int64, device=self.device)
             self._step_t = torch.zeros(1, dtype=torch.int32, device=self.device)
             # side stream for the independent target-table chain
             # (d_target GEMM -> all-reduce -> Adam) overlapping main backward
             self._side_stream = torch.cuda.Stream()
+            # second side stream: the sparse-grad hash build depends only on
+            # the input ids, so it runs under the FORWARD pass
+            self._hash_stream = torch.cuda.Stream()
 
     # ---- parameters ----
 
@@ -228,6 +232,28 @@ class Code2VecNetwork:
         cfg = self.config
         B, C = src_ids.shape
         D = cfg.CODE_VECTOR_SIZE
+        dt = cfg.TOKEN_EMBEDDINGS_SIZE
+
+        # Launch the sparse-grad hash build (claim/compact/lookup over the
+        # token and path ids) on its own stream NOW: it has no gradient
+        # dependency, so it hides under the forward GEMMs instead of
+        # serializing in the backward tail (~0.45 ms/step on Zipf ids).
+        tok_ids = torch.cat([src_ids.reshape(-1), tgt_ids.reshape(-1)])
+        path_ids_flat = path_ids.reshape(-1)
+        tok_state = path_state = None
+        hash_done = None
+        if self._hash_stream is not None:
+            ev = torch.cuda.Event()
+            ev.record()
+            with torch.cuda.stream(self._hash_stream):
+                self._hash_stream.wait_event(ev)
+                tok_state = F.sparse_hash_build(tok_ids)
+                path_state = F.sparse_hash_build(path_ids_flat)
+                hash_done = torch.cuda.Event()
+                hash_done.record()
+            if not torch.cuda.is_current_stream_capturing():
+                tok_ids.record_stream(self._hash_stream)
+                path_ids_flat.record_stream(self._hash_stream)
 
         st = self.forward(src_ids, path_ids, tgt_ids, valid_mask, training=True)
         code_c = st.code.to(self.compute_dtype)
@@ -360,11 +386,14 @@ class Code2VecNetwork:
         # all-gathered (not dense-all-reduced) — SURVEY §2.4. Single-process
         # runs skip row materialization entirely (grads read straight out of
         # the d_ctx slices by the scatter-Adam kernel).
-        dt = cfg.TOKEN_EMBEDDINGS_SIZE
-        tok_ids = torch.cat([src_ids.reshape(-1), tgt_ids.reshape(-1)])
-        path_ids_flat = path_ids.reshape(-1)
         ctx_direct = reducer.world_size == 1
         pending_gather = None
+        if hash_done is not None:
+            torch.cuda.current_stream().wait_event(hash_done)
+            if not torch.cuda.is_current_stream_capturing():
+                cur = torch.cuda.current_stream()
+                for t in (tok_state or ()) + (path_state or ()):
+                    t.record_stream(cur)
         if not ctx_direct:
             import os as _os
             if _os.environ.get('C2V_DP_DEDUP', '1') == '1':
@@ -372,9 +401,16 @@ class Code2VecNetwork:
                 # row once (3-5x fewer xGMI bytes on Zipf-shaped real ids);
                 # the gather is launched async and waited just before the
                 # sparse Adam so the dense w/a chain overlaps the comm
-                tok_e = F.sparse_dedup_sum_ctx(tok_ids, d_ctx, 0, 2 * dt, 2, dt)
-                path_e = F.sparse_dedup_sum_ctx(path_ids_flat, d_ctx,
-                                                dt, dt, 1, dt)
+                if tok_state is not None:
+                    tok_e = F.sparse_dedup_sum_ctx_pre(tok_state, d_ctx,
+                                                       0, 2 * dt, 2, dt)
+                    path_e = F.sparse_dedup_sum_ctx_pre(path_state, d_ctx,
+                                                        dt, dt, 1, dt)
+                else:
+                    tok_e = F.sparse_dedup_sum_ctx(tok_ids, d_ctx,
+                                                   0, 2 * dt, 2, dt)
+                    path_e = F.sparse_dedup_sum_ctx(path_ids_flat, d_ctx,
+                                                    dt, dt, 1, dt)
                 pending_gather = reducer.allgather_sparse_dedup(
                     [tok_e, path_e])
             else:
@@ -399,14 +435,24 @@ class Code2VecNetwork:
         t, lr = self.adam_step, cfg.ADAM_LR
         b1, b2, eps = cfg.ADAM_BETA1, cfg.ADAM_BETA2, cfg.ADAM_EPS
         if ctx_direct:
-            F.adam_sparse_rows_from_ctx(
-                self.tok_table, tok_ids, d_ctx, 0, 2 * dt, 2, dt,
-                self._adam_m['tok_table'], self._adam_v['tok_table'],
-                t, lr, b1, b2, eps, lrt_t=st_t)
-            F.adam_sparse_rows_from_ctx(
-                self.path_table, path_ids_flat, d_ctx, dt, dt, 1, dt,
-                self._adam_m['path_table'], self._adam_v['path_table'],
-                t, lr, b1, b2, eps, lrt_t=st_t)
+            if tok_state is not None:
+                F.adam_sparse_rows_from_ctx_pre(
+                    self.tok_table, tok_state, d_ctx, 0, 2 * dt, 2, dt,
+                    self._adam_m['tok_table'], self._adam_v['tok_table'],
+                    t, lr, b1, b2, eps, lrt_t=st_t)
+                F.adam_sparse_rows_from_ctx_pre(
+                    self.path_table, path_state, d_ctx, dt, dt, 1, dt,
+                    self._adam_m['path_table'], self._adam_v['path_table'],
+                    t, lr, b1, b2, eps, lrt_t=st_t)
+            else:
+                F.adam_sparse_rows_from_ctx(
+                    self.tok_table, tok_ids, d_ctx, 0, 2 * dt, 2, dt,
+                    self._adam_m['tok_table'], self._adam_v['tok_table'],
+                    t, lr, b1, b2, eps, lrt_t=st_t)
+                F.adam_sparse_rows_from_ctx(
+                    self.path_table, path_ids_flat, d_ctx, dt, dt, 1, dt,
+                    self._adam_m['path_table'], self._adam_v['path_table'],
+                    t, lr, b1, b2, eps, lrt_t=st_t)
         # dense w/a Adam before the sparse-table updates: under DP this work
         # (and its small all-reduce waits) overlaps the in-flight sparse
         # gather instead of queueing behind it

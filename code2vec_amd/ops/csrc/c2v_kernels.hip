@@ -3211,6 +3211,60 @@ void adam_sparse_rows_hash_ctx(torch::Tensor p, torch::Tensor ids,
       (float)eps, lrt_ptr);
 }
 
+// Phase split for forward-overlap: the hash build (claim/compact/lookup)
+// depends only on the ids — launch it on a side stream at step START so it
+// hides under the forward GEMMs; the accumulate/Adam phases consume the
+// prebuilt state after d_ctx exists (~450 us off the backward tail).
+std::vector<torch::Tensor> sparse_hash_build(torch::Tensor ids) {
+  auto ids_c = ids.contiguous();
+  auto st = hash_dedup_ids(ids_c);
+  return {st.uniq, st.inverse, st.n_uniq, st.hot2cidx, st.n_hot};
+}
+
+static DedupState state_from(torch::Tensor uniq, torch::Tensor inverse,
+                             torch::Tensor n_uniq, torch::Tensor hot2cidx,
+                             torch::Tensor n_hot) {
+  return {uniq, inverse, n_uniq, hot2cidx, n_hot};
+}
+
+void adam_sparse_rows_hash_ctx_pre(
+    torch::Tensor p, torch::Tensor uniq, torch::Tensor inverse,
+    torch::Tensor n_uniq, torch::Tensor hot2cidx, torch::Tensor n_hot,
+    torch::Tensor d_ctx, int64_t off0, int64_t off1, int64_t n_seg,
+    int64_t d, torch::Tensor m, torch::Tensor v, int64_t step, double lr,
+    double beta1, double beta2, double eps, torch::Tensor lrt_t) {
+  CHECK_DEV(p); CHECK_CONT(p); CHECK_DEV(d_ctx); CHECK_CONT(d_ctx);
+  const float* lrt_ptr = (lrt_t.defined() && lrt_t.numel() == 1)
+                             ? lrt_t.data_ptr<float>() : nullptr;
+  const long n = inverse.numel();
+  const long n_per_seg = d_ctx.size(0);
+  TORCH_CHECK(n == n_per_seg * n_seg, "ids length mismatch");
+  auto st = state_from(uniq, inverse, n_uniq, hot2cidx, n_hot);
+  auto acc = accum_ctx_common(st, d_ctx, (int)off0, (int)off1, n_per_seg,
+                              (int)n_seg, (int)d);
+  const float lr_t = (float)(lr * std::sqrt(1.0 - std::pow(beta2, (double)step)) /
+                             (1.0 - std::pow(beta1, (double)step)));
+  k_adam_rows_dyn<<<grid_1d(n * d, 256), 256, 0, cur_stream()>>>(
+      p.data_ptr<float>(), st.uniq.data_ptr<long>(), acc.data_ptr<float>(),
+      m.data_ptr<float>(), v.data_ptr<float>(), nullptr,
+      st.n_uniq.data_ptr<int>(), (int)d, lr_t, (float)beta1, (float)beta2,
+      (float)eps, lrt_ptr);
+}
+
+std::vector<torch::Tensor> sparse_dedup_sum_ctx_pre(
+    torch::Tensor uniq, torch::Tensor inverse, torch::Tensor n_uniq,
+    torch::Tensor hot2cidx, torch::Tensor n_hot, torch::Tensor d_ctx,
+    int64_t off0, int64_t off1, int64_t n_seg, int64_t d) {
+  CHECK_DEV(d_ctx); CHECK_CONT(d_ctx);
+  const long n = inverse.numel();
+  const long n_per_seg = d_ctx.size(0);
+  TORCH_CHECK(n == n_per_seg * n_seg, "ids length mismatch");
+  auto st = state_from(uniq, inverse, n_uniq, hot2cidx, n_hot);
+  auto acc = accum_ctx_common(st, d_ctx, (int)off0, (int)off1, n_per_seg,
+                              (int)n_seg, (int)d);
+  return {st.uniq, acc, st.n_uniq};
+}
+
 std::vector<torch::Tensor> sparse_dedup_sum_ctx(torch::Tensor ids,
                                                 torch::Tensor d_ctx,
                                                 int64_t off0, int64_t off1,
@@ -3309,6 +3363,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("adam_sparse_rows_hash", &adam_sparse_rows_hash);
   mod.def("adam_sparse_rows_hash_ctx", &adam_sparse_rows_hash_ctx);
   mod.def("sparse_dedup_sum_ctx", &sparse_dedup_sum_ctx);
+  mod.def("sparse_hash_build", &sparse_hash_build);
+  mod.def("adam_sparse_rows_hash_ctx_pre", &adam_sparse_rows_hash_ctx_pre);
+  mod.def("sparse_dedup_sum_ctx_pre", &sparse_dedup_sum_ctx_pre);
   mod.def("sparse_dedup_sum_rows", &sparse_dedup_sum_rows);
   mod.def("sampled_ce_fwd", &sampled_ce_fwd);
   mod.def("sampled_ce_bwd", &sampled_ce_bwd);

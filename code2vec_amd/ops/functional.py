@@ -312,6 +312,37 @@ def adam_sparse_rows_from_ctx(p, ids, d_ctx, off0, off1, n_seg, d, m, v,
     ref.adam_sparse_rows_step(p, ids, rows, m, v, step, lr, beta1, beta2, eps)
 
 
+def sparse_hash_build(ids):
+    """Prebuild the sparse-grad dedup hash state from the ids alone (they
+    are known at step START, so this launches on a side stream under the
+    forward pass — the claim/compact/lookup passes leave the backward
+    tail). Returns an opaque state tuple on GPU; None on CPU (the eager
+    reference path recomputes its dedup inline, deterministically)."""
+    if backend_for(ids) == 'hip':
+        return tuple(hip_ext(True).sparse_hash_build(ids))
+    return None
+
+
+def adam_sparse_rows_from_ctx_pre(p, state, d_ctx, off0, off1, n_seg, d,
+                                  m, v, step: int, lr: float, beta1: float,
+                                  beta2: float, eps: float, lrt_t=None):
+    """adam_sparse_rows_from_ctx consuming a prebuilt hash state."""
+    hip_ext(True).adam_sparse_rows_hash_ctx_pre(
+        p, state[0], state[1], state[2], state[3], state[4], d_ctx,
+        int(off0), int(off1), int(n_seg), int(d), m, v, int(step), float(lr),
+        float(beta1), float(beta2), float(eps),
+        lrt_t if lrt_t is not None else torch.empty(0))
+
+
+def sparse_dedup_sum_ctx_pre(state, d_ctx, off0: int, off1: int, n_seg: int,
+                             d: int):
+    """sparse_dedup_sum_ctx consuming a prebuilt hash state (GPU only)."""
+    uniq, acc, n_uniq = hip_ext(True).sparse_dedup_sum_ctx_pre(
+        state[0], state[1], state[2], state[3], state[4], d_ctx,
+        int(off0), int(off1), int(n_seg), int(d))
+    return uniq, acc, n_uniq
+
+
 def sparse_dedup_sum_ctx(ids, d_ctx, off0: int, off1: int, n_seg: int, d: int):
     """Rank-local dedup+sum of embedding grad rows read straight from the
     (N,3d) d_ctx layout (DP wire-volume reduction — SURVEY §2.4). Returns
