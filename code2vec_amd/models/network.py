@@ -242,7 +242,9 @@ class Code2VecNetwork:
         path_ids_flat = path_ids.reshape(-1)
         tok_state = path_state = None
         hash_done = None
-        if self._hash_stream is not None:
+        import os as _os0
+        if (self._hash_stream is not None
+                and _os0.environ.get('C2V_HASH_OVERLAP', '1') == '1'):
             ev = torch.cuda.Event()
             ev.record()
             with torch.cuda.stream(self._hash_stream):
