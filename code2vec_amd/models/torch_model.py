@@ -94,6 +94,21 @@ class Code2VecModel(Code2VecModelBase):
         'model/ATTENTION': 'a',
     }
 
+    def _save_tf_checkpoint(self, prefix: str):
+        """Write the 5 model tensors as a TF Saver V2 bundle (weights-only
+        release form; Adam slots deliberately omitted like the reference's
+        `--release`)."""
+        import numpy as np
+        from ..utils.tf_bundle import write_checkpoint
+        net = self.network
+        tensors = {}
+        for tf_name, param_name in self.TF_NAME_MAP.items():
+            arr = net.get_param(param_name).detach().cpu().numpy()
+            if param_name == 'a':
+                arr = arr.reshape(-1, 1)        # TF stores ATTENTION (D, 1)
+            tensors[tf_name] = np.ascontiguousarray(arr, dtype=np.float32)
+        write_checkpoint(prefix, tensors)
+
     def _load_tf_checkpoint(self, prefix: str):
         """Load a reference-format TF V2 checkpoint (entire-model with Adam
         slots, or a weights-only `.release`) into the engine."""
@@ -136,7 +151,12 @@ class Code2VecModel(Code2VecModelBase):
             # weights only, optimizer state stripped (reference release flow)
             torch.save({'model': self.network.weights_state_dict(),
                         'epoch': self._epochs_trained}, path + '.release')
-            self.log('Released model saved to: %s' % (path + '.release'))
+            # also emit the TF Saver V2 form so the ORIGINAL reference code
+            # (tf.compat.v1.train.Saver.restore) can load models trained
+            # here — the inverse of the released-model import path
+            self._save_tf_checkpoint(path + '.release')
+            self.log('Released model saved to: %s (+ TF-format .index/.data)'
+                     % (path + '.release'))
             return
         torch.save({'model': self.network.state_dict(),
                     'epoch': self._epochs_trained},

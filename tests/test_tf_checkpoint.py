@@ -208,3 +208,33 @@ def test_reader_handles_snappy_compressed_blocks(tmp_path):
     r = tfb.TFCheckpointReader(prefix)
     got = r.get_tensor('model/TRANSFORM', verify=True)
     assert np.array_equal(got, tensors['model/TRANSFORM'])
+
+
+def test_release_emits_tf_format(tmp_path):
+    """`--release` also writes the Saver V2 files so the ORIGINAL reference
+    restores models trained here; the emitted bundle round-trips with the
+    expected graph-variable names and shapes."""
+    import sys
+    sys.path.insert(0, os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))))
+    from tests.test_model_cpu import tiny_train_config, write_dataset
+    from code2vec_amd.models.torch_model import Code2VecModel
+
+    prefix = write_dataset(tmp_path, n_train=16, n_test=4)
+    cfg = tiny_train_config(tmp_path, prefix)
+    cfg.RELEASE = True
+    model = Code2VecModel(cfg)
+    save_path = str(tmp_path / 'models' / 'rel_model')
+    os.makedirs(os.path.dirname(save_path), exist_ok=True)
+    model._save_inner_model(save_path)
+
+    assert os.path.isfile(save_path + '.release')          # torch form
+    assert os.path.isfile(save_path + '.release.index')    # TF form
+    r = tfb.TFCheckpointReader(save_path + '.release')
+    for tf_name, pname in Code2VecModel.TF_NAME_MAP.items():
+        arr = r.get_tensor(tf_name, verify=True)
+        want = model.network.get_param(pname).detach().cpu().numpy()
+        if pname == 'a':
+            want = want.reshape(-1, 1)
+        assert arr.shape == want.shape, tf_name
+        assert np.array_equal(arr, want), tf_name
