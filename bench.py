@@ -343,8 +343,11 @@ def main():
     distributed = world_size > 1
 
     if torch.cuda.is_available():
-        torch.cuda.set_device(local_rank)
-        device = 'cuda:%d' % local_rank
+        # modulo lets N ranks share fewer GPUs (RCCL-path validation on a
+        # 1-GPU box); on a full node local_rank < device_count anyway
+        dev_idx = local_rank % max(1, torch.cuda.device_count())
+        torch.cuda.set_device(dev_idx)
+        device = 'cuda:%d' % dev_idx
     else:
         device = 'cpu'
 
