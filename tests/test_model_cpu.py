@@ -93,6 +93,9 @@ def test_training_learns(trained):
     # the tiny dataset is nearly deterministic — top-10 accuracy should be high
     assert results.topk_acc[-1] > 0.8
     assert results.subtoken_f1 > 0.5
+    # eval loss is reported (reference Keras backend behavior) and sane for
+    # a model that has learned the 5-target task
+    assert results.loss is not None and 0.0 < results.loss < 2.0
 
 
 def test_checkpoint_roundtrip(trained, tmp_path):
