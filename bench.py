@@ -245,6 +245,11 @@ def run_end_to_end(args, device, rank, world_size, reducer, distributed):
         if device.startswith('cuda'):
             torch.cuda.synchronize()
 
+    if args.e2e_reader_only:
+        # isolate the data pipeline: consume batches without stepping
+        def do_step(b):  # noqa: F811
+            return None
+
     for _ in range(args.warmup):
         do_step(next(batches))
     barrier_sync()
@@ -289,6 +294,7 @@ def run_end_to_end(args, device, rank, world_size, reducer, distributed):
                 'target_vocab': cfg.MAX_TARGET_VOCAB_SIZE,
                 'id_dist': 'zipf-%s' % ZIPF_S,
                 'end_to_end': True,
+                'reader_only': bool(args.e2e_reader_only),
             },
         }))
     # orderly shutdown: signal the reader threads, then join the prefetch
@@ -318,6 +324,9 @@ def main():
                          'instead of resident synthetic batches')
     ap.add_argument('--e2e-rows', type=int, default=250000,
                     help='rows of on-disk data to generate for --end-to-end')
+    ap.add_argument('--e2e-reader-only', action='store_true',
+                    help='with --end-to-end: consume batches without '
+                         'stepping (isolates the data-pipeline ceiling)')
     args = ap.parse_args()
 
     # --gpus N without a torchrun rendezvous: self-launch one rank per GPU
