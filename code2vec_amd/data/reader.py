@@ -326,10 +326,17 @@ class PathContextReader:
                 # STRAIGHT INTO pinned memory: batches become views of one
                 # pinned block — no per-batch pin copy, and the ~1 ms/batch
                 # of GIL-holding torch CPU copies stops serializing against
-                # the training loop's launch thread
+                # the training loop's launch thread.
+                # Allocation size is ROUNDED UP to a fixed bucket: drain
+                # sizes vary by a chunk each time, and distinct sizes defeat
+                # the pinned caching allocator — every drain then pays a
+                # fresh multi-MB cudaHostAlloc (page-locking), which capped
+                # the whole pipeline at ~260K rows/s.
                 perm = torch.randperm(n, generator=g)
-                cat = [torch.empty((n,) + tuple(pool[0][i].shape[1:]),
-                                   dtype=pool[0][i].dtype, pin_memory=pin)
+                alloc_n = -(-n // 16384) * 16384
+                cat = [torch.empty((alloc_n,) + tuple(pool[0][i].shape[1:]),
+                                   dtype=pool[0][i].dtype,
+                                   pin_memory=pin)[:n]
                        for i in range(5)]
                 nat_gather([[p[i] for p in pool] for i in range(5)],
                            perm, cat, 6)
