@@ -154,16 +154,44 @@ struct Parser {
   // split here and parsed thread-parallel — no per-line python objects.
   py::tuple parse_buffer(py::bytes data) const {
     std::string_view buf = std::string_view(data);
-    // line offsets
+    return parse_views(buf, std::string_view());
+  }
+
+  // Zero-copy variant for the streaming reader: `body[:body_len]` is a raw
+  // read() chunk ending at a line boundary and `head` is the previous
+  // chunk's partial last line (prepended to the FIRST line here). The
+  // python-side `carry + chunk` concat and `buf[:last_nl+1]` slice this
+  // replaces moved ~8-12 MB per 4 MB chunk on one thread and capped the
+  // whole pipeline at ~260K rows/s.
+  py::tuple parse_buffer2(py::bytes body, long body_len,
+                          py::bytes head) const {
+    std::string_view body_sv = std::string_view(body).substr(
+        0, (size_t)std::max<long>(0, body_len));
+    return parse_views(body_sv, std::string_view(head));
+  }
+
+  py::tuple parse_views(std::string_view buf, std::string_view head) const {
+    // line offsets; a non-empty head splices onto buf's first line
+    std::string first_line;
     std::vector<std::pair<size_t, size_t>> lines;
+    bool has_first = false;
     size_t start = 0;
+    if (!head.empty()) {
+      const size_t nl = buf.find('\n');
+      first_line.assign(head);
+      first_line.append(buf.substr(0, nl == std::string_view::npos
+                                          ? buf.size() : nl));
+      has_first = !first_line.empty();
+      start = (nl == std::string_view::npos) ? buf.size() : nl + 1;
+    }
     while (start < buf.size()) {
       size_t nl = buf.find('\n', start);
       size_t end = (nl == std::string_view::npos) ? buf.size() : nl;
       if (end > start) lines.push_back({start, end - start});
       start = end + 1;
     }
-    const int64_t B = (int64_t)lines.size();
+    const int64_t nplain = (int64_t)lines.size();
+    const int64_t B = nplain + (has_first ? 1 : 0);
     const int64_t C = max_contexts;
     auto opts = torch::TensorOptions().dtype(torch::kInt32);
     auto src = torch::empty({B, C}, opts);
@@ -180,12 +208,16 @@ struct Parser {
       py::gil_scoped_release release;
       const int nt = (int)std::min<int64_t>(n_threads, std::max<int64_t>(1, B));
       std::atomic<int64_t> next(0);
+      const int64_t off = has_first ? 1 : 0;
+      if (has_first)
+        tidx_p[0] = parse_line(first_line, src_p, pth_p, tgt_p, mask_p);
       auto work = [&]() {
         int64_t i;
-        while ((i = next.fetch_add(1)) < B) {
-          tidx_p[i] = parse_line(buf.substr(lines[i].first, lines[i].second),
-                                 src_p + i * C, pth_p + i * C, tgt_p + i * C,
-                                 mask_p + i * C);
+        while ((i = next.fetch_add(1)) < nplain) {
+          const int64_t o = i + off;
+          tidx_p[o] = parse_line(buf.substr(lines[i].first, lines[i].second),
+                                 src_p + o * C, pth_p + o * C, tgt_p + o * C,
+                                 mask_p + o * C);
         }
       };
       std::vector<std::thread> pool;
@@ -317,5 +349,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
                     const std::unordered_map<std::string, int>&, int, int, int,
                     int, int, int, int>())
       .def("parse_batch", &Parser::parse_batch)
-      .def("parse_buffer", &Parser::parse_buffer);
+      .def("parse_buffer", &Parser::parse_buffer)
+      .def("parse_buffer2", &Parser::parse_buffer2);
 }
