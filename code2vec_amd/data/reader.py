@@ -430,6 +430,9 @@ class PathContextReader:
         raw_q: 'queue.Queue' = queue.Queue(maxsize=2 * n_workers + 2)
         out_q: 'queue.Queue' = queue.Queue(maxsize=2 * n_workers + 2)
         stop = threading.Event()
+        import time as _time
+        stats = ({'io': 0.0, 'parse': 0.0, 'drain': 0.0, 'emit': 0}
+                 if os.environ.get('C2V_READER_STATS') == '1' else None)
 
         def q_put(q, item) -> bool:
             while not stop.is_set():
@@ -452,6 +455,7 @@ class PathContextReader:
                     chunk_i = 0
                     with open(data_path, 'rb') as f:
                         carry = b''
+                        t_io = _time.perf_counter()
                         while not stop.is_set():
                             chunk = f.read(chunk_bytes)
                             if not chunk:
@@ -469,7 +473,12 @@ class PathContextReader:
                                 continue
                             item = ((head, chunk, last_nl + 1) if have_pb2
                                     else head + chunk[:last_nl + 1])
-                            if not q_put(raw_q, item):
+                            if stats is not None:
+                                stats['io'] += _time.perf_counter() - t_io
+                            ok = q_put(raw_q, item)
+                            if stats is not None:
+                                t_io = _time.perf_counter()
+                            if not ok:
                                 return
                         if carry.strip() and not stop.is_set() and (
                                 not chunk_shard
@@ -491,7 +500,10 @@ class PathContextReader:
                     if buf is None:
                         q_put(out_q, None)
                         return
+                    t0 = _time.perf_counter() if stats is not None else 0
                     tup = parse_filter(buf)
+                    if stats is not None:
+                        stats['parse'] += _time.perf_counter() - t0
                     if tup is not None and not q_put(out_q, tup):
                         return
             except BaseException as exc:  # noqa: BLE001
@@ -520,11 +532,24 @@ class PathContextReader:
                     raise item
                 pool.append(item)
                 pool_rows += item[0].shape[0]
-                yield from drain()
+                t0 = _time.perf_counter() if stats is not None else 0
+                for b in drain():
+                    if stats is not None:
+                        stats['drain'] += _time.perf_counter() - t0
+                        stats['emit'] += 1
+                    yield b
+                    t0 = _time.perf_counter() if stats is not None else 0
+                if stats is not None:
+                    stats['drain'] += _time.perf_counter() - t0
             if not stop.is_set():
                 yield from drain(final=True)
         finally:
             stop.set()
+            if stats is not None:
+                print('[reader stats] io_busy=%.2fs parse_busy=%.2fs '
+                      'drain_busy=%.2fs batches=%d' %
+                      (stats['io'], stats['parse'], stats['drain'],
+                       stats['emit']), flush=True)
 
     def stop_streaming(self, join: bool = True, timeout: float = 5.0):
         """Stop the stream-path threads. Call before interpreter shutdown:
