@@ -163,14 +163,15 @@ struct Parser {
   // python-side `carry + chunk` concat and `buf[:last_nl+1]` slice this
   // replaces moved ~8-12 MB per 4 MB chunk on one thread and capped the
   // whole pipeline at ~260K rows/s.
-  py::tuple parse_buffer2(py::bytes body, long body_len,
-                          py::bytes head) const {
+  py::tuple parse_buffer2(py::bytes body, long body_len, py::bytes head,
+                          long nt_override) const {
     std::string_view body_sv = std::string_view(body).substr(
         0, (size_t)std::max<long>(0, body_len));
-    return parse_views(body_sv, std::string_view(head));
+    return parse_views(body_sv, std::string_view(head), (int)nt_override);
   }
 
-  py::tuple parse_views(std::string_view buf, std::string_view head) const {
+  py::tuple parse_views(std::string_view buf, std::string_view head,
+                        int nt_override = 0) const {
     // line offsets; a non-empty head splices onto buf's first line
     std::string first_line;
     std::vector<std::pair<size_t, size_t>> lines;
@@ -206,7 +207,8 @@ struct Parser {
     int64_t* tidx_p = tidx.data_ptr<int64_t>();
     {
       py::gil_scoped_release release;
-      const int nt = (int)std::min<int64_t>(n_threads, std::max<int64_t>(1, B));
+      const int want_nt = nt_override > 0 ? nt_override : n_threads;
+      const int nt = (int)std::min<int64_t>(want_nt, std::max<int64_t>(1, B));
       std::atomic<int64_t> next(0);
       const int64_t off = has_first ? 1 : 0;
       if (has_first)
@@ -220,9 +222,14 @@ struct Parser {
                                  mask_p + o * C);
         }
       };
-      std::vector<std::thread> pool;
-      for (int t = 0; t < nt; ++t) pool.emplace_back(work);
-      for (auto& th : pool) th.join();
+      if (nt <= 1) {
+        work();  // caller-side parallelism (outer worker threads): no
+                 // per-call thread spawn/join
+      } else {
+        std::vector<std::thread> pool;
+        for (int t = 0; t < nt; ++t) pool.emplace_back(work);
+        for (auto& th : pool) th.join();
+      }
     }
     return py::make_tuple(src, pth, tgt, mask, tidx);
   }

@@ -406,8 +406,10 @@ class PathContextReader:
             nonlocal line_base
             if isinstance(use, tuple):
                 head, body, blen = use
+                # nt=1: parallelism comes from the OUTER worker threads; a
+                # per-call 6-thread spawn/join was the hidden per-chunk cost
                 src, pth, tgt, mask, tidx = self._native.parse_buffer2(
-                    body, blen, head)
+                    body, blen, head, 1)
             else:
                 src, pth, tgt, mask, tidx = self._native.parse_buffer(use)
             n = src.shape[0]
