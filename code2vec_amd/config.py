@@ -113,7 +113,7 @@ class Config:
         self.ADAM_EPS = 1e-8
         self.DP_BUCKET_BYTES = 4 << 20  # dense-grad all-reduce bucket size over xGMI
         self.READER_QUEUE_DEPTH = 8     # prefetched batches on the H2D copy stream
-        self.READER_WORKERS = 4         # parallel parse_buffer calls (stream path)
+        self.READER_WORKERS = 8         # parallel parse_buffer calls (stream path)
 
     def load_from_args(self):
         args = self.arguments_parser().parse_args()
@@ -184,7 +184,7 @@ class Config:
         self.ADAM_EPS: float = 1e-8
         self.DP_BUCKET_BYTES: int = 4 << 20
         self.READER_QUEUE_DEPTH: int = 8
-        self.READER_WORKERS: int = 4
+        self.READER_WORKERS: int = 8
 
         # Filled by the model base when datasets are counted
         self.NUM_TRAIN_EXAMPLES: int = 0

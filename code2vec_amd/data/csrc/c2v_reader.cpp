@@ -193,6 +193,10 @@ struct Parser {
 
   py::tuple parse_views(std::string_view buf, std::string_view head,
                         int nt_override = 0) const {
+    // GIL released for the whole body: line scanning, tensor allocation and
+    // parsing are pure C++ (the string_views borrow python buffers the
+    // caller keeps referenced); only the return-tuple build needs the GIL
+    py::gil_scoped_release release_all;
     // line offsets; a non-empty head splices onto buf's first line
     std::string first_line;
     std::vector<std::pair<size_t, size_t>> lines;
@@ -227,7 +231,6 @@ struct Parser {
     float* mask_p = mask.data_ptr<float>();
     int64_t* tidx_p = tidx.data_ptr<int64_t>();
     {
-      py::gil_scoped_release release;
       const int want_nt = nt_override > 0 ? nt_override : n_threads;
       const int nt = (int)std::min<int64_t>(want_nt, std::max<int64_t>(1, B));
       std::atomic<int64_t> next(0);
@@ -252,6 +255,7 @@ struct Parser {
         for (auto& th : pool) th.join();
       }
     }
+    py::gil_scoped_acquire acquire;
     return py::make_tuple(src, pth, tgt, mask, tidx);
   }
 
