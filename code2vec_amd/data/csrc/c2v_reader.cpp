@@ -120,6 +120,11 @@ struct Parser {
   }
 
   // Parse one line into the row buffers. Returns target index.
+  // Two-pass structure: tokenize first, then run the hash lookups with a
+  // software-prefetch window — the 3C lookups per row are independent
+  // random DRAM reads (the vocab tables dwarf every cache), and issuing
+  // them ~4 contexts ahead turns a serial miss chain into overlapped
+  // misses.
   int parse_line(std::string_view line, int* src_row, int* path_row,
                  int* tgt_row, float* mask_row) const {
     const int C = max_contexts;
@@ -138,6 +143,13 @@ struct Parser {
                                                  ? line.size() : pos);
     int target_idx = target.empty() ? tgt_oov : look(tgt, target, tgt_oov);
 
+    // pass 1: split fields into per-part views (thread-local scratch:
+    // parse_line runs on many threads; a per-row heap alloc would churn)
+    thread_local std::vector<std::string_view> f0, f1, f2;
+    f0.assign(C, {});
+    f1.assign(C, {});
+    f2.assign(C, {});
+    int n_ctx = 0;
     int c = 0;
     while (pos != std::string_view::npos && c < C) {
       size_t start = pos + 1;
@@ -145,28 +157,43 @@ struct Parser {
       std::string_view field = line.substr(
           start, (pos == std::string_view::npos ? line.size() : pos) - start);
       if (field.empty()) { ++c; continue; }
-      // split on ',' into up to 3 parts; missing parts stay PAD
       size_t c1 = field.find(',');
       size_t c2 = c1 == std::string_view::npos ? std::string_view::npos
                                                : field.find(',', c1 + 1);
-      std::string_view p0 = field.substr(0, c1);
-      std::string_view p1 = c1 == std::string_view::npos
-                                ? std::string_view()
-                                : field.substr(c1 + 1,
-                                               (c2 == std::string_view::npos
-                                                    ? field.size()
-                                                    : c2) - c1 - 1);
-      std::string_view p2 = c2 == std::string_view::npos
-                                ? std::string_view()
-                                : field.substr(c2 + 1);
-      int si = p0.empty() ? tok_pad : look(tok, p0, tok_oov);
-      int pi = p1.empty() ? path_pad : look(path, p1, path_oov);
-      int ti = p2.empty() ? tok_pad : look(tok, p2, tok_oov);
-      src_row[c] = si;
-      path_row[c] = pi;
-      tgt_row[c] = ti;
-      mask_row[c] = (si != tok_pad || ti != tok_pad || pi != path_pad) ? 1.f : 0.f;
+      f0[c] = field.substr(0, c1);
+      f1[c] = c1 == std::string_view::npos
+                  ? std::string_view()
+                  : field.substr(c1 + 1, (c2 == std::string_view::npos
+                                              ? field.size() : c2) - c1 - 1);
+      f2[c] = c2 == std::string_view::npos ? std::string_view()
+                                           : field.substr(c2 + 1);
       ++c;
+      n_ctx = c;
+    }
+
+    // pass 2: lookups with a prefetch window
+    constexpr int PF = 4;
+    auto prefetch3 = [&](int j) {
+      if (j >= n_ctx) return;
+      if (!f0[j].empty())
+        __builtin_prefetch(&tok.slots[StrMap::hash(f0[j]) & tok.mask]);
+      if (!f1[j].empty())
+        __builtin_prefetch(&path.slots[StrMap::hash(f1[j]) & path.mask]);
+      if (!f2[j].empty())
+        __builtin_prefetch(&tok.slots[StrMap::hash(f2[j]) & tok.mask]);
+    };
+    for (int j = 0; j < std::min(PF, n_ctx); ++j) prefetch3(j);
+    for (int j = 0; j < n_ctx; ++j) {
+      prefetch3(j + PF);
+      if (f0[j].empty() && f1[j].empty() && f2[j].empty()) continue;
+      int si = f0[j].empty() ? tok_pad : look(tok, f0[j], tok_oov);
+      int pi = f1[j].empty() ? path_pad : look(path, f1[j], path_oov);
+      int ti = f2[j].empty() ? tok_pad : look(tok, f2[j], tok_oov);
+      src_row[j] = si;
+      path_row[j] = pi;
+      tgt_row[j] = ti;
+      mask_row[j] = (si != tok_pad || ti != tok_pad || pi != path_pad)
+                        ? 1.f : 0.f;
     }
     return target_idx;
   }
