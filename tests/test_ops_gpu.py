@@ -678,3 +678,26 @@ def test_sampled_train_step_gpu_vs_cpu():
             assert abs(lg - lc) < 0.06 * max(1.0, abs(lc)), (step, lg, lc)
     finally:
         RR2.sample_log_uniform = orig_draw
+
+
+def test_sparse_hash_build_pre_matches_inline():
+    """The phase-split dedup (sparse_hash_build + sparse_dedup_sum_ctx_pre,
+    the DP path under the forward-overlap scheme) must produce the same
+    (unique ids, summed rows) as the single-call sparse_dedup_sum_ctx."""
+    e = ext()
+    torch.manual_seed(77)
+    N, d = 512, 32
+    d_ctx = (torch.randn(N, 3 * d, device='cuda') * 0.1).to(torch.bfloat16)
+    ids = torch.randint(0, 300, (2 * N,), dtype=torch.int32, device='cuda')
+    ids[::5] = 7
+    u1, acc1, c1 = e.sparse_dedup_sum_ctx(ids, d_ctx, 0, 2 * d, 2, d)
+    state = e.sparse_hash_build(ids)
+    u2, acc2, c2 = e.sparse_dedup_sum_ctx_pre(state[0], state[1], state[2],
+                                              state[3], state[4], d_ctx,
+                                              0, 2 * d, 2, d)
+    n1, n2 = int(c1.item()), int(c2.item())
+    assert n1 == n2
+    o1 = torch.argsort(u1[:n1])
+    o2 = torch.argsort(u2[:n2])
+    assert torch.equal(u1[:n1][o1], u2[:n2][o2])
+    assert (acc1[:n1][o1] - acc2[:n2][o2]).abs().max().item() < 1e-4
