@@ -19,6 +19,7 @@ target-table grad all-reduce is launched before the rest of backward runs).
 """
 
 import math
+import os
 from typing import Dict, NamedTuple, Optional
 
 import torch
@@ -242,9 +243,8 @@ class Code2VecNetwork:
         path_ids_flat = path_ids.reshape(-1)
         tok_state = path_state = None
         hash_done = None
-        import os as _os0
         if (self._hash_stream is not None
-                and _os0.environ.get('C2V_HASH_OVERLAP', '1') == '1'):
+                and os.environ.get('C2V_HASH_OVERLAP', '1') == '1'):
             ev = torch.cuda.Event()
             ev.record()
             with torch.cuda.stream(self._hash_stream):
@@ -284,9 +284,8 @@ class Code2VecNetwork:
             d_cand = F.sampled_ce_bwd(logits_cand, labels, sampled, corr_true,
                                       corr_samp, lse, 1.0 / B)
             d_target_rows = F.sampled_bwd_target_rows(d_cand, code_c)
-            import os as _os
             if (reducer.world_size > 1
-                    and _os.environ.get('C2V_DP_DEDUP', '1') == '1'):
+                    and os.environ.get('C2V_DP_DEDUP', '1') == '1'):
                 # labels repeat across the candidate set: dedup before gather
                 (cand_g, target_rows_g), = reducer.allgather_sparse_dedup(
                     [F.sparse_dedup_sum(cand, d_target_rows)]).wait()
@@ -397,8 +396,7 @@ class Code2VecNetwork:
                 for t in (tok_state or ()) + (path_state or ()):
                     t.record_stream(cur)
         if not ctx_direct:
-            import os as _os
-            if _os.environ.get('C2V_DP_DEDUP', '1') == '1':
+            if os.environ.get('C2V_DP_DEDUP', '1') == '1':
                 # rank-local dedup+sum before the gather: ships each unique
                 # row once (3-5x fewer xGMI bytes on Zipf-shaped real ids);
                 # the gather is launched async and waited just before the
