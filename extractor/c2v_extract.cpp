@@ -1460,6 +1460,7 @@ class Parser {
       return n;
     }
     if (is_kw("null")) {
+      lx_.advance();
       Node* n = ast_.mk("NullLiteralExpr");
       n->isNullLit = true;
       n->text = "null";

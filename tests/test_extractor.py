@@ -359,3 +359,69 @@ class In {
     name, ctxs = contexts_of(lines[0])
     assert name == 'pick'
     assert any('TryStmt' in c for c in ctxs)
+
+
+def test_null_literal_everywhere(extractor, tmp_path):
+    """Regression: the null-literal parse consumed no token, so ANY file
+    containing `null` produced zero output (whole-file loss)."""
+    code = '''
+class In {
+    Object plain() { return null; }
+    String concat(Object o) { return "x" + o + null; }
+    boolean isNull(Object o) { return o == null; }
+    void assign() { Object a = null; a = null; }
+}
+'''
+    lines = run_extract(extractor, code, tmp_path)
+    names = [contexts_of(l)[0] for l in lines]
+    assert names == ['plain', 'concat', 'is|null', 'assign']
+    assert any('NullLiteralExpr' in p for p in _paths(lines))
+
+
+def test_enum_ctor_inner_class_static_init_labels(extractor, tmp_path):
+    """Enums with methods, constructors, static initializers, inner classes
+    and labeled break/continue neither crash nor leak non-method members as
+    examples (reference extracts MethodDeclarations only)."""
+    code = '''
+enum Color { RED, GREEN;
+    int shade(int base) { return base * 2; } }
+class In {
+    static int counter;
+    static { counter = 5; }
+    In(int x) { counter += x; }
+    class Inner { int get() { return counter; } }
+    void loops() {
+        outer:
+        for (int i = 0; i < 3; i++) {
+            for (int j = 0; j < 3; j++) {
+                if (j == 1) continue outer;
+                if (i == 2) break outer;
+            }
+        }
+    }
+}
+'''
+    lines = run_extract(extractor, code, tmp_path)
+    names = [contexts_of(l)[0] for l in lines]
+    assert names == ['shade', 'get', 'loops']
+    assert any('LabeledStmt' in p for p in _paths(lines))
+
+
+def test_instanceof_cast_dowhile_escapes(extractor, tmp_path):
+    code = '''
+class In {
+    boolean check(Object o) { return o instanceof String && ((String) o).isEmpty(); }
+    int dw(int n) { int s = 0; do { s += n; n--; } while (n > 0); return s; }
+    String esc() { return "a,b\\n\\"c\\""; }
+}
+'''
+    lines = run_extract(extractor, code, tmp_path)
+    names = [contexts_of(l)[0] for l in lines]
+    assert names == ['check', 'dw', 'esc']
+    assert any('InstanceOfExpr' in p for p in _paths(lines))
+    assert any('DoStmt' in p for p in _paths(lines))
+    # string-literal tokens are normalized: no raw commas can corrupt the
+    # ctx1,ctx2,ctx3 wire format
+    for _, ctxs in map(contexts_of, lines):
+        for c in ctxs:
+            assert c.count(',') == 2, c
