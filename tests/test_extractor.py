@@ -425,3 +425,38 @@ class In {
     for _, ctxs in map(contexts_of, lines):
         for c in ctxs:
             assert c.count(',') == 2, c
+
+
+def test_empty_and_bodyless_methods_skipped(extractor, tmp_path):
+    """Reference parity: empty bodies have method length 0 and min_code_len
+    is 1 (FunctionVisitor.getMethodLength + ExtractFeaturesTask filter), so
+    `void e() {}` and abstract/native declarations yield no example."""
+    code = '''
+abstract class In {
+    void empty() { }
+    native void nat();
+    abstract int abs(int x);
+    int real() { return 1; }
+}
+'''
+    lines = run_extract(extractor, code, tmp_path)
+    assert [contexts_of(l)[0] for l in lines] == ['real']
+
+
+def test_deep_generics_multicatch_array_inits(extractor, tmp_path):
+    code = '''
+import java.util.*;
+class In {
+    Map<String, List<Map<Integer, String>>> deep() { return new HashMap<>(); }
+    void multicatch() {
+        try { deep(); } catch (RuntimeException | Error e) { e.toString(); }
+    }
+    int[][] init2() { int[][] a = {{1, 2}, {3}}; return a; }
+    void bounds(List<? super Integer> l, List<? extends Number> u) { l.add(1); }
+}
+'''
+    lines = run_extract(extractor, code, tmp_path)
+    # note: normalizeName/subtoken split drops digits -> init2 reads "init"
+    assert [contexts_of(l)[0] for l in lines] == ['deep', 'multicatch',
+                                                  'init', 'bounds']
+    assert any('ArrayInitializerExpr' in p for p in _paths(lines))
