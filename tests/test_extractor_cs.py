@@ -130,3 +130,32 @@ class P {
     assert out.returncode == 0, out.stderr
     names = [l.split(' ')[0] for l in out.stdout.strip().split('\n') if l]
     assert names == ['add', 'greet']
+
+
+def test_csharp_idiom_sweep(cs_extractor, tmp_path):
+    """Null literals, string switch, collection initializers, lambdas,
+    expression-bodied members, generic constraints, null-coalescing and
+    nullable types all extract (one example per method)."""
+    code = '''
+using System;
+using System.Collections.Generic;
+class Q {
+    object N() { return null; }
+    int Sw(string k) { switch (k) { case "a": return 1; default: return 0; } }
+    List<int> Gen() { var l = new List<int> { 1, 2 }; return l; }
+    int Lam(Func<int,int> f) { return f(3); }
+    void Use() { Lam(x => x * 2); }
+    int Tern(int a) => a > 0 ? 1 : 0;
+    T Cast<T>(object o) where T : class { return o as T; }
+    int Nullc(int? v) { return v ?? -1; }
+}
+'''
+    src = tmp_path / 'Q.cs'
+    src.write_text(code)
+    import subprocess
+    out = subprocess.run([cs_extractor, '--path', str(src), '--max_length',
+                          '8', '--max_width', '2', '--no_hash'],
+                         capture_output=True, text=True)
+    assert out.returncode == 0, out.stderr
+    names = [l.split(' ')[0] for l in out.stdout.strip().split('\n') if l]
+    assert names == ['n', 'sw', 'gen', 'lam', 'use', 'tern', 'cast', 'nullc']
