@@ -87,6 +87,16 @@ class Lexer {
     }
   }
 
+  // position save/restore for error recovery (the reference member makes
+  // the default copy-assignment unusable; copy-construct + restore_from
+  // covers the rewind use case)
+  void restore_from(const Lexer& other) {
+    cur_ = other.cur_;
+    peek_ = other.peek_;
+    has_peek_ = other.has_peek_;
+    i_ = other.i_;
+  }
+
  private:
   Token lex() {
     skip_ws_comments();
@@ -300,6 +310,11 @@ class Parser {
  public:
   Parser(const std::string& src, Ast& ast) : lx_(src), ast_(ast) {}
 
+  // members dropped by the per-member error recovery; the 3-stage retry
+  // treats a parse with skips as soft-failed (a later wrapping stage may
+  // recover MORE of the snippet)
+  int skipped_members() const { return skipped_members_; }
+
   Node* parse_compilation_unit() {
     Node* cu = ast_.mk("CompilationUnit");
     if (is_kw("package")) {
@@ -470,11 +485,50 @@ class Parser {
     std::vector<Node*> members;
     expect("{");
     while (!is_punct("}") && !at_end()) {
-      Node* m = parse_member();
-      if (m) members.push_back(m);
+      // Per-member error recovery: a construct this grammar doesn't know
+      // costs only its own member, not the whole file (JavaParser fails
+      // whole-file too, but ITS grammar has no gaps — ours might). On a
+      // parse error, rewind and skip one balanced member.
+      Lexer save = lx_;
+      try {
+        Node* m = parse_member();
+        if (m) members.push_back(m);
+      } catch (const ParseError&) {
+        lx_.restore_from(save);
+        skip_one_member();
+        ++skipped_members_;
+      }
     }
     expect("}");
     return members;
+  }
+
+  // Skip one member: advance to the first ';' at brace depth 0 or past one
+  // balanced '{...}' block (whichever comes first).
+  void skip_one_member() {
+    int depth = 0;
+    bool moved = false;
+    while (!at_end()) {
+      const Token& t = lx_.cur();
+      if (t.kind == Tok::Punct) {
+        if (t.text == "{") {
+          ++depth;
+        } else if (t.text == "}") {
+          if (depth == 0) return;  // enclosing class brace: stop before it
+          --depth;
+          lx_.advance();
+          if (depth == 0) return;  // finished the member's block
+          moved = true;
+          continue;
+        } else if (t.text == ";" && depth == 0) {
+          lx_.advance();
+          return;
+        }
+      }
+      lx_.advance();
+      moved = true;
+    }
+    (void)moved;
   }
 
   Node* parse_member() {
@@ -1535,6 +1589,7 @@ class Parser {
   }
 
   Lexer lx_;
+  int skipped_members_ = 0;
   Ast& ast_;
 };
 
@@ -1755,23 +1810,41 @@ static std::string extract_file_content(const std::string& code,
                                         const ExtractorOptions& opt) {
   Ast ast;
   Node* cu = nullptr;
-  // 3-stage parse retry (FeatureExtractor.java:51-75)
+  // 3-stage parse retry (FeatureExtractor.java:51-75). A stage that parsed
+  // but DROPPED members via the per-member recovery is only accepted if no
+  // later stage does better (most methods wins; clean parses win outright).
   std::vector<std::string> attempts = {
       code,
       "public class Test {SomeUnknownReturnType f() {" + code +
           "return noSuchReturnValue; }}",
       "public class Test {" + code + "}"};
+  Ast best_ast;
+  Node* best_cu = nullptr;
+  size_t best_methods = 0;
   for (const std::string& attempt : attempts) {
     Ast fresh;
     try {
       Parser p(attempt, fresh);
       Node* parsed = p.parse_compilation_unit();
-      ast.nodes_.swap(fresh.nodes_);
-      cu = parsed;
-      break;
+      if (p.skipped_members() == 0) {
+        ast.nodes_.swap(fresh.nodes_);
+        cu = parsed;
+        break;
+      }
+      std::vector<Node*> ms;
+      find_methods(parsed, ms);
+      if (best_cu == nullptr || ms.size() > best_methods) {
+        best_ast.nodes_.swap(fresh.nodes_);
+        best_cu = parsed;
+        best_methods = ms.size();
+      }
     } catch (const ParseError&) {
       continue;
     }
+  }
+  if (!cu && best_cu) {
+    ast.nodes_.swap(best_ast.nodes_);
+    cu = best_cu;
   }
   if (!cu) return "";
 

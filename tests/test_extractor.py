@@ -460,3 +460,18 @@ class In {
     assert [contexts_of(l)[0] for l in lines] == ['deep', 'multicatch',
                                                   'init', 'bounds']
     assert any('ArrayInitializerExpr' in p for p in _paths(lines))
+
+
+def test_unknown_construct_costs_only_its_method(extractor, tmp_path):
+    """Per-member error recovery: a grammar gap (here a Java-14 switch
+    expression) drops only the containing method; siblings survive. The
+    3-stage retry still prefers stages that parse cleanly."""
+    code = '''
+class In {
+    int good1() { return 1; }
+    void weird() { int x = switch (1) { case 1 -> 2; default -> 3; }; }
+    int good2() { return 2; }
+}
+'''
+    lines = run_extract(extractor, code, tmp_path)
+    assert [contexts_of(l)[0] for l in lines] == ['good', 'good']
