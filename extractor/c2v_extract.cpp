@@ -1833,7 +1833,9 @@ static std::string extract_file_content(const std::string& code,
       }
       std::vector<Node*> ms;
       find_methods(parsed, ms);
-      if (best_cu == nullptr || ms.size() > best_methods) {
+      // keep the FIRST stage that recovered any methods: later stages wrap
+      // the snippet in synthetic members that would leak into the output
+      if (best_cu == nullptr && !ms.empty()) {
         best_ast.nodes_.swap(fresh.nodes_);
         best_cu = parsed;
         best_methods = ms.size();
