@@ -396,7 +396,24 @@ class Code2VecNetwork:
                 for t in (tok_state or ()) + (path_state or ()):
                     t.record_stream(cur)
         if not ctx_direct:
-            if os.environ.get('C2V_DP_DEDUP', '1') == '1':
+            sparse_mode = os.environ.get('C2V_DP_SPARSE', 'dedup')
+            if sparse_mode == 'owner':
+                # owner-sharded reduce (opt-in): id%N all-to-all + disjoint
+                # reduced-shard gather — ~half the ring bytes of the dedup
+                # gather at java14m Zipf shapes (ddp.reduce_sparse_owner)
+                if tok_state is not None:
+                    tok_e = F.sparse_dedup_sum_ctx_pre(tok_state, d_ctx,
+                                                       0, 2 * dt, 2, dt)
+                    path_e = F.sparse_dedup_sum_ctx_pre(path_state, d_ctx,
+                                                        dt, dt, 1, dt)
+                else:
+                    tok_e = F.sparse_dedup_sum_ctx(tok_ids, d_ctx,
+                                                   0, 2 * dt, 2, dt)
+                    path_e = F.sparse_dedup_sum_ctx(path_ids_flat, d_ctx,
+                                                    dt, dt, 1, dt)
+                (tok_ids, tok_rows), (path_ids_flat, path_rows) = \
+                    reducer.reduce_sparse_owner([tok_e, path_e])
+            elif os.environ.get('C2V_DP_DEDUP', '1') == '1':
                 # rank-local dedup+sum before the gather: ships each unique
                 # row once (3-5x fewer xGMI bytes on Zipf-shaped real ids);
                 # the gather is launched async and waited just before the

@@ -203,6 +203,121 @@ class Reducer:
             pending.append((h1, h2, ids_out, rows_out))
         return _PendingSparseGather(pending)
 
+    def _all_to_all_rows(self, send_ids, send_rows):
+        """Exchange per-destination (ids, rows) lists: returns concatenated
+        receives. NCCL runs true all-to-all over xGMI (per-link parallel);
+        gloo (CPU tests) emulates with an all-gather + own-slice select."""
+        ws = self.world_size
+        rank = dist.get_rank(self.group)
+        if dist.get_backend(self.group) == 'nccl':
+            send_id_sizes = [t.numel() for t in send_ids]
+            # exchange sizes on the host group
+            sz = torch.tensor(send_id_sizes, dtype=torch.int64)
+            all_sz = [torch.empty_like(sz) for _ in range(ws)]
+            dist.all_gather(all_sz, sz, group=self._host_pg())
+            recv_sizes = [int(all_sz[src][rank]) for src in range(ws)]
+            d = send_rows[0].shape[1]
+            dev = send_rows[0].device
+            recv_ids = [torch.empty(n, dtype=torch.int64, device=dev)
+                        for n in recv_sizes]
+            recv_rows = [torch.empty(n, d, dtype=send_rows[0].dtype,
+                                     device=dev) for n in recv_sizes]
+            dist.all_to_all(recv_ids, list(send_ids), group=self.group)
+            dist.all_to_all(recv_rows, list(send_rows), group=self.group)
+            return torch.cat(recv_ids), torch.cat(recv_rows)
+        # gloo: all-gather every rank's full send set, keep what's ours
+        mine_ids, mine_rows = [], []
+        for dst in range(ws):
+            ids_d, rows_d = send_ids[dst], send_rows[dst]
+            n_local = torch.tensor([ids_d.numel()], dtype=torch.int64)
+            counts = [torch.zeros_like(n_local) for _ in range(ws)]
+            dist.all_gather(counts, n_local, group=self.group)
+            n_max = max(int(c.item()) for c in counts)
+            pad_ids = torch.zeros(n_max, dtype=torch.int64)
+            pad_ids[:ids_d.numel()] = ids_d
+            pad_rows = torch.zeros(n_max, send_rows[0].shape[1],
+                                   dtype=rows_d.dtype)
+            pad_rows[:rows_d.shape[0]] = rows_d
+            out_i = [torch.empty_like(pad_ids) for _ in range(ws)]
+            out_r = [torch.empty_like(pad_rows) for _ in range(ws)]
+            dist.all_gather(out_i, pad_ids, group=self.group)
+            dist.all_gather(out_r, pad_rows, group=self.group)
+            if dst == rank:
+                for src in range(ws):
+                    c = int(counts[src].item())
+                    mine_ids.append(out_i[src][:c])
+                    mine_rows.append(out_r[src][:c])
+        return torch.cat(mine_ids), torch.cat(mine_rows)
+
+    def reduce_sparse_owner(self, entries):
+        """Owner-sharded sparse reduction (opt-in alternative to the dedup
+        all-gather, C2V_DP_SPARSE=owner): each rank sends its deduped rows
+        to owner rank `id % N` (all-to-all: per-link parallel over xGMI),
+        owners sum their shard, and the DISJOINT reduced sets are
+        all-gathered. Wire bytes ~ V_local*(N-1)/N spread over all links
+        plus V_global/N per rank on the gather — at java14m Zipf shapes
+        roughly half the dedup-gather's ring bytes (profiles/
+        r02_dp_volume.md). Returns [(ids, rows)] like
+        allgather_sparse_dedup(...).wait()."""
+        from ..ops import functional as F
+        ws = self.world_size
+        assert ws > 1
+        results = []
+        for ids, rows, cnt in entries:
+            if torch.is_tensor(cnt):
+                c = int(cnt.item())
+            else:
+                c = int(cnt)
+            ids_l = ids[:c]
+            rows_l = rows[:c] * (1.0 / ws)
+            if ids_l.is_cuda:
+                rows_l = rows_l.to(torch.bfloat16)
+            owner = (ids_l % ws).to(torch.int64)
+            send_ids, send_rows = [], []
+            for dst in range(ws):
+                sel = (owner == dst).nonzero(as_tuple=True)[0]
+                send_ids.append(ids_l.index_select(0, sel))
+                send_rows.append(rows_l.index_select(0, sel))
+            got_ids, got_rows = self._all_to_all_rows(send_ids, send_rows)
+            # owner-side reduce of the shard (ids collide across ranks)
+            if got_ids.numel():
+                ru, racc, rc = F.sparse_dedup_sum(got_ids, got_rows)
+                rc = int(rc.item()) if torch.is_tensor(rc) else int(rc)
+                red_ids, red_rows = ru[:rc], racc[:rc]
+                if ids_l.is_cuda:
+                    red_rows = red_rows.to(torch.bfloat16)
+            else:
+                red_ids = got_ids
+                red_rows = got_rows
+            # gather the disjoint reduced shards
+            n_local = torch.tensor([red_ids.numel()], dtype=torch.int64)
+            counts = [torch.zeros_like(n_local) for _ in range(ws)]
+            dist.all_gather(counts, n_local, group=self._host_pg())
+            n_max = max(int(x.item()) for x in counts)
+            dev = red_ids.device
+            # padding uses id -1 (never a vocab id); consumers filter it
+            # out after the gather — unlike the dedup-gather path there is
+            # no guaranteed local id to replicate (a shard can be empty)
+            wire_ids = torch.full((n_max,), -1, dtype=torch.int64,
+                                  device=dev)
+            wire_rows = torch.zeros(n_max, rows_l.shape[1],
+                                    dtype=red_rows.dtype, device=dev)
+            if red_ids.numel():
+                wire_ids[:red_ids.numel()] = red_ids
+                wire_rows[:red_rows.shape[0]] = red_rows
+            ids_out = [torch.empty_like(wire_ids) for _ in range(ws)]
+            rows_out = [torch.empty_like(wire_rows) for _ in range(ws)]
+            dist.all_gather(ids_out, wire_ids, group=self.group)
+            dist.all_gather(rows_out, wire_rows, group=self.group)
+            all_ids = torch.cat(ids_out)
+            all_rows = torch.cat(rows_out)
+            keep = (all_ids >= 0).nonzero(as_tuple=True)[0]
+            if keep.numel() != all_ids.numel():
+                all_ids = all_ids.index_select(0, keep)
+                all_rows = all_rows.index_select(0, keep)
+            results.append((all_ids, all_rows))
+        return results
+
     # -- termination consensus --------------------------------------------
     # Ranks can end an epoch with unequal batch counts (the DP shard split is
     # not batch-aligned); a rank stepping once more than its peers would

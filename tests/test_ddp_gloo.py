@@ -233,3 +233,47 @@ def test_dp2_sampled_softmax_matches_dp1(tmp_path):
     dps = torch.load(str(tmp_path / 'dps.pt'), weights_only=False)
     for name in net1.param_names():
         assert torch.allclose(net1.get_param(name), dps[name], atol=1e-5), name
+
+
+def _worker_owner(rank, world_size, init_file, result_dir):
+    os.environ['C2V_DP_SPARSE'] = 'owner'
+    dist.init_process_group('gloo', init_method='file://' + init_file,
+                            rank=rank, world_size=world_size)
+    from code2vec_amd.parallel.ddp import Reducer
+    torch.manual_seed(7)
+    net = Code2VecNetwork(tiny_cfg(), V_TOK, V_PATH, V_TGT, device='cpu')
+    src, pth, tgt, mask, labels = make_batch()
+    sl = slice(rank * (B // world_size), (rank + 1) * (B // world_size))
+    reducer = Reducer()
+    for _ in range(3):
+        net.train_step(src[sl], pth[sl], tgt[sl], mask[sl], labels[sl],
+                       reducer=reducer)
+    if rank == 0:
+        torch.save(net.state_dict(), os.path.join(result_dir, 'dpo.pt'))
+    dist.barrier()
+    dist.destroy_process_group()
+    os.environ.pop('C2V_DP_SPARSE', None)
+
+
+@pytest.mark.timeout(240)
+@pytest.mark.parametrize('world_size', [2, 4])
+def test_dpn_owner_sharded_matches_dp1(tmp_path, world_size):
+    """Owner-sharded sparse reduce (C2V_DP_SPARSE=owner): id%N all-to-all +
+    owner reduce + disjoint-shard gather must match DP=1 like the dedup
+    gather does."""
+    os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+    init_file = str(tmp_path / 'pg_init_o')
+    mp.spawn(_worker_owner, args=(world_size, init_file, str(tmp_path)),
+             nprocs=world_size, join=True)
+
+    torch.manual_seed(7)
+    net1 = Code2VecNetwork(tiny_cfg(), V_TOK, V_PATH, V_TGT, device='cpu')
+    src, pth, tgt, mask, labels = make_batch()
+    for _ in range(3):
+        net1.train_step(src, pth, tgt, mask, labels)
+
+    dpo = torch.load(str(tmp_path / 'dpo.pt'), weights_only=False)
+    for name in net1.param_names():
+        assert torch.allclose(net1.get_param(name), dpo[name], atol=1e-6), name
+        assert torch.allclose(net1._adam_m[name], dpo['adam_m.' + name],
+                              atol=1e-6), name
