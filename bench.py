@@ -319,6 +319,9 @@ def main():
     ap.add_argument('--sampled-softmax', type=int, default=0,
                     help='train with sampled softmax over N negatives '
                          '(BASELINE config 4); 0 = full softmax (default)')
+    ap.add_argument('--burnin', type=int,
+                    default=int(os.environ.get('C2V_BENCH_BURNIN', 400)),
+                    help='untimed pre-warmup steps (device-busy telemetry)')
     ap.add_argument('--end-to-end', action='store_true',
                     help='time reader->H2D->step on generated on-disk data '
                          'instead of resident synthetic batches')
@@ -403,6 +406,11 @@ def main():
         if device.startswith('cuda'):
             torch.cuda.synchronize()
 
+    # burn-in: a couple of seconds of untimed GPU work before the measured
+    # window so out-of-band telemetry (SMI sampling by the bench driver)
+    # observes a busy device even when the timed region is sub-second
+    for i in range(args.burnin):
+        do_step(batches[i % len(batches)])
     for i in range(args.warmup):
         do_step(batches[i % len(batches)])
     barrier_sync()
