@@ -69,8 +69,25 @@ def main():
         'draws Zipf-shaped ids (bench.py synth_batches).',
         '',
         'At DP=8 a ring all-gather of 31 MB/rank costs ~(N-1)*31MB/153GB/s',
-        '= 1.4 ms vs 7.2 ms for the raw 157 MB — under the ~3.3 ms step and',
-        'overlappable with the sparse-Adam accumulate work.',
+        '= 1.4 ms vs 7.2 ms for the raw 157 MB — under the ~3.5-3.9 ms step,',
+        'launched async and overlapped with the dense w/a Adam chain.',
+        '',
+        '## DP=8 weak-scaling projection (post-overlap step ~3.85 ms mid-box)',
+        '',
+        'Per-rank, per-step xGMI link time at DP=8, full softmax:',
+        '- target-table all-reduce (200 MB bf16, ring): 2*(7/8)*200/153 = 2.3 ms',
+        '  (side stream, launched right after the fused CE backward);',
+        '- sparse gather (deduped, Zipf-1.1): ~1.4 ms; C2V_DP_SPARSE=owner',
+        '  halves it to ~0.9 ms if needed;',
+        '- w/a all-reduce: negligible (0.6 MB).',
+        '',
+        'Total link occupancy ~3.7 ms vs ~3.6 ms device compute: the step',
+        'projects to max(compute, comm)+join = 3.9-4.4 ms => 87-98% weak',
+        'scaling with good overlap, ~80% if the single NCCL stream',
+        'serializes poorly at the dependency points. Sampled softmax drops',
+        'the 200 MB all-reduce (sparse target rows) and projects >95%.',
+        'Measured numbers await the driver 8-GPU node; every collective',
+        'path is pinned by the gloo ws=2/4 exactness tests.',
         '',
     ]
     out = '\n'.join(lines)
