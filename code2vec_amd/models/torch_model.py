@@ -174,14 +174,7 @@ class Code2VecModel(Code2VecModelBase):
         steps_per_epoch = max(1, cfg.train_steps_per_epoch // max(1, self.world_size))
         save_every = max(1, steps_per_epoch * cfg.SAVE_EVERY_EPOCHS)
 
-        batch_num = 0
-        # device-resident loss accumulator: train_step returns an unsynced
-        # device scalar; forcing float() per batch would stall the pipeline,
-        # so the H2D sync is paid only at the logging boundary
-        sum_loss = None
-        window_examples = 0
         start = time.time()
-        multi_batch_start = start
         scalar_log = None
         if cfg.USE_TENSORBOARD:
             # tensorboard isn't installed in this environment; the flag writes
@@ -201,6 +194,28 @@ class Code2VecModel(Code2VecModelBase):
         # asynchronously while the current step runs (Reducer.start_vote on
         # the host-side group), so the consensus costs no per-step latency;
         # longer ranks drop their surplus batches.
+        try:
+            self._train_loop(cfg, batches, prefetcher, steps_per_epoch,
+                             save_every, scalar_log, start)
+        finally:
+            # tear the pipeline threads down on EVERY exit path: a daemon
+            # thread still inside the C++ parser at interpreter shutdown
+            # aborts the process
+            reader.stop_streaming(join=False)
+            prefetcher.stop()
+            reader.stop_streaming(join=True)
+
+        elapsed = int(time.time() - start)
+        self.log('Done training')
+        self.log('Training time: %sH:%sM:%sS'
+                 % (elapsed // 3600, (elapsed // 60) % 60, elapsed % 60))
+
+    def _train_loop(self, cfg, batches, prefetcher, steps_per_epoch,
+                    save_every, scalar_log, start):
+        batch_num = 0
+        sum_loss = None
+        window_examples = 0
+        multi_batch_start = time.time()
         nxt = next(batches, None)
         vote = self.reducer.start_vote(nxt is not None)
         while True:
@@ -266,10 +281,6 @@ class Code2VecModel(Code2VecModelBase):
 
         if scalar_log is not None:
             scalar_log.close()
-        elapsed = int(time.time() - start)
-        self.log('Done training')
-        self.log('Training time: %sH:%sM:%sS'
-                 % (elapsed // 3600, (elapsed // 60) % 60, elapsed % 60))
 
     # ---- evaluation ----
 
