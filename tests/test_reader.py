@@ -117,3 +117,60 @@ def test_dp_sharding(vocabs_and_cfg, tmp_path):
     n0 = sum(b.source_token_indices.shape[0] for b in r0.iter_batches(str(data)))
     n1 = sum(b.source_token_indices.shape[0] for b in r1.iter_batches(str(data)))
     assert n0 == 5 and n1 == 5
+
+
+def test_reader_crlf_long_lines_empty_file(tmp_path):
+    """Format robustness: CRLF line endings parse like LF; a line longer
+    than the 4 MB IO chunk survives the carry logic; an empty dataset
+    yields zero batches without hanging."""
+    import os
+    from code2vec_amd.config import Config
+    from code2vec_amd.data.reader import EstimatorAction, PathContextReader
+    from code2vec_amd.vocabularies import (
+        Vocab, VocabType, _SpecialVocabWords_JoinedOovPad)
+    from types import SimpleNamespace
+
+    toks = ['t%d' % i for i in range(50)]
+    paths = ['p%d' % i for i in range(50)]
+    tgts = ['alpha', 'beta']
+    vocabs = SimpleNamespace(
+        token_vocab=Vocab(VocabType.Token, toks,
+                          _SpecialVocabWords_JoinedOovPad),
+        path_vocab=Vocab(VocabType.Path, paths,
+                         _SpecialVocabWords_JoinedOovPad),
+        target_vocab=Vocab(VocabType.Target, tgts,
+                           _SpecialVocabWords_JoinedOovPad))
+    cfg = Config(set_defaults=True)
+    cfg.MAX_CONTEXTS = 8
+    cfg.TRAIN_BATCH_SIZE = 4
+    cfg.NUM_TRAIN_EPOCHS = 1
+    cfg.SHUFFLE_BUFFER_SIZE = 0
+    cfg.VERBOSE_MODE = 0
+
+    # CRLF + a giant line (> chunk) + normal lines
+    os.environ['C2V_READER_CHUNK_BYTES'] = str(1 << 16)  # 64 KB chunks
+    try:
+        big_ctxs = ' '.join('t1,p1,t2' for _ in range(12000))  # ~100 KB line
+        data = tmp_path / 'x.train.c2v'
+        with open(data, 'wb') as f:
+            f.write(b'alpha t1,p1,t2 t3,p3,t4\r\n')
+            f.write(('beta ' + big_ctxs + '\n').encode())
+            f.write(b'alpha t5,p5,t6\r\n')
+        reader = PathContextReader(vocabs=vocabs, config=cfg,
+                                   estimator_action=EstimatorAction.Train)
+        rows = 0
+        targets = []
+        for b in reader.iter_batches(data_path=str(data)):
+            rows += b.source_token_indices.shape[0]
+            targets.extend(b.target_index.tolist())
+        assert rows == 3
+        reader.stop_streaming()
+
+        empty = tmp_path / 'e.train.c2v'
+        empty.write_bytes(b'')
+        reader2 = PathContextReader(vocabs=vocabs, config=cfg,
+                                    estimator_action=EstimatorAction.Train)
+        assert list(reader2.iter_batches(data_path=str(empty))) == []
+        reader2.stop_streaming()
+    finally:
+        os.environ.pop('C2V_READER_CHUNK_BYTES', None)
