@@ -238,3 +238,25 @@ def test_release_emits_tf_format(tmp_path):
             want = want.reshape(-1, 1)
         assert arr.shape == want.shape, tf_name
         assert np.array_equal(arr, want), tf_name
+
+
+def test_reader_rejects_malformed_files(tmp_path):
+    """Loud failures on truncated / non-checkpoint index files and missing
+    data shards (silent garbage loads would be far worse)."""
+    prefix = str(tmp_path / 'ck')
+    tfb.write_checkpoint(prefix, {'a': np.zeros((2, 2), np.float32)})
+    raw = open(prefix + '.index', 'rb').read()
+
+    open(prefix + '.index', 'wb').write(raw[:10])
+    with pytest.raises(Exception):
+        tfb.read_index_file(prefix + '.index')
+
+    open(prefix + '.index', 'wb').write(b'not a table at all' * 4)
+    with pytest.raises(ValueError, match='not a TensorFlow checkpoint'):
+        tfb.read_index_file(prefix + '.index')
+
+    open(prefix + '.index', 'wb').write(raw)
+    os.remove(prefix + '.data-00000-of-00001')
+    r = tfb.TFCheckpointReader(prefix)
+    with pytest.raises(FileNotFoundError):
+        r.get_tensor('a')
