@@ -86,17 +86,27 @@ def zipf_ids(vocab_size, shape, g, s=ZIPF_S):
 
 
 def synth_batches(cfg, device, batch_size, n_batches=8, seed=0):
-    """Pool of synthetic java14m-shaped batches, resident on device."""
+    """Pool of synthetic java14m-shaped batches, resident on device.
+    C2V_BENCH_ID_DIST=uniform switches to round-1's uniform ids (for
+    apples-to-apples comparisons against the r01 numbers)."""
     g = torch.Generator(device='cpu').manual_seed(seed)
     batches = []
     V_tok = cfg.MAX_TOKEN_VOCAB_SIZE + 1
     V_path = cfg.MAX_PATH_VOCAB_SIZE + 1
     V_tgt = cfg.MAX_TARGET_VOCAB_SIZE + 1
     C = cfg.MAX_CONTEXTS
+    uniform = os.environ.get('C2V_BENCH_ID_DIST') == 'uniform'
+
+    def draw(vocab, shape):
+        if uniform:
+            return torch.randint(1, vocab, shape, generator=g,
+                                 dtype=torch.int32)
+        return zipf_ids(vocab, shape, g)
+
     for _ in range(n_batches):
-        src = zipf_ids(V_tok, (batch_size, C), g)
-        pth = zipf_ids(V_path, (batch_size, C), g)
-        tgt = zipf_ids(V_tok, (batch_size, C), g)
+        src = draw(V_tok, (batch_size, C))
+        pth = draw(V_path, (batch_size, C))
+        tgt = draw(V_tok, (batch_size, C))
         # realistic context-count distribution: valid prefix of U[64, 200]
         lo = min(64, max(1, C // 2))
         n_valid = torch.randint(lo, C + 1, (batch_size,), generator=g)
@@ -455,7 +465,9 @@ def main():
                 'token_vocab': cfg.MAX_TOKEN_VOCAB_SIZE,
                 'path_vocab': cfg.MAX_PATH_VOCAB_SIZE,
                 'target_vocab': cfg.MAX_TARGET_VOCAB_SIZE,
-                'id_dist': 'zipf-%s' % ZIPF_S,
+                'id_dist': ('uniform'
+                            if os.environ.get('C2V_BENCH_ID_DIST') == 'uniform'
+                            else 'zipf-%s' % ZIPF_S),
             },
         }
         print(json.dumps(out))
