@@ -311,10 +311,16 @@ class Code2VecModel(Code2VecModelBase):
         log_file = open('log.txt', 'w')
         total_loss, total_rows = 0.0, 0
         start = time.time()
+        # prefetch like training (the reference eval pipeline is the same
+        # tf.data graph incl. prefetch, path_context_reader.py:150); batches
+        # arrive device-resident with the H2D on the copy stream
+        from ..data.prefetcher import BatchPrefetcher
+        prefetcher = BatchPrefetcher(reader.iter_batches(), device,
+                                     depth=cfg.READER_QUEUE_DEPTH)
         try:
             nr_examples = 0
-            for batch in reader.iter_batches():
-                b = batch.to(device) if device.type != 'cpu' else batch
+            for batch in prefetcher:
+                b = batch
                 if b.target_index is not None:
                     indices, scores, code, _alpha, loss_sum = \
                         self.network.eval_batch(
@@ -330,7 +336,7 @@ class Code2VecModel(Code2VecModelBase):
                 idx_np = indices.cpu().numpy()
                 top_words = [[index_to_word.get(int(i), special_words.OOV)
                               for i in row] for row in idx_np]
-                pairs = list(zip(batch.target_string, top_words))
+                pairs = list(zip(b.target_string, top_words))
                 topk_metric.update_batch(pairs)
                 subtoken_metric.update_batch(pairs)
                 nr_examples += len(pairs)
@@ -349,6 +355,9 @@ class Code2VecModel(Code2VecModelBase):
             log_file.close()
             if export_file is not None:
                 export_file.close()
+            reader.stop_streaming(join=False)
+            prefetcher.stop()
+            reader.stop_streaming(join=True)
 
         elapsed = int(time.time() - start)
         self.log('Done evaluating, epoch reached. Evaluated %d examples in %ds'
