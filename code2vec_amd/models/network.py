@@ -199,14 +199,15 @@ class Code2VecNetwork:
     # ---- forward ----
 
     def forward(self, src_ids, path_ids, tgt_ids, valid_mask,
-                training: bool) -> ForwardState:
+                training: bool, _seed_preadvanced: bool = False) -> ForwardState:
         B, C = src_ids.shape
         D = self.config.CODE_VECTOR_SIZE
         self._step_ctr += 1
         seed = (self._seed_counter + self._step_ctr * 2654435761) & 0x7FFFFFFFFFFFFFFF
         seed_t = None
         if training and self._seed_t is not None:
-            self._seed_t.add_(2654435761)   # in-graph advance (capture-safe)
+            if not _seed_preadvanced:
+                self._seed_t.add_(2654435761)  # in-graph advance (capture-safe)
             seed_t = self._seed_t
             seed = 0
         keep = self.config.DROPOUT_KEEP_RATE if training else 1.0
@@ -241,6 +242,15 @@ class Code2VecNetwork:
         # serializing in the backward tail (~0.45 ms/step on Zipf ids).
         tok_ids = torch.cat([src_ids.reshape(-1), tgt_ids.reshape(-1)])
         path_ids_flat = path_ids.reshape(-1)
+        # Advance the dropout seed on the main stream BEFORE the hash build
+        # is enqueued: the 1-block add_ kernel gets CU-starved under the
+        # claim kernel's atomic storm when launched concurrently (profiled
+        # 361 us in gpurun_out/prof_r02samp while the forward, which depends
+        # on the seed, sat blocked behind it).
+        seed_pre = False
+        if self._seed_t is not None:
+            self._seed_t.add_(2654435761)
+            seed_pre = True
         tok_state = path_state = None
         hash_done = None
         if (self._hash_stream is not None
@@ -257,7 +267,8 @@ class Code2VecNetwork:
                 tok_ids.record_stream(self._hash_stream)
                 path_ids_flat.record_stream(self._hash_stream)
 
-        st = self.forward(src_ids, path_ids, tgt_ids, valid_mask, training=True)
+        st = self.forward(src_ids, path_ids, tgt_ids, valid_mask, training=True,
+                          _seed_preadvanced=seed_pre)
         code_c = st.code.to(self.compute_dtype)
         S = int(cfg.SAMPLED_SOFTMAX_SIZE)
         V = self.target_table.shape[0]
