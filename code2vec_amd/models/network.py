@@ -188,9 +188,10 @@ class Code2VecNetwork:
             self._step_t.fill_(self.adam_step)
         if self._seed_t is not None:
             # Restore the GPU dropout seed stream to where the uninterrupted
-            # run would be: forward() advances _seed_t by 2654435761/step from
-            # 0, so after _step_ctr steps it holds _step_ctr*2654435761 with
-            # int64 wraparound semantics.
+            # run would be: each training step advances _seed_t by 2654435761
+            # (train_step pre-advances; forward advances when called outside
+            # train_step), so after _step_ctr steps it holds
+            # _step_ctr*2654435761 with int64 wraparound semantics.
             v = (self._step_ctr * 2654435761) & 0xFFFFFFFFFFFFFFFF
             if v >= 1 << 63:
                 v -= 1 << 64
