@@ -249,7 +249,8 @@ class Code2VecNetwork:
         # 361 us in gpurun_out/prof_r02samp while the forward, which depends
         # on the seed, sat blocked behind it).
         seed_pre = False
-        if self._seed_t is not None:
+        if (self._seed_t is not None
+                and os.environ.get('C2V_SEED_PRE', '1') == '1'):
             self._seed_t.add_(2654435761)
             seed_pre = True
         tok_state = path_state = None
