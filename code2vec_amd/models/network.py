@@ -200,7 +200,8 @@ class Code2VecNetwork:
     # ---- forward ----
 
     def forward(self, src_ids, path_ids, tgt_ids, valid_mask,
-                training: bool, _seed_preadvanced: bool = False) -> ForwardState:
+                training: bool, _seed_preadvanced: bool = False,
+                _post_ctx_hook=None) -> ForwardState:
         B, C = src_ids.shape
         D = self.config.CODE_VECTOR_SIZE
         self._step_ctr += 1
@@ -215,6 +216,8 @@ class Code2VecNetwork:
         ctx = F.gather_concat_fwd(self.tok_table, self.path_table, src_ids,
                                   path_ids, tgt_ids, keep, seed, training,
                                   out_dtype=self.compute_dtype, seed_t=seed_t)
+        if _post_ctx_hook is not None:
+            _post_ctx_hook()
         comb = F.transform_tanh_fwd(ctx, self.w_oi)                  # (B*C, D)
         code, alpha = F.attention_fwd(comb.reshape(B, C, D), self.a_c, valid_mask)
         return ForwardState(ctx=ctx, comb=comb, code=code, alpha=alpha,
@@ -255,22 +258,44 @@ class Code2VecNetwork:
             seed_pre = True
         tok_state = path_state = None
         hash_done = None
-        if (self._hash_stream is not None
-                and os.environ.get('C2V_HASH_OVERLAP', '1') == '1'):
+        hash_box = {}
+
+        def _enqueue_hash_build():
             ev = torch.cuda.Event()
             ev.record()
             with torch.cuda.stream(self._hash_stream):
                 self._hash_stream.wait_event(ev)
-                tok_state = F.sparse_hash_build(tok_ids)
-                path_state = F.sparse_hash_build(path_ids_flat)
-                hash_done = torch.cuda.Event()
-                hash_done.record()
+                hash_box['tok'] = F.sparse_hash_build(tok_ids)
+                hash_box['path'] = F.sparse_hash_build(path_ids_flat)
+                done = torch.cuda.Event()
+                done.record()
+                hash_box['done'] = done
             if not torch.cuda.is_current_stream_capturing():
                 tok_ids.record_stream(self._hash_stream)
                 path_ids_flat.record_stream(self._hash_stream)
 
+        hash_overlap = (self._hash_stream is not None
+                        and os.environ.get('C2V_HASH_OVERLAP', '1') == '1')
+        # The claim kernel's gather/atomic traffic thrashes with the
+        # (equally HBM-gather-bound) embedding gather+concat when the two
+        # run concurrently (394 vs 163 us for the gather, profiled in
+        # gpurun_out/prof_r02samp2), but coexists well with the MFMA-bound
+        # transform GEMM — so by default the build is enqueued right after
+        # the gather stage via the forward hook. C2V_HASH_POS=start keeps
+        # the step-start placement for measurement.
+        hash_at_start = os.environ.get('C2V_HASH_POS', 'postgather') == 'start'
+        post_ctx_hook = None
+        if hash_overlap and hash_at_start:
+            _enqueue_hash_build()
+        elif hash_overlap:
+            post_ctx_hook = _enqueue_hash_build
+
         st = self.forward(src_ids, path_ids, tgt_ids, valid_mask, training=True,
-                          _seed_preadvanced=seed_pre)
+                          _seed_preadvanced=seed_pre,
+                          _post_ctx_hook=post_ctx_hook)
+        tok_state = hash_box.get('tok')
+        path_state = hash_box.get('path')
+        hash_done = hash_box.get('done')
         code_c = st.code.to(self.compute_dtype)
         S = int(cfg.SAMPLED_SOFTMAX_SIZE)
         V = self.target_table.shape[0]
