@@ -246,14 +246,19 @@ class Code2VecNetwork:
         # serializing in the backward tail (~0.45 ms/step on Zipf ids).
         tok_ids = torch.cat([src_ids.reshape(-1), tgt_ids.reshape(-1)])
         path_ids_flat = path_ids.reshape(-1)
-        # Advance the dropout seed on the main stream BEFORE the hash build
-        # is enqueued: the 1-block add_ kernel gets CU-starved under the
-        # claim kernel's atomic storm when launched concurrently (profiled
-        # 361 us in gpurun_out/prof_r02samp while the forward, which depends
-        # on the seed, sat blocked behind it).
+        # Scheduling note (measured, counterintuitive): the seed add_ kernel
+        # launched AFTER the hash build gets CU-starved under the claim
+        # kernel's atomic storm (~360 us in gpurun_out/prof_r02samp), and the
+        # forward's gather waits on it — but that implicit serialization is
+        # FASTER than releasing the gather early: gather+claim thrash when
+        # concurrent (394 vs 163 us gather), and a later claim start pushes
+        # the hash tail into the sparse-Adam critical path. Same-box A/B
+        # (gpurun_out/r02z_ab.log, r02z2_ab.log): pre-advancing the seed
+        # costs ~45 us/step; post-gather hash placement costs ~150 us/step.
+        # Both variants stay env-gated for re-measurement on new hardware.
         seed_pre = False
         if (self._seed_t is not None
-                and os.environ.get('C2V_SEED_PRE', '1') == '1'):
+                and os.environ.get('C2V_SEED_PRE', '0') == '1'):
             self._seed_t.add_(2654435761)
             seed_pre = True
         tok_state = path_state = None
@@ -276,14 +281,10 @@ class Code2VecNetwork:
 
         hash_overlap = (self._hash_stream is not None
                         and os.environ.get('C2V_HASH_OVERLAP', '1') == '1')
-        # The claim kernel's gather/atomic traffic thrashes with the
-        # (equally HBM-gather-bound) embedding gather+concat when the two
-        # run concurrently (394 vs 163 us for the gather, profiled in
-        # gpurun_out/prof_r02samp2), but coexists well with the MFMA-bound
-        # transform GEMM — so by default the build is enqueued right after
-        # the gather stage via the forward hook. C2V_HASH_POS=start keeps
-        # the step-start placement for measurement.
-        hash_at_start = os.environ.get('C2V_HASH_POS', 'postgather') == 'start'
+        # Step-start placement measured fastest (see scheduling note above);
+        # C2V_HASH_POS=postgather enqueues the build after the gather stage
+        # via the forward hook instead, for re-measurement.
+        hash_at_start = os.environ.get('C2V_HASH_POS', 'start') == 'start'
         post_ctx_hook = None
         if hash_overlap and hash_at_start:
             _enqueue_hash_build()
