@@ -2145,6 +2145,22 @@ __global__ void k_hot_fold(const float* __restrict__ hot_acc,
   }
 }
 
+// One-launch init of the hash-build workspace: [-1 x n_neg | 0 x rest].
+// Replaces the five torch fills (tbl_id/tbl_cidx full(-1), tbl_cnt/n_hot/
+// n_uniq zeros) that cost ~15 launches/step across the three builds.
+__global__ void k_hash_ws_init(int* __restrict__ ws, long n_neg, long n_tot) {
+  const long n4 = n_neg / 4;
+  for (long i4 = blockIdx.x * blockDim.x + threadIdx.x; i4 < n4;
+       i4 += (long)gridDim.x * blockDim.x)
+    *reinterpret_cast<int4*>(ws + i4 * 4) = int4{-1, -1, -1, -1};
+  for (long i = n4 * 4 + blockIdx.x * blockDim.x + threadIdx.x; i < n_neg;
+       i += (long)gridDim.x * blockDim.x)
+    ws[i] = -1;
+  for (long i = n_neg + blockIdx.x * blockDim.x + threadIdx.x; i < n_tot;
+       i += (long)gridDim.x * blockDim.x)
+    ws[i] = 0;
+}
+
 __global__ void k_zero_rows_dyn(float* __restrict__ acc,
                                 const int* __restrict__ n_uniq_ptr, int d,
                                 int max_rows = 0x7FFFFFFF) {
@@ -3033,14 +3049,22 @@ static DedupState hash_dedup_ids(const torch::Tensor& ids_c) {
   while (cap < (u32)want) cap <<= 1;
   auto opts_i32 = torch::TensorOptions().device(ids_c.device())
                       .dtype(torch::kInt32);
-  auto tbl_id = torch::full({(long)cap}, -1, opts_i32);
-  auto tbl_cidx = torch::full({(long)cap}, -1, opts_i32);
-  auto tbl_cnt = torch::zeros({(long)cap}, opts_i32);
+  // Single workspace + one init kernel instead of five torch fills:
+  // [tbl_id(cap) | tbl_cidx(cap) | tbl_cnt(cap) | n_hot | n_uniq].
+  // n_hot/n_uniq are narrow() views, so the returned DedupState keeps the
+  // workspace alive.
+  const long ws_len = 3 * (long)cap + 2;
+  auto ws = torch::empty({ws_len}, opts_i32);
+  auto tbl_id = ws.narrow(0, 0, cap);
+  auto tbl_cidx = ws.narrow(0, cap, cap);
+  auto tbl_cnt = ws.narrow(0, 2 * (long)cap, cap);
+  auto n_hot = ws.narrow(0, 3 * (long)cap, 1);
+  auto n_uniq = ws.narrow(0, 3 * (long)cap + 1, 1);
+  k_hash_ws_init<<<grid_1d(ws_len / 4 + 1, 256), 256, 0, cur_stream()>>>(
+      ws.data_ptr<int>(), 2 * (long)cap, ws_len);
   auto tbl_hot = torch::empty({(long)cap}, opts_i32);
   auto hot2cidx = torch::empty({HOT_CAP}, opts_i32);
-  auto n_hot = torch::zeros({1}, opts_i32);
   auto uniq = torch::empty({n}, opts_i32.dtype(torch::kInt64));
-  auto n_uniq = torch::zeros({1}, opts_i32);
   auto inverse = torch::empty({n}, opts_i32);
   const u32 mask_ = cap - 1;
   if (ids_c.scalar_type() == torch::kInt32) {
