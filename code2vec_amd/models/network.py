@@ -307,21 +307,19 @@ class Code2VecNetwork:
             # [B local labels | S shared log-uniform negatives]; the target
             # table gets SPARSE row grads (all-gathered under DP like the
             # embedding tables) instead of the dense 200 MB all-reduce.
-            from ..ops.reference import log_uniform_probs, sample_log_uniform
+            from ..ops.reference import sample_log_uniform
             sampled = sample_log_uniform(S, V, code_c.device)
             cand = torch.cat([labels, sampled])                  # (B+S,)
             # candidate logits/backward GEMMs run with the candidate-row
             # gather fused into the kernels' B staging — no w_cand
-            # materialization, no torch GEMM in the sampled path
+            # materialization, no torch GEMM in the sampled path; the
+            # log-uniform corrections are computed inline in the CE kernels
             logits_cand = F.sampled_logits_gemm(code_c, self.target_shadow,
                                                 cand)            # (B, B+S)
-            corr_true = torch.log(log_uniform_probs(labels, V) * S)
-            corr_samp = torch.log(log_uniform_probs(sampled, V) * S)
-            loss_rows, lse = F.sampled_ce_fwd(logits_cand, labels, sampled,
-                                              corr_true, corr_samp)
+            loss_rows, lse = F.sampled_ce_fwd(logits_cand, labels, sampled, V)
             loss = loss_rows.float().mean()
-            d_cand = F.sampled_ce_bwd(logits_cand, labels, sampled, corr_true,
-                                      corr_samp, lse, 1.0 / B)
+            d_cand = F.sampled_ce_bwd(logits_cand, labels, sampled, V,
+                                      lse, 1.0 / B)
             d_target_rows = F.sampled_bwd_target_rows(d_cand, code_c)
             if (reducer.world_size > 1
                     and os.environ.get('C2V_DP_DEDUP', '1') == '1'):

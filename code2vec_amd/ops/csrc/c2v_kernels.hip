@@ -1724,12 +1724,20 @@ __global__ void k_ce_bwd(const ushort* __restrict__ logits,
 // {L[b,B+j]-corr_samp[j] | sampled[j] != labels[b]} (accidental hits masked).
 // ---------------------------------------------------------------------------
 
+// log-uniform sampling correction log(S*q(id)) computed inline instead of
+// via ~12 torch elementwise launches/step: q = log((id+2)/(id+1))/log(V+1),
+// so log(S*q) = log(s_over_logv1 * log((id+2)/(id+1))) with
+// s_over_logv1 = S/log(V+1) precomputed on the host.
+__device__ __forceinline__ float lu_corr(long id, float s_over_logv1) {
+  return __logf(s_over_logv1 *
+                __logf(((float)id + 2.f) / ((float)id + 1.f)));
+}
+
 __launch_bounds__(256)
 __global__ void k_sampled_ce_fwd(const ushort* __restrict__ L,
                                  const long* __restrict__ labels,
                                  const long* __restrict__ sampled,
-                                 const float* __restrict__ corr_true,
-                                 const float* __restrict__ corr_samp,
+                                 float s_over_logv1,
                                  float* __restrict__ loss,
                                  float* __restrict__ lse_out, int B, int S) {
   __shared__ float sm[4], ss[4];
@@ -1740,13 +1748,13 @@ __global__ void k_sampled_ce_fwd(const ushort* __restrict__ L,
   const int wid = tid >> 6;
   const ushort* row = L + (long)b * T;
   const long lab = labels[b];
-  const float z0 = bf2f(row[b]) - corr_true[b];
+  const float z0 = bf2f(row[b]) - lu_corr(lab, s_over_logv1);
 
   float m = z0, s = 1.f;  // thread 0's stream starts with z0; others -inf
   if (tid != 0) { m = -3.0e38f; s = 0.f; }
   for (int j = tid; j < S; j += blockDim.x) {
     if (sampled[j] == lab) continue;
-    const float x = bf2f(row[B + j]) - corr_samp[j];
+    const float x = bf2f(row[B + j]) - lu_corr(sampled[j], s_over_logv1);
     if (x > m) { s = s * __expf(m - x) + 1.f; m = x; }
     else s += __expf(x - m);
   }
@@ -1775,8 +1783,7 @@ __global__ void k_sampled_ce_fwd(const ushort* __restrict__ L,
 __global__ void k_sampled_ce_bwd(const ushort* __restrict__ L,
                                  const long* __restrict__ labels,
                                  const long* __restrict__ sampled,
-                                 const float* __restrict__ corr_true,
-                                 const float* __restrict__ corr_samp,
+                                 float s_over_logv1,
                                  const float* __restrict__ lse,
                                  ushort* __restrict__ dL, float scale, int B,
                                  int S) {
@@ -1790,12 +1797,17 @@ __global__ void k_sampled_ce_bwd(const ushort* __restrict__ L,
        j += gridDim.x * blockDim.x) {
     float p = 0.f;
     if (sampled[j] != lab)
-      p = __expf(bf2f(row[B + j]) - corr_samp[j] - l);
+      p = __expf(bf2f(row[B + j]) - lu_corr(sampled[j], s_over_logv1) - l);
     drow[B + j] = f2bf(p * scale);
   }
-  if (blockIdx.x == 0 && threadIdx.x == 0) {
-    const float z0 = bf2f(row[b]) - corr_true[b];
-    drow[b] = f2bf((__expf(z0 - l) - 1.f) * scale);
+  if (blockIdx.x == 0) {
+    // the B true-label columns: zero except this row's own (column b)
+    for (int j = threadIdx.x; j < B; j += blockDim.x) drow[j] = f2bf(0.f);
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      const float z0 = bf2f(row[b]) - lu_corr(lab, s_over_logv1);
+      drow[b] = f2bf((__expf(z0 - l) - 1.f) * scale);
+    }
   }
 }
 
@@ -3052,16 +3064,30 @@ static DedupState hash_dedup_ids(const torch::Tensor& ids_c) {
   // Single workspace + one init kernel instead of five torch fills:
   // [tbl_id(cap) | tbl_cidx(cap) | tbl_cnt(cap) | n_hot | n_uniq].
   // n_hot/n_uniq are narrow() views, so the returned DedupState keeps the
-  // workspace alive.
-  const long ws_len = 3 * (long)cap + 2;
-  auto ws = torch::empty({ws_len}, opts_i32);
-  auto tbl_id = ws.narrow(0, 0, cap);
-  auto tbl_cidx = ws.narrow(0, cap, cap);
-  auto tbl_cnt = ws.narrow(0, 2 * (long)cap, cap);
-  auto n_hot = ws.narrow(0, 3 * (long)cap, 1);
-  auto n_uniq = ws.narrow(0, 3 * (long)cap + 1, 1);
-  k_hash_ws_init<<<grid_1d(ws_len / 4 + 1, 256), 256, 0, cur_stream()>>>(
-      ws.data_ptr<int>(), 2 * (long)cap, ws_len);
+  // workspace alive. C2V_HASH_WS=0 restores the per-tensor allocations
+  // (A/B: the consolidation is suspected of hurting claim locality).
+  static const bool use_ws = [] {
+    const char* e = getenv("C2V_HASH_WS");
+    return !(e && e[0] == '0');
+  }();
+  torch::Tensor tbl_id, tbl_cidx, tbl_cnt, n_hot, n_uniq;
+  if (use_ws) {
+    const long ws_len = 3 * (long)cap + 2;
+    auto ws = torch::empty({ws_len}, opts_i32);
+    tbl_id = ws.narrow(0, 0, cap);
+    tbl_cidx = ws.narrow(0, cap, cap);
+    tbl_cnt = ws.narrow(0, 2 * (long)cap, cap);
+    n_hot = ws.narrow(0, 3 * (long)cap, 1);
+    n_uniq = ws.narrow(0, 3 * (long)cap + 1, 1);
+    k_hash_ws_init<<<grid_1d(ws_len / 4 + 1, 256), 256, 0, cur_stream()>>>(
+        ws.data_ptr<int>(), 2 * (long)cap, ws_len);
+  } else {
+    tbl_id = torch::full({(long)cap}, -1, opts_i32);
+    tbl_cidx = torch::full({(long)cap}, -1, opts_i32);
+    tbl_cnt = torch::zeros({(long)cap}, opts_i32);
+    n_hot = torch::zeros({1}, opts_i32);
+    n_uniq = torch::zeros({1}, opts_i32);
+  }
   auto tbl_hot = torch::empty({(long)cap}, opts_i32);
   auto hot2cidx = torch::empty({HOT_CAP}, opts_i32);
   auto uniq = torch::empty({n}, opts_i32.dtype(torch::kInt64));
@@ -3320,41 +3346,39 @@ std::vector<torch::Tensor> sparse_dedup_sum_rows(torch::Tensor ids,
 std::vector<torch::Tensor> sampled_ce_fwd(torch::Tensor logits_cand,
                                           torch::Tensor labels,
                                           torch::Tensor sampled,
-                                          torch::Tensor corr_true,
-                                          torch::Tensor corr_samp) {
+                                          int64_t vocab) {
   CHECK_DEV(logits_cand); CHECK_CONT(logits_cand);
   const int B = logits_cand.size(0);
   const int S = (int)sampled.numel();
   TORCH_CHECK(logits_cand.size(1) == B + S, "candidate layout mismatch");
   auto labels_c = labels.contiguous();
   auto sampled_c = sampled.contiguous();
-  auto ct = corr_true.contiguous();
-  auto cs = corr_samp.contiguous();
+  const float sol = (float)((double)S / std::log((double)vocab + 1.0));
   auto loss = torch::empty({B}, logits_cand.options().dtype(torch::kFloat32));
   auto lse = torch::empty({B}, logits_cand.options().dtype(torch::kFloat32));
   k_sampled_ce_fwd<<<B, 256, 0, cur_stream()>>>(
       bf_ptr(logits_cand), labels_c.data_ptr<long>(),
-      sampled_c.data_ptr<long>(), ct.data_ptr<float>(), cs.data_ptr<float>(),
+      sampled_c.data_ptr<long>(), sol,
       loss.data_ptr<float>(), lse.data_ptr<float>(), B, S);
   return {loss, lse};
 }
 
 torch::Tensor sampled_ce_bwd(torch::Tensor logits_cand, torch::Tensor labels,
-                             torch::Tensor sampled, torch::Tensor corr_true,
-                             torch::Tensor corr_samp, torch::Tensor lse,
-                             double scale) {
+                             torch::Tensor sampled, int64_t vocab,
+                             torch::Tensor lse, double scale) {
   CHECK_DEV(logits_cand); CHECK_CONT(logits_cand);
   const int B = logits_cand.size(0);
   const int S = (int)sampled.numel();
   auto labels_c = labels.contiguous();
   auto sampled_c = sampled.contiguous();
-  auto ct = corr_true.contiguous();
-  auto cs = corr_samp.contiguous();
-  auto d = torch::zeros_like(logits_cand);
+  const float sol = (float)((double)S / std::log((double)vocab + 1.0));
+  // torch::empty, not zeros: the kernel writes every column (blockIdx.x==0
+  // zeroes the B true-label columns), saving the 10 MB fill launch
+  auto d = torch::empty_like(logits_cand);
   dim3 grid(grid_1d(S, 256, 32), B);
   k_sampled_ce_bwd<<<grid, 256, 0, cur_stream()>>>(
       bf_ptr(logits_cand), labels_c.data_ptr<long>(),
-      sampled_c.data_ptr<long>(), ct.data_ptr<float>(), cs.data_ptr<float>(),
+      sampled_c.data_ptr<long>(), sol,
       lse.data_ptr<float>(), bf_ptr_mut(d), (float)scale, B, S);
   return d;
 }

@@ -399,18 +399,28 @@ def sampled_bwd_target_rows(d_cand, code_c):
     return d_cand.t() @ code_c
 
 
-def sampled_ce_fwd(logits_cand, labels, sampled, corr_true, corr_samp):
+def _lu_corrections(labels, sampled, vocab: int):
+    """log(S*q(id)) corrections for the log-uniform candidate sampler
+    (reference-path helper; the HIP kernels compute these inline)."""
+    S = int(sampled.numel())
+    ct = torch.log(ref.log_uniform_probs(labels, vocab) * S)
+    cs = torch.log(ref.log_uniform_probs(sampled, vocab) * S)
+    return ct, cs
+
+
+def sampled_ce_fwd(logits_cand, labels, sampled, vocab: int):
     if backend_for(logits_cand) == 'hip':
         return hip_ext(True).sampled_ce_fwd(logits_cand, labels, sampled,
-                                            corr_true, corr_samp)
-    return ref.sampled_ce_fwd(logits_cand, labels, sampled, corr_true, corr_samp)
+                                            int(vocab))
+    ct, cs = _lu_corrections(labels, sampled, vocab)
+    return ref.sampled_ce_fwd(logits_cand, labels, sampled, ct, cs)
 
 
-def sampled_ce_bwd(logits_cand, labels, sampled, corr_true, corr_samp, lse,
+def sampled_ce_bwd(logits_cand, labels, sampled, vocab: int, lse,
                    scale: float):
     if backend_for(logits_cand) == 'hip':
         return hip_ext(True).sampled_ce_bwd(logits_cand, labels, sampled,
-                                            corr_true, corr_samp, lse,
-                                            float(scale))
-    return ref.sampled_ce_bwd(logits_cand, labels, sampled, corr_true,
-                              corr_samp, lse, scale)
+                                            int(vocab), lse, float(scale))
+    ct, cs = _lu_corrections(labels, sampled, vocab)
+    return ref.sampled_ce_bwd(logits_cand, labels, sampled, ct, cs, lse,
+                              scale)
