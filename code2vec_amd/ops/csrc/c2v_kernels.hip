@@ -3061,14 +3061,18 @@ static DedupState hash_dedup_ids(const torch::Tensor& ids_c) {
   while (cap < (u32)want) cap <<= 1;
   auto opts_i32 = torch::TensorOptions().device(ids_c.device())
                       .dtype(torch::kInt32);
-  // Single workspace + one init kernel instead of five torch fills:
-  // [tbl_id(cap) | tbl_cidx(cap) | tbl_cnt(cap) | n_hot | n_uniq].
-  // n_hot/n_uniq are narrow() views, so the returned DedupState keeps the
-  // workspace alive. C2V_HASH_WS=0 restores the per-tensor allocations
-  // (A/B: the consolidation is suspected of hurting claim locality).
+  // C2V_HASH_WS=1: single workspace + one init kernel instead of five
+  // torch fills ([tbl_id(cap) | tbl_cidx(cap) | tbl_cnt(cap) | n_hot |
+  // n_uniq], counters as narrow() views keeping it alive). Measured 300
+  // us/step SLOWER on the sampled bench (same-box A/B, gpurun_out/
+  // r02t/bo3z7vnuo): placing tbl_id and tbl_cnt at an exact power-of-two
+  // byte offset (cap*4) makes the claim kernel's paired id/cnt accesses
+  // alias the same HBM channels; the allocator's independent placements
+  // break that alignment. Default OFF — the ~15 fill launches it saved
+  // are ~60 us, the aliasing costs 5x that.
   static const bool use_ws = [] {
     const char* e = getenv("C2V_HASH_WS");
-    return !(e && e[0] == '0');
+    return e && e[0] == '1';
   }();
   torch::Tensor tbl_id, tbl_cidx, tbl_cnt, n_hot, n_uniq;
   if (use_ws) {
