@@ -3076,15 +3076,19 @@ static DedupState hash_dedup_ids(const torch::Tensor& ids_c) {
   }();
   torch::Tensor tbl_id, tbl_cidx, tbl_cnt, n_hot, n_uniq;
   if (use_ws) {
-    const long ws_len = 3 * (long)cap + 2;
+    // 4.25 KB pads between the sections break the exact power-of-two
+    // offsets (the aliasing suspect); pad bytes land in the -1/0 fill
+    // regions and are never read
+    const long P = 1088;
+    const long ws_len = 3 * (long)cap + 2 * P + 2;
     auto ws = torch::empty({ws_len}, opts_i32);
     tbl_id = ws.narrow(0, 0, cap);
-    tbl_cidx = ws.narrow(0, cap, cap);
-    tbl_cnt = ws.narrow(0, 2 * (long)cap, cap);
-    n_hot = ws.narrow(0, 3 * (long)cap, 1);
-    n_uniq = ws.narrow(0, 3 * (long)cap + 1, 1);
+    tbl_cidx = ws.narrow(0, cap + P, cap);
+    tbl_cnt = ws.narrow(0, 2 * ((long)cap + P), cap);
+    n_hot = ws.narrow(0, 3 * (long)cap + 2 * P, 1);
+    n_uniq = ws.narrow(0, 3 * (long)cap + 2 * P + 1, 1);
     k_hash_ws_init<<<grid_1d(ws_len / 4 + 1, 256), 256, 0, cur_stream()>>>(
-        ws.data_ptr<int>(), 2 * (long)cap, ws_len);
+        ws.data_ptr<int>(), 2 * ((long)cap + P), ws_len);
   } else {
     tbl_id = torch::full({(long)cap}, -1, opts_i32);
     tbl_cidx = torch::full({(long)cap}, -1, opts_i32);
