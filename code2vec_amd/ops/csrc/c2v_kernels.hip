@@ -3062,14 +3062,14 @@ static DedupState hash_dedup_ids(const torch::Tensor& ids_c) {
   auto opts_i32 = torch::TensorOptions().device(ids_c.device())
                       .dtype(torch::kInt32);
   // C2V_HASH_WS=1: single workspace + one init kernel instead of five
-  // torch fills ([tbl_id(cap) | tbl_cidx(cap) | tbl_cnt(cap) | n_hot |
+  // torch fills ([tbl_id | pad | tbl_cidx | pad | tbl_cnt | n_hot |
   // n_uniq], counters as narrow() views keeping it alive). Measured 300
-  // us/step SLOWER on the sampled bench (same-box A/B, gpurun_out/
-  // r02t/bo3z7vnuo): placing tbl_id and tbl_cnt at an exact power-of-two
-  // byte offset (cap*4) makes the claim kernel's paired id/cnt accesses
-  // alias the same HBM channels; the allocator's independent placements
-  // break that alignment. Default OFF — the ~15 fill launches it saved
-  // are ~60 us, the aliasing costs 5x that.
+  // us/step SLOWER on the sampled bench across two same-box A/Bs — and
+  // the 4.25 KB inter-section pads did NOT recover it, so the obvious
+  // power-of-two channel-aliasing explanation is falsified. The cost is
+  // some unidentified sensitivity of the claim kernel to the captured
+  // allocation layout. Default OFF: the ~15 fill launches it saves are
+  // worth ~60 us, the regression costs 5x that.
   static const bool use_ws = [] {
     const char* e = getenv("C2V_HASH_WS");
     return e && e[0] == '1';
