@@ -10,6 +10,9 @@
 
 #include <torch/extension.h>
 
+#include <sys/syscall.h>
+#include <unistd.h>
+
 #include <atomic>
 #include <cstring>
 #include <string>
@@ -19,6 +22,28 @@
 #include <vector>
 
 namespace {
+
+// Best-effort NUMA interleave of an already-populated buffer (raw mbind —
+// no libnuma in the image). The vocab tables are built single-threaded, so
+// first-touch puts every page on the builder's node and parser threads on
+// the other socket pay remote-DRAM latency on each (random) lookup;
+// MPOL_MF_MOVE migrates the pages round-robin across the allowed nodes.
+// Silently a no-op on single-node machines or where the syscall is denied.
+void interleave_pages(void* p, size_t n) {
+  if (n < (size_t)1 << 20) return;                 // not worth it under 1 MB
+  const long page = sysconf(_SC_PAGESIZE);
+  uintptr_t a = ((uintptr_t)p + page - 1) & ~(uintptr_t)(page - 1);
+  uintptr_t end = ((uintptr_t)p + n) & ~(uintptr_t)(page - 1);
+  if (end <= a) return;
+  unsigned long mems[16] = {0};
+  // MPOL_F_MEMS_ALLOWED(4) fills the mask of nodes this thread may use
+  if (syscall(SYS_get_mempolicy, nullptr, mems, sizeof(mems) * 8, nullptr,
+              4L) != 0)
+    return;
+  // MPOL_INTERLEAVE(3), MPOL_MF_MOVE(2)
+  syscall(SYS_mbind, (void*)a, (unsigned long)(end - a), 3L, mems,
+          (unsigned long)(sizeof(mems) * 8), 2UL);
+}
 
 // allocation-free open-addressing string->int map. Short keys (<= 19 B —
 // the overwhelming majority of code tokens and hash-int path strings) live
@@ -69,6 +94,9 @@ struct StrMap {
         long_keys.push_back(k);
       }
     }
+    const char* e = getenv("C2V_NUMA_INTERLEAVE");
+    if (!(e && e[0] == '0'))
+      interleave_pages(slots.data(), slots.size() * sizeof(Slot));
   }
 
   int look(std::string_view s, int dflt) const {
