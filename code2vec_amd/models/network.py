@@ -366,9 +366,7 @@ class Code2VecNetwork:
             st_t_pre = None
             if self._step_t is not None:
                 self._step_t.add_(1)
-                tf32 = self._step_t.to(torch.float32)
-                st_t_pre = (lr * torch.sqrt(1.0 - torch.exp(tf32 * math.log(b2)))
-                            / (1.0 - torch.exp(tf32 * math.log(b1)))).reshape(1)
+                st_t_pre = F.adam_lrt(self._step_t, lr, b1, b2)
             if self._side_stream is not None:
                 # the whole target chain — the big GEMM, the (DP) all-reduce
                 # and the 100M-param Adam — is independent of the rest of
@@ -480,10 +478,8 @@ class Code2VecNetwork:
             st_t = None
             if self._step_t is not None:
                 self._step_t.add_(1)
-                tf32 = self._step_t.to(torch.float32)
-                lr0, b10, b20 = cfg.ADAM_LR, cfg.ADAM_BETA1, cfg.ADAM_BETA2
-                st_t = (lr0 * torch.sqrt(1.0 - torch.exp(tf32 * math.log(b20)))
-                        / (1.0 - torch.exp(tf32 * math.log(b10)))).reshape(1)
+                st_t = F.adam_lrt(self._step_t, cfg.ADAM_LR, cfg.ADAM_BETA1,
+                                  cfg.ADAM_BETA2)
         else:
             st_t = st_t_pre  # counters already advanced pre-backward
         t, lr = self.adam_step, cfg.ADAM_LR

@@ -1820,6 +1820,17 @@ __global__ void k_sampled_ce_bwd(const ushort* __restrict__ L,
 // lazy update of p/m/v.
 // ---------------------------------------------------------------------------
 
+// One-thread fusion of the Adam bias-correction scalar chain
+// lr_t = lr*sqrt(1-b2^t)/(1-b1^t): the torch version is ~7 elementwise
+// launches on 1-element tensors per step inside the captured graph
+// (~35 us of serialized ~5 us dispatches in the steady-state trace).
+__global__ void k_adam_lrt(const long* __restrict__ step_t,
+                           float* __restrict__ out, float lr, float log_b1,
+                           float log_b2) {
+  const float t = (float)*step_t;
+  out[0] = lr * sqrtf(1.f - __expf(t * log_b2)) / (1.f - __expf(t * log_b1));
+}
+
 template <typename G, bool NT = false>
 __global__ void k_adam_dense(float* __restrict__ p, const G* __restrict__ g,
                              float* __restrict__ m, float* __restrict__ v,
@@ -3351,6 +3362,16 @@ std::vector<torch::Tensor> sparse_dedup_sum_rows(torch::Tensor ids,
   return {st.uniq, acc, st.n_uniq};
 }
 
+torch::Tensor adam_lrt(torch::Tensor step_t, double lr, double beta1,
+                       double beta2) {
+  CHECK_DEV(step_t);
+  auto out = torch::empty({1}, step_t.options().dtype(torch::kFloat32));
+  k_adam_lrt<<<1, 1, 0, cur_stream()>>>(
+      step_t.data_ptr<long>(), out.data_ptr<float>(), (float)lr,
+      (float)std::log(beta1), (float)std::log(beta2));
+  return out;
+}
+
 std::vector<torch::Tensor> sampled_ce_fwd(torch::Tensor logits_cand,
                                           torch::Tensor labels,
                                           torch::Tensor sampled,
@@ -3415,6 +3436,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("ce_fwd", &ce_fwd);
   mod.def("ce_bwd", &ce_bwd);
   mod.def("adam_dense_step", &adam_dense_step);
+  mod.def("adam_lrt", &adam_lrt);
   mod.def("adam_sparse_rows_step", &adam_sparse_rows_step);
   mod.def("adam_sparse_rows_hash", &adam_sparse_rows_hash);
   mod.def("adam_sparse_rows_hash_ctx", &adam_sparse_rows_hash_ctx);

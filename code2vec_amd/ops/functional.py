@@ -2,6 +2,7 @@
 the pure-torch reference on CPU. The HIP path is mandatory on GPU (see
 ops/__init__.backend_for) — no silent eager fallback."""
 
+import math
 from typing import Tuple
 
 import torch
@@ -262,6 +263,18 @@ def topk(logits, k: int):
         return vals, idx
     vals, idx = torch.topk(logits.float(), k=k, dim=1)
     return vals, idx
+
+
+def adam_lrt(step_t, lr: float, beta1: float, beta2: float):
+    """Bias-corrected lr_t = lr*sqrt(1-b2^t)/(1-b1^t) from the device step
+    counter, as ONE kernel (the torch scalar chain is ~7 launches inside
+    the captured graph). Falls back to the torch chain off-GPU."""
+    if backend_for(step_t) == 'hip':
+        return hip_ext(True).adam_lrt(step_t, float(lr), float(beta1),
+                                      float(beta2))
+    t = step_t.to(torch.float32)
+    return (lr * torch.sqrt(1.0 - torch.exp(t * math.log(beta2)))
+            / (1.0 - torch.exp(t * math.log(beta1)))).reshape(1)
 
 
 def adam_dense_step(p, g, m, v, step: int, lr: float, beta1: float,
