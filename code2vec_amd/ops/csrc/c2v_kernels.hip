@@ -1824,7 +1824,8 @@ __global__ void k_sampled_ce_bwd(const ushort* __restrict__ L,
 // lr_t = lr*sqrt(1-b2^t)/(1-b1^t): the torch version is ~7 elementwise
 // launches on 1-element tensors per step inside the captured graph
 // (~35 us of serialized ~5 us dispatches in the steady-state trace).
-__global__ void k_adam_lrt(const long* __restrict__ step_t,
+template <typename T>
+__global__ void k_adam_lrt(const T* __restrict__ step_t,
                            float* __restrict__ out, float lr, float log_b1,
                            float log_b2) {
   const float t = (float)*step_t;
@@ -3366,9 +3367,14 @@ torch::Tensor adam_lrt(torch::Tensor step_t, double lr, double beta1,
                        double beta2) {
   CHECK_DEV(step_t);
   auto out = torch::empty({1}, step_t.options().dtype(torch::kFloat32));
-  k_adam_lrt<<<1, 1, 0, cur_stream()>>>(
-      step_t.data_ptr<long>(), out.data_ptr<float>(), (float)lr,
-      (float)std::log(beta1), (float)std::log(beta2));
+  if (step_t.scalar_type() == torch::kInt64)
+    k_adam_lrt<long><<<1, 1, 0, cur_stream()>>>(
+        step_t.data_ptr<long>(), out.data_ptr<float>(), (float)lr,
+        (float)std::log(beta1), (float)std::log(beta2));
+  else
+    k_adam_lrt<int><<<1, 1, 0, cur_stream()>>>(
+        step_t.data_ptr<int>(), out.data_ptr<float>(), (float)lr,
+        (float)std::log(beta1), (float)std::log(beta2));
   return out;
 }
 
