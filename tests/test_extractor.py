@@ -475,3 +475,59 @@ class In {
 '''
     lines = run_extract(extractor, code, tmp_path)
     assert [contexts_of(l)[0] for l in lines] == ['good', 'good']
+
+
+def test_advanced_constructs_all_extract(extractor, tmp_path):
+    """Default interface methods, enum bodies (ctor/field/method), static
+    initializers, try-with-resources, multi-catch rethrow, method
+    references (instance and static), string switch and block-body
+    lambdas must all parse — every named method in the file produces an
+    output line (javaparser-3.0.0-alpha.4 handles all of these, so a
+    skip here would be a fidelity gap vs the reference)."""
+    code = '''
+import java.io.*;
+import java.util.function.*;
+public class B {
+    interface Op { default int apply(int a) { return a + 1; } }
+    enum E { ONE(1), TWO(2); final int v; E(int v) { this.v = v; }
+             int get() { return v; } }
+    static int[] table = new int[]{1, 2, 3};
+    static { table[0] = 9; }
+    public String twr(File f) throws IOException {
+        try (BufferedReader r = new BufferedReader(new FileReader(f))) {
+            return r.readLine();
+        } catch (IOException | RuntimeException e) { throw e;
+        } finally { System.gc(); }
+    }
+    public IntSupplier mref() { return table.length > 0 ? this::len : B::slen; }
+    private int len() { return table.length; }
+    private static int slen() { return 3; }
+    public int strSwitch(String s) {
+        switch (s) { case "a": return 1; default: return 0; }
+    }
+    public Function<Integer, Integer> lamBlock() {
+        return x -> { int y = x * 2; return y + 1; };
+    }
+}
+'''
+    names = [contexts_of(l)[0] for l in run_extract(extractor, code, tmp_path)]
+    assert names == ['apply', 'get', 'twr', 'mref', 'len', 'slen',
+                     'str|switch', 'lam|block'], names
+
+
+def test_post_java8_method_skipped_others_survive(extractor, tmp_path):
+    """Syntax beyond the reference parser's generation (Java 14 switch
+    expressions, var) costs only the containing method via per-member
+    recovery — sibling methods still extract, matching the reference's
+    behavior of failing only what javaparser-alpha.4 cannot parse."""
+    code = '''
+public class C {
+    public int modern(int k) {
+        var x = switch (k) { case 1 -> 10; default -> 0; };
+        return x;
+    }
+    public int classic(int k) { return k + 1; }
+}
+'''
+    names = [contexts_of(l)[0] for l in run_extract(extractor, code, tmp_path)]
+    assert names == ['classic'], names
