@@ -159,3 +159,38 @@ class Q {
     assert out.returncode == 0, out.stderr
     names = [l.split(' ')[0] for l in out.stdout.strip().split('\n') if l]
     assert names == ['n', 'sw', 'gen', 'lam', 'use', 'tern', 'cast', 'nullc']
+
+
+def test_cs_advanced_constructs(cs_extractor, tmp_path):
+    """Generic constraints, expression-bodied methods, using statements,
+    try/finally and nested classes all extract; operators, property
+    accessors and events produce NO lines — the reference's Roslyn
+    walker only takes MethodDeclarationSyntax (Extractor.cs), so those
+    members are correctly excluded rather than mis-extracted."""
+    lines = run_cs(cs_extractor, '''
+using System;
+using System.Collections.Generic;
+namespace N {
+    public class P<T> where T : IComparable<T> {
+        public event EventHandler Changed;
+        public int Count { get; private set; }
+        public static P<T> operator +(P<T> a, P<T> b) { return a; }
+        public int ExprBody(int x) => x * 2 + 1;
+        public T Max(List<T> items) {
+            T best = items[0];
+            foreach (var it in items) if (it.CompareTo(best) > 0) best = it;
+            return best;
+        }
+        public void UsingStmt() {
+            using (var d = new System.IO.MemoryStream()) { d.WriteByte(1); }
+        }
+        public int TryFin(int k) {
+            try { return 10 / k; } catch (DivideByZeroException) { return 0; }
+            finally { Count++; }
+        }
+        private class Inner { public int Val() { return 7; } }
+    }
+}
+''', tmp_path)
+    names = [l.split(' ')[0] for l in lines]
+    assert names == ['expr|body', 'max', 'using|stmt', 'try|fin', 'val'], names
