@@ -105,3 +105,55 @@ def test_extract_preprocess_train_predict(java_corpus, tmp_path):
     assert preds and preds[0].original_name == 'get|value'
     words = list(preds[0].topk_predicted_words)
     assert 'get|value' in words[:3], words
+
+
+def test_interactive_predictor_loop(tmp_path, monkeypatch, capsys):
+    """InteractivePredictor's REPL loop with stubbed model/extractor: one
+    'ready' keypress predicts Input.java and prints the top-k names,
+    attention contexts (paths unhashed) and the code vector; 'q' exits
+    (reference behavior: interactive_predict.py:28-57)."""
+    import numpy as np
+
+    from code2vec_amd.models.model_base import ModelPredictionResults
+    from code2vec_amd.serving import interactive_predict as ip
+
+    raw = ModelPredictionResults(
+        original_name='get|value',
+        topk_predicted_words=np.array(['get|value', 'set|value']),
+        topk_predicted_words_scores=np.array([0.9, 0.1]),
+        attention_per_context={('this', '1234', 'value'): 0.75,
+                               ('value', '77', 'one'): 0.25},
+        code_vector=np.array([0.5, -0.5]))
+
+    class StubVocab:
+        class target_vocab:  # noqa: N801 — attribute-shaped stub
+            class special_words:  # noqa: N801
+                OOV = '<OOV>'
+
+    class StubModel:
+        vocabs = StubVocab
+        def predict(self, lines):
+            return [raw] if lines else []
+
+    class StubExtractor:
+        def __init__(self, *a, **k):
+            pass
+        def extract_paths(self, filename):
+            assert filename == ip.INPUT_FILENAME
+            return ['get|value this,1234,value'], {'1234': 'unhashed^path'}
+
+    monkeypatch.setattr(ip, 'Extractor', StubExtractor)
+
+    class Cfg:
+        EXPORT_CODE_VECTORS = True
+
+    pred = ip.InteractivePredictor(Cfg(), StubModel())
+    answers = iter(['', 'q'])
+    monkeypatch.setattr('builtins.input', lambda: next(answers))
+    pred.predict()
+    out = capsys.readouterr().out
+    assert 'Original name:\tget|value' in out
+    assert 'predicted: get|value' in out or "predicted: ['get', 'value']" in out
+    assert 'unhashed^path' in out          # unhash dict applied
+    assert 'Code vector:' in out and '0.5' in out
+    assert 'Exiting...' in out
