@@ -103,10 +103,14 @@ class Lexer {
     Token t;
     if (i_ >= s_.size()) return t;
     char c = s_[i_];
-    if (isalpha((unsigned char)c) || c == '_' || c == '$') {
+    // bytes >= 0x80 are UTF-8 lead/continuation bytes: Java allows unicode
+    // identifiers (javaparser accepts them), so fold whole multi-byte
+    // sequences into the identifier token instead of failing the member
+    if (isalpha((unsigned char)c) || c == '_' || c == '$' ||
+        (unsigned char)c >= 0x80) {
       size_t j = i_;
       while (j < s_.size() && (isalnum((unsigned char)s_[j]) || s_[j] == '_' ||
-                               s_[j] == '$'))
+                               s_[j] == '$' || (unsigned char)s_[j] >= 0x80))
         ++j;
       t.text = s_.substr(i_, j - i_);
       t.kind = kKeywords.count(t.text) ? Tok::Keyword : Tok::Ident;

@@ -531,3 +531,40 @@ public class C {
 '''
     names = [contexts_of(l)[0] for l in run_extract(extractor, code, tmp_path)]
     assert names == ['classic'], names
+
+
+def test_labeled_loops_wildcards_unicode(extractor, tmp_path):
+    """Labeled break/continue, bounded wildcards with cast, bit-shift
+    chains, char literals and UNICODE identifiers (Java permits them;
+    the lexer folds multi-byte UTF-8 into the identifier token) all
+    extract."""
+    code = '''
+public class D {
+    public int labeled(int[][] m) {
+        int s = 0;
+        outer:
+        for (int i = 0; i < m.length; i++) {
+            for (int j = 0; j < m[i].length; j++) {
+                if (m[i][j] < 0) continue outer;
+                if (m[i][j] == 99) break outer;
+                s += m[i][j];
+            }
+        }
+        return s;
+    }
+    public char charLit(String s) { return s.isEmpty() ? '\\0' : s.charAt(0); }
+    public boolean inst(Object o) {
+        return o instanceof String && ((String) o).length() > 2;
+    }
+    @SuppressWarnings({"unchecked", "rawtypes"})
+    public java.util.List<? extends Number> wild(
+            java.util.List<? super Integer> in) {
+        return (java.util.List) in;
+    }
+    public long bitops(long a, long b) { return (a << 3) | (b >>> 2) ^ ~a & b; }
+    public double unicodeId() { double π = 3.14159; return π; }
+}
+'''
+    names = [contexts_of(l)[0] for l in run_extract(extractor, code, tmp_path)]
+    assert names == ['labeled', 'char|lit', 'inst', 'wild', 'bitops',
+                     'unicode|id'], names
