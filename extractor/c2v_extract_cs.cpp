@@ -122,10 +122,15 @@ class CsLexer {
     Token t;
     if (i_ >= s_.size()) return t;
     char c = s_[i_];
-    if (isalpha((unsigned char)c) || c == '_' || c == '@') {
+    // bytes >= 0x80 are UTF-8 sequences: C# (like Roslyn) permits unicode
+    // identifiers, and with no per-member recovery here a lex failure
+    // would cost the whole file
+    if (isalpha((unsigned char)c) || c == '_' || c == '@' ||
+        (unsigned char)c >= 0x80) {
       size_t j = i_ + (c == '@' ? 1 : 0);
       size_t b = j;
-      while (j < s_.size() && (isalnum((unsigned char)s_[j]) || s_[j] == '_'))
+      while (j < s_.size() && (isalnum((unsigned char)s_[j]) || s_[j] == '_' ||
+                               (unsigned char)s_[j] >= 0x80))
         ++j;
       t.text = s_.substr(b, j - b);
       t.kind = (c != '@' && kCsKeywords.count(t.text)) ? Tok::Keyword : Tok::Ident;

@@ -194,3 +194,17 @@ namespace N {
 ''', tmp_path)
     names = [l.split(' ')[0] for l in lines]
     assert names == ['expr|body', 'max', 'using|stmt', 'try|fin', 'val'], names
+
+
+def test_cs_unicode_identifier(cs_extractor, tmp_path):
+    """Unicode identifiers lex as single tokens (Roslyn permits them);
+    without this the C# parser — which has file-level, not per-member,
+    error recovery — lost the entire file."""
+    lines = run_cs(cs_extractor, '''
+public class U {
+    public double UniId() { double π = 3.14; return π; }
+    public int Plain(int x) { return x + 1; }
+}
+''', tmp_path)
+    names = [l.split(' ')[0] for l in lines]
+    assert names == ['uni|id', 'plain'], names
