@@ -568,3 +568,36 @@ public class D {
     names = [contexts_of(l)[0] for l in run_extract(extractor, code, tmp_path)]
     assert names == ['labeled', 'char|lit', 'inst', 'wild', 'bitops',
                      'unicode|id'], names
+
+
+def test_member_kind_exclusions_and_generics(extractor, tmp_path):
+    """Instance initializers, generic constructors, interface/abstract/
+    native methods (no body) produce NO lines — the reference visits
+    only MethodDeclaration nodes with bodies (FunctionVisitor.java:25-55)
+    — while class literals, static imports, conditional array
+    initializers, generic methods with throws, and strictfp all
+    extract."""
+    code = '''
+import static java.lang.Math.max;
+public class E {
+    { counter = 1; }
+    int counter;
+    <T> E(T seed) { counter = seed.hashCode(); }
+    interface Cb { void fire(Class<?> cls); }
+    public Class<?> classLit() { return String.class; }
+    public int statImport(int a, int b) { return max(a, b); }
+    public int[] arrInit() { return new int[]{1 > 0 ? 1 : 2, 3}; }
+    public <K, V> java.util.Map<K, V> genMethod(K k, V v)
+            throws IllegalStateException {
+        java.util.Map<K, V> m = new java.util.HashMap<>();
+        m.put(k, v);
+        return m;
+    }
+    public strictfp double sfp(double d) { return d * 2.0; }
+    public final native int nat(int x);
+    public abstract static class Inner { public abstract int go(); }
+}
+'''
+    names = [contexts_of(l)[0] for l in run_extract(extractor, code, tmp_path)]
+    assert names == ['class|lit', 'stat|import', 'arr|init', 'gen|method',
+                     'sfp'], names
