@@ -174,3 +174,36 @@ def test_reader_crlf_long_lines_empty_file(tmp_path):
         reader2.stop_streaming()
     finally:
         os.environ.pop('C2V_READER_CHUNK_BYTES', None)
+
+
+def test_malformed_rows_degrade_not_crash(vocabs_and_cfg, tmp_path):
+    """Adversarial .c2v input: malformed contexts (2 or 4 comma fields),
+    over-long rows, non-UTF8 bytes, empty lines and context-less rows.
+    Valid rows stream through; empty/context-less/OOV-target rows are
+    dropped by the train filters; malformed contexts degrade to PAD/OOV
+    per-field (the reference's decode_csv+split is equally permissive)
+    — and nothing crashes the native parser."""
+    vocabs, cfg = vocabs_and_cfg
+    rows = [
+        b'name|one s1,p1,t1 s2,p2,s1',      # valid
+        b'other s1,p1',                     # 2-field context -> padded
+        b'',                                # dropped
+        b'name|one',                        # no contexts -> dropped
+        b'other s1,p1,t1,zzz s2,p2,s1',     # 4-field context
+        b'name|one ' + b's1,p1,t1 ' * 9,    # > MAX_CONTEXTS, truncated
+        b'nope s1,p1,t1',                   # OOV target -> dropped (train)
+        b'other \xff\xfe,p1,t1 s2,p2,s1',   # non-UTF8 token bytes
+    ]
+    data = tmp_path / 'mal.train.c2v'
+    data.write_bytes(b'\n'.join(rows) + b'\n')
+    cfg.TRAIN_DATA_PATH_PREFIX = str(tmp_path / 'mal')
+    cfg.NUM_TRAIN_EPOCHS = 1
+    r = PathContextReader(vocabs, cfg, EstimatorAction.Train)
+    n = 0
+    try:
+        for b in r.iter_batches():
+            assert b.source_token_indices.shape[1] == cfg.MAX_CONTEXTS
+            n += int(b.target_index.shape[0])
+    finally:
+        r.stop_streaming(join=True)
+    assert n == 5, n   # all but the empty, context-less and OOV-target rows
